@@ -1,0 +1,90 @@
+import numpy as np
+import pytest
+
+from realhf_amd.base.datapack import (
+    flat2d,
+    min_abs_diff_partition,
+    partition_balanced,
+    reorder_to_balanced_batches,
+)
+from realhf_amd.base.topology import PipeDataTensorTopology, ProcessTopology
+
+
+def test_topology_rank_coord_roundtrip():
+    topo = PipeDataTensorTopology(num_pp=2, num_dp=4, num_tp=2)
+    assert topo.world_size() == 16
+    for r in range(16):
+        c = topo.get_coord(r)
+        assert topo.get_rank(pipe=c.pipe, data=c.data, tensor=c.tensor) == r
+    # tensor axis is fastest-varying: ranks 0 and 1 share (pipe, data)
+    c0, c1 = topo.get_coord(0), topo.get_coord(1)
+    assert (c0.pipe, c0.data) == (c1.pipe, c1.data)
+    assert c0.tensor == 0 and c1.tensor == 1
+
+
+def test_topology_filter_match():
+    topo = PipeDataTensorTopology(num_pp=2, num_dp=2, num_tp=2)
+    tp_group = topo.filter_match(pipe=0, data=0)
+    assert tp_group == [0, 1]
+    dp_group = topo.filter_match(pipe=0, tensor=1)
+    assert dp_group == [1, 3]
+    pp_group = topo.filter_match(data=1, tensor=0)
+    assert pp_group == [2, 6]
+
+
+@pytest.mark.parametrize("n,k", [(8, 2), (10, 3), (100, 8), (5, 5), (17, 4)])
+def test_min_abs_diff_partition(n, k):
+    rng = np.random.RandomState(n * 100 + k)
+    lens = rng.randint(1, 100, size=n).tolist()
+    bounds = min_abs_diff_partition(lens, k)
+    assert len(bounds) == k
+    assert bounds[0][0] == 0 and bounds[-1][1] == n
+    for (s0, e0), (s1, e1) in zip(bounds, bounds[1:]):
+        assert e0 == s1
+        assert e0 > s0
+    sums = [sum(lens[s:e]) for s, e in bounds]
+    # balanced: max group <= total (trivially) and reasonably tight
+    assert max(sums) <= sum(lens)
+    # optimal max-sum check vs brute force for small n
+    if n <= 10:
+        import itertools
+
+        best = min(
+            max(
+                sum(lens[s:e])
+                for s, e in zip((0,) + cut, cut + (n,))
+            )
+            for cut in itertools.combinations(range(1, n), k - 1)
+        )
+        assert max(sums) == best
+
+
+def test_partition_balanced_and_reorder():
+    lens = [5, 1, 1, 1, 5, 1, 1, 1]
+    groups = partition_balanced(lens, 2)
+    assert flat2d(groups) == list(range(8))
+    batches = reorder_to_balanced_batches(lens, 2)
+    s0 = sum(lens[i] for i in batches[0])
+    s1 = sum(lens[i] for i in batches[1])
+    assert abs(s0 - s1) <= 1
+
+
+def test_name_resolve_file(tmp_path):
+    from realhf_amd.base.name_resolve import (
+        FileNameRecordRepository,
+        NameEntryExistsError,
+        NameEntryNotFoundError,
+    )
+
+    repo = FileNameRecordRepository(root=str(tmp_path))
+    repo.add("a/b/c", "1")
+    assert repo.get("a/b/c") == "1"
+    with pytest.raises(NameEntryExistsError):
+        repo.add("a/b/c", "2")
+    repo.add("a/b/c", "2", replace=True)
+    assert repo.get("a/b/c") == "2"
+    repo.add("a/b/d", "3")
+    assert repo.get_subtree("a/b") == ["2", "3"]
+    repo.clear_subtree("a")
+    with pytest.raises(NameEntryNotFoundError):
+        repo.get("a/b/c")

@@ -1,0 +1,99 @@
+import numpy as np
+import pytest
+import torch
+
+from realhf_amd.api.data import PackedDataLoader, SequenceSample
+
+
+def make_sample(bs, seed=0, with_reward=True):
+    rng = np.random.RandomState(seed)
+    seqlens = rng.randint(5, 40, size=bs).tolist()
+    total = sum(seqlens)
+    data = {
+        "packed_input_ids": torch.arange(total, dtype=torch.long),
+        "packed_logprobs": torch.randn(total - bs),
+    }
+    if with_reward:
+        data["rewards"] = torch.randn(bs)
+    ids = [f"s{seed}_{i}" for i in range(bs)]
+    return SequenceSample.from_default(ids=ids, seqlens=seqlens, data=data)
+
+
+def test_from_default_key_rules():
+    s = make_sample(4)
+    assert s.seqlens["packed_logprobs"] == [[l[0] - 1] for l in s.seqlens["packed_input_ids"]]
+    assert s.seqlens["rewards"] == [[1]] * 4
+
+
+def test_gather_split_roundtrip():
+    parts = [make_sample(3, seed=i) for i in range(4)]
+    g = SequenceSample.gather(parts)
+    assert g.bs == 12
+    back = g.split_with_spec([(0, 3), (3, 6), (6, 9), (9, 12)])
+    for orig, b in zip(parts, back):
+        assert orig.ids == b.ids
+        for k in orig.keys:
+            assert torch.equal(orig.data[k], b.data[k])
+
+
+@pytest.mark.parametrize("dp", [1, 2, 3, 4, 8])
+def test_balanced_split(dp):
+    s = make_sample(16, seed=dp)
+    shards = s.split(dp)
+    assert sum(x.bs for x in shards) == 16
+    recon = SequenceSample.gather(shards)
+    for k in s.keys:
+        assert torch.equal(s.data[k], recon.data[k])
+    # balance check
+    tok = [sum(x.main_seqlens()) for x in shards]
+    assert max(tok) - min(tok) <= max(s.main_seqlens())
+
+
+def test_unpack_and_meta():
+    s = make_sample(5)
+    singles = s.unpack()
+    assert len(singles) == 5
+    assert all(x.bs == 1 for x in singles)
+    m = s.meta()
+    assert m.data is None
+    assert m.seqlens == s.seqlens
+    assert m.dtypes["packed_input_ids"] == torch.long
+
+
+def test_update_and_remap():
+    s = make_sample(4, with_reward=False)
+    total = sum(s.main_seqlens())
+    other = SequenceSample(
+        keys=("values",),
+        ids=list(s.ids),
+        seqlens={"values": s.seqlens["packed_input_ids"]},
+        data={"values": torch.randn(total)},
+    )
+    s.update_(other)
+    assert "values" in s.keys
+    s.remap_keys_({"values": "old_values"})
+    assert "old_values" in s.keys and "values" not in s.keys
+
+
+def test_select():
+    s = make_sample(6)
+    sub = s.select_idx([5, 0, 3])
+    assert sub.ids == [s.ids[5], s.ids[0], s.ids[3]]
+    k = s.select_keys(["packed_input_ids"])
+    assert k.keys == ("packed_input_ids",)
+
+
+def test_packed_dataloader():
+    class DS(torch.utils.data.Dataset):
+        def __len__(self):
+            return 10
+
+        def __getitem__(self, i):
+            return make_sample(1, seed=100 + i)
+
+    dl = PackedDataLoader(DS(), batch_n_seqs=4, shuffle=True, seed=1)
+    batches = list(dl)
+    assert len(batches) == 3
+    assert batches[0].bs == 4
+    all_ids = [i for b in batches for i in b.ids]
+    assert len(set(all_ids)) == 10
