@@ -30,6 +30,7 @@ class GenerationHyperparameters:
 class MoEConfig:
     num_experts: int = 8
     top_k: int = 2
+    norm_topk_prob: bool = True  # renormalize top-k routing weights (mixtral)
     routing_type: str = "aux_loss"  # aux_loss | sinkhorn | none
     aux_loss_coef: float = 1e-2
     z_loss_coef: float = 0.0
@@ -72,6 +73,8 @@ class ReaLModelConfig:
     # norm details
     scale_attn_by_inverse_layer_idx: bool = False
     qk_layernorm: bool = False
+    # gemma-style sqrt(hidden) embedding scaling
+    embedding_multiplier: Optional[float] = None
     # dropout (0 for RLHF)
     attn_pdrop: float = 0.0
     resid_pdrop: float = 0.0

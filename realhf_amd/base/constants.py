@@ -80,6 +80,11 @@ def model_scope(model_name: str):
         _model_scope = prev
 
 
+def has_current() -> bool:
+    """True when inside a model_scope (parallel state available)."""
+    return _model_scope is not None
+
+
 def current_model_name() -> str:
     assert _model_scope is not None, "not inside a model_scope"
     return _model_scope

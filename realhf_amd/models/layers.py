@@ -1,0 +1,283 @@
+"""Transformer layer modules over flat-buffer parameter views.
+
+Reference semantics: realhf/impl/model/nn/real_llm_base.py (ReaLModelBlock:103,
+VocabPositionEmbedding:260, OutputHead:346, SequenceParallelCriticHead:356,
+ParallelActorHead:370) and modules/{attn,mlp}.py.
+
+Every module receives its parameters as a dict of tensors that are VIEWS
+into the owning ReaLModel's contiguous flat buffer; modules never allocate
+weights.  TP behavior comes from the current model scope's grid
+(realhf_amd.base.constants); with tp_size == 1 every collective is a no-op
+so the same code runs single-process on CPU.
+
+MI355X specifics: wq/wk/wv (and gate/up) flat regions are adjacent, so the
+module runs ONE hipBLASLt GEMM over the concatenated view; RMSNorm / RoPE /
+SwiGLU / attention are the hand-written HIP kernels in realhf_amd.ops.
+"""
+import math
+from typing import Dict, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from realhf_amd.api.model import ReaLModelConfig
+from realhf_amd.base import constants
+from realhf_amd.ops import functional as ops
+from realhf_amd.parallel import mappings
+from realhf_amd.utils.functional import compute_varlen_position_indices
+
+
+def _maybe_merged(params: Dict[str, torch.Tensor], names, dim_out_total):
+    """If the named 2-D params are contiguous consecutive views of the flat
+    buffer, return a single merged [sum_out, in] view; else None."""
+    ts = [params[n] for n in names]
+    base = ts[0]
+    if any(t.dtype != base.dtype or t.device != base.device for t in ts):
+        return None
+    try:
+        storages = {t.untyped_storage().data_ptr() for t in ts}
+    except RuntimeError:
+        return None
+    if len(storages) != 1:
+        return None
+    off = base.storage_offset()
+    for t in ts:
+        if t.storage_offset() != off or not t.is_contiguous():
+            return None
+        off += t.numel()
+    in_dim = base.shape[1]
+    merged = base.as_strided((dim_out_total, in_dim), (in_dim, 1), base.storage_offset())
+    return merged
+
+
+def _linear(x, w, b=None):
+    return F.linear(x, w, b)
+
+
+class VocabEmbedding(nn.Module):
+    """Vocab-parallel token embedding (+ optional learned positions).
+
+    TP: each rank holds rows [r*V/tp, (r+1)*V/tp); out-of-range ids embed
+    to zero and the partial results all-reduce over the TP group
+    (reference: model_parallel/modules.py:53 ParallelEmbedding)."""
+
+    def __init__(self, cfg: ReaLModelConfig, params: Dict[str, torch.Tensor]):
+        super().__init__()
+        self.cfg = cfg
+        self.wte = params["0.wte.weight"]
+        self.wpe = params.get("0.wpe.weight")
+
+    def forward(self, packed_input_ids: torch.Tensor, positions: torch.Tensor):
+        tp = constants.tp_world_size() if constants.has_current() else 1
+        if tp > 1:
+            r = constants.tp_rank()
+            n = self.wte.shape[0]
+            lo, hi = r * n, (r + 1) * n
+            mask = (packed_input_ids >= lo) & (packed_input_ids < hi)
+            local_ids = (packed_input_ids - lo).clamp(0, n - 1)
+            h = F.embedding(local_ids, self.wte)
+            h = h * mask.unsqueeze(-1).to(h.dtype)
+            h = mappings.reduce_from_tp_region(h)
+        else:
+            h = F.embedding(packed_input_ids, self.wte)
+        if self.cfg.embedding_multiplier is not None:
+            h = h * self.cfg.embedding_multiplier
+        if self.wpe is not None:
+            h = h + F.embedding(positions, self.wpe)
+        if constants.has_current() and constants.sequence_parallel():
+            h = mappings.scatter_to_sp_region(h)
+        return h
+
+
+def _norm(cfg: ReaLModelConfig, x, w, b):
+    if cfg.norm_type == "rms":
+        return ops.rms_norm(x, w, cfg.layer_norm_epsilon)
+    if cfg.norm_type == "gemma_rms":
+        return ops.rms_norm(x, w, cfg.layer_norm_epsilon, gemma_style=True)
+    return F.layer_norm(
+        x.float(), (x.shape[-1],), w.float(), b.float() if b is not None else None,
+        cfg.layer_norm_epsilon,
+    ).to(x.dtype)
+
+
+class ReaLModelBlock(nn.Module):
+    """Pre-LN transformer block on packed sequences (reference:
+    real_llm_base.py:103, modules/attn.py:31, modules/mlp.py)."""
+
+    def __init__(self, cfg: ReaLModelConfig, layer_idx: int, params: Dict[str, torch.Tensor], tp_size: int = 1):
+        super().__init__()
+        self.cfg = cfg
+        self.i = layer_idx
+        self.p = params
+        self.tp_size = tp_size
+        self.nq = cfg.n_heads // tp_size
+        self.nkv = max(cfg.n_kv_heads // tp_size, 1)
+        assert cfg.n_kv_heads % tp_size == 0 or tp_size % cfg.n_kv_heads == 0
+        self.hd = cfg.head_dim
+        i = layer_idx
+        self._qkv_names = [f"{i}.attn.wq.weight", f"{i}.attn.wk.weight", f"{i}.attn.wv.weight"]
+        self._gu_names = (
+            [f"{i}.mlp.gate.weight", f"{i}.mlp.up.weight"]
+            if cfg.activation in ("silu", "geglu") and cfg.moe is None
+            else None
+        )
+        self.moe = None
+        if cfg.moe is not None:
+            from realhf_amd.models.moe import MoELayer
+
+            self.moe = MoELayer(cfg, layer_idx, params, tp_size)
+
+    # -- attention --------------------------------------------------------
+    def _qkv(self, x):
+        qkv_dim = (self.nq + 2 * self.nkv) * self.hd
+        merged = _maybe_merged(self.p, self._qkv_names, qkv_dim)
+        if merged is not None:
+            qkv = _linear(x, merged)
+        else:
+            qkv = torch.cat([_linear(x, self.p[n]) for n in self._qkv_names], dim=-1)
+        i = self.i
+        if f"{i}.attn.wq.bias" in self.p:
+            bias = torch.cat(
+                [self.p[f"{i}.attn.w{c}.bias"] for c in "qkv"], dim=0
+            )
+            qkv = qkv + bias
+        q, k, v = qkv.split(
+            [self.nq * self.hd, self.nkv * self.hd, self.nkv * self.hd], dim=-1
+        )
+        t = x.shape[0]
+        return (
+            q.view(t, self.nq, self.hd),
+            k.view(t, self.nkv, self.hd),
+            v.view(t, self.nkv, self.hd),
+        )
+
+    def forward(
+        self,
+        x: torch.Tensor,  # [total, h]
+        cu_seqlens: torch.Tensor,
+        max_seqlen: int,
+        positions: torch.Tensor,  # [total]
+        k_cache: Optional[torch.Tensor] = None,  # [bs, maxlen, nkv, hd]
+        v_cache: Optional[torch.Tensor] = None,
+        cache_seqlens: Optional[torch.Tensor] = None,  # [bs]
+        decode: bool = False,
+    ):
+        i = self.i
+        cfg = self.cfg
+        h = _norm(cfg, x, self.p[f"{i}.attn.ln.weight"], self.p.get(f"{i}.attn.ln.bias"))
+        sp = constants.has_current() and constants.sequence_parallel()
+        if sp:
+            h = mappings.gather_from_sp_region(h)
+        else:
+            h = mappings.copy_to_tp_region(h)
+        q, k, v = self._qkv(h)
+        if cfg.apply_rotary:
+            cos, sin = ops.rotary_cache.get(
+                self.hd,
+                int(positions.max().item()) + 1 if positions.numel() else 1,
+                cfg.rotary_base,
+                x.device,
+                scaling=cfg.rotary_scaling,
+            )
+            q = ops.apply_rotary(q, cos, sin, positions, cfg.rotary_interleaved)
+            k = ops.apply_rotary(k, cos, sin, positions, cfg.rotary_interleaved)
+
+        scale = 1.0 / math.sqrt(self.hd)
+        if cfg.scale_attn_by_inverse_layer_idx:
+            scale = scale / float(self.i)
+
+        if decode:
+            # one new token per sequence; write into cache then attend
+            bs = k_cache.shape[0]
+            assert q.shape[0] == bs
+            idx = (cache_seqlens.long() - 1).clamp(min=0)
+            b_idx = torch.arange(bs, device=x.device)
+            k_cache[b_idx, idx] = k.to(k_cache.dtype)
+            v_cache[b_idx, idx] = v.to(v_cache.dtype)
+            attn_out = ops.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
+        else:
+            if k_cache is not None:
+                # prefill: write all tokens into the cache
+                bs = k_cache.shape[0]
+                seq_id = torch.bucketize(
+                    torch.arange(x.shape[0], device=x.device),
+                    cu_seqlens[1:].long(),
+                    right=True,
+                )
+                k_cache[seq_id, positions] = k.detach().to(k_cache.dtype)
+                v_cache[seq_id, positions] = v.detach().to(v_cache.dtype)
+            attn_out = ops.attn_varlen(
+                q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=scale
+            )
+        attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
+        o = _linear(attn_out, self.p[f"{i}.attn.wo.weight"])
+        if sp:
+            o = mappings.reduce_scatter_to_sp_region(o)
+        else:
+            o = mappings.reduce_from_tp_region(o)
+        if f"{i}.attn.wo.bias" in self.p:
+            o = o + self.p[f"{i}.attn.wo.bias"]
+        x = x + o
+
+        # -- MLP ----------------------------------------------------------
+        h = _norm(cfg, x, self.p[f"{i}.mlp.ln.weight"], self.p.get(f"{i}.mlp.ln.bias"))
+        if self.moe is not None:
+            m = self.moe(h)
+            x = x + m
+            return x
+        if sp:
+            h = mappings.gather_from_sp_region(h)
+        else:
+            h = mappings.copy_to_tp_region(h)
+        if cfg.activation in ("silu", "geglu"):
+            idim_local = self.p[f"{i}.mlp.gate.weight"].shape[0]
+            merged = _maybe_merged(self.p, self._gu_names, 2 * idim_local)
+            if merged is not None:
+                gu = _linear(h, merged)
+            else:
+                gu = torch.cat(
+                    [_linear(h, self.p[n]) for n in self._gu_names], dim=-1
+                )
+            if cfg.activation == "silu":
+                act = ops.swiglu(gu)
+            else:  # geglu (gemma)
+                gate, up = gu.chunk(2, dim=-1)
+                act = (
+                    F.gelu(gate.float(), approximate="tanh") * up.float()
+                ).to(gu.dtype)
+        else:
+            up = _linear(h, self.p[f"{i}.mlp.up.weight"], self.p.get(f"{i}.mlp.up.bias"))
+            act = F.gelu(up, approximate="tanh")
+        down = _linear(act, self.p[f"{i}.mlp.down.weight"])
+        if sp:
+            down = mappings.reduce_scatter_to_sp_region(down)
+        else:
+            down = mappings.reduce_from_tp_region(down)
+        if f"{i}.mlp.down.bias" in self.p:
+            down = down + self.p[f"{i}.mlp.down.bias"]
+        return x + down
+
+
+class OutputHead(nn.Module):
+    """Final norm + LM head (vocab-parallel) or critic head (replicated
+    1-dim output) (reference: real_llm_base.py:346-392)."""
+
+    def __init__(self, cfg: ReaLModelConfig, params: Dict[str, torch.Tensor],
+                 tied_embedding_weight: Optional[torch.Tensor] = None):
+        super().__init__()
+        self.cfg = cfg
+        self.p = params
+        self.tied_w = tied_embedding_weight
+        self.i = cfg.n_layers + 1
+
+    def forward(self, x):
+        cfg = self.cfg
+        h = _norm(cfg, x, self.p[f"{self.i}.ln_f.weight"], self.p.get(f"{self.i}.ln_f.bias"))
+        if constants.has_current() and constants.sequence_parallel():
+            h = mappings.gather_from_sp_region(h)
+        w = self.tied_w if self.tied_w is not None else self.p.get(f"{self.i}.head.weight")
+        if cfg.is_critic:
+            return _linear(h.float(), w.float())  # [total, 1] fp32
+        h = mappings.copy_to_tp_region(h)
+        return _linear(h, w)  # [total, vocab/tp] — vocab-parallel logits

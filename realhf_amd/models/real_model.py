@@ -1,0 +1,260 @@
+"""ReaLModel — the flat-parameter packed-sequence transformer.
+
+Reference semantics: realhf/impl/model/nn/real_llm_api.py (ReaLModel:82,
+instantiate:183, forward:384, state_dict remap:511, async_offload:274).
+
+All parameters of this shard (its TP shard of its PP stage's layers) live
+in ONE contiguous buffer (`flat_param`); layer modules hold views.  This
+is the contract that makes ZeRO-1 sharding, offload, and parameter
+reallocation interval math over a single tensor (SURVEY.md §2.1
+"Flat-param spec").
+"""
+import dataclasses
+from typing import Dict, List, Optional
+
+import torch
+import torch.nn as nn
+
+from realhf_amd.api.model import ReaLModelConfig
+from realhf_amd.base import constants, logging
+from realhf_amd.models import param_layout as PL
+from realhf_amd.models.layers import OutputHead, ReaLModelBlock, VocabEmbedding
+from realhf_amd.utils.functional import compute_varlen_position_indices
+
+logger = logging.getLogger("model")
+
+
+class ReaLModel(nn.Module):
+    def __init__(
+        self,
+        config: ReaLModelConfig,
+        device=None,
+        dtype: Optional[torch.dtype] = None,
+        tp_rank: int = 0,
+        tp_size: int = 1,
+        pp_rank: int = 0,
+        pp_size: int = 1,
+        instantiate: bool = True,
+    ):
+        super().__init__()
+        self.config = config
+        self.device = torch.device(device or ("cuda" if torch.cuda.is_available() else "cpu"))
+        self.dtype = dtype or config.torch_dtype
+        self.tp_rank, self.tp_size = tp_rank, tp_size
+        self.pp_rank, self.pp_size = pp_rank, pp_size
+        self.pipeline_partition = PL.partition_pipeline_layers(config, pp_size)
+        lo, hi = self.pipeline_partition[pp_rank]
+        self.layer_indices = list(range(lo, hi))
+        self.layout = PL.build_flat_layout(config, self.layer_indices, tp_rank, tp_size)
+
+        self.flat_param: Optional[torch.Tensor] = None
+        self.flat_grad: Optional[torch.Tensor] = None
+        self._params: Dict[str, torch.Tensor] = {}
+        self.layers = nn.ModuleList()
+        self._offload_buf: Optional[torch.Tensor] = None
+        self._offloaded = False
+        if instantiate:
+            self.instantiate()
+
+    # ------------------------------------------------------------------
+    @property
+    def is_first_stage(self):
+        return self.pp_rank == 0
+
+    @property
+    def is_last_stage(self):
+        return self.pp_rank == self.pp_size - 1
+
+    @property
+    def num_layers_total(self):
+        return self.config.n_layers + 2
+
+    def instantiate(self):
+        self.flat_param = torch.empty(
+            self.layout.total_numel, dtype=self.dtype, device=self.device
+        )
+        self._map_params()
+        self._build_modules()
+
+    def _map_params(self):
+        self._params = {}
+        for k in self.layout.keys:
+            s = self.layout.specs[k]
+            self._params[k] = self.flat_param[s.start : s.end].view(s.shape)
+
+    def _build_modules(self):
+        cfg = self.config
+        self.layers = nn.ModuleList()
+        for idx in self.layer_indices:
+            if idx == 0:
+                self.layers.append(VocabEmbedding(cfg, self._params))
+            elif idx == cfg.n_layers + 1:
+                tied_w = None
+                if cfg.tied_embedding and not cfg.is_critic:
+                    assert "0.wte.weight" in self._params or self.pp_size > 1, (
+                        "tied embedding with pp>1 needs the embedding group sync"
+                    )
+                    tied_w = self._params.get("0.wte.weight")
+                self.layers.append(OutputHead(cfg, self._params, tied_w))
+            else:
+                self.layers.append(
+                    ReaLModelBlock(cfg, idx, self._params, self.tp_size)
+                )
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def random_init(self, std: float = 0.02):
+        """Random init (bench path: no network, random weights)."""
+        for k, t in self._params.items():
+            if t.ndim >= 2:
+                t.normal_(0.0, std)
+            else:
+                if k.endswith("ln.weight") or k.endswith("ln_f.weight"):
+                    t.fill_(1.0)
+                else:
+                    t.zero_()
+
+    # ------------------------------------------------------------------ fwd
+    def forward(
+        self,
+        packed_input_ids: Optional[torch.Tensor] = None,
+        cu_seqlens: Optional[torch.Tensor] = None,
+        max_seqlen: Optional[int] = None,
+        hidden_states: Optional[torch.Tensor] = None,
+        positions: Optional[torch.Tensor] = None,
+        kv_caches: Optional[List[torch.Tensor]] = None,  # [(k,v)] per local block
+        cache_seqlens: Optional[torch.Tensor] = None,
+        decode: bool = False,
+    ):
+        """Run this PP stage.  First stage consumes packed_input_ids, later
+        stages consume hidden_states.  Returns hidden states (mid stages)
+        or logits/values (last stage)."""
+        assert self._offloaded is False, "model is offloaded; reload first"
+        cfg = self.config
+        if decode:
+            assert cache_seqlens is not None
+            positions = cache_seqlens.long() - 1
+            cu = None
+            mx = None
+        else:
+            total = (
+                packed_input_ids.shape[0]
+                if packed_input_ids is not None
+                else hidden_states.shape[0]
+            )
+            if positions is None:
+                positions = compute_varlen_position_indices(total, cu_seqlens)
+            cu = cu_seqlens
+            mx = max_seqlen
+
+        x = hidden_states
+        cache_i = 0
+        use_ckpt = (
+            constants.has_current()
+            and constants.gradient_checkpointing()
+            and torch.is_grad_enabled()
+            and not decode
+        )
+        for idx, layer in zip(self.layer_indices, self.layers):
+            if idx == 0:
+                x = layer(packed_input_ids, positions)
+            elif idx == cfg.n_layers + 1:
+                x = layer(x)
+            else:
+                kc = vc = None
+                if kv_caches is not None:
+                    kc, vc = kv_caches[cache_i]
+                    cache_i += 1
+                if use_ckpt:
+                    x = torch.utils.checkpoint.checkpoint(
+                        layer, x, cu, mx, positions, kc, vc, cache_seqlens, decode,
+                        use_reentrant=False,
+                    )
+                else:
+                    x = layer(
+                        x, cu, mx, positions,
+                        k_cache=kc, v_cache=vc,
+                        cache_seqlens=cache_seqlens, decode=decode,
+                    )
+        return x
+
+    # ---------------------------------------------------------- state dict
+    def state_dict(self, *args, **kwargs):  # noqa: D102 — canonical keys
+        return {k: v for k, v in self._params.items()}
+
+    def load_state_dict(self, sd: Dict[str, torch.Tensor], strict: bool = True):
+        with torch.no_grad():
+            for k, v in self._params.items():
+                if k in sd:
+                    v.copy_(sd[k].to(v.dtype))
+                elif strict:
+                    raise KeyError(k)
+        return self
+
+    def named_parameters(self, *a, **kw):
+        for k, v in self._params.items():
+            yield k, v
+
+    @property
+    def n_local_params(self):
+        return self.layout.total_numel
+
+    # ------------------------------------------------------------- offload
+    def async_offload(self):
+        """Copy flat params to pinned host memory and free the device
+        buffer (reference: real_llm_api.py:274).  Used for non-trainable
+        roles between their MFCs."""
+        if self._offloaded:
+            return
+        if self._offload_buf is None:
+            self._offload_buf = torch.empty(
+                self.flat_param.shape, dtype=self.dtype, device="cpu",
+                pin_memory=torch.cuda.is_available(),
+            )
+        self._offload_buf.copy_(self.flat_param, non_blocking=False)
+        self.flat_param = None
+        self._params = {}
+        self.layers = nn.ModuleList()
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+        self._offloaded = True
+
+    def reload_from_offload(self):
+        if not self._offloaded:
+            return
+        self.flat_param = self._offload_buf.to(self.device, non_blocking=False)
+        self._map_params()
+        self._build_modules()
+        self._offloaded = False
+
+    # -------------------------------------------------------------- resize
+    def allocate_grad_buffer(self):
+        if self.flat_grad is None or self.flat_grad.numel() != self.flat_param.numel():
+            self.flat_grad = torch.zeros_like(self.flat_param)
+        return self.flat_grad
+
+    def grad_view(self, key: str) -> torch.Tensor:
+        s = self.layout.specs[key]
+        return self.flat_grad[s.start : s.end].view(s.shape)
+
+    def param_view(self, key: str) -> torch.Tensor:
+        return self._params[key]
+
+
+# ---------------------------------------------------------------------------
+def make_real_model(
+    config: ReaLModelConfig,
+    model_name: Optional[str] = None,
+    device=None,
+    dtype=None,
+) -> ReaLModel:
+    """Factory reading TP/PP geometry from the registered grid (when a
+    model scope/grid exists) or building a single-shard model."""
+    if model_name is not None and constants.has_model(model_name):
+        g = constants.grid_of(model_name)
+        return ReaLModel(
+            config, device=device, dtype=dtype,
+            tp_rank=g.tp_rank, tp_size=g.tp_size,
+            pp_rank=g.pp_rank, pp_size=g.pp_size,
+        )
+    return ReaLModel(config, device=device, dtype=dtype)

@@ -1,0 +1,428 @@
+"""Hot ops: HIP kernels with torch reference fallbacks.
+
+HIP kernels (realhf_amd/ops/csrc/, reference op inventory SURVEY.md §2.2):
+  rms_norm           — fused RMSNorm fwd/bwd (reference used TransformerEngine)
+  apply_rotary       — RoPE on packed varlen qk (reference: flash-attn fused rotary)
+  swiglu             — silu(gate)*up fused fwd/bwd
+  attn_varlen        — flash-attention packed prefill (reference: flash_attn_varlen_func)
+  attn_decode        — single-token GQA decode over contiguous KV cache
+                       (reference: flash_attn_with_kvcache)
+  gae                — packed GAE scan (reference: realhf._C.cugae)
+  slice_intervals /
+  set_intervals      — flat-param interval gather/scatter (realhf._C.interval_op_cuda)
+  fused_adamw        — flat-buffer AdamW (reference: apex fused adam via Megatron)
+
+All torch reference paths are fp32-upcast where the reference math is fp32.
+"""
+import math
+from typing import List, Optional, Tuple
+
+import torch
+
+from realhf_amd import ops as _ops
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+def rms_norm_ref(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    xf = x.float()
+    var = xf.pow(2).mean(-1, keepdim=True)
+    out = xf * torch.rsqrt(var + eps)
+    return (out * weight.float()).to(x.dtype)
+
+
+def gemma_rms_norm_ref(x, weight, eps):
+    xf = x.float()
+    var = xf.pow(2).mean(-1, keepdim=True)
+    out = xf * torch.rsqrt(var + eps)
+    return (out * (1.0 + weight.float())).to(x.dtype)
+
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        C = _ops.require_hip()
+        out, rstd = C.rmsnorm_fwd(x, weight, eps)
+        ctx.save_for_backward(x, weight, rstd)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        C = _ops.require_hip()
+        x, weight, rstd = ctx.saved_tensors
+        dx, dw = C.rmsnorm_bwd(grad_out.contiguous(), x, weight, rstd)
+        return dx, dw, None
+
+
+def rms_norm(x, weight, eps: float = 1e-5, gemma_style: bool = False):
+    if _ops.use_hip(x) and not gemma_style:
+        return _RMSNormFn.apply(x.contiguous(), weight, eps)
+    if gemma_style:
+        return gemma_rms_norm_ref(x, weight, eps)
+    return rms_norm_ref(x, weight, eps)
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+class RotaryCache:
+    """Host-precomputed cos/sin tables (Appendix B: never sinf/cosf on
+    device).  Shapes: [max_len, head_dim/2]."""
+
+    def __init__(self):
+        self._cache = {}
+
+    def get(
+        self,
+        head_dim: int,
+        max_len: int,
+        base: float,
+        device,
+        dtype=torch.float32,
+        scaling: Optional[float] = None,
+    ):
+        key = (head_dim, base, str(device), dtype, scaling)
+        cos, sin, cached_len = self._cache.get(key, (None, None, 0))
+        if cached_len < max_len:
+            max_len = max(max_len, 2 * cached_len, 2048)
+            inv_freq = 1.0 / (
+                base ** (torch.arange(0, head_dim, 2, dtype=torch.float64) / head_dim)
+            )
+            t = torch.arange(max_len, dtype=torch.float64)
+            if scaling is not None:
+                t = t / scaling
+            freqs = torch.outer(t, inv_freq)
+            cos = freqs.cos().to(dtype).to(device)
+            sin = freqs.sin().to(dtype).to(device)
+            self._cache[key] = (cos, sin, max_len)
+        cos, sin, _ = self._cache[key]
+        return cos, sin
+
+
+rotary_cache = RotaryCache()
+
+
+def apply_rotary_ref(
+    x: torch.Tensor,  # [total, n_heads, head_dim]
+    cos: torch.Tensor,  # [max_len, head_dim/2]
+    sin: torch.Tensor,
+    positions: torch.Tensor,  # [total] int32/64
+    interleaved: bool = False,
+) -> torch.Tensor:
+    c = cos[positions].unsqueeze(1)  # [total, 1, hd/2]
+    s = sin[positions].unsqueeze(1)
+    xf = x.float()
+    hd = x.shape[-1]
+    if interleaved:
+        x1, x2 = xf[..., 0::2], xf[..., 1::2]
+        o1 = x1 * c - x2 * s
+        o2 = x2 * c + x1 * s
+        out = torch.stack([o1, o2], dim=-1).flatten(-2)
+    else:
+        x1, x2 = xf[..., : hd // 2], xf[..., hd // 2 :]
+        o1 = x1 * c - x2 * s
+        o2 = x2 * c + x1 * s
+        out = torch.cat([o1, o2], dim=-1)
+    return out.to(x.dtype)
+
+
+class _RotaryFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos, sin, positions, interleaved):
+        C = _ops.require_hip()
+        out = C.rope_fwd(x, cos, sin, positions, interleaved, False)
+        ctx.save_for_backward(cos, sin, positions)
+        ctx.interleaved = interleaved
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        C = _ops.require_hip()
+        cos, sin, positions = ctx.saved_tensors
+        dx = C.rope_fwd(grad.contiguous(), cos, sin, positions, ctx.interleaved, True)
+        return dx, None, None, None, None
+
+
+def apply_rotary(x, cos, sin, positions, interleaved=False):
+    if _ops.use_hip(x):
+        return _RotaryFn.apply(x.contiguous(), cos, sin, positions, interleaved)
+    return apply_rotary_ref(x, cos, sin, positions, interleaved)
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU epilogue: silu(gate) * up
+# ---------------------------------------------------------------------------
+def swiglu_ref(gate_up: torch.Tensor) -> torch.Tensor:
+    """gate_up: [tokens, 2*idim] with gate = [:, :idim], up = [:, idim:]."""
+    idim = gate_up.shape[-1] // 2
+    gate, up = gate_up[..., :idim], gate_up[..., idim:]
+    return (torch.nn.functional.silu(gate.float()) * up.float()).to(gate_up.dtype)
+
+
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate_up):
+        C = _ops.require_hip()
+        out = C.swiglu_fwd(gate_up)
+        ctx.save_for_backward(gate_up)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        C = _ops.require_hip()
+        (gate_up,) = ctx.saved_tensors
+        return C.swiglu_bwd(grad.contiguous(), gate_up)
+
+
+def swiglu(gate_up):
+    if _ops.use_hip(gate_up):
+        return _SwiGLUFn.apply(gate_up.contiguous())
+    return swiglu_ref(gate_up)
+
+
+# ---------------------------------------------------------------------------
+# Attention — packed varlen (prefill/training)
+# ---------------------------------------------------------------------------
+def attn_varlen_ref(
+    q: torch.Tensor,  # [total, nq, hd]
+    k: torch.Tensor,  # [total, nkv, hd]
+    v: torch.Tensor,  # [total, nkv, hd]
+    cu_seqlens: torch.Tensor,  # [bs+1] int32
+    causal: bool = True,
+    softmax_scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Per-sequence SDPA in fp32 — the numerics oracle."""
+    nq, hd = q.shape[1], q.shape[2]
+    nkv = k.shape[1]
+    scale = softmax_scale or (1.0 / math.sqrt(hd))
+    out = torch.empty_like(q)
+    rep = nq // nkv
+    cu = cu_seqlens.tolist()
+    for i in range(len(cu) - 1):
+        s, e = cu[i], cu[i + 1]
+        qi = q[s:e].transpose(0, 1).float()  # [nq, L, hd]
+        ki = k[s:e].transpose(0, 1).float()
+        vi = v[s:e].transpose(0, 1).float()
+        if rep > 1:
+            ki = ki.repeat_interleave(rep, dim=0)
+            vi = vi.repeat_interleave(rep, dim=0)
+        scores = torch.matmul(qi, ki.transpose(-1, -2)) * scale
+        L = e - s
+        if causal and L > 1:
+            mask = torch.triu(
+                torch.ones(L, L, dtype=torch.bool, device=q.device), diagonal=1
+            )
+            scores = scores.masked_fill(mask, float("-inf"))
+        probs = torch.softmax(scores, dim=-1)
+        o = torch.matmul(probs, vi)  # [nq, L, hd]
+        out[s:e] = o.transpose(0, 1).to(q.dtype)
+    return out
+
+
+def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale):
+    """Memory-bounded torch implementation on GPU (used for backward
+    recompute until the hand-written HIP backward lands): processes each
+    sequence with SDPA-like math in torch (rocBLAS GEMMs on MFMA)."""
+    return attn_varlen_ref(q, k, v, cu_seqlens, causal, scale)
+
+
+class _AttnVarlenFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, cu_seqlens, max_seqlen, causal, scale):
+        C = _ops.require_hip()
+        out, lse = C.attn_varlen_fwd(q, k, v, cu_seqlens, int(max_seqlen), causal, scale)
+        ctx.save_for_backward(q, k, v, out, lse, cu_seqlens)
+        ctx.causal, ctx.scale, ctx.max_seqlen = causal, scale, max_seqlen
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        q, k, v, out, lse, cu_seqlens = ctx.saved_tensors
+        C = _ops.require_hip()
+        dq, dk, dv = C.attn_varlen_bwd(
+            grad_out.contiguous(), q, k, v, out, lse, cu_seqlens,
+            int(ctx.max_seqlen), ctx.causal, ctx.scale,
+        )
+        return dq, dk, dv, None, None, None, None
+
+
+def attn_varlen(q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=None):
+    scale = softmax_scale or (1.0 / math.sqrt(q.shape[-1]))
+    if _ops.use_hip(q):
+        return _AttnVarlenFn.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(),
+            cu_seqlens, max_seqlen, causal, scale,
+        )
+    return attn_varlen_ref(q, k, v, cu_seqlens, causal, scale)
+
+
+# ---------------------------------------------------------------------------
+# Attention — single-token decode over contiguous KV cache
+# ---------------------------------------------------------------------------
+def attn_decode_ref(
+    q: torch.Tensor,  # [bs, nq, hd] — the new token's q
+    k_cache: torch.Tensor,  # [bs, max_len, nkv, hd]
+    v_cache: torch.Tensor,
+    cache_seqlens: torch.Tensor,  # [bs] int32 — valid length INCLUDING new token
+    softmax_scale: Optional[float] = None,
+) -> torch.Tensor:
+    bs, nq, hd = q.shape
+    nkv = k_cache.shape[2]
+    rep = nq // nkv
+    scale = softmax_scale or (1.0 / math.sqrt(hd))
+    out = torch.empty_like(q)
+    for b in range(bs):
+        L = int(cache_seqlens[b])
+        kb = k_cache[b, :L].transpose(0, 1).float()  # [nkv, L, hd]
+        vb = v_cache[b, :L].transpose(0, 1).float()
+        if rep > 1:
+            kb = kb.repeat_interleave(rep, dim=0)
+            vb = vb.repeat_interleave(rep, dim=0)
+        qb = q[b].unsqueeze(1).float()  # [nq, 1, hd]
+        scores = torch.matmul(qb, kb.transpose(-1, -2)) * scale
+        probs = torch.softmax(scores, dim=-1)
+        out[b] = torch.matmul(probs, vb).squeeze(1).to(q.dtype)
+    return out
+
+
+def attn_decode(q, k_cache, v_cache, cache_seqlens, softmax_scale=None):
+    scale = softmax_scale or (1.0 / math.sqrt(q.shape[-1]))
+    if _ops.use_hip(q):
+        C = _ops.require_hip()
+        return C.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
+    return attn_decode_ref(q, k_cache, v_cache, cache_seqlens, scale)
+
+
+# ---------------------------------------------------------------------------
+# GAE over packed sequences
+# ---------------------------------------------------------------------------
+def gae_ref(
+    rewards: torch.Tensor,  # [total] fp32 — per-token rewards (shifted: len-1 per seq)
+    values: torch.Tensor,  # [total + bs] fp32 — values with one extra bootstrap per seq
+    cu_seqlens: torch.Tensor,  # [bs+1] — over the REWARD lengths
+    bootstrap: torch.Tensor,  # [bs] bool — whether the last value bootstraps
+    gamma: float,
+    lam: float,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """advantages [total], returns [total].  Mirrors cugae1d_nolp_misalign
+    (reference csrc/cugae/gae.cu:10): values for sequence i live at
+    [cu[i]+i, cu[i+1]+i+1) — one longer than rewards."""
+    adv = torch.zeros_like(rewards)
+    ret = torch.zeros_like(rewards)
+    cu = cu_seqlens.tolist()
+    bs = len(cu) - 1
+    for i in range(bs):
+        rs, re = cu[i], cu[i + 1]
+        vs = rs + i
+        lastgae = 0.0
+        L = re - rs
+        for t in range(L - 1, -1, -1):
+            nex = values[vs + t + 1]
+            if t == L - 1 and not bool(bootstrap[i]):
+                nex = values.new_zeros(())
+            delta = rewards[rs + t] + gamma * nex - values[vs + t]
+            lastgae = delta + gamma * lam * lastgae
+            adv[rs + t] = lastgae
+            ret[rs + t] = lastgae + values[vs + t]
+    return adv, ret
+
+
+def gae(rewards, values, cu_seqlens, bootstrap, gamma, lam):
+    if _ops.use_hip(rewards):
+        C = _ops.require_hip()
+        return C.gae_1d(rewards, values, cu_seqlens, bootstrap, float(gamma), float(lam))
+    return gae_ref(rewards, values, cu_seqlens, bootstrap, gamma, lam)
+
+
+# ---------------------------------------------------------------------------
+# Flat-parameter interval gather/scatter
+# ---------------------------------------------------------------------------
+def slice_intervals_ref(src: torch.Tensor, intervals: torch.Tensor) -> torch.Tensor:
+    outs = [src[s:e] for s, e in intervals.tolist()]
+    return torch.cat(outs) if outs else src.new_empty(0)
+
+
+def set_intervals_ref(src: torch.Tensor, dst: torch.Tensor, intervals: torch.Tensor):
+    off = 0
+    for s, e in intervals.tolist():
+        n = e - s
+        dst[s:e] = src[off : off + n]
+        off += n
+    return dst
+
+
+def slice_intervals(src, intervals):
+    if _ops.use_hip(src) and intervals.shape[0] >= 128:
+        C = _ops.require_hip()
+        return C.slice_intervals(src, intervals.to(src.device))
+    return slice_intervals_ref(src, intervals)
+
+
+def set_intervals(src, dst, intervals):
+    if _ops.use_hip(dst) and intervals.shape[0] >= 128:
+        C = _ops.require_hip()
+        C.set_intervals(src, dst, intervals.to(dst.device))
+        return dst
+    return set_intervals_ref(src, dst, intervals)
+
+
+def merge_intervals(intervals: List[Tuple[int, int]]) -> List[Tuple[int, int]]:
+    """Coalesce adjacent/overlapping [a,b) pairs (reference:
+    csrc/interval_op/interval_op.cpp merge_intervals)."""
+    if not intervals:
+        return []
+    intervals = sorted(intervals)
+    out = [list(intervals[0])]
+    for s, e in intervals[1:]:
+        if s <= out[-1][1]:
+            out[-1][1] = max(out[-1][1], e)
+        else:
+            out.append([s, e])
+    return [tuple(x) for x in out]
+
+
+# ---------------------------------------------------------------------------
+# Fused AdamW on flat buffers
+# ---------------------------------------------------------------------------
+def fused_adamw_ref(
+    param_f32: torch.Tensor,
+    grad: torch.Tensor,
+    exp_avg: torch.Tensor,
+    exp_avg_sq: torch.Tensor,
+    lr: float,
+    beta1: float,
+    beta2: float,
+    eps: float,
+    weight_decay: float,
+    step: int,
+    bf16_out: Optional[torch.Tensor] = None,
+):
+    g = grad.float()
+    exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    bc1 = 1 - beta1**step
+    bc2 = 1 - beta2**step
+    denom = (exp_avg_sq / bc2).sqrt_().add_(eps)
+    param_f32.mul_(1 - lr * weight_decay)
+    param_f32.addcdiv_(exp_avg, denom, value=-lr / bc1)
+    if bf16_out is not None:
+        bf16_out.copy_(param_f32)
+    return param_f32
+
+
+def fused_adamw(param_f32, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps,
+                weight_decay, step, bf16_out=None):
+    if _ops.use_hip(param_f32):
+        C = _ops.require_hip()
+        C.fused_adamw(
+            param_f32, grad, exp_avg, exp_avg_sq,
+            bf16_out if bf16_out is not None else param_f32.new_empty(0).to(torch.bfloat16),
+            float(lr), float(beta1), float(beta2), float(eps),
+            float(weight_decay), int(step), bf16_out is not None,
+        )
+        return param_f32
+    return fused_adamw_ref(
+        param_f32, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps,
+        weight_decay, step, bf16_out,
+    )
