@@ -1,0 +1,217 @@
+"""Packed-batch generation engine.
+
+Reference semantics: realhf/impl/model/nn/real_llm_generate.py (generate:252,
+genstep:26, prepare_generate_inputs:144, maybe_capture_cudagraph:214).
+
+Design (MI355X-first):
+- prefill runs the packed varlen forward once, writing the contiguous
+  per-layer KV caches [bs, prompt_max + max_new, nkv_local, hd];
+- decode is one fixed-shape forward per token (bs tokens), hipGraph-captured
+  after the first step (launch-bound otherwise: ~32 small kernels/layer);
+- sampling: temperature → top-k → top-p → Categorical, identical across TP
+  ranks via a shared-seed torch.Generator (no sampling collective — the
+  reference all-reduces sampled tokens instead, an extra hop per token that
+  a shared RNG makes unnecessary);
+- termination: all sequences hit EOS or max_new_tokens; min_new_tokens
+  masks EOS early.
+"""
+import dataclasses
+from typing import List, Optional, Tuple
+
+import torch
+
+from realhf_amd.api.model import GenerationHyperparameters
+from realhf_amd.base import logging
+from realhf_amd.models.real_model import ReaLModel
+from realhf_amd.utils.functional import top_k_top_p_logits
+
+logger = logging.getLogger("generate")
+
+
+@dataclasses.dataclass
+class GenerationOutput:
+    gen_tokens: torch.Tensor  # [bs, gen_len] (right-padded with pad_id)
+    gen_logprobs: torch.Tensor  # [bs, gen_len] — logprob of each sampled token
+    gen_lengths: torch.Tensor  # [bs] — actual generated length (incl. eos)
+    no_eos_mask: torch.Tensor  # [bs] bool — True if never hit EOS
+
+
+def _sample_from_logits(
+    logits: torch.Tensor,  # [bs, vocab] fp32
+    gconfig: GenerationHyperparameters,
+    generator: Optional[torch.Generator],
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if gconfig.temperature != 1.0 and not gconfig.greedy:
+        logits = logits / max(gconfig.temperature, 1e-5)
+    if gconfig.greedy:
+        tokens = logits.argmax(dim=-1)
+        logp = torch.log_softmax(logits, dim=-1)
+        return tokens, logp.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
+    filtered = top_k_top_p_logits(logits, gconfig.top_k, gconfig.top_p)
+    probs = torch.softmax(filtered, dim=-1)
+    tokens = torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+    logp_all = torch.log_softmax(logits, dim=-1)  # logprob under UNFILTERED dist
+    return tokens, logp_all.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
+
+
+class DecodeGraph:
+    """hipGraph capture of the decode-step forward.
+
+    Static buffers: input token ids [bs], cache_seqlens [bs], output
+    logits.  The KV caches are already static.  Replay = copy tokens in,
+    bump cache_seqlens, one graph launch.
+    """
+
+    def __init__(self, model: ReaLModel, kv_caches, bs: int):
+        self.model = model
+        self.kv_caches = kv_caches
+        self.bs = bs
+        self.graph: Optional[torch.cuda.CUDAGraph] = None
+        self.in_tokens = None
+        self.in_cache_seqlens = None
+        self.out = None
+
+    def _eager(self, tokens, cache_seqlens):
+        return self.model(
+            packed_input_ids=tokens,
+            kv_caches=self.kv_caches,
+            cache_seqlens=cache_seqlens,
+            decode=True,
+        )
+
+    def capture(self, tokens, cache_seqlens):
+        self.in_tokens = tokens.clone()
+        self.in_cache_seqlens = cache_seqlens.clone()
+        torch.cuda.synchronize()
+        # warmup on a side stream (required before capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._eager(self.in_tokens, self.in_cache_seqlens)
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.out = self._eager(self.in_tokens, self.in_cache_seqlens)
+        logger.debug("decode hipGraph captured (bs=%d)", self.bs)
+
+    def step(self, tokens, cache_seqlens):
+        if self.graph is None:
+            return self._eager(tokens, cache_seqlens)
+        self.in_tokens.copy_(tokens)
+        self.in_cache_seqlens.copy_(cache_seqlens)
+        self.graph.replay()
+        return self.out
+
+    def destroy(self):
+        self.graph = None
+
+
+@torch.no_grad()
+def generate(
+    model: ReaLModel,
+    packed_prompts: torch.Tensor,  # [total_prompt] int64
+    cu_seqlens: torch.Tensor,  # [bs+1] int32
+    gconfig: GenerationHyperparameters,
+    eos_token_id: Optional[int] = None,
+    pad_token_id: int = 0,
+    generator: Optional[torch.Generator] = None,
+) -> GenerationOutput:
+    assert model.pp_size == 1, "pp>1 generation goes through the pipe engine"
+    cfg = model.config
+    device = packed_prompts.device
+    bs = cu_seqlens.shape[0] - 1
+    prompt_lens = (cu_seqlens[1:] - cu_seqlens[:-1]).to(device)
+    max_prompt = int(prompt_lens.max())
+    max_new = gconfig.max_new_tokens
+    cache_len = max_prompt + max_new
+
+    nkv_local = max(cfg.n_kv_heads // model.tp_size, 1)
+    n_blocks = sum(1 for i in model.layer_indices if 1 <= i <= cfg.n_layers)
+    kv_dtype = model.dtype
+    kv_caches = [
+        (
+            torch.zeros(bs, cache_len, nkv_local, cfg.head_dim, dtype=kv_dtype, device=device),
+            torch.zeros(bs, cache_len, nkv_local, cfg.head_dim, dtype=kv_dtype, device=device),
+        )
+        for _ in range(n_blocks)
+    ]
+
+    # ---- prefill --------------------------------------------------------
+    logits = model(
+        packed_input_ids=packed_prompts,
+        cu_seqlens=cu_seqlens,
+        max_seqlen=max_prompt,
+        kv_caches=kv_caches,
+    )
+    from realhf_amd.parallel import mappings
+
+    last_idx = (cu_seqlens[1:].long() - 1)
+    last_logits = logits[last_idx]  # [bs, vocab/tp]
+    last_logits = mappings.gather_from_tp_region(last_logits).float()
+
+    gen_tokens = torch.full((bs, max_new), pad_token_id, dtype=torch.long, device=device)
+    gen_logprobs = torch.zeros(bs, max_new, dtype=torch.float32, device=device)
+    done = torch.zeros(bs, dtype=torch.bool, device=device)
+    gen_lengths = torch.zeros(bs, dtype=torch.long, device=device)
+
+    cache_seqlens = prompt_lens.to(torch.int32).clone()
+
+    decoder = DecodeGraph(model, kv_caches, bs)
+    use_graph = gconfig.use_hip_graph and device.type == "cuda"
+
+    cur_logits = last_logits
+    for t in range(max_new):
+        if eos_token_id is not None and t < gconfig.min_new_tokens:
+            cur_logits[:, eos_token_id] = float("-inf")
+        tokens, logp = _sample_from_logits(cur_logits, gconfig, generator)
+        tokens = torch.where(done, torch.full_like(tokens, pad_token_id), tokens)
+        gen_tokens[:, t] = tokens
+        gen_logprobs[:, t] = torch.where(done, torch.zeros_like(logp), logp)
+        gen_lengths += (~done).long()
+        if eos_token_id is not None:
+            done = done | (tokens == eos_token_id)
+        if bool(done.all()) or t == max_new - 1:
+            break
+
+        cache_seqlens += 1  # the new token's slot
+        if use_graph and decoder.graph is None:
+            decoder.capture(tokens, cache_seqlens)
+        logits_step = decoder.step(tokens, cache_seqlens)
+        logits_step = mappings.gather_from_tp_region(logits_step).float()
+        cur_logits = logits_step
+
+    decoder.destroy()
+    return GenerationOutput(
+        gen_tokens=gen_tokens[:, : int(gen_lengths.max())],
+        gen_logprobs=gen_logprobs[:, : int(gen_lengths.max())],
+        gen_lengths=gen_lengths,
+        no_eos_mask=~done,
+    )
+
+
+def concat_prompt_to_generation_output(
+    packed_prompts: torch.Tensor,
+    prompt_cu_seqlens: torch.Tensor,
+    out: GenerationOutput,
+) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Pack prompt+generation into one packed batch.
+
+    Returns (packed_input_ids, cu_seqlens, prompt_mask) — reference:
+    real_llm_generate.py:451."""
+    bs = prompt_cu_seqlens.shape[0] - 1
+    seqs, masks = [], []
+    for i in range(bs):
+        s, e = int(prompt_cu_seqlens[i]), int(prompt_cu_seqlens[i + 1])
+        gl = int(out.gen_lengths[i])
+        seq = torch.cat([packed_prompts[s:e], out.gen_tokens[i, :gl]])
+        seqs.append(seq)
+        m = torch.zeros(seq.shape[0], dtype=torch.bool, device=seq.device)
+        m[: e - s] = True
+        masks.append(m)
+    lens = [int(x.shape[0]) for x in seqs]
+    cum = [0]
+    for l in lens:
+        cum.append(cum[-1] + l)
+    cu = torch.tensor(cum, dtype=torch.int32, device=packed_prompts.device)
+    return torch.cat(seqs), cu, torch.cat(masks)
