@@ -34,6 +34,7 @@ class GenerationOutput:
     gen_logprobs: torch.Tensor  # [bs, gen_len] — logprob of each sampled token
     gen_lengths: torch.Tensor  # [bs] — actual generated length (incl. eos)
     no_eos_mask: torch.Tensor  # [bs] bool — True if never hit EOS
+    prompt_logprobs: Optional[torch.Tensor] = None  # [total_prompt - bs] packed
 
 
 def _sample_from_logits(
@@ -116,6 +117,7 @@ def generate(
     eos_token_id: Optional[int] = None,
     pad_token_id: int = 0,
     generator: Optional[torch.Generator] = None,
+    return_prompt_logprobs: bool = False,
 ) -> GenerationOutput:
     assert model.pp_size == 1, "pp>1 generation goes through the pipe engine"
     cfg = model.config
@@ -145,6 +147,12 @@ def generate(
         kv_caches=kv_caches,
     )
     from realhf_amd.parallel import mappings
+
+    prompt_logprobs = None
+    if return_prompt_logprobs:
+        from realhf_amd.parallel.tp import packed_shifted_logprobs
+
+        prompt_logprobs = packed_shifted_logprobs(logits, cu_seqlens, packed_prompts)
 
     last_idx = (cu_seqlens[1:].long() - 1)
     last_logits = logits[last_idx]  # [bs, vocab/tp]
@@ -187,6 +195,7 @@ def generate(
         gen_logprobs=gen_logprobs[:, : int(gen_lengths.max())],
         gen_lengths=gen_lengths,
         no_eos_mask=~done,
+        prompt_logprobs=prompt_logprobs,
     )
 
 
