@@ -1,0 +1,255 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: PPO samples/sec, 4 x LLaMA-7B (actor/critic/ref/RM).
+
+Implements the BASELINE.json metric: the reference config is
+examples/scripts/local/ppo.sh (prompt len 128, train_bs_n_seqs 128 on 8
+GPUs -> 16 seqs/GPU weak scaling, gen len fixed 512, top_p 0.9 / top_k
+1000, 4 PPO minibatches), synthetic prompts, random-init weights, bf16.
+
+Run:  python bench.py [--gpus N] [--steps K] [--warmup W]
+Multi-GPU (driver):  python -m torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+One process per GPU over RCCL; per-model data parallelism (weak scaling:
+per-GPU batch fixed at 16 sequences).
+"""
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+def log(msg):
+    r = int(os.environ.get("RANK", "0"))
+    print(f"[bench rank{r}] {msg}", flush=True, file=__import__("sys").stderr)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--seqs-per-gpu", type=int, default=16)
+    p.add_argument("--prompt-len", type=int, default=128)
+    p.add_argument("--gen-len", type=int, default=512)
+    p.add_argument("--n-minibatches", type=int, default=4)
+    p.add_argument("--model", type=str, default="llama-7b",
+                   help="llama-7b | llama-small (debug)")
+    p.add_argument("--no-hip-graph", action="store_true")
+    p.add_argument("--offload-frozen", action="store_true", default=True)
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    assert world == args.gpus or world == 1, (world, args.gpus)
+
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:  # CPU debug path (--model llama-small); the driver always has GPUs
+        device = torch.device("cpu")
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl" if use_cuda else "gloo",
+                                rank=rank, world_size=world)
+
+    import realhf_amd.interfaces  # noqa: F401
+    import realhf_amd.models.hf  # noqa: F401
+    import realhf_amd.runtime.engine  # noqa: F401
+    from realhf_amd.api.config import Abstraction, ModelName
+    from realhf_amd.api.data import SequenceSample
+    from realhf_amd.api.model import (
+        FinetuneSpec,
+        Model,
+        make_backend,
+        make_interface,
+    )
+    from realhf_amd.base import constants, seeding
+    from realhf_amd.base.topology import ParallelGrid, PipeDataTensorTopology
+    from realhf_amd.models.hf.llama import llama7b_config, make_test_config
+    from realhf_amd.models.real_model import ReaLModel
+
+    seeding.set_random_seed(1234, rank_offset=rank)
+
+    def mkcfg(is_critic):
+        if args.model == "llama-7b":
+            return llama7b_config(is_critic=is_critic)
+        cfg = make_test_config(
+            n_layers=4, hidden_dim=512, n_heads=8, n_kv_heads=8,
+            head_dim=64, intermediate_dim=1408, vocab_size=32000,
+            max_position_embeddings=2048, is_critic=is_critic,
+        )
+        cfg.family = "llama"
+        return cfg
+
+    # grids: every model spans all ranks, pure DP (round-1 allocation)
+    names = ["actor", "critic", "ref", "rew"]
+    if world > 1:
+        for n in names:
+            topo = PipeDataTensorTopology(num_pp=1, num_dp=world, num_tp=1)
+            constants.set_grid(n, ParallelGrid(topo))
+
+    def scope(n):
+        import contextlib
+
+        if constants.has_model(n):
+            return constants.model_scope(n)
+        return contextlib.nullcontext()
+
+    log("building models (4x %s, bf16, random init)..." % args.model)
+    t0 = time.time()
+    models = {}
+    for n in names:
+        is_critic = n in ("critic", "rew")
+        cfg = mkcfg(is_critic)
+        with scope(n):
+            m = ReaLModel(cfg, device=device,
+                          dtype=torch.bfloat16 if use_cuda else torch.float32)
+            m.random_init()
+            model = Model(
+                name=ModelName(n, 0), module=m, tokenizer=None, device=device,
+                dtype=torch.bfloat16,
+            )
+            if n in ("actor", "critic"):
+                backend = make_backend(
+                    Abstraction("zero1", {"optimizer": {"lr": 1e-6,
+                                                        "warmup_steps_proportion": 0.0}})
+                )
+            else:
+                backend = make_backend(Abstraction("inference"))
+            models[n] = backend.initialize(model, FinetuneSpec(1, 512, 128))
+        if use_cuda:
+            torch.cuda.synchronize()
+    mem = torch.cuda.memory_allocated() / 2**30 if use_cuda else 0.0
+    log(f"models built in {time.time() - t0:.1f}s; mem={mem:.1f} GiB")
+
+    gconfig = dict(
+        max_new_tokens=args.gen_len, min_new_tokens=args.gen_len,
+        greedy=False, top_k=1000, top_p=0.9, temperature=1.0,
+        use_hip_graph=not args.no_hip_graph and torch.cuda.is_available(),
+    )
+    actor_iface = make_interface(
+        Abstraction("ppo_actor", {
+            "n_minibatches": args.n_minibatches, "gconfig": gconfig,
+            "kl_ctl": 0.1, "adv_norm": True,
+        })
+    )
+    critic_iface = make_interface(
+        Abstraction("ppo_critic", {"n_minibatches": args.n_minibatches})
+    )
+    rew_iface = make_interface(Abstraction("paired_rw"))
+
+    vocab = models["actor"].module.module.config.vocab_size
+    rng = np.random.RandomState(1000 + rank)
+
+    def make_prompt_batch():
+        bs = args.seqs_per_gpu
+        toks = torch.from_numpy(
+            rng.randint(10, vocab - 10, size=bs * args.prompt_len)
+        ).long().to(device)
+        return SequenceSample(
+            keys=("packed_prompts",),
+            ids=[f"r{rank}-{i}" for i in range(bs)],
+            seqlens={"packed_prompts": [[args.prompt_len]] * bs},
+            data={"packed_prompts": toks},
+        )
+
+    frozen = ("ref", "rew")
+
+    def ppo_step():
+        batch = make_prompt_batch()
+        with scope("actor"):
+            rollout = actor_iface.generate(models["actor"], batch)
+        sample = rollout
+        seq_only = sample.select_keys(["packed_input_ids"])
+        with scope("rew"):
+            models["rew"].module.model.reload_from_offload()
+            sample.update_(rew_iface.inference(models["rew"], seq_only))
+            if args.offload_frozen:
+                models["rew"].module.model.async_offload()
+        with scope("ref"):
+            models["ref"].module.model.reload_from_offload()
+            sample.update_(actor_iface.inference(models["ref"], seq_only))
+            if args.offload_frozen:
+                models["ref"].module.model.async_offload()
+        with scope("critic"):
+            sample.update_(critic_iface.inference(models["critic"], seq_only))
+        with scope("actor"):
+            astats = actor_iface.train_step(models["actor"], sample)
+        with scope("critic"):
+            cstats = critic_iface.train_step(models["critic"], sample)
+        return astats, cstats
+
+    if args.offload_frozen:
+        for n in frozen:
+            models[n].module.model.async_offload()
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    log(f"warmup x{args.warmup}...")
+    for i in range(args.warmup):
+        ppo_step()
+        m2 = torch.cuda.max_memory_allocated() / 2**30 if use_cuda else 0.0
+        log(f"warmup step {i} done; peak mem={m2:.1f} GiB")
+    barrier_sync()
+
+    t0 = time.time()
+    for i in range(args.steps):
+        astats, cstats = ppo_step()
+        log(f"step {i}: actor_loss={astats.get('actor_loss', 0):.4f} "
+            f"kl={astats.get('kl', 0):.4f}")
+    barrier_sync()
+    elapsed = time.time() - t0
+
+    # MAX elapsed over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t)
+
+    n_gpus = world
+    total_samples = args.seqs_per_gpu * n_gpus * args.steps
+    samples_per_sec = total_samples / elapsed
+    ms_per_step = elapsed / args.steps * 1000
+
+    if rank == 0:
+        result = {
+            "metric": "PPO samples/sec (whole node), 4xLLaMA-7B actor/critic/ref/RM",
+            "value": samples_per_sec,
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "4x" + args.model,
+                "global_batch": args.seqs_per_gpu * n_gpus,
+                "seq_len": args.prompt_len + args.gen_len,
+                "prompt_len": args.prompt_len,
+                "gen_len": args.gen_len,
+                "parallelism": f"dp{n_gpus}",
+                "ppo_n_minibatches": args.n_minibatches,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

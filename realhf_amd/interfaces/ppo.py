@@ -211,11 +211,8 @@ class PPOActorInterface(ModelInterface):
         train_sample.update_(extra)
 
         all_stats: Dict[str, float] = {}
-        mbs = (
-            train_sample.split(self.n_minibatches)
-            if self.n_minibatches > 1
-            else [train_sample]
-        )
+        n_mb = min(self.n_minibatches, train_sample.bs)
+        mbs = train_sample.split(n_mb) if n_mb > 1 else [train_sample]
         for mb in mbs:
             stats = model.module.train_batch(
                 mb, self._loss_fn, version_steps=model.version.global_step,
@@ -332,11 +329,8 @@ class PPOCriticInterface(ModelInterface):
         train_sample.update_(extra)
 
         all_stats: Dict[str, float] = {}
-        mbs = (
-            train_sample.split(self.n_minibatches)
-            if self.n_minibatches > 1
-            else [train_sample]
-        )
+        n_mb = min(self.n_minibatches, train_sample.bs)
+        mbs = train_sample.split(n_mb) if n_mb > 1 else [train_sample]
         for mb in mbs:
             stats = model.module.train_batch(
                 mb, self._loss_fn, version_steps=model.version.global_step,

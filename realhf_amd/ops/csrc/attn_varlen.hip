@@ -1,0 +1,279 @@
+// Flash-attention packed-varlen causal forward for gfx950 (MFMA).
+// Replaces flash_attn_varlen_func in the reference training/prefill path
+// (reference: realhf/impl/model/modules/attn.py:255).
+//
+// Structure (correctness-first instance of the guide's §B prefill recipe):
+//   grid = (q_block, q_head); workgroup = 256 threads = 4 waves.
+//   Each workgroup owns 64 query rows of one sequence+head (wave w: 16
+//   rows).  KV tiles of 64 keys are staged cooperatively in LDS — K
+//   row-major [64][136] (8-elem pad), V TRANSPOSED [128][72] so the PV
+//   B-fragment is a contiguous ds_read_b128.  Online softmax state (m, l)
+//   is held per C-row in registers, reduced with shfl_xor over the 16-lane
+//   column groups.
+//
+// MFMA: v_mfma_f32_16x16x32_bf16.  Lane maps (guide §3):
+//   A[i][k]:  i = lane&15, k = (lane>>4)*8 + j   (j = 0..7, 4 VGPRs bf16)
+//   B[k][n]:  n = lane&15, k = (lane>>4)*8 + j
+//   C/D[i][j]: j = lane&15, i = (lane>>4)*4 + r  (r = 0..3, fp32)
+// `mfma_probe` below verifies this mapping numerically on device.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+#define QBLK 64      // queries per workgroup
+#define QW 16        // queries per wave
+#define KVBLK 64     // keys per LDS tile
+#define HDMAX 128
+#define KPAD 8       // K tile row pad (bf16 elems)
+#define VPAD 8       // vT tile row pad
+
+DEVINL float group16_max(float x) {
+  // reduce over the 16 lanes that share lane>>4
+  x = fmaxf(x, __shfl_xor(x, 1, 64));
+  x = fmaxf(x, __shfl_xor(x, 2, 64));
+  x = fmaxf(x, __shfl_xor(x, 4, 64));
+  x = fmaxf(x, __shfl_xor(x, 8, 64));
+  return x;
+}
+DEVINL float group16_sum(float x) {
+  x += __shfl_xor(x, 1, 64);
+  x += __shfl_xor(x, 2, 64);
+  x += __shfl_xor(x, 4, 64);
+  x += __shfl_xor(x, 8, 64);
+  return x;
+}
+
+template <int HD>
+__global__ __launch_bounds__(256, 2) void attn_varlen_fwd_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const int* __restrict__ cu_seqlens,
+    const int* __restrict__ blk_seq, const int* __restrict__ blk_qstart,
+    bf16* __restrict__ out, float* __restrict__ lse,
+    int nq, int nkv, float scale, bool causal) {
+  constexpr int HDCH = HD / 32;  // mfma K-chunks over head_dim
+  const int blk = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int kvh = qh / (nq / nkv);
+  const int seq = blk_seq[blk];
+  const int q0_local = blk_qstart[blk];  // local query start in seq
+  const int s0 = cu_seqlens[seq], s1 = cu_seqlens[seq + 1];
+  const int L = s1 - s0;
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int i16 = lane & 15;   // A-row / B-col / C-col index
+  const int g = lane >> 4;     // k-chunk group / C row group
+
+  __shared__ __bf16 k_s[KVBLK][HD + KPAD];
+  __shared__ __bf16 vt_s[HD][KVBLK + VPAD];
+  __shared__ __bf16 p_s[4][QW][KVBLK + VPAD];
+
+  // ---- load Q fragments (registers, whole kernel) --------------------
+  // wave w owns query rows qrow_local = q0_local + w*QW + i16 (A layout)
+  bf16x8 qfrag[HDCH];
+  const int my_qrow_a = q0_local + w * QW + i16;  // A-layout row
+  {
+    const bf16* qrow = q + ((long)(s0 + min(my_qrow_a, L - 1)) * nq + qh) * HD;
+    #pragma unroll
+    for (int c = 0; c < HDCH; c++)
+      qfrag[c] = *(const bf16x8*)(qrow + c * 32 + g * 8);
+  }
+
+  // online state for the 4 C-rows this lane touches
+  float m_r[4], l_r[4];
+  f32x4 o_acc[HD / 16];
+  #pragma unroll
+  for (int r = 0; r < 4; r++) { m_r[r] = -1e30f; l_r[r] = 0.f; }
+  #pragma unroll
+  for (int t = 0; t < HD / 16; t++) o_acc[t] = {0.f, 0.f, 0.f, 0.f};
+
+  // causal: keys needed up to q0_local + QBLK - 1 (inclusive); else all L
+  const int kv_end = causal ? min(L, q0_local + QBLK) : L;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
+    const int kchunk = min(KVBLK, kv_end - kv0);
+    // ---- stage K and V(T) tiles ------------------------------------
+    __syncthreads();
+    // K: 64 rows x HD; thread strides rows. 256 threads, each loads 16B
+    for (int idx = threadIdx.x; idx < KVBLK * (HD / 8); idx += 256) {
+      int row = idx / (HD / 8);
+      int col8 = (idx % (HD / 8)) * 8;
+      bf16x8 val = {};
+      if (row < kchunk)
+        val = *(const bf16x8*)(k + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
+      *(bf16x8*)(&k_s[row][col8]) = val;
+      // V transposed: vt_s[col][row]
+      bf16x8 vv = {};
+      if (row < kchunk)
+        vv = *(const bf16x8*)(v + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
+      #pragma unroll
+      for (int j = 0; j < 8; j++) vt_s[col8 + j][row] = vv[j];
+    }
+    __syncthreads();
+
+    // ---- S = Q K^T over 4 key subtiles ------------------------------
+    f32x4 s_sub[4];
+    #pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int c = 0; c < HDCH; c++) {
+        // B fragment: K[key = ks*16 + i16][c*32 + g*8 + j]
+        bf16x8 bfrag = *(const bf16x8*)(&k_s[ks * 16 + i16][c * 32 + g * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfrag, acc, 0, 0, 0);
+      }
+      s_sub[ks] = acc;
+    }
+
+    // ---- softmax over the 64-key tile -------------------------------
+    float tile_max[4];
+    #pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float mx = -1e30f;
+      const int qrow = q0_local + w * QW + g * 4 + r;  // C-layout row
+      #pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        int kidx = kv0 + ks * 16 + i16;
+        bool ok = (kidx < kv_end) && (!causal || kidx <= qrow) && (qrow < L);
+        float sv = ok ? s_sub[ks][r] * scale : -1e30f;
+        s_sub[ks][r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+      tile_max[r] = group16_max(mx);
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float nm = fmaxf(m_r[r], tile_max[r]);
+      if (nm < -1e29f) nm = 0.f;  // fully-masked row guard
+      float f = __expf(m_r[r] - nm);
+      if (m_r[r] < -1e29f) f = 0.f;
+      m_r[r] = nm;
+      l_r[r] *= f;
+      #pragma unroll
+      for (int t = 0; t < HD / 16; t++) o_acc[t][r] *= f;
+      float rowsum = 0.f;
+      #pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        float p = (s_sub[ks][r] > -1e29f) ? __expf(s_sub[ks][r] - nm) : 0.f;
+        s_sub[ks][r] = p;
+        rowsum += p;
+      }
+      l_r[r] += group16_sum(rowsum);
+    }
+    // ---- P -> LDS (bf16, C layout -> A-readable) --------------------
+    #pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      #pragma unroll
+      for (int r = 0; r < 4; r++)
+        p_s[w][g * 4 + r][ks * 16 + i16] = (__bf16)s_sub[ks][r];
+    }
+    __syncthreads();
+
+    // ---- O += P V ----------------------------------------------------
+    #pragma unroll
+    for (int kk = 0; kk < 2; kk++) {  // two K=32 chunks over 64 keys
+      // A fragment: P[qrow = i16][kk*32 + g*8 + j]
+      bf16x8 pa = *(const bf16x8*)(&p_s[w][i16][kk * 32 + g * 8]);
+      #pragma unroll
+      for (int t = 0; t < HD / 16; t++) {
+        // B fragment: vT[n = t*16 + i16 (hd)][kk*32 + g*8 + j (key)]
+        bf16x8 vb = *(const bf16x8*)(&vt_s[t * 16 + i16][kk * 32 + g * 8]);
+        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue ------------------------------------------------------
+  #pragma unroll
+  for (int r = 0; r < 4; r++) {
+    const int qrow = q0_local + w * QW + g * 4 + r;
+    if (qrow >= L) continue;
+    float inv = (l_r[r] > 0.f) ? 1.f / l_r[r] : 0.f;
+    bf16* orow = out + ((long)(s0 + qrow) * nq + qh) * HD;
+    #pragma unroll
+    for (int t = 0; t < HD / 16; t++)
+      orow[t * 16 + i16] = __float2bfloat16(o_acc[t][r] * inv);
+    if (lse && i16 == 0)
+      lse[(long)(s0 + qrow) * nq + qh] = m_r[r] + __logf(fmaxf(l_r[r], 1e-30f));
+  }
+}
+
+std::vector<torch::Tensor> attn_varlen_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
+              "attn_varlen_fwd: bf16 only");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  int total = q.size(0), nq = q.size(1), hd = q.size(2);
+  int nkv = k.size(1);
+  TORCH_CHECK(nq % nkv == 0);
+  auto cu_cpu = cu_seqlens.to(torch::kInt).cpu();
+  auto cu_dev = cu_seqlens.to(torch::kInt).to(q.device());
+  int bs = cu_cpu.numel() - 1;
+  const int* cu = cu_cpu.data_ptr<int>();
+  std::vector<int> bseq, bqs;
+  for (int i = 0; i < bs; i++) {
+    int L = cu[i + 1] - cu[i];
+    for (int qs = 0; qs < L; qs += QBLK) {
+      bseq.push_back(i);
+      bqs.push_back(qs);
+    }
+  }
+  auto opts = torch::TensorOptions().dtype(torch::kInt).device(q.device());
+  auto bseq_t = torch::from_blob(bseq.data(), {(long)bseq.size()},
+                                 torch::kInt).to(q.device());
+  auto bqs_t = torch::from_blob(bqs.data(), {(long)bqs.size()},
+                                torch::kInt).to(q.device());
+  auto out = torch::empty_like(q);
+  auto lse = torch::empty({total, nq}, q.options().dtype(torch::kFloat));
+  dim3 grid((unsigned)bseq.size(), nq);
+  if (hd == 128) {
+    hipLaunchKernelGGL((attn_varlen_fwd_kernel<128>), grid, dim3(256), 0,
+      cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+      (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
+      bseq_t.data_ptr<int>(), bqs_t.data_ptr<int>(), (bf16*)out.data_ptr(),
+      lse.data_ptr<float>(), nq, nkv, (float)scale, causal);
+  } else if (hd == 64) {
+    hipLaunchKernelGGL((attn_varlen_fwd_kernel<64>), grid, dim3(256), 0,
+      cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+      (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
+      bseq_t.data_ptr<int>(), bqs_t.data_ptr<int>(), (bf16*)out.data_ptr(),
+      lse.data_ptr<float>(), nq, nkv, (float)scale, causal);
+  } else {
+    TORCH_CHECK(false, "unsupported head_dim ", hd);
+  }
+  CHECK_CUDA_OK();
+  return {out, lse};
+}
+
+// ---------------------------------------------------------------------------
+// mfma_probe: D = A(16x32) @ B(32x16) with the assumed lane mapping —
+// numerics check for the fragment layout (guide §3 "A=I-check with
+// ASYMMETRIC B").
+// ---------------------------------------------------------------------------
+__global__ void mfma_probe_kernel(const bf16* A, const bf16* B, float* D) {
+  int lane = threadIdx.x & 63;
+  int i16 = lane & 15, g = lane >> 4;
+  bf16x8 a, b;
+  #pragma unroll
+  for (int j = 0; j < 8; j++) {
+    a[j] = (__bf16)A[i16 * 32 + g * 8 + j];      // A[i][k] row-major
+    b[j] = (__bf16)B[(g * 8 + j) * 16 + i16];    // B[k][n] row-major
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  #pragma unroll
+  for (int r = 0; r < 4; r++) D[(g * 4 + r) * 16 + i16] = c[r];
+}
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}));
+  TORCH_CHECK(B.sizes() == torch::IntArrayRef({32, 16}));
+  auto Ac = A.to(torch::kBFloat16).contiguous();
+  auto Bc = B.to(torch::kBFloat16).contiguous();
+  auto D = torch::zeros({16, 16}, A.options().dtype(torch::kFloat).device(A.device()));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+    (const bf16*)Ac.data_ptr(), (const bf16*)Bc.data_ptr(), D.data_ptr<float>());
+  CHECK_CUDA_OK();
+  return D;
+}

@@ -1,0 +1,40 @@
+// pybind bindings for realhf_amd._C — the hand-written gfx950 kernels.
+#include <torch/extension.h>
+#include <vector>
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor rstd);
+torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cosb, torch::Tensor sinb,
+                       torch::Tensor positions, bool interleaved, bool conj);
+torch::Tensor swiglu_fwd(torch::Tensor gu);
+torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu);
+std::vector<torch::Tensor> gae_1d(torch::Tensor rewards, torch::Tensor values,
+                                  torch::Tensor cu_seqlens, torch::Tensor bootstrap,
+                                  double gamma, double lam);
+torch::Tensor slice_intervals(torch::Tensor src, torch::Tensor intervals);
+void set_intervals(torch::Tensor src, torch::Tensor dst, torch::Tensor intervals);
+void fused_adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                 torch::Tensor v, torch::Tensor out_bf16, double lr, double b1,
+                 double b2, double eps, double wd, long step, bool write_bf16);
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                          torch::Tensor cache_seqlens, double scale);
+std::vector<torch::Tensor> attn_varlen_fwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale);
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope_fwd", &rope_fwd);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("gae_1d", &gae_1d);
+  m.def("slice_intervals", &slice_intervals);
+  m.def("set_intervals", &set_intervals);
+  m.def("fused_adamw", &fused_adamw);
+  m.def("attn_decode", &attn_decode);
+  m.def("attn_varlen_fwd", &attn_varlen_fwd);
+  m.def("mfma_probe", &mfma_probe);
+}

@@ -1,0 +1,190 @@
+"""HIP kernel numerics vs plain-torch fp32 references (reference test
+style: tests/cpp_extensions/*).  All tests require an MI355X."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():
+    pytest.skip("needs GPU", allow_module_level=True)
+
+from realhf_amd.ops import functional as F  # noqa: E402
+import realhf_amd._C as C  # noqa: E402 — must exist on a GPU box
+
+
+def test_mfma_fragment_layout():
+    """Asymmetric A/B probe: validates the assumed 16x16x32 lane mapping."""
+    torch.manual_seed(0)
+    A = torch.randn(16, 32) * 0.5
+    B = torch.randn(32, 16) * 0.5
+    D = C.mfma_probe(A.cuda(), B.cuda()).cpu()
+    ref = A.to(torch.bfloat16).float() @ B.to(torch.bfloat16).float()
+    torch.testing.assert_close(D, ref, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("rows,H", [(7, 128), (640, 4096), (1000, 11008 // 2 * 2)])
+def test_rmsnorm_fwd_bwd(rows, H):
+    torch.manual_seed(1)
+    x = torch.randn(rows, H, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(H, dtype=torch.bfloat16, device="cuda")
+    x_ref = x.clone().float().requires_grad_(True)
+    w_ref = w.clone().float().requires_grad_(True)
+    ref = F.rms_norm_ref(x_ref, w_ref, 1e-5)
+    xg = x.clone().requires_grad_(True)
+    wg = w.clone().requires_grad_(True)
+    out = F._RMSNormFn.apply(xg, wg, 1e-5)
+    torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    out.backward(g.to(torch.bfloat16))
+    torch.testing.assert_close(xg.grad.float(), x_ref.grad, atol=1e-1, rtol=1e-1)
+    torch.testing.assert_close(
+        wg.grad.float(), w_ref.grad, atol=0.5, rtol=5e-2
+    )
+
+
+def test_rope():
+    torch.manual_seed(2)
+    total, nh, hd = 300, 8, 128
+    x = torch.randn(total, nh, hd, dtype=torch.bfloat16, device="cuda")
+    pos = torch.randint(0, 500, (total,), device="cuda")
+    cos, sin = F.rotary_cache.get(hd, 512, 10000.0, torch.device("cuda"))
+    out = C.rope_fwd(x, cos, sin, pos.long(), False, False)
+    ref = F.apply_rotary_ref(x.float(), cos, sin, pos, False)
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+    # conj is the inverse rotation
+    back = C.rope_fwd(out, cos, sin, pos.long(), False, True)
+    torch.testing.assert_close(back.float(), x.float(), atol=5e-2, rtol=5e-2)
+
+
+def test_swiglu_fwd_bwd():
+    torch.manual_seed(3)
+    gu = torch.randn(257, 2 * 2816, dtype=torch.bfloat16, device="cuda")
+    ref_in = gu.clone().float().requires_grad_(True)
+    ref = F.swiglu_ref(ref_in)
+    out = C.swiglu_fwd(gu)
+    torch.testing.assert_close(out.float(), ref, atol=3e-2, rtol=3e-2)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    dgu = C.swiglu_bwd(g.to(torch.bfloat16), gu)
+    torch.testing.assert_close(dgu.float(), ref_in.grad, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("bs", [1, 4, 16])
+def test_gae(bs):
+    rng = np.random.RandomState(bs)
+    lens = rng.randint(3, 50, size=bs)
+    total = int(lens.sum())
+    cu = np.concatenate([[0], np.cumsum(lens)])
+    rewards = torch.randn(total, device="cuda")
+    values = torch.randn(total + bs, device="cuda")
+    bootstrap = torch.tensor(rng.rand(bs) > 0.5, device="cuda")
+    cut = torch.tensor(cu, dtype=torch.int32, device="cuda")
+    adv, ret = C.gae_1d(rewards, values, cut, bootstrap, 0.99, 0.95)
+    adv_ref, ret_ref = F.gae_ref(
+        rewards.cpu(), values.cpu(), cut.cpu(), bootstrap.cpu(), 0.99, 0.95
+    )
+    torch.testing.assert_close(adv.cpu(), adv_ref, atol=1e-4, rtol=1e-4)
+    torch.testing.assert_close(ret.cpu(), ret_ref, atol=1e-4, rtol=1e-4)
+
+
+def test_intervals():
+    torch.manual_seed(4)
+    n = 10_000_000
+    src = torch.randn(n, dtype=torch.bfloat16, device="cuda")
+    rng = np.random.RandomState(0)
+    starts = np.sort(rng.choice(n - 2000, size=500, replace=False))
+    ivs = []
+    for s in starts:
+        ivs.append((int(s), int(s + rng.randint(1, 1500))))
+    ivs = F.merge_intervals(ivs)
+    iv_t = torch.tensor(ivs, dtype=torch.long, device="cuda")
+    out = C.slice_intervals(src, iv_t)
+    ref = F.slice_intervals_ref(src, iv_t.cpu())
+    assert torch.equal(out.cpu(), ref.cpu())
+    dst = torch.zeros_like(src)
+    C.set_intervals(out, dst, iv_t)
+    dst_ref = torch.zeros_like(src.cpu())
+    F.set_intervals_ref(out.cpu(), dst_ref, iv_t.cpu())
+    assert torch.equal(dst.cpu(), dst_ref)
+
+
+def test_fused_adamw():
+    torch.manual_seed(5)
+    n = 4096 * 8
+    p = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda") * 0.1
+    m = torch.randn(n, device="cuda").abs() * 0.01
+    v = torch.randn(n, device="cuda").abs() * 0.001
+    bf = torch.zeros(n, dtype=torch.bfloat16, device="cuda")
+    p2, m2, v2 = p.clone(), m.clone(), v.clone()
+    C.fused_adamw(p, g, m, v, bf, 1e-3, 0.9, 0.95, 1e-5, 0.1, 7, True)
+    F.fused_adamw_ref(p2, g, m2, v2, 1e-3, 0.9, 0.95, 1e-5, 0.1, 7)
+    torch.testing.assert_close(p, p2, atol=1e-6, rtol=1e-5)
+    torch.testing.assert_close(m, m2, atol=1e-6, rtol=1e-5)
+    torch.testing.assert_close(v, v2, atol=1e-7, rtol=1e-5)
+    torch.testing.assert_close(bf.float(), p2, atol=1e-2, rtol=1e-2)
+
+
+@pytest.mark.parametrize("nq,nkv", [(8, 8), (8, 2)])
+@pytest.mark.parametrize("lens", [[64], [128, 64, 200], [1, 17, 330]])
+def test_attn_varlen_fwd(nq, nkv, lens):
+    torch.manual_seed(6)
+    hd = 128
+    total = sum(lens)
+    cu = torch.tensor([0] + list(np.cumsum(lens)), dtype=torch.int32, device="cuda")
+    q = torch.randn(total, nq, hd, dtype=torch.bfloat16, device="cuda") * 0.5
+    k = torch.randn(total, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
+    v = torch.randn(total, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
+    scale = 1.0 / np.sqrt(hd)
+    out, lse = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale)
+    ref = F.attn_varlen_ref(
+        q.float().cpu(), k.float().cpu(), v.float().cpu(), cu.cpu(), True, scale
+    )
+    torch.testing.assert_close(out.float().cpu(), ref, atol=3e-2, rtol=3e-2)
+    assert torch.isfinite(lse[:, 0]).all()
+
+
+@pytest.mark.parametrize("nq,nkv,hd", [(32, 32, 128), (8, 2, 128), (8, 8, 64)])
+def test_attn_decode(nq, nkv, hd):
+    torch.manual_seed(7)
+    bs, maxlen = 5, 400
+    lens = torch.tensor([1, 50, 200, 399, 64], dtype=torch.int32, device="cuda")
+    q = torch.randn(bs, nq, hd, dtype=torch.bfloat16, device="cuda") * 0.5
+    kc = torch.randn(bs, maxlen, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
+    vc = torch.randn(bs, maxlen, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
+    scale = 1.0 / np.sqrt(hd)
+    out = C.attn_decode(q, kc, vc, lens, scale)
+    ref = F.attn_decode_ref(
+        q.float().cpu(), kc.float().cpu(), vc.float().cpu(), lens.cpu(), scale
+    )
+    torch.testing.assert_close(out.float().cpu(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_attn_varlen_autograd_bwd():
+    """The autograd path (HIP fwd + recompute bwd) vs full torch fp32."""
+    torch.manual_seed(8)
+    lens = [96, 130]
+    total = sum(lens)
+    nq = nkv = 4
+    hd = 128
+    cu = torch.tensor([0] + list(np.cumsum(lens)), dtype=torch.int32, device="cuda")
+    q = (torch.randn(total, nq, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    k = (torch.randn(total, nkv, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    v = (torch.randn(total, nkv, hd, device="cuda") * 0.5).to(torch.bfloat16)
+    qg = q.clone().requires_grad_(True)
+    kg = k.clone().requires_grad_(True)
+    vg = v.clone().requires_grad_(True)
+    out = F.attn_varlen(qg, kg, vg, cu, max(lens), True, None)
+    g = torch.randn_like(out)
+    out.backward(g)
+    qr = q.float().cpu().requires_grad_(True)
+    kr = k.float().cpu().requires_grad_(True)
+    vr = v.float().cpu().requires_grad_(True)
+    ref = F.attn_varlen_ref(qr, kr, vr, cu.cpu(), True, 1.0 / np.sqrt(hd))
+    ref.backward(g.float().cpu())
+    torch.testing.assert_close(out.float().cpu(), ref, atol=4e-2, rtol=4e-2)
+    torch.testing.assert_close(qg.grad.float().cpu(), qr.grad, atol=8e-2, rtol=8e-2)
+    torch.testing.assert_close(kg.grad.float().cpu(), kr.grad, atol=8e-2, rtol=8e-2)
+    torch.testing.assert_close(vg.grad.float().cpu(), vr.grad, atol=8e-2, rtol=8e-2)
