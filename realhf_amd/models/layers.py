@@ -173,9 +173,14 @@ class ReaLModelBlock(nn.Module):
             h = mappings.copy_to_tp_region(h)
         q, k, v = self._qkv(h)
         if cfg.apply_rotary:
+            # graph-capture-safe length bound: never read positions back
+            if decode:
+                rot_len = int(k_cache.shape[1])
+            else:
+                rot_len = max(int(max_seqlen), cfg.max_position_embeddings)
             cos, sin = ops.rotary_cache.get(
                 self.hd,
-                int(positions.max().item()) + 1 if positions.numel() else 1,
+                rot_len,
                 cfg.rotary_base,
                 x.device,
                 scaling=cfg.rotary_scaling,
