@@ -161,28 +161,46 @@ def main():
 
     frozen = ("ref", "rew")
 
+    phase_t = {}
+
+    def mark(name, t0):
+        if use_cuda:
+            torch.cuda.synchronize()
+        now = time.time()
+        phase_t[name] = phase_t.get(name, 0.0) + (now - t0)
+        return now
+
     def ppo_step():
+        t = time.time()
         batch = make_prompt_batch()
         with scope("actor"):
             rollout = actor_iface.generate(models["actor"], batch)
+        t = mark("actor_gen", t)
         sample = rollout
         seq_only = sample.select_keys(["packed_input_ids"])
         with scope("rew"):
             models["rew"].module.model.reload_from_offload()
+            t = mark("rew_reload", t)
             sample.update_(rew_iface.inference(models["rew"], seq_only))
             if args.offload_frozen:
                 models["rew"].module.model.async_offload()
+        t = mark("rew_inf", t)
         with scope("ref"):
             models["ref"].module.model.reload_from_offload()
+            t = mark("ref_reload", t)
             sample.update_(actor_iface.inference(models["ref"], seq_only))
             if args.offload_frozen:
                 models["ref"].module.model.async_offload()
+        t = mark("ref_inf", t)
         with scope("critic"):
             sample.update_(critic_iface.inference(models["critic"], seq_only))
+        t = mark("critic_inf", t)
         with scope("actor"):
             astats = actor_iface.train_step(models["actor"], sample)
+        t = mark("actor_train", t)
         with scope("critic"):
             cstats = critic_iface.train_step(models["critic"], sample)
+        t = mark("critic_train", t)
         return astats, cstats
 
     if args.offload_frozen:
@@ -202,6 +220,7 @@ def main():
         log(f"warmup step {i} done; peak mem={m2:.1f} GiB")
     barrier_sync()
 
+    phase_t.clear()
     t0 = time.time()
     for i in range(args.steps):
         astats, cstats = ppo_step()
@@ -209,6 +228,8 @@ def main():
             f"kl={astats.get('kl', 0):.4f}")
     barrier_sync()
     elapsed = time.time() - t0
+    log("phase breakdown (s/step): " + json.dumps(
+        {k: round(v / args.steps, 3) for k, v in phase_t.items()}))
 
     # MAX elapsed over ranks
     if world > 1:
