@@ -108,6 +108,37 @@ class DecodeGraph:
         self.graph = None
 
 
+class GenerationSession:
+    """Persistent KV caches + captured decode graph, reused across generate
+    calls of the same shape (the capture costs ~100 eager layer launches;
+    re-capturing per call was ~2 s/step at the 7B bench shape)."""
+
+    def __init__(self, kv_caches, decoder):
+        self.kv_caches = kv_caches
+        self.decoder = decoder
+
+
+def _get_session(model: ReaLModel, bs, cache_len, nkv_local, device):
+    cfg = model.config
+    key = (bs, cache_len, id(model.flat_param))
+    sess = getattr(model, "_gen_session", None)
+    if sess is not None and sess[0] == key:
+        return sess[1]
+    n_blocks = sum(1 for i in model.layer_indices if 1 <= i <= cfg.n_layers)
+    kv_caches = [
+        (
+            torch.zeros(bs, cache_len, nkv_local, cfg.head_dim,
+                        dtype=model.dtype, device=device),
+            torch.zeros(bs, cache_len, nkv_local, cfg.head_dim,
+                        dtype=model.dtype, device=device),
+        )
+        for _ in range(n_blocks)
+    ]
+    s = GenerationSession(kv_caches, DecodeGraph(model, kv_caches, bs))
+    model._gen_session = (key, s)
+    return s
+
+
 @torch.no_grad()
 def generate(
     model: ReaLModel,
@@ -126,18 +157,13 @@ def generate(
     prompt_lens = (cu_seqlens[1:] - cu_seqlens[:-1]).to(device)
     max_prompt = int(prompt_lens.max())
     max_new = gconfig.max_new_tokens
-    cache_len = max_prompt + max_new
+    # bucket the cache length so different max_prompt values reuse the
+    # same session/graph
+    cache_len = ((max_prompt + max_new + 127) // 128) * 128
 
     nkv_local = max(cfg.n_kv_heads // model.tp_size, 1)
-    n_blocks = sum(1 for i in model.layer_indices if 1 <= i <= cfg.n_layers)
-    kv_dtype = model.dtype
-    kv_caches = [
-        (
-            torch.zeros(bs, cache_len, nkv_local, cfg.head_dim, dtype=kv_dtype, device=device),
-            torch.zeros(bs, cache_len, nkv_local, cfg.head_dim, dtype=kv_dtype, device=device),
-        )
-        for _ in range(n_blocks)
-    ]
+    sess = _get_session(model, bs, cache_len, nkv_local, device)
+    kv_caches = sess.kv_caches
 
     # ---- prefill --------------------------------------------------------
     logits = model(
@@ -165,7 +191,7 @@ def generate(
 
     cache_seqlens = prompt_lens.to(torch.int32).clone()
 
-    decoder = DecodeGraph(model, kv_caches, bs)
+    decoder = sess.decoder
     use_graph = gconfig.use_hip_graph and device.type == "cuda"
 
     cur_logits = last_logits
@@ -179,7 +205,10 @@ def generate(
         gen_lengths += (~done).long()
         if eos_token_id is not None:
             done = done | (tokens == eos_token_id)
-        if bool(done.all()) or t == max_new - 1:
+            # early-exit check costs a host sync — amortize it
+            if (t & 15) == 15 and bool(done.all()):
+                break
+        if t == max_new - 1:
             break
 
         cache_seqlens += 1  # the new token's slot
@@ -189,7 +218,6 @@ def generate(
         logits_step = mappings.gather_from_tp_region(logits_step).float()
         cur_logits = logits_step
 
-    decoder.destroy()
     return GenerationOutput(
         gen_tokens=gen_tokens[:, : int(gen_lengths.max())],
         gen_logprobs=gen_logprobs[:, : int(gen_lengths.max())],
