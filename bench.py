@@ -17,6 +17,9 @@ import json
 import os
 import time
 
+# avoid allocator fragmentation at the 270+ GiB working set
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import numpy as np
 import torch
 import torch.distributed as dist

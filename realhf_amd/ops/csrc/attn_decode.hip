@@ -1,22 +1,26 @@
 // Single-token GQA decode attention over a contiguous KV cache (gfx950).
 // Replaces flash_attn_with_kvcache in the reference hot loop
-// (realhf/impl/model/nn/real_llm_generate.py:?38 decode step).
+// (realhf/impl/model/nn/real_llm_generate.py decode step).
 //
 // Shapes: q [bs, nq, hd], k/v_cache [bs, maxlen, nkv, hd], cache_seqlens[bs].
 // Memory-bound: the job is to stream each sequence's KV exactly once at
-// full HBM bandwidth.  One 256-thread workgroup (4 waves) per (batch,
+// full HBM bandwidth.  One 512-thread workgroup (8 waves) per (batch,
 // kv-head); the workgroup computes ALL rep = nq/nkv query heads of that
 // kv head so the KV stream is read once regardless of GQA ratio.
 //
-// Per 64-key chunk (one key per lane):
-//   lane = key: score_r = q_r . k  (16-byte vectorized k loads, 256 B/lane
-//   contiguous -> full lines); online-softmax per wave; p -> LDS; then the
-//   lane role flips to "2 output dims per lane" and V is accumulated with
-//   coalesced 4-byte loads.  Wave partials (m, s, acc) combine through LDS.
+// Per 64-key chunk (wave w takes chunks w, w+8, ...):
+//   lane = key: the lane streams its key's K row AND V row with 16-byte
+//   vector loads (256 B contiguous per lane — full cache lines); V goes
+//   to this wave's LDS tile.  Online softmax per wave (shfl reduces);
+//   then the lane role flips to "DPL output dims per lane" and V is
+//   accumulated from LDS (conflict-free b32 reads), p broadcast by shfl.
+//   Wave partials (m, s, acc) combine through LDS at the end.
 #include "common.h"
 
+#define DEC_WAVES 8
+
 template <int HD, int REP>
-__global__ void attn_decode_kernel(
+__global__ __launch_bounds__(64 * DEC_WAVES) void attn_decode_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
     const bf16* __restrict__ vc, const int* __restrict__ cache_seqlens,
     bf16* __restrict__ out, int bs, int nq, int nkv, long maxlen, float scale) {
@@ -28,10 +32,10 @@ __global__ void attn_decode_kernel(
   const int w = threadIdx.x / WAVE;
 
   __shared__ float q_s[REP][HD];
-  __shared__ float m_s[4], s_s[4];
-  __shared__ float acc_s[4][HD];
+  __shared__ __bf16 v_s[DEC_WAVES][WAVE][HD];  // per-wave V chunk tile
+  __shared__ float m_s[DEC_WAVES], s_s[DEC_WAVES];
+  __shared__ float acc_s[DEC_WAVES][HD];
 
-  // stage q (scaled) into LDS
   for (int i = threadIdx.x; i < REP * HD; i += blockDim.x) {
     int r = i / HD, d = i % HD;
     q_s[r][d] = __bfloat162float(q[((long)b * nq + kvh * REP + r) * HD + d]) * scale;
@@ -51,18 +55,20 @@ __global__ void attn_decode_kernel(
   const bf16* kb = kc + (long)b * maxlen * kv_stride + (long)kvh * HD;
   const bf16* vb = vc + (long)b * maxlen * kv_stride + (long)kvh * HD;
 
-  for (int base = w * WAVE; base < L; base += 4 * WAVE) {
-    int l = base + lane;
-    bool valid = l < L;
-    // scores for this lane's key against all REP q heads
+  for (int base = w * WAVE; base < L; base += DEC_WAVES * WAVE) {
+    const int l = base + lane;
+    const bool valid = l < L;
     float sc[REP];
     #pragma unroll
     for (int r = 0; r < REP; r++) sc[r] = 0.f;
     if (valid) {
       const bf16* krow = kb + (long)l * kv_stride;
+      const bf16* vrow = vb + (long)l * kv_stride;
       #pragma unroll
       for (int i = 0; i < HD / 8; i++) {
         short8 kv8 = *(const short8*)((const short*)krow + i * 8);
+        short8 vv8 = *(const short8*)((const short*)vrow + i * 8);
+        *(short8*)(&v_s[w][lane][i * 8]) = vv8;
         float kf[8];
         #pragma unroll
         for (int j = 0; j < 8; j++) kf[j] = bf2f(kv8[j]);
@@ -73,6 +79,7 @@ __global__ void attn_decode_kernel(
         }
       }
     }
+    const int nvalid = min(WAVE, L - base);
     #pragma unroll
     for (int r = 0; r < REP; r++) {
       if (!valid) sc[r] = -1e30f;
@@ -85,21 +92,18 @@ __global__ void attn_decode_kernel(
       for (int d = 0; d < DPL; d++) acc_w[r][d] *= f;
       float p = valid ? __expf(sc[r] - nm) : 0.f;
       s_w[r] += wave_sum(p);
-      // role flip: lane owns DPL output dims; broadcast p via shfl
-      #pragma unroll 4
-      for (int j = 0; j < WAVE; j++) {
-        int lj = base + j;
-        if (lj >= L) break;
+      // role flip: lane owns DPL output dims; V comes from LDS,
+      // p broadcast by shfl (no cross-wave LDS -> in-wave visibility)
+      for (int j = 0; j < nvalid; j++) {
         float pj = __shfl(p, j, 64);
-        const bf16* vrow = vb + (long)lj * kv_stride;
         #pragma unroll
         for (int d = 0; d < DPL; d++)
-          acc_w[r][d] += pj * __bfloat162float(vrow[lane * DPL + d]);
+          acc_w[r][d] += pj * (float)v_s[w][j][lane * DPL + d];
       }
     }
   }
 
-  // combine the 4 waves' partials per rep head
+  // combine the wave partials per rep head
   #pragma unroll
   for (int r = 0; r < REP; r++) {
     if (lane == 0) { m_s[w] = m_w[r]; s_s[w] = s_w[r]; }
@@ -107,13 +111,15 @@ __global__ void attn_decode_kernel(
     for (int d = 0; d < DPL; d++) acc_s[w][lane * DPL + d] = acc_w[r][d];
     __syncthreads();
     if (w == 0) {
-      float M = fmaxf(fmaxf(m_s[0], m_s[1]), fmaxf(m_s[2], m_s[3]));
+      float M = -1e30f;
+      #pragma unroll
+      for (int ww = 0; ww < DEC_WAVES; ww++) M = fmaxf(M, m_s[ww]);
       float S = 0.f;
       float o[DPL];
       #pragma unroll
       for (int d = 0; d < DPL; d++) o[d] = 0.f;
       #pragma unroll
-      for (int ww = 0; ww < 4; ww++) {
+      for (int ww = 0; ww < DEC_WAVES; ww++) {
         float f = __expf(m_s[ww] - M);
         S += f * s_s[ww];
         #pragma unroll
@@ -142,7 +148,7 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   dim3 grid(bs * nkv);
   auto launch = [&](auto hd_c, auto rep_c) {
     hipLaunchKernelGGL((attn_decode_kernel<hd_c.value, rep_c.value>), grid,
-      dim3(256), 0, cur_stream(), (const bf16*)q.data_ptr(),
+      dim3(64 * DEC_WAVES), 0, cur_stream(), (const bf16*)q.data_ptr(),
       (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
       cs.data_ptr<int>(), (bf16*)out.data_ptr(), bs, nq, nkv, maxlen,
       (float)scale);

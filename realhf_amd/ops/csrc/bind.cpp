@@ -16,7 +16,8 @@ torch::Tensor slice_intervals(torch::Tensor src, torch::Tensor intervals);
 void set_intervals(torch::Tensor src, torch::Tensor dst, torch::Tensor intervals);
 void fused_adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                  torch::Tensor v, torch::Tensor out_bf16, double lr, double b1,
-                 double b2, double eps, double wd, long step, bool write_bf16);
+                 double b2, double eps, double wd, long step, double gscale,
+                 bool write_bf16);
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
                           torch::Tensor cache_seqlens, double scale);
 std::vector<torch::Tensor> attn_varlen_fwd(

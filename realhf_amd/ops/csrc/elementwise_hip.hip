@@ -170,11 +170,43 @@ torch::Tensor swiglu_bwd(torch::Tensor dy, torch::Tensor gu) {
 // Fused AdamW on the flat fp32 master shard, optional bf16 write-back.
 // Replaces apex fused adam (reference Megatron dep).
 // ---------------------------------------------------------------------------
+template <typename TG>
 __global__ void fused_adamw_kernel(
+    float* __restrict__ p, const TG* __restrict__ g,
+    float* __restrict__ m, float* __restrict__ v, bf16* __restrict__ out_bf16,
+    long n, float lr, float b1, float b2, float eps, float wd,
+    float bc1, float bc2, float gscale, bool write_bf16) {
+  for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
+       i += (long)gridDim.x * blockDim.x * 4) {
+    float4v pv = *(float4v*)(p + i);
+    TG gv4[4];
+    *(short4v*)gv4 = *(const short4v*)((const short*)(g + i));  // 8B for bf16
+    float4v mv = *(float4v*)(m + i);
+    float4v vv = *(float4v*)(v + i);
+    short ob[4];
+    #pragma unroll
+    for (int j = 0; j < 4; j++) {
+      float gj = to_f32<TG>(gv4[j]) * gscale;
+      mv[j] = b1 * mv[j] + (1.f - b1) * gj;
+      vv[j] = b2 * vv[j] + (1.f - b2) * gj * gj;
+      float denom = sqrtf(vv[j] / bc2) + eps;
+      pv[j] = pv[j] * (1.f - lr * wd) - lr / bc1 * mv[j] / denom;
+      ob[j] = f2bf(pv[j]);
+    }
+    *(float4v*)(p + i) = pv;
+    *(float4v*)(m + i) = mv;
+    *(float4v*)(v + i) = vv;
+    if (write_bf16) *(short4v*)((short*)out_bf16 + i) = *(short4v*)ob;
+  }
+}
+
+// fp32-grad specialization needs 16B loads
+template <>
+__global__ void fused_adamw_kernel<float>(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ m, float* __restrict__ v, bf16* __restrict__ out_bf16,
     long n, float lr, float b1, float b2, float eps, float wd,
-    float bc1, float bc2, bool write_bf16) {
+    float bc1, float bc2, float gscale, bool write_bf16) {
   for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
        i += (long)gridDim.x * blockDim.x * 4) {
     float4v pv = *(float4v*)(p + i);
@@ -184,8 +216,9 @@ __global__ void fused_adamw_kernel(
     short ob[4];
     #pragma unroll
     for (int j = 0; j < 4; j++) {
-      mv[j] = b1 * mv[j] + (1.f - b1) * gv[j];
-      vv[j] = b2 * vv[j] + (1.f - b2) * gv[j] * gv[j];
+      float gj = gv[j] * gscale;
+      mv[j] = b1 * mv[j] + (1.f - b1) * gj;
+      vv[j] = b2 * vv[j] + (1.f - b2) * gj * gj;
       float denom = sqrtf(vv[j] / bc2) + eps;
       pv[j] = pv[j] * (1.f - lr * wd) - lr / bc1 * mv[j] / denom;
       ob[j] = f2bf(pv[j]);
@@ -199,17 +232,29 @@ __global__ void fused_adamw_kernel(
 
 void fused_adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                  torch::Tensor v, torch::Tensor out_bf16, double lr, double b1,
-                 double b2, double eps, double wd, long step, bool write_bf16) {
+                 double b2, double eps, double wd, long step, double gscale,
+                 bool write_bf16) {
   long n = p.numel();
   TORCH_CHECK(n % 4 == 0, "flat shard must be 4-aligned");
   TORCH_CHECK(p.is_cuda() && g.is_cuda());
   float bc1 = 1.f - powf((float)b1, (float)step);
   float bc2 = 1.f - powf((float)b2, (float)step);
   int grid = (int)std::min<long>((n / 4 + 255) / 256, 4096);
-  hipLaunchKernelGGL(fused_adamw_kernel, dim3(grid), dim3(256), 0, cur_stream(),
-    p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
-    v.data_ptr<float>(), write_bf16 ? (bf16*)out_bf16.data_ptr() : nullptr,
-    n, (float)lr, (float)b1, (float)b2, (float)eps, (float)wd, bc1, bc2,
-    write_bf16);
+  if (g.scalar_type() == torch::kFloat) {
+    hipLaunchKernelGGL(fused_adamw_kernel<float>, dim3(grid), dim3(256), 0,
+      cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+      m.data_ptr<float>(), v.data_ptr<float>(),
+      write_bf16 ? (bf16*)out_bf16.data_ptr() : nullptr,
+      n, (float)lr, (float)b1, (float)b2, (float)eps, (float)wd, bc1, bc2,
+      (float)gscale, write_bf16);
+  } else {
+    TORCH_CHECK(g.scalar_type() == torch::kBFloat16);
+    hipLaunchKernelGGL(fused_adamw_kernel<bf16>, dim3(grid), dim3(256), 0,
+      cur_stream(), p.data_ptr<float>(), (const bf16*)g.data_ptr(),
+      m.data_ptr<float>(), v.data_ptr<float>(),
+      write_bf16 ? (bf16*)out_bf16.data_ptr() : nullptr,
+      n, (float)lr, (float)b1, (float)b2, (float)eps, (float)wd, bc1, bc2,
+      (float)gscale, write_bf16);
+  }
   CHECK_CUDA_OK();
 }

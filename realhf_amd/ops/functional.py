@@ -221,10 +221,45 @@ def attn_varlen_ref(
 
 
 def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale):
-    """Memory-bounded torch implementation on GPU (used for backward
-    recompute until the hand-written HIP backward lands): processes each
-    sequence with SDPA-like math in torch (rocBLAS GEMMs on MFMA)."""
-    return attn_varlen_ref(q, k, v, cu_seqlens, causal, scale)
+    """Batched-padded torch implementation on GPU (used for backward
+    recompute until the hand-written HIP backward lands): one set of
+    batched rocBLAS GEMMs instead of a per-sequence python loop (the loop
+    was CPU-launch-bound: bs x layers x ~15 kernels)."""
+    total, nq, hd = q.shape
+    nkv = k.shape[1]
+    rep = nq // nkv
+    lens = (cu_seqlens[1:] - cu_seqlens[:-1]).long()
+    bs = lens.shape[0]
+    Lmax = int(lens.max())
+    device = q.device
+    pos = torch.arange(total, device=device)
+    seq_id = torch.bucketize(pos, cu_seqlens[1:].long(), right=True)
+    local = pos - cu_seqlens[:-1].long()[seq_id]
+
+    def pad(x, nh):
+        xp = x.new_zeros(bs, Lmax, nh, hd)
+        xp[seq_id, local] = x
+        return xp.permute(0, 2, 1, 3).float()  # [bs, nh, Lmax, hd]
+
+    qp = pad(q, nq)
+    kp = pad(k, nkv)
+    vp = pad(v, nkv)
+    if rep > 1:
+        kp = kp.repeat_interleave(rep, dim=1)
+        vp = vp.repeat_interleave(rep, dim=1)
+    scores = torch.matmul(qp, kp.transpose(-1, -2)) * scale
+    kpos = torch.arange(Lmax, device=device)
+    valid = kpos.unsqueeze(0) < lens.unsqueeze(1)  # [bs, Lmax]
+    mask = valid.unsqueeze(1).unsqueeze(2)  # [bs,1,1,L]
+    if causal:
+        cm = kpos.unsqueeze(0) <= kpos.unsqueeze(1)  # [Lq, Lk]
+        mask = mask & cm.unsqueeze(0).unsqueeze(0)
+    scores = scores.masked_fill(~mask, float("-inf"))
+    probs = torch.softmax(scores, dim=-1)
+    probs = torch.nan_to_num(probs, nan=0.0)
+    op = torch.matmul(probs, vp)  # [bs, nh, Lmax, hd]
+    op = op.permute(0, 2, 1, 3)  # [bs, Lmax, nh, hd]
+    return op[seq_id, local].to(q.dtype)
 
 
 class _AttnVarlenFn(torch.autograd.Function):
@@ -404,8 +439,9 @@ def fused_adamw_ref(
     weight_decay: float,
     step: int,
     bf16_out: Optional[torch.Tensor] = None,
+    grad_scale: float = 1.0,
 ):
-    g = grad.float()
+    g = grad.float() * grad_scale
     exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
     exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
     bc1 = 1 - beta1**step
@@ -419,17 +455,18 @@ def fused_adamw_ref(
 
 
 def fused_adamw(param_f32, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps,
-                weight_decay, step, bf16_out=None):
+                weight_decay, step, bf16_out=None, grad_scale=1.0):
     if _ops.use_hip(param_f32):
         C = _ops.require_hip()
         C.fused_adamw(
             param_f32, grad, exp_avg, exp_avg_sq,
             bf16_out if bf16_out is not None else param_f32.new_empty(0).to(torch.bfloat16),
             float(lr), float(beta1), float(beta2), float(eps),
-            float(weight_decay), int(step), bf16_out is not None,
+            float(weight_decay), int(step), float(grad_scale),
+            bf16_out is not None,
         )
         return param_f32
     return fused_adamw_ref(
         param_f32, grad, exp_avg, exp_avg_sq, lr, beta1, beta2, eps,
-        weight_decay, step, bf16_out,
+        weight_decay, step, bf16_out, grad_scale,
     )

@@ -147,17 +147,19 @@ class ZeRO1Optimizer:
                 self.grad_shard, self.grad_padded, op=dist.ReduceOp.AVG,
                 group=self.dp_group,
             )
+            gshard = self.grad_shard
         else:
-            self.grad_shard.copy_(self.grad_padded)
+            gshard = self.grad_padded  # no copy needed at dp==1
 
-        gs32 = self.grad_shard.float()
-
-        # 2. grad-norm clip (norm over DP shards, reduced across DP;
+        # 2. grad-norm clip — computed WITHOUT materializing an fp32 copy
+        #    (a 7B fp32 grad copy is a 28 GiB spike).  The clip factor is
+        #    folded into the AdamW kernel's grad_scale.
         #    NOTE with TP>1 replicated-param grads are counted tp× —
-        #    conservative overestimate, acceptable for clipping)
+        #    conservative overestimate, acceptable for clipping.
         grad_norm = None
+        gscale = 1.0
         if cfg.gradient_clipping and cfg.gradient_clipping > 0:
-            sq = gs32.pow(2).sum()
+            sq = torch.linalg.vector_norm(gshard, dtype=torch.float32) ** 2
             if self.dp_size > 1:
                 dist.all_reduce(sq, group=self.dp_group)
             grad_norm = float(sq.sqrt())
@@ -165,20 +167,19 @@ class ZeRO1Optimizer:
                 logger.warning("non-finite grad norm %s — skipping step", grad_norm)
                 return {"lr": self._lr(), "grad_norm": grad_norm, "skipped": 1.0}
             clip = cfg.gradient_clipping / (grad_norm + 1e-6)
-            if clip < 1.0:
-                gs32.mul_(clip)
+            gscale = min(1.0, clip)
 
-        # 3. AdamW on this rank's shard
+        # 3. AdamW on this rank's shard (bf16 grads read directly)
         lr = self._lr()
         s0, s1 = self.shard_bounds
         param_shard = self._param_padded[s0:s1]
-        if cfg.offload:
-            gs32 = gs32.cpu()
+        g_in = gshard.float().cpu() if cfg.offload else gshard
         ops.fused_adamw(
-            self.master, gs32, self.exp_avg, self.exp_avg_sq,
+            self.master, g_in, self.exp_avg, self.exp_avg_sq,
             lr=lr, beta1=cfg.beta1, beta2=cfg.beta2, eps=cfg.eps,
             weight_decay=cfg.weight_decay, step=self.step_count,
             bf16_out=None if cfg.offload else param_shard,
+            grad_scale=gscale,
         )
         if cfg.offload:
             param_shard.copy_(self.master.to(param_shard.dtype))
