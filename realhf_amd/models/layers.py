@@ -171,6 +171,55 @@ class ReaLModelBlock(nn.Module):
             h = mappings.gather_from_sp_region(h)
         else:
             h = mappings.copy_to_tp_region(h)
+
+        scale = 1.0 / math.sqrt(self.hd)
+        if cfg.scale_attn_by_inverse_layer_idx:
+            scale = scale / float(self.i)
+
+        # ---- fused decode fast path: one kernel for qkv-split + bias +
+        # RoPE + KV append (HIP only, bf16, no qk-layernorm)
+        import os as _os
+
+        if (
+            decode
+            and h.is_cuda
+            and h.dtype == torch.bfloat16
+            and not cfg.qk_layernorm
+            and _os.environ.get("REALHF_AMD_NO_FUSED_DECODE") != "1"
+        ):
+            from realhf_amd import ops as _ops_pkg
+
+            C = _ops_pkg.require_hip()
+            qkv_dim = (self.nq + 2 * self.nkv) * self.hd
+            merged = _maybe_merged(self.p, self._qkv_names, qkv_dim)
+            if merged is not None:
+                qkv_raw = _linear(h, merged)
+            else:
+                qkv_raw = torch.cat(
+                    [_linear(h, self.p[n]) for n in self._qkv_names], dim=-1
+                )
+            bias = None
+            if f"{i}.attn.wq.bias" in self.p:
+                bias = torch.cat(
+                    [self.p[f"{i}.attn.w{c}.bias"] for c in "qkv"], dim=0
+                ).to(qkv_raw.dtype)
+            cos, sin = ops.rotary_cache.get(
+                self.hd, int(k_cache.shape[1]), cfg.rotary_base, x.device,
+                scaling=cfg.rotary_scaling,
+            )
+            q = C.rope_qkv_decode(
+                qkv_raw, bias, k_cache, v_cache,
+                cache_seqlens, cos, sin, self.nq, cfg.apply_rotary,
+            )
+            attn_out = ops.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
+            attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
+            o = _linear(attn_out, self.p[f"{i}.attn.wo.weight"])
+            o = mappings.reduce_from_tp_region(o)
+            if f"{i}.attn.wo.bias" in self.p:
+                o = o + self.p[f"{i}.attn.wo.bias"]
+            x = x + o
+            return self._mlp(x, sp)
+
         q, k, v = self._qkv(h)
         if cfg.apply_rotary:
             # graph-capture-safe length bound: never read positions back
@@ -187,10 +236,6 @@ class ReaLModelBlock(nn.Module):
             )
             q = ops.apply_rotary(q, cos, sin, positions, cfg.rotary_interleaved)
             k = ops.apply_rotary(k, cos, sin, positions, cfg.rotary_interleaved)
-
-        scale = 1.0 / math.sqrt(self.hd)
-        if cfg.scale_attn_by_inverse_layer_idx:
-            scale = scale / float(self.i)
 
         if decode:
             # one new token per sequence; write into cache then attend
@@ -224,8 +269,11 @@ class ReaLModelBlock(nn.Module):
         if f"{i}.attn.wo.bias" in self.p:
             o = o + self.p[f"{i}.attn.wo.bias"]
         x = x + o
+        return self._mlp(x, sp)
 
-        # -- MLP ----------------------------------------------------------
+    def _mlp(self, x, sp):
+        cfg = self.cfg
+        i = self.i
         h = _norm(cfg, x, self.p[f"{i}.mlp.ln.weight"], self.p.get(f"{i}.mlp.ln.bias"))
         if self.moe is not None:
             m = self.moe(h)

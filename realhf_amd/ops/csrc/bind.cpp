@@ -24,6 +24,10 @@ std::vector<torch::Tensor> attn_varlen_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor rope_qkv_decode(
+    torch::Tensor qkv, c10::optional<torch::Tensor> bias, torch::Tensor kcache,
+    torch::Tensor vcache, torch::Tensor cache_seqlens, torch::Tensor cosb,
+    torch::Tensor sinb, long nq, bool apply_rope);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
@@ -38,4 +42,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode);
   m.def("attn_varlen_fwd", &attn_varlen_fwd);
   m.def("mfma_probe", &mfma_probe);
+  m.def("rope_qkv_decode", &rope_qkv_decode);
 }

@@ -19,6 +19,7 @@ SRC = [
     "realhf_amd/ops/csrc/gae.hip",
     "realhf_amd/ops/csrc/interval.hip",
     "realhf_amd/ops/csrc/attn_decode.hip",
+    "realhf_amd/ops/csrc/rope_decode.hip",
     "realhf_amd/ops/csrc/attn_varlen.hip",
 ]
 

@@ -188,3 +188,49 @@ def test_attn_varlen_autograd_bwd():
     torch.testing.assert_close(qg.grad.float().cpu(), qr.grad, atol=8e-2, rtol=8e-2)
     torch.testing.assert_close(kg.grad.float().cpu(), kr.grad, atol=8e-2, rtol=8e-2)
     torch.testing.assert_close(vg.grad.float().cpu(), vr.grad, atol=8e-2, rtol=8e-2)
+
+
+def test_fused_decode_layer_matches_eager():
+    """The fused rope_qkv_decode path vs the composed eager decode path."""
+    import os
+
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.models.real_model import ReaLModel
+
+    fam = hf_reg.get_family("llama")
+    cfg = fam.make_test_config(
+        n_layers=2, hidden_dim=256, n_heads=4, n_kv_heads=2, head_dim=64,
+        vocab_size=128,
+    )
+    torch.manual_seed(11)
+    model = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
+    model.random_init()
+    bs, maxlen = 3, 32
+    kv = [
+        (
+            torch.zeros(bs, maxlen, 2, 64, dtype=torch.bfloat16, device="cuda"),
+            torch.zeros(bs, maxlen, 2, 64, dtype=torch.bfloat16, device="cuda"),
+        )
+        for _ in range(cfg.n_layers)
+    ]
+    kv2 = [(k.clone(), v.clone()) for k, v in kv]
+    tokens = torch.randint(0, 128, (bs,), device="cuda")
+    cache_seqlens = torch.tensor([5, 9, 2], dtype=torch.int32, device="cuda")
+    with torch.no_grad():
+        out_fused = model(
+            packed_input_ids=tokens, kv_caches=kv, cache_seqlens=cache_seqlens,
+            decode=True,
+        )
+        os.environ["REALHF_AMD_NO_FUSED_DECODE"] = "1"
+        try:
+            out_eager = model(
+                packed_input_ids=tokens, kv_caches=kv2,
+                cache_seqlens=cache_seqlens, decode=True,
+            )
+        finally:
+            del os.environ["REALHF_AMD_NO_FUSED_DECODE"]
+    torch.testing.assert_close(out_fused.float(), out_eager.float(),
+                               atol=3e-2, rtol=3e-2)
+    for (k1, v1), (k2, v2) in zip(kv, kv2):
+        torch.testing.assert_close(k1.float(), k2.float(), atol=2e-2, rtol=2e-2)
+        torch.testing.assert_close(v1.float(), v2.float(), atol=2e-2, rtol=2e-2)
