@@ -119,12 +119,21 @@ def test_fused_adamw():
     v = torch.randn(n, device="cuda").abs() * 0.001
     bf = torch.zeros(n, dtype=torch.bfloat16, device="cuda")
     p2, m2, v2 = p.clone(), m.clone(), v.clone()
-    C.fused_adamw(p, g, m, v, bf, 1e-3, 0.9, 0.95, 1e-5, 0.1, 7, True)
+    C.fused_adamw(p, g, m, v, bf, 1e-3, 0.9, 0.95, 1e-5, 0.1, 7, 1.0, True)
     F.fused_adamw_ref(p2, g, m2, v2, 1e-3, 0.9, 0.95, 1e-5, 0.1, 7)
     torch.testing.assert_close(p, p2, atol=1e-6, rtol=1e-5)
     torch.testing.assert_close(m, m2, atol=1e-6, rtol=1e-5)
     torch.testing.assert_close(v, v2, atol=1e-7, rtol=1e-5)
     torch.testing.assert_close(bf.float(), p2, atol=1e-2, rtol=1e-2)
+    # bf16 grads + clip scale
+    p3, m3, v3 = p.clone(), m.clone(), v.clone()
+    p4, m4, v4 = p.clone(), m.clone(), v.clone()
+    gb = g.to(torch.bfloat16)
+    C.fused_adamw(p3, gb, m3, v3, bf, 1e-3, 0.9, 0.95, 1e-5, 0.1, 8, 0.5, True)
+    F.fused_adamw_ref(p4, gb, m4, v4, 1e-3, 0.9, 0.95, 1e-5, 0.1, 8,
+                      grad_scale=0.5)
+    torch.testing.assert_close(p3, p4, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(m3, m4, atol=1e-6, rtol=1e-5)
 
 
 @pytest.mark.parametrize("nq,nkv", [(8, 8), (8, 2)])
