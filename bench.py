@@ -19,6 +19,13 @@ import time
 
 # avoid allocator fragmentation at the 270+ GiB working set
 os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+# pre-tuned hipBLASLt/rocBLAS GEMM solutions for the bench shapes (gfx950)
+_TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      "realhf_amd", "data", "tunableop_gfx950.csv")
+if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED
 
 import numpy as np
 import torch
@@ -176,6 +183,8 @@ def main():
     def ppo_step():
         t = time.time()
         batch = make_prompt_batch()
+        # prefetch the reward model's weights: H2D overlaps generation
+        models["rew"].module.model.start_reload()
         with scope("actor"):
             rollout = actor_iface.generate(models["actor"], batch)
         t = mark("actor_gen", t)
@@ -184,6 +193,7 @@ def main():
         with scope("rew"):
             models["rew"].module.model.reload_from_offload()
             t = mark("rew_reload", t)
+            models["ref"].module.model.start_reload()  # overlap with rew_inf
             sample.update_(rew_iface.inference(models["rew"], seq_only))
             if args.offload_frozen:
                 models["rew"].module.model.async_offload()

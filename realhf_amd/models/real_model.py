@@ -200,10 +200,17 @@ class ReaLModel(nn.Module):
         return self.layout.total_numel
 
     # ------------------------------------------------------------- offload
-    def async_offload(self):
+    def _side_stream(self):
+        if not hasattr(self, "_offload_stream"):
+            self._offload_stream = torch.cuda.Stream()
+        return self._offload_stream
+
+    def async_offload(self, non_blocking: bool = True):
         """Copy flat params to pinned host memory and free the device
         buffer (reference: real_llm_api.py:274).  Used for non-trainable
-        roles between their MFCs."""
+        roles between their MFCs.  With non_blocking=True the D2H runs on
+        a side stream overlapped with subsequent compute; the allocator
+        defers block reuse via record_stream."""
         if self._offloaded:
             return
         if self._offload_buf is None:
@@ -211,18 +218,48 @@ class ReaLModel(nn.Module):
                 self.flat_param.shape, dtype=self.dtype, device="cpu",
                 pin_memory=torch.cuda.is_available(),
             )
-        self._offload_buf.copy_(self.flat_param, non_blocking=False)
+        if non_blocking and self.flat_param.is_cuda:
+            s = self._side_stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                self._offload_buf.copy_(self.flat_param, non_blocking=True)
+            self.flat_param.record_stream(s)
+        else:
+            self._offload_buf.copy_(self.flat_param, non_blocking=False)
         self.flat_param = None
         self._params = {}
         self.layers = nn.ModuleList()
-        if torch.cuda.is_available():
-            torch.cuda.empty_cache()
         self._offloaded = True
+
+    def start_reload(self):
+        """Kick the H2D reload on the side stream (overlaps with whatever
+        compute is running); finish_reload() must be called before use."""
+        if not self._offloaded or getattr(self, "_reload_pending", None):
+            return
+        if not torch.cuda.is_available() or not self.device.type == "cuda":
+            return
+        s = self._side_stream()
+        with torch.cuda.stream(s):
+            flat = torch.empty(
+                self._offload_buf.shape, dtype=self.dtype, device=self.device
+            )
+            flat.copy_(self._offload_buf, non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record(s)
+        self._reload_pending = (flat, ev)
 
     def reload_from_offload(self):
         if not self._offloaded:
             return
-        self.flat_param = self._offload_buf.to(self.device, non_blocking=False)
+        pending = getattr(self, "_reload_pending", None)
+        if pending is not None:
+            flat, ev = pending
+            torch.cuda.current_stream().wait_event(ev)
+            flat.record_stream(torch.cuda.current_stream())
+            self.flat_param = flat
+            self._reload_pending = None
+        else:
+            self.flat_param = self._offload_buf.to(self.device, non_blocking=False)
         self._map_params()
         self._build_modules()
         self._offloaded = False
