@@ -23,9 +23,15 @@ os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
 _TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                       "realhf_amd", "data", "tunableop_gfx950.csv")
 if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
+    # TunableOp appends the device ordinal before .csv on read — stage a
+    # copy per possible ordinal under /tmp
+    import shutil as _sh
+
+    for _d in range(8):
+        _sh.copy(_TUNED, f"/tmp/realhf_tunableop{_d}.csv")
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
-    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TUNED
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = "/tmp/realhf_tunableop.csv"
 
 import numpy as np
 import torch
@@ -49,6 +55,8 @@ def main():
     p.add_argument("--model", type=str, default="llama-7b",
                    help="llama-7b | llama-small (debug)")
     p.add_argument("--no-hip-graph", action="store_true")
+    p.add_argument("--phase-timing", action="store_true",
+                   help="per-phase sync+timing (serializes side-stream overlap)")
     p.add_argument("--offload-frozen", action="store_true", default=True)
     args = p.parse_args()
 
@@ -174,6 +182,8 @@ def main():
     phase_t = {}
 
     def mark(name, t0):
+        if not args.phase_timing:
+            return t0
         if use_cuda:
             torch.cuda.synchronize()
         now = time.time()
@@ -241,8 +251,9 @@ def main():
             f"kl={astats.get('kl', 0):.4f}")
     barrier_sync()
     elapsed = time.time() - t0
-    log("phase breakdown (s/step): " + json.dumps(
-        {k: round(v / args.steps, 3) for k, v in phase_t.items()}))
+    if args.phase_timing:
+        log("phase breakdown (s/step): " + json.dumps(
+            {k: round(v / args.steps, 3) for k, v in phase_t.items()}))
 
     # MAX elapsed over ranks
     if world > 1:
