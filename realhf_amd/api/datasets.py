@@ -4,7 +4,6 @@ benchmarking (no-network environment).
 
 All datasets produce SequenceSamples with unique ids.
 """
-import uuid
 from typing import Dict, List, Optional
 
 import numpy as np
@@ -26,7 +25,8 @@ class PromptDataset(torch.utils.data.Dataset):
         self.records = load_shuffle_split_dataset(path, seed, dp_rank, world_size)
         self.max_prompt_len = max_prompt_len
         self.tokenizer = tokenizer
-        self.ids = [str(uuid.uuid4()) for _ in self.records]
+        # deterministic ids: every SPMD rank must assign identical ids
+        self.ids = [f"prompt-{seed}-{dp_rank}-{i}" for i in range(len(self.records))]
 
     def __len__(self):
         return len(self.records)
@@ -57,7 +57,7 @@ class PromptAnswerDataset(torch.utils.data.Dataset):
         self.records = load_shuffle_split_dataset(path, seed, dp_rank, world_size)
         self.max_seqlen = max_seqlen
         self.tokenizer = tokenizer
-        self.ids = [str(uuid.uuid4()) for _ in self.records]
+        self.ids = [f"pa-{seed}-{dp_rank}-{i}" for i in range(len(self.records))]
 
     def __len__(self):
         return len(self.records)
@@ -97,7 +97,7 @@ class RewardModelingPairedDataset(torch.utils.data.Dataset):
         self.max_seqlen = max_seqlen
         self.max_pairs = max_pairs_per_prompt
         self.tokenizer = tokenizer
-        self.ids = [str(uuid.uuid4()) for _ in self.records]
+        self.ids = [f"rw-{seed}-{dp_rank}-{i}" for i in range(len(self.records))]
         self.rng = np.random.RandomState(seed)
 
     def __len__(self):

@@ -294,13 +294,32 @@ class ParallelGrid:
 
 
 class FakeGrid:
-    """Grid math without process groups — for CPU tests and the allocation
-    planner (reference: topology.py:593)."""
+    """Grid math without process groups — for CPU tests, single-process
+    runs and the allocation planner (reference: topology.py:593)."""
 
     def __init__(self, rank: int, topo: PipeDataTensorTopology):
         self.topo = topo
         self.rank = rank
         self._coord = topo.get_coord(rank)
+
+    # group getters are no-ops (single process)
+    def tp_group(self):
+        return None
+
+    def dp_group(self):
+        return None
+
+    def dp_cpu_group(self):
+        return None
+
+    def pp_group(self):
+        return None
+
+    def model_group(self):
+        return None
+
+    def embedding_group(self):
+        return None
 
     @property
     def coord(self):

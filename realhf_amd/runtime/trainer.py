@@ -1,0 +1,481 @@
+"""Experiment builder + training loop (the SPMD 'controller').
+
+Reference semantics: realhf/experiments/common/*.py (graph construction
+ppo_exp.py:261-377, resolve_replica_ids/resolve_rpc_hooks utils.py:126/143)
++ the master worker's epoch/save/eval loop (master_worker.py:1273-1499),
+re-done as a deterministic SPMD program (see runtime/executor.py).
+"""
+import dataclasses
+import json
+import os
+import pickle
+import time
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+import realhf_amd.interfaces  # noqa: F401 — register interfaces
+import realhf_amd.runtime.engine  # noqa: F401 — register backends
+from realhf_amd.api import datasets as ds_impl  # noqa: F401 — register datasets
+from realhf_amd.api.config import (
+    Abstraction,
+    ModelInterfaceType,
+    ModelName,
+    ParallelismConfig,
+    parse_parallelism,
+)
+from realhf_amd.api.data import PackedDataLoader, SequenceSample, make_dataset
+from realhf_amd.api.dfg import DFG, MFCDef, OffloadHook, ParamReallocHook, build_graph
+from realhf_amd.api.experiment import (
+    CommonExperimentConfig,
+    DPOConfig,
+    GenerationConfig,
+    GRPOConfig,
+    PPOConfig,
+    RWConfig,
+    SFTConfig,
+)
+from realhf_amd.api.model import (
+    FinetuneSpec,
+    Model,
+    make_backend,
+    make_interface,
+)
+from realhf_amd.base import constants, logging, seeding
+from realhf_amd.base.topology import ParallelGrid, PipeDataTensorTopology
+from realhf_amd.models import hf as hf_reg
+from realhf_amd.models.real_model import ReaLModel
+from realhf_amd.parallel.realloc import ParallelStrategy
+from realhf_amd.runtime.executor import DFGExecutor, MFCAllocation
+
+logger = logging.getLogger("trainer")
+
+
+def _strategy_for(par: ParallelismConfig, world: int) -> ParallelStrategy:
+    ws = par.world_size
+    assert ws <= world, (par, world)
+    return ParallelStrategy.make(
+        par.pipeline_parallel_size, par.data_parallel_size,
+        par.tensor_parallel_size, ranks=list(range(ws)),
+    )
+
+
+def _resolve_parallel(cfg: CommonExperimentConfig, model_par: ParallelismConfig,
+                      world: int) -> ParallelismConfig:
+    mode = cfg.allocation_mode
+    if mode == "global" or mode == "heuristic":
+        # heuristic for one MI355X node: pure DP fits models <= ~13B whole
+        # (288 GB HBM); larger models should use manual/tp strategies
+        return ParallelismConfig(data_parallel_size=world)
+    if mode == "manual":
+        return model_par
+    return parse_parallelism(mode)
+
+
+@dataclasses.dataclass
+class BuiltExperiment:
+    graph: DFG
+    allocations: Dict[str, MFCAllocation]
+    model_strategies: Dict[ModelName, ParallelStrategy]
+    model_cfgs: Dict[ModelName, object]  # ReaLModelConfig
+    model_roles: Dict[ModelName, object]  # ModelTrainEvalConfig
+    interfaces: Dict[str, object]
+    trainable: List[ModelName]
+
+
+def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment:
+    T = ModelInterfaceType
+
+    def model_cfg_of(mc, name):
+        if mc.path:
+            rcfg = hf_reg.config_from_hf_path(mc.family, mc.path)
+        else:
+            fam = hf_reg.get_family(mc.family)
+            rcfg = fam.make_test_config()
+            rcfg.family = mc.family
+        rcfg.is_critic = mc.is_critic
+        rcfg.dtype = mc.dtype
+        return rcfg
+
+    mfcs: List[MFCDef] = []
+    interfaces: Dict[str, object] = {}
+    model_strategies: Dict[ModelName, ParallelStrategy] = {}
+    model_cfgs: Dict[ModelName, object] = {}
+    model_roles: Dict[ModelName, object] = {}
+    allocations: Dict[str, MFCAllocation] = {}
+    trainable: List[ModelName] = []
+
+    def add_model(role, mc, replica=0):
+        name = ModelName(role, replica)
+        par = _resolve_parallel(cfg, mc.parallel, world)
+        model_strategies[name] = _strategy_for(par, world)
+        model_cfgs[name] = model_cfg_of(mc, name)
+        model_roles[name] = mc
+        return name, par
+
+    def add_mfc(name, model_name, itype, iface_cfg, inp, out, mc, par,
+                n_mbs=None):
+        mfcs.append(
+            MFCDef(
+                name=name, model_name=model_name, interface_type=itype,
+                interface_impl=iface_cfg, input_keys=tuple(inp),
+                output_keys=tuple(out), n_seqs=cfg.dataset.train_bs_n_seqs,
+            )
+        )
+        interfaces[name] = make_interface(iface_cfg)
+        allocations[name] = MFCAllocation(
+            strategy=model_strategies[model_name],
+            sequence_parallel=par.sequence_parallel,
+            gradient_checkpointing=mc.gradient_checkpointing,
+            n_mbs=n_mbs,
+        )
+
+    if isinstance(cfg, SFTConfig):
+        name, par = add_model("default", cfg.model)
+        trainable.append(name)
+        add_mfc("train", name, T.TRAIN_STEP, Abstraction("sft"),
+                ["packed_input_ids", "prompt_mask"], [], cfg.model, par)
+    elif isinstance(cfg, RWConfig):
+        name, par = add_model("default", cfg.model)
+        trainable.append(name)
+        add_mfc("train", name, T.TRAIN_STEP, Abstraction("paired_rw"),
+                ["packed_input_ids"], [], cfg.model, par)
+    elif isinstance(cfg, DPOConfig):
+        actor, apar = add_model("actor", cfg.actor)
+        ref, rpar = add_model("ref", cfg.ref)
+        trainable.append(actor)
+        add_mfc("ref_inf", ref, T.INFERENCE, Abstraction("dpo", {"beta": cfg.beta}),
+                ["packed_input_ids", "prompt_mask"], ["seqlogp"], cfg.ref, rpar)
+        add_mfc("dpo_train", actor, T.TRAIN_STEP,
+                Abstraction("dpo", {"beta": cfg.beta}),
+                ["packed_input_ids", "prompt_mask", "seqlogp"], [],
+                cfg.actor, apar)
+        if cfg.ref.offload:
+            g = [m for m in mfcs if m.name == "ref_inf"][0]
+            g.post_hooks.append(OffloadHook())
+    elif isinstance(cfg, GRPOConfig):
+        _build_grpo(cfg, world, add_model, add_mfc, mfcs, trainable)
+    elif isinstance(cfg, PPOConfig):
+        _build_ppo(cfg, world, add_model, add_mfc, mfcs, trainable)
+    elif isinstance(cfg, GenerationConfig):
+        name, par = add_model("default", cfg.model)
+        add_mfc("gen", name, T.GENERATE,
+                Abstraction("generation", {"gconfig": dataclasses.asdict(cfg.gen)}),
+                ["packed_prompts"], ["packed_input_ids", "prompt_mask"],
+                cfg.model, par)
+    else:
+        raise TypeError(cfg)
+
+    graph = build_graph(mfcs)
+    return BuiltExperiment(
+        graph=graph, allocations=allocations,
+        model_strategies=model_strategies, model_cfgs=model_cfgs,
+        model_roles=model_roles, interfaces=interfaces, trainable=trainable,
+    )
+
+
+def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
+    T = ModelInterfaceType
+    ppo = cfg.ppo
+    actor, apar = add_model("actor", cfg.actor)
+    critic, cpar = add_model("critic", cfg.critic)
+    ref, refpar = add_model("ref", cfg.ref)
+    rew, rewpar = add_model("rew", cfg.rew)
+    trainable += [actor, critic]
+
+    gen_cfg = dataclasses.asdict(ppo.gen)
+    actor_iface = Abstraction("ppo_actor", {
+        "n_minibatches": ppo.ppo_n_minibatches, "gconfig": gen_cfg,
+        "kl_ctl": ppo.kl_ctl, "adaptive_kl_ctl": ppo.use_adaptive_kl_ctl,
+        "eps_clip": ppo.eps_clip, "max_reward_clip": ppo.max_reward_clip,
+        "discount": ppo.discount, "gae_lambda": ppo.gae_lambda,
+        "adv_norm": ppo.adv_norm,
+        "early_stop_imp_ratio": ppo.early_stop_imp_ratio,
+    })
+    critic_iface = Abstraction("ppo_critic", {
+        "n_minibatches": ppo.ppo_n_minibatches,
+        "value_eps_clip": ppo.value_eps_clip, "kl_ctl": ppo.kl_ctl,
+        "max_reward_clip": ppo.max_reward_clip, "discount": ppo.discount,
+        "gae_lambda": ppo.gae_lambda, "value_norm": ppo.value_norm,
+    })
+    rw_iface = Abstraction("paired_rw", {
+        "output_scaling": ppo.reward_output_scaling,
+        "output_bias": ppo.reward_output_bias,
+    })
+
+    # separate generation replica when gen uses a different strategy —
+    # the parameter-reallocation flow (reference: resolve_replica_ids +
+    # resolve_rpc_hooks, experiments/common/utils.py:126/143)
+    gen_model = actor
+    gen_par = apar
+    if cfg.actor.gen_parallel is not None and cfg.allocation_mode == "manual":
+        gen_mc = dataclasses.replace(cfg.actor, parallel=cfg.actor.gen_parallel)
+        gen_model, gen_par = add_model("actor", gen_mc, replica=1)
+    add_mfc("actor_gen", gen_model, T.GENERATE, actor_iface,
+            ["packed_prompts"],
+            ["packed_input_ids", "packed_logprobs", "prompt_mask",
+             "seq_no_eos_mask"], cfg.actor, gen_par)
+    if gen_model != actor:
+        gen_mfc = mfcs[-1]
+        gen_mfc.pre_hooks.append(
+            ParamReallocHook(source=actor, target=gen_model)
+        )
+        gen_mfc.post_hooks.append(OffloadHook())
+    add_mfc("rew_inf", rew, T.INFERENCE, rw_iface,
+            ["packed_input_ids"], ["rewards"], cfg.rew, rewpar)
+    add_mfc("ref_inf", ref, T.INFERENCE, actor_iface,
+            ["packed_input_ids"], ["packed_ref_logprobs"], cfg.ref, refpar)
+    add_mfc("critic_inf", critic, T.INFERENCE, critic_iface,
+            ["packed_input_ids"], ["values"], cfg.critic, cpar)
+    train_keys = ["packed_input_ids", "packed_logprobs", "packed_ref_logprobs",
+                  "rewards", "values", "prompt_mask", "seq_no_eos_mask"]
+    add_mfc("actor_train", actor, T.TRAIN_STEP, actor_iface, train_keys, [],
+            cfg.actor, apar)
+    add_mfc("critic_train", critic, T.TRAIN_STEP, critic_iface, train_keys, [],
+            cfg.critic, cpar)
+
+    for m in mfcs:
+        if m.name == "rew_inf" and cfg.rew.offload:
+            m.post_hooks.append(OffloadHook())
+        if m.name == "ref_inf" and cfg.ref.offload:
+            m.post_hooks.append(OffloadHook())
+
+
+def _build_grpo(cfg: GRPOConfig, world, add_model, add_mfc, mfcs, trainable):
+    T = ModelInterfaceType
+    ppo = cfg.ppo
+    actor, apar = add_model("actor", cfg.actor)
+    ref, refpar = add_model("ref", cfg.ref)
+    rew, rewpar = add_model("rew", cfg.rew)
+    trainable.append(actor)
+    iface = Abstraction("grpo", {
+        "n_minibatches": ppo.ppo_n_minibatches,
+        "gconfig": dataclasses.asdict(ppo.gen),
+        "kl_ctl": ppo.kl_ctl, "eps_clip": ppo.eps_clip,
+        "group_size": cfg.group_size, "kl_in_loss_coef": cfg.kl_in_loss_coef,
+    })
+    rw_iface = Abstraction("paired_rw", {
+        "output_scaling": ppo.reward_output_scaling,
+        "output_bias": ppo.reward_output_bias,
+    })
+    add_mfc("actor_gen", actor, T.GENERATE, iface, ["packed_prompts"],
+            ["packed_input_ids", "packed_logprobs", "prompt_mask",
+             "seq_no_eos_mask"], cfg.actor, apar)
+    add_mfc("rew_inf", rew, T.INFERENCE, rw_iface, ["packed_input_ids"],
+            ["rewards"], cfg.rew, rewpar)
+    add_mfc("ref_inf", ref, T.INFERENCE, iface, ["packed_input_ids"],
+            ["packed_ref_logprobs"], cfg.ref, refpar)
+    add_mfc("actor_train", actor, T.TRAIN_STEP, iface,
+            ["packed_input_ids", "packed_logprobs", "packed_ref_logprobs",
+             "rewards", "prompt_mask"], [], cfg.actor, apar)
+    for m in mfcs:
+        if m.name == "rew_inf" and cfg.rew.offload:
+            m.post_hooks.append(OffloadHook())
+        if m.name == "ref_inf" and cfg.ref.offload:
+            m.post_hooks.append(OffloadHook())
+
+
+# ---------------------------------------------------------------------------
+class Trainer:
+    def __init__(self, cfg: CommonExperimentConfig):
+        self.cfg = cfg
+        self.rank = dist.get_rank() if dist.is_initialized() else 0
+        self.world = dist.get_world_size() if dist.is_initialized() else 1
+        seeding.set_random_seed(cfg.seed)
+        constants.set_experiment_trial_names(cfg.experiment_name, cfg.trial_name)
+        self.built = build_experiment(cfg, self.world)
+        self.device = torch.device(
+            "cuda", int(os.environ.get("LOCAL_RANK", 0))
+        ) if torch.cuda.is_available() else torch.device("cpu")
+        self.tokenizer = self._load_tokenizer()
+        self.models = self._build_models()
+        self.executor = DFGExecutor(
+            self.built.graph, self.built.allocations, self.models,
+            self.built.interfaces, self.built.model_strategies,
+        )
+        self.global_step = 0
+
+    def _load_tokenizer(self):
+        path = self.cfg.tokenizer_path
+        if path is None:
+            for mc in self.built.model_roles.values():
+                if mc.path:
+                    path = mc.path
+                    break
+        if path is None:
+            return None
+        try:
+            import transformers
+
+            return transformers.AutoTokenizer.from_pretrained(path)
+        except Exception as e:  # pragma: no cover
+            logger.warning("tokenizer load failed: %s", e)
+            return None
+
+    def _build_models(self) -> Dict[ModelName, Model]:
+        models = {}
+        for name, strat in self.built.model_strategies.items():
+            my_coord = None
+            for (p, d, t), r in strat.rank_map:
+                if r == self.rank:
+                    my_coord = (p, d, t)
+            # register grid (group creation is collective — all ranks join)
+            topo = PipeDataTensorTopology(
+                num_pp=strat.pp, num_dp=strat.dp, num_tp=strat.tp,
+                sequence_parallel=False,
+            )
+            rank_mapping = {
+                topo.get_rank(pipe=p, data=d, tensor=t): r
+                for (p, d, t), r in strat.rank_map
+            }
+            if dist.is_initialized():
+                grid = ParallelGrid(topo, rank_mapping)
+            else:
+                from realhf_amd.base.topology import FakeGrid
+
+                grid = FakeGrid(0, topo)
+            constants.set_grid(str(name), grid)
+            if my_coord is None:
+                continue
+            p_, d_, t_ = my_coord
+            rcfg = self.built.model_cfgs[name]
+            mc = self.built.model_roles[name]
+            with constants.model_scope(str(name)):
+                m = ReaLModel(
+                    rcfg, device=self.device, dtype=rcfg.torch_dtype,
+                    tp_rank=t_, tp_size=strat.tp, pp_rank=p_, pp_size=strat.pp,
+                )
+                if mc.path:
+                    hf_reg.load_from_hf(m, mc.family, mc.path)
+                else:
+                    m.random_init()
+                model = Model(
+                    name=name, module=m, tokenizer=self.tokenizer,
+                    device=self.device, dtype=rcfg.torch_dtype,
+                )
+                spec = FinetuneSpec(
+                    self.cfg.exp_ctrl.total_train_epochs,
+                    10_000,
+                    self.cfg.dataset.train_bs_n_seqs,
+                )
+                if name in self.built.trainable:
+                    backend = make_backend(
+                        Abstraction("zero1", {"optimizer": dataclasses.asdict(mc.optimizer)})
+                    )
+                else:
+                    backend = make_backend(Abstraction("inference"))
+                models[name] = backend.initialize(model, spec)
+            if mc.offload and name not in self.built.trainable:
+                real = models[name].module.model
+                real.async_offload(non_blocking=False)
+        return models
+
+    def _build_dataloader(self):
+        d = self.cfg.dataset
+        ds_cfg = Abstraction(d.type_, dict(d.args))
+        if d.type_ == "prompt":
+            ds_cfg.args.setdefault("path", d.path)
+            ds_cfg.args.setdefault("max_prompt_len", d.max_prompt_len)
+        elif d.type_ == "prompt_answer":
+            ds_cfg.args.setdefault("path", d.path)
+            ds_cfg.args.setdefault("max_seqlen", d.max_seqlen)
+        elif d.type_ == "rw_paired":
+            ds_cfg.args.setdefault("path", d.path)
+            ds_cfg.args.setdefault("max_seqlen", d.max_seqlen)
+        # SPMD: every rank builds the identical dataset/loader (same seed)
+        dataset = make_dataset(ds_cfg, seed=self.cfg.seed, dp_rank=0,
+                               world_size=1, tokenizer=self.tokenizer)
+        return PackedDataLoader(
+            dataset, batch_n_seqs=d.train_bs_n_seqs, shuffle=True,
+            seed=self.cfg.seed,
+        )
+
+    # ----------------------------------------------------------------- run
+    def run(self):
+        cfg = self.cfg
+        dl = self._build_dataloader()
+        ctrl = cfg.exp_ctrl
+        bench_t0 = None
+        recover = self._maybe_load_recover()
+        start_epoch, start_step = (recover or (0, 0))
+        for epoch in range(start_epoch, ctrl.total_train_epochs):
+            for i, batch in enumerate(dl):
+                if epoch == start_epoch and i < start_step:
+                    continue
+                t0 = time.time()
+                stats = self.executor.run_step(batch)
+                dt = time.time() - t0
+                self.global_step += 1
+                if self.rank == 0:
+                    logger.info(
+                        "epoch %d step %d (global %d): %.2fs %s",
+                        epoch, i, self.global_step, dt,
+                        {k: round(v, 4) for k, v in stats.items()
+                         if isinstance(v, float)},
+                    )
+                if ctrl.save_freq_steps and self.global_step % ctrl.save_freq_steps == 0:
+                    self.save()
+                if ctrl.benchmark_steps:
+                    if bench_t0 is None:
+                        bench_t0 = time.time()
+                        bench_step0 = self.global_step
+                    elif self.global_step - bench_step0 >= ctrl.benchmark_steps:
+                        el = time.time() - bench_t0
+                        n = self.global_step - bench_step0
+                        if self.rank == 0:
+                            logger.info(
+                                "benchmark: %d steps, %.3fs/step, %.2f samples/s",
+                                n, el / n,
+                                n * cfg.dataset.train_bs_n_seqs / el,
+                            )
+                        self._save_recover_info(epoch, i)
+                        return
+            self._save_recover_info(epoch + 1, 0)
+        if ctrl.save_freq_steps:
+            self.save()
+
+    def save(self):
+        for name in self.built.trainable:
+            if name not in self.models:
+                continue
+            save_dir = os.path.join(
+                constants.MODEL_SAVE_ROOT(self.cfg.experiment_name, self.cfg.trial_name),
+                name.role, f"globalstep{self.global_step}",
+            )
+            # find the mfc interface owning this model to save via it
+            for mfc in self.built.graph.mfcs:
+                if mfc.model_name == name and mfc.interface_type == ModelInterfaceType.TRAIN_STEP:
+                    with constants.model_scope(str(name)):
+                        self.built.interfaces[mfc.name].save(
+                            self.models[name], save_dir
+                        )
+                    if self.rank == 0:
+                        logger.info("saved %s -> %s", name, save_dir)
+                    break
+
+    # -------------------------------------------------------------- recover
+    def _recover_path(self):
+        return os.path.join(
+            constants.RECOVER_ROOT(self.cfg.experiment_name, self.cfg.trial_name),
+            "recover_info.pkl",
+        )
+
+    def _save_recover_info(self, epoch, step):
+        if self.cfg.recover_mode == "disabled" or self.rank != 0:
+            return
+        with open(self._recover_path(), "wb") as f:
+            pickle.dump({"epoch": epoch, "step": step,
+                         "global_step": self.global_step}, f)
+
+    def _maybe_load_recover(self):
+        if self.cfg.recover_mode not in ("auto", "resume"):
+            return None
+        p = self._recover_path()
+        if not os.path.exists(p):
+            return None
+        with open(p, "rb") as f:
+            info = pickle.load(f)
+        self.global_step = info["global_step"]
+        logger.info("recovering from %s", info)
+        return (info["epoch"], info["step"])

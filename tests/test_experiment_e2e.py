@@ -1,0 +1,176 @@
+"""End-to-end experiment runs through the CLI config path on CPU
+(reference milestone: config #1 'GPT-2 small SFT on CPU' + the quickstart
+CLI, SURVEY.md §7 stage 4)."""
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from realhf_amd.apps.quickstart import parse_cli
+from realhf_amd.base.testing import LocalMultiProcessTest
+
+
+def _write_sft_data(path, n=16, vocab=64):
+    rng = np.random.RandomState(0)
+    with open(path, "w") as f:
+        for _ in range(n):
+            rec = {
+                "prompt_ids": rng.randint(0, vocab, size=rng.randint(3, 6)).tolist(),
+                "answer_ids": rng.randint(0, vocab, size=rng.randint(4, 9)).tolist(),
+            }
+            f.write(json.dumps(rec) + "\n")
+
+
+def _write_prompt_data(path, n=16, vocab=64):
+    rng = np.random.RandomState(1)
+    with open(path, "w") as f:
+        for _ in range(n):
+            rec = {"input_ids": rng.randint(3, vocab - 3, size=rng.randint(4, 8)).tolist()}
+            f.write(json.dumps(rec) + "\n")
+
+
+def _write_rw_data(path, n=16, vocab=64):
+    rng = np.random.RandomState(2)
+    with open(path, "w") as f:
+        for _ in range(n):
+            rec = {
+                "pos_ids": rng.randint(0, vocab, size=rng.randint(5, 10)).tolist(),
+                "neg_ids": rng.randint(0, vocab, size=rng.randint(5, 10)).tolist(),
+            }
+            f.write(json.dumps(rec) + "\n")
+
+
+def test_cli_override_parsing():
+    exp, cfg = parse_cli([
+        "ppo", "experiment_name=x", "n_gpus=4",
+        "ppo.gen.max_new_tokens=64", "ppo.kl_ctl=0.05",
+        "actor.optimizer.lr=0.0001", "exp_ctrl.save_freq_steps=null",
+        "dataset.train_bs_n_seqs=8", "actor.gradient_checkpointing=true",
+    ])
+    assert exp == "ppo"
+    assert cfg.n_gpus == 4
+    assert cfg.ppo.gen.max_new_tokens == 64
+    assert cfg.ppo.kl_ctl == 0.05
+    assert cfg.actor.optimizer.lr == 1e-4
+    assert cfg.exp_ctrl.save_freq_steps is None
+    assert cfg.actor.gradient_checkpointing is True
+
+
+def test_sft_experiment_single_process(tmp_path):
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "sft.jsonl")
+    _write_sft_data(data)
+    cfg = SFTConfig(
+        experiment_name="t-sft", trial_name="cpu", n_gpus=1,
+    )
+    cfg.model.dtype = "float32"
+    cfg.dataset.type_ = "prompt_answer"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.exp_ctrl.total_train_epochs = 1
+    cfg.exp_ctrl.benchmark_steps = 2
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
+
+
+def test_ppo_experiment_single_process(tmp_path):
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data)
+    cfg = PPOConfig(experiment_name="t-ppo", trial_name="cpu", n_gpus=1)
+    for mc in (cfg.actor, cfg.critic, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 6
+    cfg.ppo.gen.min_new_tokens = 2
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
+
+
+def test_rw_experiment_with_save(tmp_path):
+    from realhf_amd.api.experiment import RWConfig
+    from realhf_amd.base import constants
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "rw.jsonl")
+    _write_rw_data(data)
+    cfg = RWConfig(experiment_name="t-rw", trial_name="cpu", n_gpus=1)
+    cfg.model.dtype = "float32"
+    cfg.dataset.type_ = "rw_paired"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.exp_ctrl.benchmark_steps = 1
+    cfg.exp_ctrl.save_freq_steps = 1
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
+    root = constants.MODEL_SAVE_ROOT("t-rw", "cpu")
+    saved = [d for d, _, fs in os.walk(root) if any(f.endswith(".safetensors") for f in fs)]
+    assert saved, f"no checkpoint written under {root}"
+
+
+def _sft_dist_worker(data, fileroot):
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = SFTConfig(experiment_name="t-sft2", trial_name="dist", n_gpus=2)
+    cfg.model.dtype = "float32"
+    cfg.allocation_mode = "global"
+    cfg.dataset.type_ = "prompt_answer"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.exp_ctrl.benchmark_steps = 2
+    Trainer(cfg).run()
+
+
+@pytest.mark.distributed
+def test_sft_experiment_two_ranks(tmp_path):
+    data = str(tmp_path / "sft.jsonl")
+    _write_sft_data(data)
+    LocalMultiProcessTest(2, _sft_dist_worker, data, str(tmp_path / "root")).launch()
+
+
+def _ppo_realloc_worker(data, fileroot):
+    """PPO with a DIFFERENT gen strategy for the actor -> exercises the
+    realloc hooks through the executor."""
+    from realhf_amd.api.config import ParallelismConfig
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = PPOConfig(experiment_name="t-ppo2", trial_name="dist", n_gpus=2)
+    for mc in (cfg.actor, cfg.critic, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+        mc.parallel = ParallelismConfig(data_parallel_size=2)
+    cfg.allocation_mode = "manual"
+    cfg.actor.gen_parallel = ParallelismConfig(
+        data_parallel_size=1, tensor_parallel_size=2
+    )
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 5
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    Trainer(cfg).run()
+
+
+@pytest.mark.distributed
+def test_ppo_realloc_two_ranks(tmp_path):
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data)
+    LocalMultiProcessTest(2, _ppo_realloc_worker, data, str(tmp_path / "root")).launch()
