@@ -47,6 +47,8 @@ class DPOInterface(ModelInterface):
             return _seq_logps(logits, mb)
 
         seqlogp = model.module.forward(data, n_mbs=n_mbs, post_hook=post_hook)
+        if seqlogp is None:  # pp mid stage
+            return None
         # 2 sequences (pos, neg) per sample
         return SequenceSample(
             keys=("seqlogp",),

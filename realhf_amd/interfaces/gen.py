@@ -28,6 +28,8 @@ class GenerationInterface(ModelInterface):
         outs = model.module.generate(
             data, tokenizer=model.tokenizer, gconfig=self.gconfig, n_mbs=n_mbs
         )
+        if outs is None:  # pp mid stage
+            return None
         all_ids, all_pm, seqlens = [], [], []
         for gen_out, prompts, cu in outs:
             packed, cu_full, pmask = concat_prompt_to_generation_output(

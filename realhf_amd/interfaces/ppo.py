@@ -126,6 +126,8 @@ class PPOActorInterface(ModelInterface):
             data, tokenizer=model.tokenizer, gconfig=self.gconfig,
             n_mbs=n_mbs, return_prompt_logprobs=True,
         )
+        if outs is None:  # pp mid stage: output lives on the last stage
+            return None
         all_ids, all_lp, all_pm, all_noeos, seqlens = [], [], [], [], []
         for gen_out, prompts, cu in outs:
             packed, cu_full, pmask = concat_prompt_to_generation_output(
@@ -174,6 +176,8 @@ class PPOActorInterface(ModelInterface):
             return packed_shifted_logprobs(logits, cu, ids)
 
         logp = model.module.forward(data, n_mbs=n_mbs, post_hook=post_hook)
+        if logp is None:  # pp mid stage
+            return None
         return SequenceSample(
             keys=("packed_ref_logprobs",),
             ids=list(data.ids),
@@ -289,6 +293,8 @@ class PPOCriticInterface(ModelInterface):
             return values.squeeze(-1).float()
 
         vals = model.module.forward(data, n_mbs=n_mbs, post_hook=post_hook)
+        if vals is None:  # pp mid stage
+            return None
         if self._rms is not None:
             vals = self._rms.denormalize(vals)
         return SequenceSample(

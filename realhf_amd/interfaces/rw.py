@@ -53,6 +53,8 @@ class PairedRewardInterface(ModelInterface):
             return _end_scores(values, cu)
 
         scores = model.module.forward(data, n_mbs=n_mbs, post_hook=post_hook)
+        if scores is None:  # pp mid stage
+            return None
         scores = (scores - self.output_bias) * self.output_scaling
         return SequenceSample(
             keys=("rewards",),

@@ -151,7 +151,12 @@ class ZeRO1Backend(ModelBackend):
             total_train_steps=spec.total_train_steps,
             bucket_size=self.bucket_size,
         )
-        model.module = PipelinableTrainEngine(model.module, opt)
+        if model.module.pp_size > 1:
+            from realhf_amd.parallel.pp import PipelinedEngine
+
+            model.module = PipelinedEngine(model.module, opt)
+        else:
+            model.module = PipelinableTrainEngine(model.module, opt)
         model.backend_name = "zero1"
         return model
 
@@ -159,7 +164,12 @@ class ZeRO1Backend(ModelBackend):
 @dataclasses.dataclass
 class InferenceBackend(ModelBackend):
     def _initialize(self, model: Model, spec: FinetuneSpec) -> Model:
-        model.module = PipelinableInferenceEngine(model.module)
+        if model.module.pp_size > 1:
+            from realhf_amd.parallel.pp import PipelinedEngine
+
+            model.module = PipelinedEngine(model.module, None)
+        else:
+            model.module = PipelinableInferenceEngine(model.module)
         model.backend_name = "inference"
         return model
 
