@@ -119,11 +119,14 @@ class PipeDataTensorTopology(ProcessTopology):
         sequence_parallel: bool = False,
         gradient_checkpointing: bool = False,
         max_prompt_len: Optional[int] = None,
+        ep_size: int = 1,
     ):
         super().__init__(["pipe", "data", "tensor"], [num_pp, num_dp, num_tp])
         self.sequence_parallel = sequence_parallel
         self.gradient_checkpointing = gradient_checkpointing
         self.max_prompt_len = max_prompt_len
+        assert num_dp % ep_size == 0, (num_dp, ep_size)
+        self.ep_size = ep_size
 
     @property
     def pp(self):
@@ -167,6 +170,7 @@ class ParallelGrid:
         self._model_group = None
         self._dp_cpu_group = None
         self._embedding_group = None
+        self._ep_group = None
         self._my_coord = None
 
         if process_groups:
@@ -202,6 +206,18 @@ class ParallelGrid:
             if my_rank in ranks:
                 self._dp_group = g
                 self._dp_cpu_group = g2
+        # EP groups: blocks of `ep_size` consecutive dp ranks per (pipe, tensor)
+        ep = getattr(topo, "ep_size", 1)
+        if ep > 1:
+            for p, t in itertools.product(range(topo.pp), range(topo.tp)):
+                for b in range(topo.dp // ep):
+                    ranks = [
+                        self.rank_mapping[topo.get_rank(pipe=p, data=d, tensor=t)]
+                        for d in range(b * ep, (b + 1) * ep)
+                    ]
+                    g = new_or_get_group(ranks)
+                    if my_rank in ranks:
+                        self._ep_group = g
         # PP groups: one per (data, tensor)
         for d, t in itertools.product(range(topo.dp), range(topo.tp)):
             ranks = self._global_ranks(topo.filter_match(data=d, tensor=t))
@@ -258,6 +274,17 @@ class ParallelGrid:
 
     def embedding_group(self):
         return self._embedding_group
+
+    def ep_group(self):
+        return self._ep_group
+
+    @property
+    def ep_size(self):
+        return getattr(self.topo, "ep_size", 1)
+
+    @property
+    def ep_rank(self):
+        return self.coord.data % self.ep_size if self.ep_size > 1 else 0
 
     # convenience ranks/sizes
     @property

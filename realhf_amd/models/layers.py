@@ -105,7 +105,7 @@ class ReaLModelBlock(nn.Module):
     """Pre-LN transformer block on packed sequences (reference:
     real_llm_base.py:103, modules/attn.py:31, modules/mlp.py)."""
 
-    def __init__(self, cfg: ReaLModelConfig, layer_idx: int, params: Dict[str, torch.Tensor], tp_size: int = 1):
+    def __init__(self, cfg: ReaLModelConfig, layer_idx: int, params: Dict[str, torch.Tensor], tp_size: int = 1, ep_rank: int = 0, ep_size: int = 1):
         super().__init__()
         self.cfg = cfg
         self.i = layer_idx
@@ -126,7 +126,8 @@ class ReaLModelBlock(nn.Module):
         if cfg.moe is not None:
             from realhf_amd.models.moe import MoELayer
 
-            self.moe = MoELayer(cfg, layer_idx, params, tp_size)
+            self.moe = MoELayer(cfg, layer_idx, params, tp_size,
+                                ep_rank=ep_rank, ep_size=ep_size)
 
     # -- attention --------------------------------------------------------
     def _qkv(self, x):

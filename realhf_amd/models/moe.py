@@ -1,26 +1,27 @@
-"""Mixture-of-Experts layer: top-k router + experts.
+"""Mixture-of-Experts layer: top-k router + dispatcher + experts.
 
 Reference semantics: realhf/impl/model/modules/moe/ (TopKRouter router.py:24,
 MoETokenDispatcher token_dispatcher.py:17, SequentialMLP/GroupedMLP
 experts.py, aux loss utils/moe.py:13).
 
-Beyond the reference: true expert parallelism (EP) — experts sharded over
-an EP group with all-to-all token exchange over xGMI (the reference
-replicates all experts; SURVEY.md §2.3 row EP).  EP activates when the
-model's grid has ep_size > 1 (see parallel/ep.py); otherwise experts are
-local and dispatch is a pure permutation.
+Beyond the reference: TRUE expert parallelism — the reference replicates
+all experts and stubs the expert group to self (megatron.py:108, SURVEY.md
+§2.3 row EP).  Here experts shard over an EP group (a block of DP ranks)
+with token all-to-all over xGMI; EP degree = cfg.moe.expert_parallel_size.
 
-Grouped expert GEMMs run through the hand-written HIP grouped-GEMM kernel
-on GPU (ops/csrc/grouped_gemm.hip) and a per-expert loop on CPU.
+Expert GEMMs run through the hand-written MFMA grouped-GEMM kernel
+(ops/csrc/grouped_gemm.hip) on the rollout/inference path; the training
+path uses per-expert GEMMs (rocBLAS) so autograd handles the backward
+(hand-written grouped bwd is a noted follow-up).
 """
 from typing import Dict, List, Optional
 
 import torch
+import torch.distributed as dist
 import torch.nn.functional as F
 
 from realhf_amd.api.model import ReaLModelConfig
 from realhf_amd.base import constants
-from realhf_amd.ops import functional as ops
 from realhf_amd.parallel import mappings
 
 # aux losses collected during forward; training interfaces drain this.
@@ -63,8 +64,13 @@ class TopKRouter(torch.nn.Module):
         self.weight = weight  # [n_experts, hidden]
 
     def forward(self, h: torch.Tensor):
-        logits = F.linear(h.float(), self.weight.float())
         moe = self.moe
+        if moe.input_jitter_eps and self.training:
+            noise = torch.empty_like(h).uniform_(
+                1.0 - moe.input_jitter_eps, 1.0 + moe.input_jitter_eps
+            )
+            h = h * noise
+        logits = F.linear(h.float(), self.weight.float())
         if moe.routing_type == "sinkhorn" and self.training:
             with torch.no_grad():
                 norm = sinkhorn(logits)
@@ -87,49 +93,175 @@ class TopKRouter(torch.nn.Module):
         return scores, idx
 
 
-class MoELayer(torch.nn.Module):
-    """Dispatch + expert MLPs + combine.  Token dispatch is a sort-based
-    permutation; expert GEMMs are grouped."""
+class _AllToAll(torch.autograd.Function):
+    """Autograd-wrapped all_to_all_single over the EP group: backward is
+    the reverse exchange (xGMI all-pairs — the natural fit)."""
 
-    def __init__(self, cfg: ReaLModelConfig, layer_idx: int, params: Dict[str, torch.Tensor], tp_size: int):
+    @staticmethod
+    def forward(ctx, x, out_splits, in_splits, group):
+        ctx.group = group
+        ctx.out_splits, ctx.in_splits = out_splits, in_splits
+        out = x.new_empty(sum(out_splits), *x.shape[1:])
+        dist.all_to_all_single(
+            out, x.contiguous(), output_split_sizes=out_splits,
+            input_split_sizes=in_splits, group=group,
+        )
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        back = g.new_empty(sum(ctx.in_splits), *g.shape[1:])
+        dist.all_to_all_single(
+            back, g.contiguous(), output_split_sizes=ctx.in_splits,
+            input_split_sizes=ctx.out_splits, group=ctx.group,
+        )
+        return back, None, None, None
+
+
+class MoELayer(torch.nn.Module):
+    """Dispatch + expert MLPs + combine; sort-based permutation, optional
+    EP all-to-all, grouped expert GEMMs."""
+
+    def __init__(self, cfg: ReaLModelConfig, layer_idx: int,
+                 params: Dict[str, torch.Tensor], tp_size: int,
+                 ep_rank: int = 0, ep_size: int = 1):
         super().__init__()
         self.cfg = cfg
         self.i = layer_idx
         self.p = params
         self.tp_size = tp_size
+        self.ep_rank, self.ep_size = ep_rank, ep_size
         self.router = TopKRouter(cfg, params[f"{layer_idx}.mlp.router.weight"])
         self.n_experts = cfg.moe.num_experts
+        assert self.n_experts % ep_size == 0
+        self.n_local = self.n_experts // ep_size
+        self.local_e0 = ep_rank * self.n_local
 
-    def _expert_ffn(self, e: int, x: torch.Tensor) -> torch.Tensor:
-        i = self.i
+    # ---------------------------------------------------------------- mlp
+    def _expert_ffn(self, e_local: int, x: torch.Tensor) -> torch.Tensor:
+        i, e = self.i, self.local_e0 + e_local
         gate = F.linear(x, self.p[f"{i}.mlp.experts.{e}.gate.weight"])
         up = F.linear(x, self.p[f"{i}.mlp.experts.{e}.up.weight"])
         act = (F.silu(gate.float()) * up.float()).to(x.dtype)
         return F.linear(act, self.p[f"{i}.mlp.experts.{e}.down.weight"])
 
+    def _expert_weight_stack(self, part: str):
+        """[E_local, N, K] expert-strided view when the flat layout is
+        uniform; None otherwise."""
+        i = self.i
+        ws = [
+            self.p[f"{i}.mlp.experts.{self.local_e0 + e}.{part}.weight"]
+            for e in range(self.n_local)
+        ]
+        base = ws[0]
+        if len(ws) == 1:
+            return base.unsqueeze(0)
+        stride = ws[1].storage_offset() - base.storage_offset()
+        for a, b in zip(ws, ws[1:]):
+            if b.storage_offset() - a.storage_offset() != stride:
+                return None
+        return base.as_strided(
+            (self.n_local, base.shape[0], base.shape[1]),
+            (stride, base.shape[1], 1), base.storage_offset(),
+        )
+
+    def _experts_forward(self, x_sorted: torch.Tensor, counts_cpu: torch.Tensor):
+        """x_sorted: tokens grouped by local expert; counts_cpu [n_local]."""
+        use_grouped = (
+            x_sorted.is_cuda
+            and x_sorted.dtype == torch.bfloat16
+            and not torch.is_grad_enabled()
+        )
+        if use_grouped:
+            from realhf_amd import ops as _ops_pkg
+
+            C = _ops_pkg.require_hip()
+            wg = self._expert_weight_stack("gate")
+            wu = self._expert_weight_stack("up")
+            wd = self._expert_weight_stack("down")
+            idim = wg.shape[1] if wg is not None else 0
+            if (
+                wg is not None and wu is not None and wd is not None
+                and x_sorted.shape[1] % 32 == 0 and idim % 64 == 0
+                and x_sorted.shape[1] % 64 == 0
+            ):
+                gate = C.grouped_gemm(x_sorted, wg, counts_cpu)
+                up = C.grouped_gemm(x_sorted, wu, counts_cpu)
+                act = (F.silu(gate.float()) * up.float()).to(x_sorted.dtype)
+                return C.grouped_gemm(act, wd, counts_cpu)
+        outs = torch.empty(
+            x_sorted.shape[0],
+            self.p[f"{self.i}.mlp.experts.{self.local_e0}.down.weight"].shape[0],
+            dtype=x_sorted.dtype, device=x_sorted.device,
+        )
+        start = 0
+        for e in range(self.n_local):
+            n = int(counts_cpu[e])
+            if n == 0:
+                continue
+            outs[start:start + n] = self._expert_ffn(e, x_sorted[start:start + n])
+            start += n
+        return outs
+
+    # ------------------------------------------------------------ forward
     def forward(self, h: torch.Tensor) -> torch.Tensor:
         scores, idx = self.router(h)  # [tokens, k]
         k = self.cfg.moe.top_k
-        tokens = h.shape[0]
         h_tp = mappings.copy_to_tp_region(h)
 
-        flat_idx = idx.flatten()  # [tokens*k]
+        flat_idx = idx.flatten()  # [tokens*k] global expert ids
         sort_order = torch.argsort(flat_idx, stable=True)
-        token_of = sort_order // k  # source token of each dispatched slot
+        token_of = sort_order // k
         counts = torch.bincount(flat_idx, minlength=self.n_experts)
-
         permuted = h_tp[token_of]
-        outs = torch.empty_like(permuted)
-        start = 0
-        for e in range(self.n_experts):
-            n = int(counts[e])
-            if n == 0:
-                continue
-            outs[start : start + n] = self._expert_ffn(e, permuted[start : start + n])
-            start += n
-        # un-permute and combine with routing weights
+
+        if self.ep_size > 1:
+            out_sorted, inv = self._ep_exchange_and_compute(permuted, counts)
+        else:
+            counts_cpu = counts.cpu()
+            out_sorted = self._experts_forward(permuted, counts_cpu)
+            inv = None
+
         combined = torch.zeros_like(h_tp, dtype=torch.float32)
         w = scores.to(torch.float32).flatten()[sort_order]
-        combined.index_add_(0, token_of, outs.float() * w.unsqueeze(-1))
+        combined.index_add_(0, token_of, out_sorted.float() * w.unsqueeze(-1))
         combined = combined.to(h.dtype)
         return mappings.reduce_from_tp_region(combined)
+
+    def _ep_exchange_and_compute(self, permuted, counts):
+        """All-to-all token exchange over the EP group, local expert
+        compute, reverse exchange.  Returns outputs aligned with the
+        (globally expert-sorted) `permuted` order."""
+        g = constants.grid()
+        group = g.ep_group()
+        counts_cpu = counts.cpu()
+        send_splits = [
+            int(counts_cpu[r * self.n_local:(r + 1) * self.n_local].sum())
+            for r in range(self.ep_size)
+        ]
+        # exchange counts so each rank knows what it receives per expert
+        cmat = counts.to(torch.long)
+        recv_cmat = torch.empty_like(cmat)
+        dist.all_to_all_single(
+            recv_cmat, cmat,
+            output_split_sizes=[self.n_local] * self.ep_size,
+            input_split_sizes=[self.n_local] * self.ep_size,
+            group=group,
+        )
+        recv_cmat_cpu = recv_cmat.cpu().view(self.ep_size, self.n_local)
+        recv_splits = [int(recv_cmat_cpu[r].sum()) for r in range(self.ep_size)]
+
+        x_recv = _AllToAll.apply(permuted, recv_splits, send_splits, group)
+        # re-sort received tokens (grouped by src rank, each sorted by
+        # local expert) into a single local-expert-major order
+        local_ids = torch.repeat_interleave(
+            torch.arange(self.n_local, device=permuted.device).repeat(self.ep_size),
+            recv_cmat.flatten().clamp(min=0),
+        )
+        order = torch.argsort(local_ids, stable=True)
+        local_counts_cpu = recv_cmat_cpu.sum(0)
+        y = self._experts_forward(x_recv[order], local_counts_cpu)
+        y_unsorted = torch.empty_like(y)
+        y_unsorted[order] = y
+        out = _AllToAll.apply(y_unsorted, send_splits, recv_splits, group)
+        return out, None

@@ -292,14 +292,33 @@ class FlatLayout:
         return s.start, s.end
 
 
+def filter_keys_for_ep(cfg: ReaLModelConfig, keys: List[str], ep_rank: int,
+                       ep_size: int) -> List[str]:
+    """With expert parallelism, each rank stores only its expert block."""
+    if ep_size <= 1 or cfg.moe is None:
+        return keys
+    n_local = cfg.moe.num_experts // ep_size
+    lo, hi = ep_rank * n_local, (ep_rank + 1) * n_local
+    out = []
+    for k in keys:
+        if ".experts." in k:
+            e = int(k.split(".experts.")[1].split(".")[0])
+            if not (lo <= e < hi):
+                continue
+        out.append(k)
+    return out
+
+
 def build_flat_layout(
     cfg: ReaLModelConfig,
     layer_indices: List[int],
     tp_rank: int,
     tp_size: int,
     alignment: int = 64,
+    ep_rank: int = 0,
+    ep_size: int = 1,
 ) -> FlatLayout:
-    keys = keys_of_layers(cfg, layer_indices)
+    keys = filter_keys_for_ep(cfg, keys_of_layers(cfg, layer_indices), ep_rank, ep_size)
     specs = {}
     off = 0
     for k in keys:

@@ -34,6 +34,8 @@ class ReaLModel(nn.Module):
         tp_size: int = 1,
         pp_rank: int = 0,
         pp_size: int = 1,
+        ep_rank: int = 0,
+        ep_size: int = 1,
         instantiate: bool = True,
     ):
         super().__init__()
@@ -42,10 +44,14 @@ class ReaLModel(nn.Module):
         self.dtype = dtype or config.torch_dtype
         self.tp_rank, self.tp_size = tp_rank, tp_size
         self.pp_rank, self.pp_size = pp_rank, pp_size
+        self.ep_rank, self.ep_size = ep_rank, ep_size
         self.pipeline_partition = PL.partition_pipeline_layers(config, pp_size)
         lo, hi = self.pipeline_partition[pp_rank]
         self.layer_indices = list(range(lo, hi))
-        self.layout = PL.build_flat_layout(config, self.layer_indices, tp_rank, tp_size)
+        self.layout = PL.build_flat_layout(
+            config, self.layer_indices, tp_rank, tp_size,
+            ep_rank=ep_rank, ep_size=ep_size,
+        )
 
         self.flat_param: Optional[torch.Tensor] = None
         self.flat_grad: Optional[torch.Tensor] = None
@@ -98,7 +104,8 @@ class ReaLModel(nn.Module):
                 self.layers.append(OutputHead(cfg, self._params, tied_w))
             else:
                 self.layers.append(
-                    ReaLModelBlock(cfg, idx, self._params, self.tp_size)
+                    ReaLModelBlock(cfg, idx, self._params, self.tp_size,
+                                   ep_rank=self.ep_rank, ep_size=self.ep_size)
                 )
 
     # ------------------------------------------------------------------

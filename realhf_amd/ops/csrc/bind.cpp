@@ -24,6 +24,8 @@ std::vector<torch::Tensor> attn_varlen_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
+                           torch::Tensor seg_lens_cpu);
 torch::Tensor rope_qkv_decode(
     torch::Tensor qkv, c10::optional<torch::Tensor> bias, torch::Tensor kcache,
     torch::Tensor vcache, torch::Tensor cache_seqlens, torch::Tensor cosb,
@@ -43,4 +45,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_varlen_fwd", &attn_varlen_fwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("rope_qkv_decode", &rope_qkv_decode);
+  m.def("grouped_gemm", &grouped_gemm);
 }

@@ -20,6 +20,7 @@ SRC = [
     "realhf_amd/ops/csrc/interval.hip",
     "realhf_amd/ops/csrc/attn_decode.hip",
     "realhf_amd/ops/csrc/rope_decode.hip",
+    "realhf_amd/ops/csrc/grouped_gemm.hip",
     "realhf_amd/ops/csrc/attn_varlen.hip",
 ]
 

@@ -1,0 +1,101 @@
+"""Expert parallelism tests (capability beyond the reference — SURVEY.md
+§2.3 row EP: the reference replicates experts)."""
+import numpy as np
+import pytest
+import torch
+
+from realhf_amd.base.testing import LocalMultiProcessTest
+from realhf_amd.models.hf import mixtral
+from tests.test_realloc import _fill_model_from_full, _full_reference_sd
+
+
+def _ep_worker():
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.topology import ParallelGrid, PipeDataTensorTopology
+    from realhf_amd.models.real_model import ReaLModel
+
+    cfg = mixtral.make_test_config(n_layers=2, hidden_dim=64, n_heads=8,
+                                   n_kv_heads=4, vocab_size=128)
+    cfg.dtype = "float32"
+    cfg.moe.expert_parallel_size = 2
+    sd = _full_reference_sd(cfg, seed=61)
+    rank = dist.get_rank()
+    topo = PipeDataTensorTopology(num_pp=1, num_dp=2, num_tp=1, ep_size=2)
+    grid = ParallelGrid(topo)
+    constants.set_grid("m", grid)
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                  ep_rank=grid.ep_rank, ep_size=2)
+    _fill_model_from_full(m, cfg, sd)
+    m.eval()
+
+    single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    _fill_model_from_full(single, cfg, sd)
+    single.eval()
+
+    rng = np.random.RandomState(3)
+    lens = [7, 9]
+    packed = torch.from_numpy(rng.randint(0, 128, size=sum(lens))).long()
+    cu = torch.tensor([0] + list(np.cumsum(lens)), dtype=torch.int32)
+    with torch.no_grad():
+        ref = single(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=9)
+        with constants.model_scope("m"):
+            out = m(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=9)
+    torch.testing.assert_close(out, ref, atol=1e-4, rtol=1e-4)
+    # EP halves the expert parameter footprint
+    n_full = single.layout.total_numel
+    n_ep = m.layout.total_numel
+    assert n_ep < n_full
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_ep2_forward_matches_replicated():
+    LocalMultiProcessTest(2, _ep_worker).launch()
+
+
+def _ep_train_worker():
+    """EP backward: gradients flow through the all-to-all."""
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.topology import ParallelGrid, PipeDataTensorTopology
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel.tp import packed_shifted_logprobs
+
+    cfg = mixtral.make_test_config(n_layers=2, hidden_dim=64, n_heads=8,
+                                   n_kv_heads=4, vocab_size=128)
+    cfg.dtype = "float32"
+    cfg.moe.expert_parallel_size = 2
+    sd = _full_reference_sd(cfg, seed=71)
+    topo = PipeDataTensorTopology(num_pp=1, num_dp=2, num_tp=1, ep_size=2)
+    grid = ParallelGrid(topo)
+    constants.set_grid("m", grid)
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                  ep_rank=grid.ep_rank, ep_size=2)
+    _fill_model_from_full(m, cfg, sd)
+    m.allocate_grad_buffer()
+    for k, p in m._params.items():
+        p.requires_grad_(True)
+        p.grad = m.grad_view(k)
+
+    rng = np.random.RandomState(4)
+    lens = [8, 8]
+    packed = torch.from_numpy(rng.randint(0, 128, size=16)).long()
+    cu = torch.tensor([0, 8, 16], dtype=torch.int32)
+    with constants.model_scope("m"):
+        logits = m(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=8)
+        logp = packed_shifted_logprobs(logits, cu, packed)
+        loss = -logp.mean()
+        loss.backward()
+    assert torch.isfinite(m.flat_grad).all()
+    # expert grads nonzero somewhere
+    gsum = float(m.flat_grad.abs().sum())
+    assert gsum > 0
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_ep2_backward_through_all_to_all():
+    LocalMultiProcessTest(2, _ep_train_worker).launch()
