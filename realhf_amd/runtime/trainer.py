@@ -408,9 +408,12 @@ class Trainer:
                 dt = time.time() - t0
                 self.global_step += 1
                 if self.rank == 0:
+                    toks = sum(batch.main_seqlens())
                     logger.info(
-                        "epoch %d step %d (global %d): %.2fs %s",
-                        epoch, i, self.global_step, dt,
+                        "epoch %d step %d (global %d): %.2fs (%.0f tok/s, "
+                        "%.2f samples/s) %s",
+                        epoch, i, self.global_step, dt, toks / dt,
+                        batch.bs / dt,
                         {k: round(v, 4) for k, v in stats.items()
                          if isinstance(v, float)},
                     )
