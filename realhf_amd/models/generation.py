@@ -48,9 +48,23 @@ def _sample_from_logits(
         tokens = logits.argmax(dim=-1)
         logp = torch.log_softmax(logits, dim=-1)
         return tokens, logp.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
-    filtered = top_k_top_p_logits(logits, gconfig.top_k, gconfig.top_p)
-    probs = torch.softmax(filtered, dim=-1)
-    tokens = torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+    V = logits.shape[-1]
+    k = gconfig.top_k if 0 < gconfig.top_k < V else V
+    if k < V:
+        # work in the compacted top-k space: selection + a k-long cumsum
+        # instead of a full-vocab sort per token (the old path sorted all
+        # 32k logits every decode step)
+        vals, idx = torch.topk(logits, k, dim=-1)  # sorted descending
+    else:
+        vals, idx = torch.sort(logits, descending=True, dim=-1)
+    probs = torch.softmax(vals, dim=-1)
+    if 0.0 < gconfig.top_p < 1.0:
+        cum = probs.cumsum(dim=-1)
+        remove = cum - probs > gconfig.top_p
+        vals = vals.masked_fill(remove, float("-inf"))
+        probs = torch.softmax(vals, dim=-1)
+    sel = torch.multinomial(probs, 1, generator=generator)
+    tokens = idx.gather(-1, sel).squeeze(-1)
     logp_all = torch.log_softmax(logits, dim=-1)  # logprob under UNFILTERED dist
     return tokens, logp_all.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
 

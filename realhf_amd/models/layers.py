@@ -186,6 +186,7 @@ class ReaLModelBlock(nn.Module):
             and h.is_cuda
             and h.dtype == torch.bfloat16
             and not cfg.qk_layernorm
+            and self.hd in (64, 128)
             and _os.environ.get("REALHF_AMD_NO_FUSED_DECODE") != "1"
         ):
             from realhf_amd import ops as _ops_pkg
@@ -218,6 +219,12 @@ class ReaLModelBlock(nn.Module):
             o = mappings.reduce_from_tp_region(o)
             if f"{i}.attn.wo.bias" in self.p:
                 o = o + self.p[f"{i}.attn.wo.bias"]
+            if cfg.norm_type == "rms" and self.moe is None:
+                # fused (x + o) + mlp-norm, then the MLP body
+                h2, x2 = C.add_rmsnorm_fwd(
+                    o, x, self.p[f"{i}.mlp.ln.weight"], cfg.layer_norm_epsilon
+                )
+                return x2 + self._mlp_body(h2, sp)
             x = x + o
             return self._mlp(x, sp)
 
@@ -277,9 +284,13 @@ class ReaLModelBlock(nn.Module):
         i = self.i
         h = _norm(cfg, x, self.p[f"{i}.mlp.ln.weight"], self.p.get(f"{i}.mlp.ln.bias"))
         if self.moe is not None:
-            m = self.moe(h)
-            x = x + m
-            return x
+            return x + self.moe(h)
+        return x + self._mlp_body(h, sp)
+
+    def _mlp_body(self, h, sp):
+        """norm-output -> MLP delta (no residual add)."""
+        cfg = self.cfg
+        i = self.i
         if sp:
             h = mappings.gather_from_sp_region(h)
         else:
@@ -310,7 +321,7 @@ class ReaLModelBlock(nn.Module):
             down = mappings.reduce_from_tp_region(down)
         if f"{i}.mlp.down.bias" in self.p:
             down = down + self.p[f"{i}.mlp.down.bias"]
-        return x + down
+        return down
 
 
 class OutputHead(nn.Module):

@@ -5,6 +5,8 @@
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                        torch::Tensor w, torch::Tensor rstd);
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor resid,
+                                           torch::Tensor w, double eps);
 torch::Tensor rope_fwd(torch::Tensor x, torch::Tensor cosb, torch::Tensor sinb,
                        torch::Tensor positions, bool interleaved, bool conj);
 torch::Tensor swiglu_fwd(torch::Tensor gu);
@@ -34,6 +36,7 @@ torch::Tensor rope_qkv_decode(
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("add_rmsnorm_fwd", &add_rmsnorm_fwd);
   m.def("rope_fwd", &rope_fwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);

@@ -4,10 +4,11 @@
 // Replaces the reference's TransformerEngine RMSNorm (SURVEY.md §2.2 ext deps).
 #include "common.h"
 
-template <typename T, int VEC>
+template <typename T, int VEC, bool RESID = false>
 __global__ void rmsnorm_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ out,
-    float* __restrict__ rstd, int rows, int H, float eps) {
+    float* __restrict__ rstd, int rows, int H, float eps,
+    const T* __restrict__ resid = nullptr, T* __restrict__ sum_out = nullptr) {
   __shared__ float red[8];
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + (long)row * H;
@@ -16,6 +17,14 @@ __global__ void rmsnorm_fwd_kernel(
     for (int i = threadIdx.x * VEC; i < H; i += blockDim.x * VEC) {
       T v[VEC];
       *(float4v*)v = *(const float4v*)(xr + i);  // 16B when VEC matches
+      if constexpr (RESID) {
+        T r[VEC];
+        *(float4v*)r = *(const float4v*)(resid + (long)row * H + i);
+        #pragma unroll
+        for (int j = 0; j < VEC; j++)
+          v[j] = from_f32<T>(to_f32<T>(v[j]) + to_f32<T>(r[j]));
+        *(float4v*)(sum_out + (long)row * H + i) = *(float4v*)v;
+      }
       #pragma unroll
       for (int j = 0; j < VEC; j++) { float f = to_f32<T>(v[j]); ss += f * f; }
     }
@@ -32,9 +41,10 @@ __global__ void rmsnorm_fwd_kernel(
     __syncthreads();
     float rs = rsqrtf(red[0] / H + eps);
     if (threadIdx.x == 0 && rstd) rstd[row] = rs;
+    const T* src2 = RESID ? (sum_out + (long)row * H) : xr;
     for (int i = threadIdx.x * VEC; i < H; i += blockDim.x * VEC) {
       T v[VEC], wv[VEC], o[VEC];
-      *(float4v*)v = *(const float4v*)(xr + i);
+      *(float4v*)v = *(const float4v*)(src2 + i);
       *(float4v*)wv = *(const float4v*)(w + i);
       #pragma unroll
       for (int j = 0; j < VEC; j++)
@@ -134,6 +144,30 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double 
   });
   CHECK_CUDA_OK();
   return {out, rstd};
+}
+
+std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor resid,
+                                           torch::Tensor w, double eps) {
+  // out_norm = rmsnorm(x + resid); sum_out = x + resid (the new residual)
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && resid.is_contiguous());
+  TORCH_CHECK(x.element_size() == 2);
+  int H = x.size(-1);
+  long rows = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0);
+  auto out = torch::empty_like(x);
+  auto sum_out = torch::empty_like(x);
+  int grid = (int)std::min<long>(rows, 2048);
+  DISPATCH_BF16_FP16_FP32(x.scalar_type(), "add_rmsnorm", [&] {
+    if constexpr (sizeof(scalar_t) == 2) {
+      hipLaunchKernelGGL((rmsnorm_fwd_kernel<scalar_t, 8, true>), dim3(grid),
+        dim3(256), 0, cur_stream(), (const scalar_t*)x.data_ptr(),
+        (const scalar_t*)w.data_ptr(), (scalar_t*)out.data_ptr(),
+        nullptr, (int)rows, H, (float)eps,
+        (const scalar_t*)resid.data_ptr(), (scalar_t*)sum_out.data_ptr());
+    }
+  });
+  CHECK_CUDA_OK();
+  return {out, sum_out};
 }
 
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,

@@ -291,11 +291,14 @@ class _AttnVarlenFn(torch.autograd.Function):
 
 def attn_varlen(q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=None):
     scale = softmax_scale or (1.0 / math.sqrt(q.shape[-1]))
-    if _ops.use_hip(q):
+    if _ops.use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
         return _AttnVarlenFn.apply(
             q.contiguous(), k.contiguous(), v.contiguous(),
             cu_seqlens, max_seqlen, causal, scale,
         )
+    if q.is_cuda:
+        # odd head dims / dtypes: batched rocBLAS path (still GPU)
+        return _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale)
     return attn_varlen_ref(q, k, v, cu_seqlens, causal, scale)
 
 
@@ -330,7 +333,12 @@ def attn_decode_ref(
 
 def attn_decode(q, k_cache, v_cache, cache_seqlens, softmax_scale=None):
     scale = softmax_scale or (1.0 / math.sqrt(q.shape[-1]))
-    if _ops.use_hip(q):
+    if (
+        _ops.use_hip(q)
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] in (64, 128)
+        and q.shape[1] // k_cache.shape[2] in (1, 2, 4, 8)
+    ):
         C = _ops.require_hip()
         return C.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
     return attn_decode_ref(q, k_cache, v_cache, cache_seqlens, scale)

@@ -273,8 +273,8 @@ def test_moe_grouped_path_matches_loop():
 
     fam = hf_reg.get_family("mixtral")
     cfg = fam.make_test_config(
-        n_layers=1, hidden_dim=64, n_heads=4, n_kv_heads=2, vocab_size=128,
-        intermediate_dim=128, head_dim=16,
+        n_layers=1, hidden_dim=64, n_heads=1, n_kv_heads=1, vocab_size=128,
+        intermediate_dim=128, head_dim=64,
     )
     torch.manual_seed(13)
     m = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
@@ -292,3 +292,15 @@ def test_moe_grouped_path_matches_loop():
     torch.testing.assert_close(
         out_grouped.float(), out_loop.float(), atol=5e-2, rtol=5e-2
     )
+
+
+def test_add_rmsnorm_fused():
+    torch.manual_seed(14)
+    x = torch.randn(40, 4096, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(40, 4096, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(4096, dtype=torch.bfloat16, device="cuda")
+    out, s = C.add_rmsnorm_fwd(x, res, w, 1e-5)
+    s_ref = (x.float() + res.float()).to(torch.bfloat16)
+    ref = F.rms_norm_ref(s_ref.float(), w.float(), 1e-5)
+    torch.testing.assert_close(s.float(), s_ref.float(), atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
