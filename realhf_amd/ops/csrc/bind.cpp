@@ -28,6 +28,8 @@ std::vector<torch::Tensor> attn_varlen_fwd(
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
                            torch::Tensor seg_lens_cpu);
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
+                          torch::Tensor out32_ws, long splitk);
 torch::Tensor rope_qkv_decode(
     torch::Tensor qkv, c10::optional<torch::Tensor> bias, torch::Tensor kcache,
     torch::Tensor vcache, torch::Tensor cache_seqlens, torch::Tensor cosb,
@@ -49,4 +51,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("rope_qkv_decode", &rope_qkv_decode);
   m.def("grouped_gemm", &grouped_gemm);
+  m.def("skinny_gemm", &skinny_gemm);
 }

@@ -1,0 +1,59 @@
+#!/usr/bin/env python3
+"""A/B the weight-streaming skinny GEMM vs torch.matmul (hipBLASLt) on the
+7B decode shapes.  Run on an MI355X."""
+import time
+
+import torch
+
+import realhf_amd._C as C
+
+SHAPES = [  # (name, N, K)
+    ("qkv", 12288, 4096),
+    ("o", 4096, 4096),
+    ("gateup", 22016, 4096),
+    ("down", 4096, 11008),
+    ("lmhead", 32000, 4096),
+]
+
+
+def pick_splitk(N, K):
+    sk = 1
+    while (N // 64) * sk < 512 and sk < 16:
+        sk *= 2
+    while K // sk > 1024:
+        sk *= 2
+    return sk
+
+
+def bench(fn, iters=50):
+    for _ in range(5):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6
+
+
+def main():
+    M = 16
+    ws = torch.empty(M * 32000, dtype=torch.float32, device="cuda")
+    print(f"{'shape':8} {'N':>6} {'K':>6} {'sk':>3} {'blaslt_us':>9} "
+          f"{'skinny_us':>9} {'roofline_us':>11} {'max_err':>8}")
+    for name, N, K in SHAPES:
+        x = (torch.randn(M, K, device="cuda") * 0.3).to(torch.bfloat16)
+        w = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)
+        sk = pick_splitk(N, K)
+        ref = x.float() @ w.float().t()
+        out = C.skinny_gemm(x, w, ws, sk)
+        err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+        t_blas = bench(lambda: torch.matmul(x, w.t()))
+        t_sk = bench(lambda: C.skinny_gemm(x, w, ws, sk))
+        roof = N * K * 2 / 6.3e12 * 1e6
+        print(f"{name:8} {N:6d} {K:6d} {sk:3d} {t_blas:9.1f} {t_sk:9.1f} "
+              f"{roof:11.1f} {err:8.4f}")
+
+
+if __name__ == "__main__":
+    main()
