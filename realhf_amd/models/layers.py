@@ -215,7 +215,10 @@ class ReaLModelBlock(nn.Module):
             )
             attn_out = ops.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
             attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
-            o = _linear(attn_out, self.p[f"{i}.attn.wo.weight"])
+            wo = self.p[f"{i}.attn.wo.weight"]
+            o = ops.maybe_skinny_linear(attn_out, wo)
+            if o is None:
+                o = _linear(attn_out, wo)
             o = mappings.reduce_from_tp_region(o)
             if f"{i}.attn.wo.bias" in self.p:
                 o = o + self.p[f"{i}.attn.wo.bias"]
@@ -314,7 +317,10 @@ class ReaLModelBlock(nn.Module):
         else:
             up = _linear(h, self.p[f"{i}.mlp.up.weight"], self.p.get(f"{i}.mlp.up.bias"))
             act = F.gelu(up, approximate="tanh")
-        down = _linear(act, self.p[f"{i}.mlp.down.weight"])
+        wd = self.p[f"{i}.mlp.down.weight"]
+        down = ops.maybe_skinny_linear(act, wd)
+        if down is None:
+            down = _linear(act, wd)
         if sp:
             down = mappings.reduce_scatter_to_sp_region(down)
         else:

@@ -46,15 +46,24 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
   f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
   const bf16* wrow = w + (long)(n0 + i16) * K;
   int kk = k_lo;
-  for (; kk + 64 <= k_hi; kk += 64) {
-    bf16x8v a0 = *(const bf16x8v*)(&x_s[(long)i16 * kslice + (kk - k_lo) + g * 8]);
+  // 128-deep body: issue all four W loads before the first MFMA so >=4
+  // 16B loads stay in flight per lane (one k-iter alone is latency-bound)
+  for (; kk + 128 <= k_hi; kk += 128) {
     bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
-    bf16x8v a1 = *(const bf16x8v*)(&x_s[(long)i16 * kslice + (kk - k_lo) + 32 + g * 8]);
     bf16x8v b1 = *(const bf16x8v*)(wrow + kk + 32 + g * 8);
+    bf16x8v b2 = *(const bf16x8v*)(wrow + kk + 64 + g * 8);
+    bf16x8v b3 = *(const bf16x8v*)(wrow + kk + 96 + g * 8);
+    const long xb = (long)i16 * kslice + (kk - k_lo) + g * 8;
+    bf16x8v a0 = *(const bf16x8v*)(&x_s[xb]);
+    bf16x8v a1 = *(const bf16x8v*)(&x_s[xb + 32]);
+    bf16x8v a2 = *(const bf16x8v*)(&x_s[xb + 64]);
+    bf16x8v a3 = *(const bf16x8v*)(&x_s[xb + 96]);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
     acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc1, 0, 0, 0);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b3, acc1, 0, 0, 0);
   }
-  for (; kk < k_hi; kk += 32) {
+  for (; kk + 32 <= k_hi; kk += 32) {
     bf16x8v a0 = *(const bf16x8v*)(&x_s[(long)i16 * kslice + (kk - k_lo) + g * 8]);
     bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
     acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);

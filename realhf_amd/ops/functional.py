@@ -433,6 +433,41 @@ def merge_intervals(intervals: List[Tuple[int, int]]) -> List[Tuple[int, int]]:
 
 
 # ---------------------------------------------------------------------------
+# Weight-streaming skinny GEMM (decode M <= 16); wins on square-ish
+# projections (o/down: ~1.6-2x over hipBLASLt), loses on very wide N —
+# callers pick per shape (tools/bench_skinny.py is the A/B harness).
+# ---------------------------------------------------------------------------
+def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor) -> Optional[torch.Tensor]:
+    if not (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and not torch.is_grad_enabled()
+        and x.dim() == 2
+        and x.shape[0] <= 16
+        and x.stride(1) == 1
+        and x.stride(0) == x.shape[1]
+        and w.stride(1) == 1
+        and w.shape[0] % 64 == 0
+        and w.shape[1] % 32 == 0
+        and w.shape[0] <= 2 * w.shape[1]  # wide-N shapes: hipBLASLt wins
+        and _ops.hip_available()
+    ):
+        return None
+    from realhf_amd.base.constants import get_global_memory_buffer
+
+    C = _ops.require_hip()
+    M, K = x.shape
+    N = w.shape[0]
+    ws = get_global_memory_buffer().get_tensor(
+        (16 * N,), torch.float32, "skinny_gemm_ws"
+    )
+    splitk = 8
+    while K // splitk > 1024:
+        splitk *= 2
+    return C.skinny_gemm(x, w, ws, splitk)
+
+
+# ---------------------------------------------------------------------------
 # Fused AdamW on flat buffers
 # ---------------------------------------------------------------------------
 def fused_adamw_ref(

@@ -44,15 +44,17 @@ def main():
     for name, N, K in SHAPES:
         x = (torch.randn(M, K, device="cuda") * 0.3).to(torch.bfloat16)
         w = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)
-        sk = pick_splitk(N, K)
         ref = x.float() @ w.float().t()
-        out = C.skinny_gemm(x, w, ws, sk)
-        err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
         t_blas = bench(lambda: torch.matmul(x, w.t()))
-        t_sk = bench(lambda: C.skinny_gemm(x, w, ws, sk))
         roof = N * K * 2 / 6.3e12 * 1e6
-        print(f"{name:8} {N:6d} {K:6d} {sk:3d} {t_blas:9.1f} {t_sk:9.1f} "
-              f"{roof:11.1f} {err:8.4f}")
+        for sk in sorted({pick_splitk(N, K), 2, 4, 8, 16}):
+            if K // sk > 1024 or K % 32:
+                continue
+            out = C.skinny_gemm(x, w, ws, sk)
+            err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+            t_sk = bench(lambda: C.skinny_gemm(x, w, ws, sk))
+            print(f"{name:8} {N:6d} {K:6d} {sk:3d} {t_blas:9.1f} {t_sk:9.1f} "
+                  f"{roof:11.1f} {err:8.4f}")
 
 
 if __name__ == "__main__":
