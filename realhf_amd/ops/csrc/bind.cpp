@@ -30,6 +30,12 @@ torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
                            torch::Tensor seg_lens_cpu);
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           torch::Tensor out32_ws, long splitk);
+std::pair<std::vector<int64_t>, double> mcmc_search(
+    int64_t n_gpus, std::vector<std::vector<std::vector<double>>> cand_rows,
+    std::vector<std::vector<int64_t>> parents_in,
+    std::vector<int64_t> role_in, std::vector<double> realloc_cost_in,
+    int64_t n_strategies, double mem_cap_bytes, int64_t n_chains,
+    int64_t n_steps, int64_t seed);
 torch::Tensor rope_qkv_decode(
     torch::Tensor qkv, c10::optional<torch::Tensor> bias, torch::Tensor kcache,
     torch::Tensor vcache, torch::Tensor cache_seqlens, torch::Tensor cosb,
@@ -52,4 +58,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_qkv_decode", &rope_qkv_decode);
   m.def("grouped_gemm", &grouped_gemm);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("mcmc_search", &mcmc_search);
 }

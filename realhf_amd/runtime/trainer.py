@@ -84,8 +84,66 @@ class BuiltExperiment:
     trainable: List[ModelName]
 
 
+def _apply_search_allocation(cfg, world: int):
+    """allocation_mode=search: pick per-MFC strategies with the MCMC
+    search engine, then write them back into the model configs (reference:
+    apps/main.py experiment._search() -> search_rpc_allocations)."""
+    from realhf_amd.api.dfg import build_graph as _bg
+    from realhf_amd.models import hf as _hf
+    from realhf_amd.search.engine import MFCSpec, search_allocations
+
+    if not isinstance(cfg, PPOConfig):
+        logger.warning("allocation_mode=search currently tunes PPO only; "
+                       "using heuristic for %s", type(cfg).__name__)
+        cfg.allocation_mode = "heuristic"
+        return
+
+    def spec_of(name, role, mc, itype, gen_tokens=0, avg_seqlen=None):
+        if mc.path:
+            rcfg = _hf.config_from_hf_path(mc.family, mc.path)
+        else:
+            rcfg = _hf.get_family(mc.family).make_test_config()
+        p = rcfg.param_count()
+        return MFCSpec(
+            name=name, role=role, interface_type=itype,
+            n_seqs=cfg.dataset.train_bs_n_seqs,
+            avg_seqlen=avg_seqlen or (cfg.dataset.max_prompt_len
+                                      + cfg.ppo.gen.max_new_tokens),
+            gen_tokens=gen_tokens, param_bytes=p * 2.0,
+            flops_per_token=2.0 * p,
+        )
+
+    T = ModelInterfaceType
+    tmp = build_experiment(
+        dataclasses.replace(cfg, allocation_mode="heuristic"), world
+    )
+    specs = {}
+    for m in tmp.graph.mfcs:
+        role = m.model_name.role
+        mc = {"actor": cfg.actor, "critic": cfg.critic, "ref": cfg.ref,
+              "rew": cfg.rew}[role]
+        gen_toks = cfg.ppo.gen.max_new_tokens if m.interface_type == T.GENERATE else 0
+        avg = cfg.dataset.max_prompt_len if m.interface_type == T.GENERATE else None
+        specs[m.name] = spec_of(m.name, role, mc, m.interface_type,
+                                gen_tokens=gen_toks, avg_seqlen=avg)
+    alloc, cost = search_allocations(
+        tmp.graph, specs, trainable_roles=["actor", "critic"], n_gpus=world,
+    )
+    cfg.allocation_mode = "manual"
+    cfg.actor.parallel = alloc["actor_train"]
+    cfg.critic.parallel = alloc["critic_train"]
+    cfg.ref.parallel = alloc["ref_inf"]
+    cfg.rew.parallel = alloc["rew_inf"]
+    if alloc["actor_gen"] != alloc["actor_train"]:
+        cfg.actor.gen_parallel = alloc["actor_gen"]
+    logger.info("search allocation (est %.3fs/step): %s", cost,
+                {k: str(v) for k, v in alloc.items()})
+
+
 def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment:
     T = ModelInterfaceType
+    if cfg.allocation_mode == "search":
+        _apply_search_allocation(cfg, world)
 
     def model_cfg_of(mc, name):
         if mc.path:

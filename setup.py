@@ -14,6 +14,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 SRC = [
     "realhf_amd/ops/csrc/bind.cpp",
+    "realhf_amd/ops/csrc/mcmc_search.cpp",
     "realhf_amd/ops/csrc/rmsnorm.hip",
     "realhf_amd/ops/csrc/elementwise.hip",
     "realhf_amd/ops/csrc/gae.hip",
