@@ -206,7 +206,14 @@ def generate(
     cache_seqlens = prompt_lens.to(torch.int32).clone()
 
     decoder = sess.decoder
-    use_graph = gconfig.use_hip_graph and device.type == "cuda"
+    # graph capture requires the fully HIP decode path (the torch
+    # fallback for odd head dims reads tensors on the host)
+    use_graph = (
+        gconfig.use_hip_graph
+        and device.type == "cuda"
+        and model.dtype == torch.bfloat16
+        and cfg.head_dim in (64, 128)
+    )
 
     cur_logits = last_logits
     for t in range(max_new):
@@ -226,8 +233,14 @@ def generate(
             break
 
         cache_seqlens += 1  # the new token's slot
-        if use_graph and decoder.graph is None:
-            decoder.capture(tokens, cache_seqlens)
+        if use_graph and decoder.graph is None and not getattr(decoder, "_capture_failed", False):
+            try:
+                decoder.capture(tokens, cache_seqlens)
+            except Exception as e:  # graph-unsafe fallback path in use
+                logger.warning("hipGraph capture failed (%s); eager decode", e)
+                decoder.graph = None
+                decoder._capture_failed = True
+                torch.cuda.synchronize()
         logits_step = decoder.step(tokens, cache_seqlens)
         logits_step = mappings.gather_from_tp_region(logits_step).float()
         cur_logits = logits_step
