@@ -223,11 +223,12 @@ class ReaLModelBlock(nn.Module):
             if f"{i}.attn.wo.bias" in self.p:
                 o = o + self.p[f"{i}.attn.wo.bias"]
             if cfg.norm_type == "rms" and self.moe is None:
-                # fused (x + o) + mlp-norm, then the MLP body
+                # fused (x + o) + mlp-norm, then the MLP body (the down
+                # projection folds the x2 residual into its combine)
                 h2, x2 = C.add_rmsnorm_fwd(
                     o, x, self.p[f"{i}.mlp.ln.weight"], cfg.layer_norm_epsilon
                 )
-                return x2 + self._mlp_body(h2, sp)
+                return self._mlp_body(h2, sp, residual=x2)
             x = x + o
             return self._mlp(x, sp)
 
@@ -288,10 +289,12 @@ class ReaLModelBlock(nn.Module):
         h = _norm(cfg, x, self.p[f"{i}.mlp.ln.weight"], self.p.get(f"{i}.mlp.ln.bias"))
         if self.moe is not None:
             return x + self.moe(h)
-        return x + self._mlp_body(h, sp)
+        return x + self._mlp_body(h, sp)  # residual not folded here
 
-    def _mlp_body(self, h, sp):
-        """norm-output -> MLP delta (no residual add)."""
+    def _mlp_body(self, h, sp, residual=None):
+        """norm-output -> MLP delta; with `residual` (tp==1 decode path)
+        the down-projection's combine kernel adds it in and the return
+        value is the full residual-stream output."""
         cfg = self.cfg
         i = self.i
         if sp:
@@ -318,15 +321,26 @@ class ReaLModelBlock(nn.Module):
             up = _linear(h, self.p[f"{i}.mlp.up.weight"], self.p.get(f"{i}.mlp.up.bias"))
             act = F.gelu(up, approximate="tanh")
         wd = self.p[f"{i}.mlp.down.weight"]
-        down = ops.maybe_skinny_linear(act, wd)
+        use_resid_fold = residual is not None and self.tp_size == 1
+        down = ops.maybe_skinny_linear(
+            act, wd, residual=residual if use_resid_fold else None
+        )
+        folded = down is not None and use_resid_fold
         if down is None:
             down = _linear(act, wd)
+        if folded:
+            return down  # residual already added
+        if residual is not None:
+            # fold failed (shape/grad): plain add at the end
+            pass
         if sp:
             down = mappings.reduce_scatter_to_sp_region(down)
         else:
             down = mappings.reduce_from_tp_region(down)
         if f"{i}.mlp.down.bias" in self.p:
             down = down + self.p[f"{i}.mlp.down.bias"]
+        if residual is not None:
+            return residual + down
         return down
 
 

@@ -29,7 +29,8 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
                            torch::Tensor seg_lens_cpu);
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
-                          torch::Tensor out32_ws, long splitk);
+                          torch::Tensor out32_ws, long splitk,
+                          c10::optional<torch::Tensor> residual);
 std::pair<std::vector<int64_t>, double> mcmc_search(
     int64_t n_gpus, std::vector<std::vector<std::vector<double>>> cand_rows,
     std::vector<std::vector<int64_t>> parents_in,

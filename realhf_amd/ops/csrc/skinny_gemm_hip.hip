@@ -69,41 +69,59 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
     bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
     acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
   }
+  // non-atomic per-K-slice partial: out32[ks][m][n]
   #pragma unroll
   for (int r = 0; r < 4; r++) {
     int m = g * 4 + r;
     float vsum = acc0[r] + acc1[r];
     if (m < M)
-      atomicAdd(out32 + (long)m * N + n0 + i16, vsum);
+      out32[((long)ks * M + m) * N + n0 + i16] = vsum;
   }
 }
 
-__global__ void f32_to_bf16_kernel(const float* __restrict__ in,
-                                   bf16* __restrict__ out, long n) {
+// combine the nks fp32 partial slabs -> bf16 (+ optional residual add)
+__global__ void sg_combine_kernel(const float* __restrict__ parts,
+                                  const bf16* __restrict__ resid,
+                                  bf16* __restrict__ out, long n, int nks) {
   for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
        i += (long)gridDim.x * blockDim.x * 4) {
-    float4v v = *(const float4v*)(in + i);
+    float4v v = *(const float4v*)(parts + i);
+    for (int s = 1; s < nks; s++) {
+      float4v p = *(const float4v*)(parts + (long)s * n + i);
+      #pragma unroll
+      for (int j = 0; j < 4; j++) v[j] += p[j];
+    }
     short o[4];
-    #pragma unroll
-    for (int j = 0; j < 4; j++) o[j] = f2bf(v[j]);
+    if (resid) {
+      short4v rv = *(const short4v*)((const short*)resid + i);
+      #pragma unroll
+      for (int j = 0; j < 4; j++)
+        o[j] = f2bf(v[j] + __bfloat162float(((const bf16*)&rv)[j]));
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 4; j++) o[j] = f2bf(v[j]);
+    }
     *(short4v*)((short*)out + i) = *(short4v*)o;
   }
 }
 
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
-                          torch::Tensor out32_ws, long splitk) {
+                          torch::Tensor out32_ws, long splitk,
+                          c10::optional<torch::Tensor> residual) {
   // x [M, K] bf16; w [N, K] bf16 (row-major view, stride(1)==1);
-  // out32_ws: caller-provided fp32 workspace >= M*N (zeroed here).
+  // out32_ws: fp32 workspace >= splitk * M * N (per-slice partials —
+  // no zeroing, no atomics); residual (optional [M, N] bf16) is folded
+  // into the combine kernel.
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1);
   TORCH_CHECK(w.dim() == 2 && w.stride(1) == 1);
   int M = x.size(0), K = x.size(1), N = w.size(0);
   TORCH_CHECK(M <= 16 && K % 32 == 0 && N % SG_TN == 0);
   TORCH_CHECK(x.stride(0) == K, "x must be contiguous");
-  auto out32 = out32_ws.narrow(0, 0, (long)M * N).view({M, N});
-  out32.zero_();
   int kslice = (K / (int)splitk + 63) / 64 * 64;
   int nks = (K + kslice - 1) / kslice;
+  TORCH_CHECK(out32_ws.numel() >= (long)nks * M * N, "workspace too small");
+  auto out32 = out32_ws.narrow(0, 0, (long)nks * M * N);
   dim3 grid(N / SG_TN, nks);
   size_t lds = (size_t)16 * kslice * sizeof(short);
   TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
@@ -113,8 +131,14 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
   auto out = torch::empty({M, (long)N}, x.options());
   long n = (long)M * N;
   int cgrid = (int)std::min<long>((n / 4 + 255) / 256, 2048);
-  hipLaunchKernelGGL(f32_to_bf16_kernel, dim3(cgrid), dim3(256), 0,
-    cur_stream(), out32.data_ptr<float>(), (bf16*)out.data_ptr(), n);
+  const bf16* rptr = nullptr;
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->is_contiguous() && residual->numel() == n);
+    rptr = (const bf16*)residual->data_ptr();
+  }
+  hipLaunchKernelGGL(sg_combine_kernel, dim3(cgrid), dim3(256), 0,
+    cur_stream(), out32.data_ptr<float>(), rptr, (bf16*)out.data_ptr(), n,
+    nks);
   CHECK_CUDA_OK();
   return out;
 }

@@ -437,7 +437,8 @@ def merge_intervals(intervals: List[Tuple[int, int]]) -> List[Tuple[int, int]]:
 # projections (o/down: ~1.6-2x over hipBLASLt), loses on very wide N —
 # callers pick per shape (tools/bench_skinny.py is the A/B harness).
 # ---------------------------------------------------------------------------
-def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor) -> Optional[torch.Tensor]:
+def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
+                        residual: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
     if not (
         x.is_cuda
         and x.dtype == torch.bfloat16
@@ -458,13 +459,15 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor) -> Optional[torch.Tens
     C = _ops.require_hip()
     M, K = x.shape
     N = w.shape[0]
-    ws = get_global_memory_buffer().get_tensor(
-        (16 * N,), torch.float32, "skinny_gemm_ws"
-    )
     splitk = 8
     while K // splitk > 1024:
         splitk *= 2
-    return C.skinny_gemm(x, w, ws, splitk)
+    ws = get_global_memory_buffer().get_tensor(
+        (2 * splitk * 16 * N,), torch.float32, "skinny_gemm_ws"
+    )
+    if residual is not None:
+        residual = residual.contiguous()
+    return C.skinny_gemm(x, w, ws, splitk, residual)
 
 
 # ---------------------------------------------------------------------------

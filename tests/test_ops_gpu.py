@@ -304,3 +304,22 @@ def test_add_rmsnorm_fused():
     ref = F.rms_norm_ref(s_ref.float(), w.float(), 1e-5)
     torch.testing.assert_close(s.float(), s_ref.float(), atol=2e-2, rtol=2e-2)
     torch.testing.assert_close(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_skinny_gemm_with_residual():
+    torch.manual_seed(15)
+    M, K, N = 16, 11008 // 32 * 32, 4096
+    x = (torch.randn(M, K, device="cuda") * 0.3).to(torch.bfloat16)
+    w = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)
+    res = (torch.randn(M, N, device="cuda") * 0.3).to(torch.bfloat16)
+    ws = torch.empty(32 * M * N, dtype=torch.float32, device="cuda")
+    ref = x.float() @ w.float().t()
+    out = C.skinny_gemm(x, w, ws, 16, None)
+    torch.testing.assert_close(out.float(), ref, atol=0.3, rtol=3e-2)
+    out_r = C.skinny_gemm(x, w, ws, 16, res)
+    torch.testing.assert_close(out_r.float(), ref + res.float(), atol=0.3,
+                               rtol=3e-2)
+    # M < 16 tail
+    x2 = x[:5]
+    out2 = C.skinny_gemm(x2.contiguous(), w, ws, 8, None)
+    torch.testing.assert_close(out2.float(), ref[:5], atol=0.3, rtol=3e-2)

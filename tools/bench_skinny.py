@@ -38,7 +38,7 @@ def bench(fn, iters=50):
 
 def main():
     M = 16
-    ws = torch.empty(M * 32000, dtype=torch.float32, device="cuda")
+    ws = torch.empty(32 * M * 32000, dtype=torch.float32, device="cuda")
     print(f"{'shape':8} {'N':>6} {'K':>6} {'sk':>3} {'blaslt_us':>9} "
           f"{'skinny_us':>9} {'roofline_us':>11} {'max_err':>8}")
     for name, N, K in SHAPES:
@@ -50,9 +50,9 @@ def main():
         for sk in sorted({pick_splitk(N, K), 2, 4, 8, 16}):
             if K // sk > 1024 or K % 32:
                 continue
-            out = C.skinny_gemm(x, w, ws, sk)
+            out = C.skinny_gemm(x, w, ws, sk, None)
             err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
-            t_sk = bench(lambda: C.skinny_gemm(x, w, ws, sk))
+            t_sk = bench(lambda: C.skinny_gemm(x, w, ws, sk, None))
             print(f"{name:8} {N:6d} {K:6d} {sk:3d} {t_blas:9.1f} {t_sk:9.1f} "
                   f"{roof:11.1f} {err:8.4f}")
 
