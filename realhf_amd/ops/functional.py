@@ -236,10 +236,15 @@ def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale):
     seq_id = torch.bucketize(pos, cu_seqlens[1:].long(), right=True)
     local = pos - cu_seqlens[:-1].long()[seq_id]
 
+    # bf16 MFMA matmuls (fp32 accumulation inside rocBLAS) with fp32
+    # softmax — the same numerics class as a flash-attention backward;
+    # fp32 matmuls here would run at 1/16 of the bf16 MFMA rate.
+    mm_dtype = torch.bfloat16 if q.is_cuda else torch.float32
+
     def pad(x, nh):
         xp = x.new_zeros(bs, Lmax, nh, hd)
         xp[seq_id, local] = x
-        return xp.permute(0, 2, 1, 3).float()  # [bs, nh, Lmax, hd]
+        return xp.permute(0, 2, 1, 3).to(mm_dtype)  # [bs, nh, Lmax, hd]
 
     qp = pad(q, nq)
     kp = pad(k, nkv)
@@ -247,7 +252,7 @@ def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale):
     if rep > 1:
         kp = kp.repeat_interleave(rep, dim=1)
         vp = vp.repeat_interleave(rep, dim=1)
-    scores = torch.matmul(qp, kp.transpose(-1, -2)) * scale
+    scores = torch.matmul(qp * scale, kp.transpose(-1, -2)).float()
     kpos = torch.arange(Lmax, device=device)
     valid = kpos.unsqueeze(0) < lens.unsqueeze(1)  # [bs, Lmax]
     mask = valid.unsqueeze(1).unsqueeze(2)  # [bs,1,1,L]
@@ -257,7 +262,7 @@ def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale):
     scores = scores.masked_fill(~mask, float("-inf"))
     probs = torch.softmax(scores, dim=-1)
     probs = torch.nan_to_num(probs, nan=0.0)
-    op = torch.matmul(probs, vp)  # [bs, nh, Lmax, hd]
+    op = torch.matmul(probs.to(mm_dtype), vp)  # [bs, nh, Lmax, hd]
     op = op.permute(0, 2, 1, 3)  # [bs, Lmax, nh, hd]
     return op[seq_id, local].to(q.dtype)
 
