@@ -48,6 +48,7 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
   int kk = k_lo;
   // 128-deep body: issue all four W loads before the first MFMA so >=4
   // 16B loads stay in flight per lane (one k-iter alone is latency-bound)
+  // (nt loads measured 15-40% SLOWER here — keep plain loads)
   for (; kk + 128 <= k_hi; kk += 128) {
     bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
     bf16x8v b1 = *(const bf16x8v*)(wrow + kk + 32 + g * 8);
