@@ -23,7 +23,7 @@ import torch
 from realhf_amd.api.model import GenerationHyperparameters
 from realhf_amd.base import logging
 from realhf_amd.models.real_model import ReaLModel
-from realhf_amd.utils.functional import top_k_top_p_logits
+from realhf_amd.utils.functional import top_k_top_p_logits  # noqa: F401 — re-exported for interface use
 
 logger = logging.getLogger("generate")
 

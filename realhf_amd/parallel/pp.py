@@ -20,7 +20,6 @@ from realhf_amd.api.data import SequenceSample
 from realhf_amd.api.model import GenerationHyperparameters, PipelinableEngine
 from realhf_amd.base import constants, logging, seeding
 from realhf_amd.runtime.engine import sample_to_packed
-from realhf_amd.utils.functional import top_k_top_p_logits
 
 logger = logging.getLogger("pp")
 

@@ -139,10 +139,17 @@ def _key_intervals_for_shard_pair(
 _PLAN_CACHE: Dict = {}
 
 
+def _cfg_key(cfg: ReaLModelConfig):
+    import json
+
+    d = dataclasses.asdict(cfg)
+    return json.dumps(d, sort_keys=True, default=str)
+
+
 def build_realloc_plan(
     cfg: ReaLModelConfig, src: ParallelStrategy, dst: ParallelStrategy
 ) -> ReallocPlan:
-    cache_key = (id(cfg), src, dst)
+    cache_key = (_cfg_key(cfg), src, dst)
     if cache_key in _PLAN_CACHE:
         return _PLAN_CACHE[cache_key]
 

@@ -179,10 +179,15 @@ class DFGExecutor:
             else:
                 merged = gather_across_dp(local_out)
                 if merged is not None:
-                    merged = merged.select_idx(
-                        _order_by_ids(merged.ids, store.ids)
-                    )
-                    store.update_(merged)
+                    order = _order_by_ids(merged.ids, store.ids)
+                    if len(order) == merged.bs == store.bs:
+                        merged = merged.select_idx(order)
+                        store.update_(merged)
+                    else:
+                        # the producer re-keyed the batch (e.g. GRPO group
+                        # expansion: each prompt becomes group_size
+                        # responses with fresh ids) — it becomes the store
+                        store = merged
 
             for h in mfc.post_hooks:
                 self._run_hook(h, mfc, pre=False)

@@ -379,9 +379,16 @@ class Trainer:
                 if r == self.rank:
                     my_coord = (p, d, t)
             # register grid (group creation is collective — all ranks join)
+            mc0 = self.built.model_roles[name]
+            rcfg0 = self.built.model_cfgs[name]
+            ep = 1
+            if rcfg0.moe is not None:
+                ep = min(rcfg0.moe.expert_parallel_size, strat.dp)
             topo = PipeDataTensorTopology(
                 num_pp=strat.pp, num_dp=strat.dp, num_tp=strat.tp,
-                sequence_parallel=False,
+                sequence_parallel=mc0.parallel.sequence_parallel and strat.tp > 1,
+                gradient_checkpointing=mc0.gradient_checkpointing,
+                ep_size=ep if strat.dp % max(ep, 1) == 0 else 1,
             )
             rank_mapping = {
                 topo.get_rank(pipe=p, data=d, tensor=t): r
@@ -399,10 +406,13 @@ class Trainer:
             p_, d_, t_ = my_coord
             rcfg = self.built.model_cfgs[name]
             mc = self.built.model_roles[name]
+            ep_size = getattr(topo, "ep_size", 1)
             with constants.model_scope(str(name)):
                 m = ReaLModel(
                     rcfg, device=self.device, dtype=rcfg.torch_dtype,
                     tp_rank=t_, tp_size=strat.tp, pp_rank=p_, pp_size=strat.pp,
+                    ep_rank=(d_ % ep_size) if ep_size > 1 else 0,
+                    ep_size=ep_size,
                 )
                 if mc.path:
                     hf_reg.load_from_hf(m, mc.family, mc.path)

@@ -99,3 +99,61 @@ def _ep_train_worker():
 @pytest.mark.distributed
 def test_ep2_backward_through_all_to_all():
     LocalMultiProcessTest(2, _ep_train_worker).launch()
+
+
+def _grpo_mixtral_ep_worker(data, fileroot):
+    """BASELINE config #5 shape: Mixtral GRPO with expert parallelism,
+    scaled down to a tiny model on 2 gloo ranks."""
+    import os
+
+    from realhf_amd.api.experiment import GRPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = GRPOConfig(experiment_name="t-grpo-ep", trial_name="cpu", n_gpus=2)
+    for mc in (cfg.actor, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+        mc.family = "mixtral"
+    cfg.rew.family = "llama"  # reward model stays dense
+    cfg.rew.is_critic = True
+    cfg.group_size = 2
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 2
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 4
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    # inject EP: tiny mixtral test config has 4 experts; shard over 2 ranks
+    from realhf_amd.runtime import trainer as T
+
+    orig = T.build_experiment
+
+    def patched(c, world):
+        built = orig(c, world)
+        for name, rcfg in built.model_cfgs.items():
+            if rcfg.moe is not None:
+                rcfg.moe.expert_parallel_size = 2
+        return built
+
+    T.build_experiment = patched
+    try:
+        Trainer(cfg).run()
+    finally:
+        T.build_experiment = orig
+
+
+@pytest.mark.distributed
+def test_grpo_mixtral_ep_experiment(tmp_path):
+    import json
+
+    rng = np.random.RandomState(5)
+    data = str(tmp_path / "p.jsonl")
+    with open(data, "w") as f:
+        for _ in range(8):
+            f.write(json.dumps(
+                {"input_ids": rng.randint(3, 60, size=6).tolist()}) + "\n")
+    LocalMultiProcessTest(
+        2, _grpo_mixtral_ep_worker, data, str(tmp_path / "root")
+    ).launch()
