@@ -57,7 +57,8 @@ def main():
     p.add_argument("--no-hip-graph", action="store_true")
     p.add_argument("--phase-timing", action="store_true",
                    help="per-phase sync+timing (serializes side-stream overlap)")
-    p.add_argument("--offload-frozen", action="store_true", default=True)
+    p.add_argument("--no-offload-frozen", dest="offload_frozen",
+                   action="store_false", default=True)
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -92,7 +93,9 @@ def main():
     from realhf_amd.models.hf.llama import llama7b_config, make_test_config
     from realhf_amd.models.real_model import ReaLModel
 
-    seeding.set_random_seed(1234, rank_offset=rank)
+    # model weights must be IDENTICAL across DP replicas; per-rank
+    # divergence only for data/sampling (below)
+    seeding.set_random_seed(1234, rank_offset=0)
 
     def mkcfg(is_critic):
         if args.model == "llama-7b":
@@ -128,6 +131,7 @@ def main():
         with scope(n):
             m = ReaLModel(cfg, device=device,
                           dtype=torch.bfloat16 if use_cuda else torch.float32)
+            torch.manual_seed(4242 + hash(n) % 1000)  # same weights on all ranks
             m.random_init()
             model = Model(
                 name=ModelName(n, 0), module=m, tokenizer=None, device=device,
