@@ -141,6 +141,26 @@ def llama7b_config(is_critic: bool = False) -> ReaLModelConfig:
     )
 
 
+def llama70b_config(is_critic: bool = False) -> ReaLModelConfig:
+    """LLaMA-2-70B architecture (the 288 GB HBM sizing tier: the whole
+    model fits one MI355X in bf16; train tier uses tp + optimizer
+    offload)."""
+    return ReaLModelConfig(
+        n_layers=80,
+        hidden_dim=8192,
+        n_heads=64,
+        n_kv_heads=8,
+        head_dim=128,
+        intermediate_dim=28672,
+        vocab_size=32000,
+        max_position_embeddings=4096,
+        activation="silu",
+        norm_type="rms",
+        is_critic=is_critic,
+        family="llama",
+    )
+
+
 register_family(
     HFFamily(
         name="llama",

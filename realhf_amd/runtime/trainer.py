@@ -62,12 +62,21 @@ def _strategy_for(par: ParallelismConfig, world: int) -> ParallelStrategy:
 
 
 def _resolve_parallel(cfg: CommonExperimentConfig, model_par: ParallelismConfig,
-                      world: int) -> ParallelismConfig:
+                      world: int, param_count: int = 0) -> ParallelismConfig:
     mode = cfg.allocation_mode
-    if mode == "global" or mode == "heuristic":
-        # heuristic for one MI355X node: pure DP fits models <= ~13B whole
-        # (288 GB HBM); larger models should use manual/tp strategies
+    if mode == "global":
         return ParallelismConfig(data_parallel_size=world)
+    if mode == "heuristic":
+        # 288 GB HBM per MI355X: a trainable model needs ~16 bytes/param
+        # (bf16 params+grads + fp32 master/m/v); pure DP while that fits,
+        # else TP across the node (ZeRO already shards optimizer states
+        # over dp, so the bound is params+grads+activations)
+        if param_count * 4 < 230e9 or world == 1:
+            return ParallelismConfig(data_parallel_size=world)
+        tp = min(8, world)
+        return ParallelismConfig(
+            data_parallel_size=world // tp, tensor_parallel_size=tp
+        )
     if mode == "manual":
         return model_par
     return parse_parallelism(mode)
@@ -166,9 +175,10 @@ def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment
 
     def add_model(role, mc, replica=0):
         name = ModelName(role, replica)
-        par = _resolve_parallel(cfg, mc.parallel, world)
+        rcfg = model_cfg_of(mc, name)
+        par = _resolve_parallel(cfg, mc.parallel, world, rcfg.param_count())
         model_strategies[name] = _strategy_for(par, world)
-        model_cfgs[name] = model_cfg_of(mc, name)
+        model_cfgs[name] = rcfg
         model_roles[name] = mc
         return name, par
 

@@ -56,3 +56,31 @@ def test_tiny_sft_gpu(tmp_path):
     cfg.exp_ctrl.benchmark_steps = 2
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
     Trainer(cfg).run()
+
+
+def test_llama70b_fits_one_gpu_forward():
+    """288 GB HBM sizing tier: the whole 70B fits one MI355X in bf16 for
+    inference; forward 512 packed tokens + a short decode."""
+    from realhf_amd.api.model import GenerationHyperparameters
+    from realhf_amd.models.generation import generate
+    from realhf_amd.models.hf.llama import llama70b_config
+    from realhf_amd.models.real_model import ReaLModel
+
+    cfg = llama70b_config()
+    try:
+        m = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
+    except torch.OutOfMemoryError:
+        pytest.skip("box lacks free HBM for the 70B tier test")
+    m.random_init()
+    toks = torch.randint(0, cfg.vocab_size, (512,), device="cuda")
+    cu = torch.tensor([0, 256, 512], dtype=torch.int32, device="cuda")
+    with torch.no_grad():
+        logits = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=256)
+    assert logits.shape == (512, cfg.vocab_size)
+    out = generate(
+        m, toks[:64], torch.tensor([0, 32, 64], dtype=torch.int32, device="cuda"),
+        GenerationHyperparameters(max_new_tokens=8, greedy=True),
+    )
+    assert out.gen_tokens.shape[0] == 2
+    del m
+    torch.cuda.empty_cache()
