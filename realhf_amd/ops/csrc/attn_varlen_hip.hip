@@ -4,9 +4,9 @@
 // (reference: realhf/impl/model/modules/attn.py:255).
 //
 // Structure (correctness-first instance of the guide's §B prefill recipe):
-//   grid = (q_block, q_head); workgroup = 256 threads = 4 waves.
-//   Each workgroup owns 64 query rows of one sequence+head (wave w: 16
-//   rows).  KV tiles of 64 keys are staged cooperatively in LDS — K
+//   grid = (q_block, q_head); workgroup = 512 threads = 8 waves.
+//   Each workgroup owns 128 query rows of one sequence+head (wave w: 16
+//   rows) so the cooperative K/V staging amortizes over 2x the MFMA work.  KV tiles of 64 keys are staged cooperatively in LDS — K
 //   row-major [64][136] (8-elem pad), V TRANSPOSED [128][72] so the PV
 //   B-fragment is a contiguous ds_read_b128.  Online softmax state (m, l)
 //   is held per C-row in registers, reduced with shfl_xor over the 16-lane
@@ -22,7 +22,8 @@
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 
-#define QBLK 64      // queries per workgroup
+#define QBLK 128     // queries per workgroup (8 waves x 16 rows)
+#define AV_WAVES 8
 #define QW 16        // queries per wave
 #define KVBLK 64     // keys per LDS tile
 #define HDMAX 128
@@ -46,7 +47,7 @@ DEVINL float group16_sum(float x) {
 }
 
 template <int HD>
-__global__ __launch_bounds__(256, 2) void attn_varlen_fwd_kernel(
+__global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const int* __restrict__ cu_seqlens,
     const int* __restrict__ blk_seq, const int* __restrict__ blk_qstart,
@@ -67,7 +68,7 @@ __global__ __launch_bounds__(256, 2) void attn_varlen_fwd_kernel(
 
   __shared__ __bf16 k_s[KVBLK][HD + KPAD];
   __shared__ __bf16 vt_s[HD][KVBLK + VPAD];
-  __shared__ __bf16 p_s[4][QW][KVBLK + VPAD];
+  __shared__ __bf16 p_s[AV_WAVES][QW][KVBLK + VPAD];
 
   // ---- load Q fragments (registers, whole kernel) --------------------
   // wave w owns query rows qrow_local = q0_local + w*QW + i16 (A layout)
@@ -95,8 +96,8 @@ __global__ __launch_bounds__(256, 2) void attn_varlen_fwd_kernel(
     const int kchunk = min(KVBLK, kv_end - kv0);
     // ---- stage K and V(T) tiles ------------------------------------
     __syncthreads();
-    // K: 64 rows x HD; thread strides rows. 256 threads, each loads 16B
-    for (int idx = threadIdx.x; idx < KVBLK * (HD / 8); idx += 256) {
+    // K: 64 rows x HD; all waves stage cooperatively, 16B per thread
+    for (int idx = threadIdx.x; idx < KVBLK * (HD / 8); idx += 64 * AV_WAVES) {
       int row = idx / (HD / 8);
       int col8 = (idx % (HD / 8)) * 8;
       bf16x8 val = {};
@@ -229,13 +230,13 @@ std::vector<torch::Tensor> attn_varlen_fwd(
   auto lse = torch::empty({total, nq}, q.options().dtype(torch::kFloat));
   dim3 grid((unsigned)bseq.size(), nq);
   if (hd == 128) {
-    hipLaunchKernelGGL((attn_varlen_fwd_kernel<128>), grid, dim3(256), 0,
+    hipLaunchKernelGGL((attn_varlen_fwd_kernel<128>), grid, dim3(64 * AV_WAVES), 0,
       cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
       (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
       bseq_t.data_ptr<int>(), bqs_t.data_ptr<int>(), (bf16*)out.data_ptr(),
       lse.data_ptr<float>(), nq, nkv, (float)scale, causal);
   } else if (hd == 64) {
-    hipLaunchKernelGGL((attn_varlen_fwd_kernel<64>), grid, dim3(256), 0,
+    hipLaunchKernelGGL((attn_varlen_fwd_kernel<64>), grid, dim3(64 * AV_WAVES), 0,
       cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
       (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
       bseq_t.data_ptr<int>(), bqs_t.data_ptr<int>(), (bf16*)out.data_ptr(),
