@@ -5,7 +5,7 @@ ModelShardID:101, ModelInterfaceType:45, abstractions:120-189).
 """
 import dataclasses
 import enum
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 
 @dataclasses.dataclass(frozen=True, order=True)

@@ -10,8 +10,7 @@ dims) tensors concatenated across samples, with per-key nested sequence
 lengths and one unique id per sample.  No padding anywhere.
 """
 import dataclasses
-import itertools
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Union
+from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 import torch

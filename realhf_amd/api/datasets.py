@@ -4,8 +4,6 @@ benchmarking (no-network environment).
 
 All datasets produce SequenceSamples with unique ids.
 """
-from typing import Dict, List, Optional
-
 import numpy as np
 import torch
 

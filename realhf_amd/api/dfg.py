@@ -9,7 +9,6 @@ edges are resolved from input/output data keys.  E.g. PPO:
 Implemented without networkx (plain dict adjacency + toposort).
 """
 import dataclasses
-import enum
 from typing import Any, Dict, List, Optional, Tuple
 
 from realhf_amd.api.config import (

@@ -9,9 +9,9 @@ same idea: structured configs translated into (DFG, allocations, worker
 state).
 """
 import dataclasses
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, Optional
 
-from realhf_amd.api.config import ParallelismConfig, parse_parallelism
+from realhf_amd.api.config import ParallelismConfig
 from realhf_amd.api.model import GenerationHyperparameters
 from realhf_amd.parallel.ddp import OptimizerConfig
 

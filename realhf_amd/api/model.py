@@ -6,7 +6,7 @@ Model:464, ModelBackend:513, ModelInterface:564, registries:635-738).
 """
 import abc
 import dataclasses
-from typing import Any, Callable, Dict, List, Optional, Tuple
+from typing import Any, Callable, Dict, List, Optional
 
 import torch
 

@@ -14,7 +14,6 @@ Launch modes: under torchrun (WORLD_SIZE set) runs this rank's worker;
 standalone it spawns n_gpus local workers (scheduler/local.py).
 """
 import dataclasses
-import json
 import os
 import sys
 import typing

@@ -13,7 +13,7 @@ import os
 import pickle
 import time
 from collections import defaultdict
-from typing import Dict, List, Optional
+from typing import Dict
 
 import torch
 

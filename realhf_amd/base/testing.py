@@ -7,7 +7,6 @@ multi-rank logic (topologies, TP math, realloc plans, the DFG executor)
 is tested without hardware.  On a GPU box the same harness runs with
 RCCL ("nccl" backend on ROCm).
 """
-import dataclasses
 import os
 import traceback
 from typing import Callable, Optional

@@ -7,17 +7,12 @@ the advantage of a response is its group-normalized reward, applied
 uniformly over its generated tokens.
 """
 import dataclasses
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 
 from realhf_amd.api.data import SequenceSample
-from realhf_amd.api.model import (
-    GenerationHyperparameters,
-    Model,
-    ModelInterface,
-    register_interface,
-)
+from realhf_amd.api.model import Model, register_interface
 from realhf_amd.interfaces import ppo_math
 from realhf_amd.interfaces.ppo import PPOActorInterface, _short_cu
 from realhf_amd.models import moe as moe_mod

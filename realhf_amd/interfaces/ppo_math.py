@@ -5,7 +5,6 @@ Reference semantics: realhf/impl/model/utils/ppo_functional.py
 get_packed_rewards:291) and modules value norm (ExponentialRunningMeanStd).
 GAE itself is the HIP kernel / reference scan in ops.functional.gae.
 """
-import dataclasses
 from typing import Optional, Tuple
 
 import torch

@@ -16,7 +16,7 @@ Design (MI355X-first):
   masks EOS early.
 """
 import dataclasses
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

@@ -14,7 +14,7 @@ transformers/vLLM.
 import dataclasses
 import json
 import os
-from typing import Callable, Dict, List, Optional
+from typing import Callable, Dict, List
 
 import torch
 

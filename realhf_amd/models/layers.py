@@ -25,7 +25,6 @@ from realhf_amd.api.model import ReaLModelConfig
 from realhf_amd.base import constants
 from realhf_amd.ops import functional as ops
 from realhf_amd.parallel import mappings
-from realhf_amd.utils.functional import compute_varlen_position_indices
 
 
 def _maybe_merged(params: Dict[str, torch.Tensor], names, dim_out_total):

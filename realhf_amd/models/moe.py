@@ -14,7 +14,7 @@ Expert GEMMs run through the hand-written MFMA grouped-GEMM kernel
 path uses per-expert GEMMs (rocBLAS) so autograd handles the backward
 (hand-written grouped bwd is a noted follow-up).
 """
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist

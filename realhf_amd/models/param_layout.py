@@ -16,7 +16,7 @@ regions concatenate into a single GEMM operand view (merged QKV / merged
 gate-up GEMMs without a separate merged tensor).
 """
 import dataclasses
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 

@@ -9,7 +9,6 @@ is the contract that makes ZeRO-1 sharding, offload, and parameter
 reallocation interval math over a single tensor (SURVEY.md §2.1
 "Flat-param spec").
 """
-import dataclasses
 from typing import Dict, List, Optional
 
 import torch

@@ -22,7 +22,7 @@ mixed-precision optimizer.
 """
 import dataclasses
 import math
-from typing import Dict, List, Optional
+from typing import Dict
 
 import torch
 import torch.distributed as dist

@@ -6,8 +6,6 @@ TP group; three small all-reduces (max, target-logit, sum-exp) replace
 gathering the [tokens, vocab] matrix (which at 32k vocab would be the
 largest activation in the step).
 """
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 

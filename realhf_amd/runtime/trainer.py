@@ -6,11 +6,10 @@ ppo_exp.py:261-377, resolve_replica_ids/resolve_rpc_hooks utils.py:126/143)
 re-done as a deterministic SPMD program (see runtime/executor.py).
 """
 import dataclasses
-import json
 import os
 import pickle
 import time
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist
@@ -25,7 +24,7 @@ from realhf_amd.api.config import (
     ParallelismConfig,
     parse_parallelism,
 )
-from realhf_amd.api.data import PackedDataLoader, SequenceSample, make_dataset
+from realhf_amd.api.data import PackedDataLoader, make_dataset
 from realhf_amd.api.dfg import DFG, MFCDef, OffloadHook, ParamReallocHook, build_graph
 from realhf_amd.api.experiment import (
     CommonExperimentConfig,
@@ -97,7 +96,6 @@ def _apply_search_allocation(cfg, world: int):
     """allocation_mode=search: pick per-MFC strategies with the MCMC
     search engine, then write them back into the model configs (reference:
     apps/main.py experiment._search() -> search_rpc_allocations)."""
-    from realhf_amd.api.dfg import build_graph as _bg
     from realhf_amd.models import hf as _hf
     from realhf_amd.search.engine import MFCSpec, search_allocations
 

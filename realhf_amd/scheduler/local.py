@@ -7,11 +7,10 @@ noted in README parity table)."""
 import os
 import signal
 import subprocess
-import sys
 import time
 from typing import Dict, List, Optional
 
-from realhf_amd.base import logging, name_resolve
+from realhf_amd.base import logging
 
 logger = logging.getLogger("scheduler")
 

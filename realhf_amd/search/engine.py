@@ -16,8 +16,7 @@ bench runs on this hardware):
 Memory model mirrors parallel/ddp.py + generation KV sizing.
 """
 import dataclasses
-import itertools
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from realhf_amd.api.config import ModelInterfaceType, ParallelismConfig
 from realhf_amd.api.dfg import DFG

@@ -1,0 +1,47 @@
+"""Quickstart CLI + local scheduler end-to-end (reference:
+apps/quickstart.py + scheduler/local/client.py)."""
+import json
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+
+def test_cli_launcher_spawns_workers(tmp_path):
+    """Full CLI path: launcher spawns 2 CPU workers which rendezvous over
+    gloo and run a 2-step SFT experiment."""
+    rng = np.random.RandomState(0)
+    data = tmp_path / "sft.jsonl"
+    with open(data, "w") as f:
+        for _ in range(8):
+            rec = {
+                "prompt_ids": rng.randint(0, 60, size=3).tolist(),
+                "answer_ids": rng.randint(0, 60, size=6).tolist(),
+            }
+            f.write(json.dumps(rec) + "\n")
+    env = dict(os.environ)
+    env["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    env.pop("WORLD_SIZE", None)
+    env.pop("RANK", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "realhf_amd.apps.quickstart", "sft",
+         "experiment_name=cli-sft", "trial_name=t", "n_gpus=2",
+         "model.dtype=float32",
+         "dataset.type_=prompt_answer", f"dataset.path={data}",
+         "dataset.train_bs_n_seqs=4", "exp_ctrl.benchmark_steps=2",
+         "exp_ctrl.total_train_epochs=4"],
+        env=env, capture_output=True, text=True, timeout=240,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "benchmark" in out.stderr
+
+
+def test_cli_help():
+    out = subprocess.run(
+        [sys.executable, "-m", "realhf_amd.apps.quickstart", "--help"],
+        capture_output=True, text=True, timeout=60,
+    )
+    assert out.returncode == 0
+    assert "ppo" in out.stdout
