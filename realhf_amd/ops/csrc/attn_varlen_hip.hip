@@ -279,3 +279,36 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   CHECK_CUDA_OK();
   return D;
 }
+
+// ---------------------------------------------------------------------------
+// tr16_probe: empirical map of ds_read_b64_tr_b16 — each lane passes
+// addr = lds_base + 2*addr_elem[lane] (bf16 elems); the kernel stages
+// lds[i] = i and dumps what (lane, j) receives, so the host can derive
+// the lane->element mapping for MFMA B-fragment use (guide §2 T10).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
+typedef __attribute__((address_space(3))) bf16x4v lds_b64_t;
+
+__global__ void tr16_probe_kernel(const int* __restrict__ addr_elem,
+                                  float* __restrict__ out) {
+  __shared__ __bf16 lds[1024];
+  int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 1024; i += 64)
+    lds[i] = (__bf16)(float)i;
+  __syncthreads();
+  bf16x4v r = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_b64_t*)&lds[addr_elem[lane]]);
+  #pragma unroll
+  for (int j = 0; j < 4; j++) out[lane * 4 + j] = (float)r[j];
+}
+
+torch::Tensor tr16_probe(torch::Tensor addr_elem) {
+  auto a = addr_elem.to(torch::kInt).cuda().contiguous();
+  TORCH_CHECK(a.numel() == 64);
+  auto out = torch::zeros({64, 4}, torch::TensorOptions()
+                          .dtype(torch::kFloat).device(a.device()));
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+    a.data_ptr<int>(), out.data_ptr<float>());
+  CHECK_CUDA_OK();
+  return out;
+}

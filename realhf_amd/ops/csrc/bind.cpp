@@ -26,6 +26,7 @@ std::vector<torch::Tensor> attn_varlen_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor tr16_probe(torch::Tensor addr_elem);
 torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
                            torch::Tensor seg_lens_cpu);
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
@@ -56,6 +57,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode);
   m.def("attn_varlen_fwd", &attn_varlen_fwd);
   m.def("mfma_probe", &mfma_probe);
+  m.def("tr16_probe", &tr16_probe);
   m.def("rope_qkv_decode", &rope_qkv_decode);
   m.def("grouped_gemm", &grouped_gemm);
   m.def("skinny_gemm", &skinny_gemm);

@@ -84,3 +84,47 @@ def test_llama70b_fits_one_gpu_forward():
     assert out.gen_tokens.shape[0] == 2
     del m
     torch.cuda.empty_cache()
+
+
+def test_tiny_ppo_with_rccl_groups(tmp_path):
+    """Single-rank RCCL process group: exercises grid construction, the
+    executor's gathers and ZeRO collectives on the nccl(=RCCL) backend —
+    the same code the 8-GPU scale run executes."""
+    import torch.distributed as dist
+
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29719")
+    torch.cuda.set_device(0)
+    created = False
+    if not dist.is_initialized():
+        dist.init_process_group("nccl", rank=0, world_size=1)
+        created = True
+    try:
+        rng = np.random.RandomState(3)
+        data = str(tmp_path / "p.jsonl")
+        with open(data, "w") as f:
+            for _ in range(8):
+                f.write(json.dumps(
+                    {"input_ids": rng.randint(3, 60, size=8).tolist()}) + "\n")
+        cfg = PPOConfig(experiment_name="t-ppo-rccl", trial_name="g", n_gpus=1)
+        cfg.dataset.type_ = "prompt"
+        cfg.dataset.path = data
+        cfg.dataset.train_bs_n_seqs = 4
+        cfg.dataset.max_prompt_len = 8
+        cfg.ppo.gen.max_new_tokens = 8
+        cfg.ppo.gen.use_hip_graph = False
+        cfg.ppo.ppo_n_minibatches = 2
+        cfg.exp_ctrl.benchmark_steps = 1
+        os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+        Trainer(cfg).run()
+    finally:
+        if created:
+            dist.destroy_process_group()
+        from realhf_amd.base import constants
+        from realhf_amd.base.topology import clear_group_cache
+
+        constants.clear_grids()
+        clear_group_cache()
