@@ -46,6 +46,9 @@ DEVINL float group16_sum(float x) {
   return x;
 }
 
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
+typedef __attribute__((address_space(3))) bf16x4v lds_b64_t;
+
 template <int HD>
 __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
@@ -67,7 +70,14 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
   const int g = lane >> 4;     // k-chunk group / C row group
 
   __shared__ __bf16 k_s[KVBLK][HD + KPAD];
-  __shared__ __bf16 vt_s[HD][KVBLK + VPAD];
+  // V stays ROW-major: the PV B-fragment is gathered by the hardware
+  // transpose read ds_read_b64_tr_b16 (mapping derived by tr16_probe:
+  // in each 16-lane group, result lane c elem j = source-lane
+  // (4j + c/4)'s element (c%4); so lane i loading the 8-byte chunk at
+  // v_s[kbase + 4t + i/4][hd0 + (i%4)*4] makes the group end up with
+  // exactly B[k = 4t + j][n = c]).  VROWPAD = 16 elems keeps the four
+  // row-steps on distinct bank sets.
+  __shared__ __bf16 v_s[KVBLK][HD + 16];
   __shared__ __bf16 p_s[AV_WAVES][QW][KVBLK + VPAD];
 
   // ---- load Q fragments (registers, whole kernel) --------------------
@@ -101,15 +111,13 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
       int row = idx / (HD / 8);
       int col8 = (idx % (HD / 8)) * 8;
       bf16x8 val = {};
-      if (row < kchunk)
-        val = *(const bf16x8*)(k + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
-      *(bf16x8*)(&k_s[row][col8]) = val;
-      // V transposed: vt_s[col][row]
       bf16x8 vv = {};
-      if (row < kchunk)
+      if (row < kchunk) {
+        val = *(const bf16x8*)(k + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
         vv = *(const bf16x8*)(v + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
-      #pragma unroll
-      for (int j = 0; j < 8; j++) vt_s[col8 + j][row] = vv[j];
+      }
+      *(bf16x8*)(&k_s[row][col8]) = val;
+      *(bf16x8*)(&v_s[row][col8]) = vv;
     }
     __syncthreads();
 
@@ -176,10 +184,18 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
     for (int kk = 0; kk < 2; kk++) {  // two K=32 chunks over 64 keys
       // A fragment: P[qrow = i16][kk*32 + g*8 + j]
       bf16x8 pa = *(const bf16x8*)(&p_s[w][i16][kk * 32 + g * 8]);
+      const int krow0 = kk * 32 + g * 8 + (i16 >> 2);
+      const int vcol4 = (i16 & 3) * 4;
       #pragma unroll
       for (int t = 0; t < HD / 16; t++) {
-        // B fragment: vT[n = t*16 + i16 (hd)][kk*32 + g*8 + j (key)]
-        bf16x8 vb = *(const bf16x8*)(&vt_s[t * 16 + i16][kk * 32 + g * 8]);
+        // B fragment via two hardware-transpose reads (see v_s comment)
+        bf16x4v r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_b64_t*)&v_s[krow0][t * 16 + vcol4]);
+        bf16x4v r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_b64_t*)&v_s[krow0 + 4][t * 16 + vcol4]);
+        bf16x8 vb;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) { vb[j] = r0[j]; vb[4 + j] = r1[j]; }
         o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[t], 0, 0, 0);
       }
     }
@@ -286,9 +302,6 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
 // lds[i] = i and dumps what (lane, j) receives, so the host can derive
 // the lane->element mapping for MFMA B-fragment use (guide §2 T10).
 // ---------------------------------------------------------------------------
-typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
-typedef __attribute__((address_space(3))) bf16x4v lds_b64_t;
-
 __global__ void tr16_probe_kernel(const int* __restrict__ addr_elem,
                                   float* __restrict__ out) {
   __shared__ __bf16 lds[1024];
