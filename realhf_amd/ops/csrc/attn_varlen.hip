@@ -101,46 +101,24 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
   // causal: keys needed up to q0_local + QBLK - 1 (inclusive); else all L
   const int kv_end = causal ? min(L, q0_local + QBLK) : L;
 
-  // ---- async-stage split (guide T14): each thread owns NCH 16-byte
-  // chunks of the K and V tiles; global loads for tile t+1 issue at the
-  // top of tile t's compute and land in LDS after the post-PV barrier.
-  constexpr int NCH = KVBLK * (HD / 8) / (64 * AV_WAVES);
-  bf16x8 kreg[NCH], vreg[NCH];
-
-  auto stage_load = [&](int kv0) {
-    #pragma unroll
-    for (int ci = 0; ci < NCH; ci++) {
-      int idx = threadIdx.x + ci * 64 * AV_WAVES;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
+    const int kchunk = min(KVBLK, kv_end - kv0);
+    // ---- stage K and V(T) tiles ------------------------------------
+    __syncthreads();
+    // K: 64 rows x HD; all waves stage cooperatively, 16B per thread
+    for (int idx = threadIdx.x; idx < KVBLK * (HD / 8); idx += 64 * AV_WAVES) {
       int row = idx / (HD / 8);
       int col8 = (idx % (HD / 8)) * 8;
       bf16x8 val = {};
       bf16x8 vv = {};
-      if (kv0 + row < kv_end) {
+      if (row < kchunk) {
         val = *(const bf16x8*)(k + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
         vv = *(const bf16x8*)(v + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
       }
-      kreg[ci] = val;
-      vreg[ci] = vv;
+      *(bf16x8*)(&k_s[row][col8]) = val;
+      *(bf16x8*)(&v_s[row][col8]) = vv;
     }
-  };
-  auto stage_write = [&]() {
-    #pragma unroll
-    for (int ci = 0; ci < NCH; ci++) {
-      int idx = threadIdx.x + ci * 64 * AV_WAVES;
-      int row = idx / (HD / 8);
-      int col8 = (idx % (HD / 8)) * 8;
-      *(bf16x8*)(&k_s[row][col8]) = kreg[ci];
-      *(bf16x8*)(&v_s[row][col8]) = vreg[ci];
-    }
-  };
-
-  stage_load(0);
-  stage_write();
-  __syncthreads();
-
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
-    if (kv0 + KVBLK < kv_end)
-      stage_load(kv0 + KVBLK);  // HBM latency hides under this tile's MFMAs
+    __syncthreads();
 
     // ---- S = Q K^T over 4 key subtiles ------------------------------
     f32x4 s_sub[4];
@@ -198,7 +176,7 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
       for (int r = 0; r < 4; r++)
         p_s[w][g * 4 + r][ks * 16 + i16] = (__bf16)s_sub[ks][r];
     }
-    __syncthreads();  // p_s visible; k_s/v_s reads for this tile also done
+    __syncthreads();
 
     // ---- O += P V ----------------------------------------------------
     #pragma unroll
@@ -219,11 +197,6 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
         for (int j = 0; j < 4; j++) { vb[j] = r0[j]; vb[4 + j] = r1[j]; }
         o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[t], 0, 0, 0);
       }
-    }
-    if (kv0 + KVBLK < kv_end) {
-      __syncthreads();  // everyone done reading v_s
-      stage_write();    // land tile t+1 (T14: write AFTER the barrier)
-      __syncthreads();
     }
   }
 
