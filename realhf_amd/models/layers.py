@@ -194,7 +194,9 @@ class ReaLModelBlock(nn.Module):
             qkv_dim = (self.nq + 2 * self.nkv) * self.hd
             merged = _maybe_merged(self.p, self._qkv_names, qkv_dim)
             if merged is not None:
-                qkv_raw = _linear(h, merged)
+                qkv_raw = ops.maybe_skinny_linear(h, merged)
+                if qkv_raw is None:
+                    qkv_raw = _linear(h, merged)
             else:
                 qkv_raw = torch.cat(
                     [_linear(h, self.p[n]) for n in self._qkv_names], dim=-1
@@ -304,7 +306,9 @@ class ReaLModelBlock(nn.Module):
             idim_local = self.p[f"{i}.mlp.gate.weight"].shape[0]
             merged = _maybe_merged(self.p, self._gu_names, 2 * idim_local)
             if merged is not None:
-                gu = _linear(h, merged)
+                gu = ops.maybe_skinny_linear(h, merged)
+                if gu is None:
+                    gu = _linear(h, merged)
             else:
                 gu = torch.cat(
                     [_linear(h, self.p[n]) for n in self._gu_names], dim=-1

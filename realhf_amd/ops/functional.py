@@ -15,6 +15,7 @@ HIP kernels (realhf_amd/ops/csrc/, reference op inventory SURVEY.md §2.2):
 All torch reference paths are fp32-upcast where the reference math is fp32.
 """
 import math
+import os
 from typing import List, Optional, Tuple
 
 import torch
@@ -455,7 +456,10 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
         and w.stride(1) == 1
         and w.shape[0] % 64 == 0
         and w.shape[1] % 32 == 0
-        and w.shape[0] <= 2 * w.shape[1]  # wide-N shapes: hipBLASLt wins
+        and (
+            w.shape[0] <= 2 * w.shape[1]  # wide-N shapes: hipBLASLt wins
+            or os.environ.get("REALHF_AMD_SKINNY_WIDE") == "1"
+        )
         and _ops.hip_available()
     ):
         return None
