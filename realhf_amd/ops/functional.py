@@ -456,9 +456,12 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
         and w.stride(1) == 1
         and w.shape[0] % 64 == 0
         and w.shape[1] % 32 == 0
+        # in-context the weight-streaming kernel also edges out hipBLASLt
+        # on the wide-N shapes (A/B: 3.09 vs 3.05 samples/s); opt out with
+        # REALHF_AMD_NO_SKINNY_WIDE=1
         and (
-            w.shape[0] <= 2 * w.shape[1]  # wide-N shapes: hipBLASLt wins
-            or os.environ.get("REALHF_AMD_SKINNY_WIDE") == "1"
+            w.shape[0] <= 2 * w.shape[1]
+            or os.environ.get("REALHF_AMD_NO_SKINNY_WIDE") != "1"
         )
         and _ops.hip_available()
     ):
