@@ -165,6 +165,12 @@ def generate(
     return_prompt_logprobs: bool = False,
 ) -> GenerationOutput:
     assert model.pp_size == 1, "pp>1 generation goes through the pipe engine"
+    from realhf_amd.base import constants as _c
+
+    assert not (_c.has_current() and _c.sequence_parallel()), (
+        "sequence parallelism is not supported during generation "
+        "(reference: pipe_runner.py:856) — use a gen replica without SP"
+    )
     cfg = model.config
     device = packed_prompts.device
     bs = cu_seqlens.shape[0] - 1
