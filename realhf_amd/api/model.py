@@ -75,6 +75,9 @@ class ReaLModelConfig:
     qk_layernorm: bool = False
     # gemma-style sqrt(hidden) embedding scaling
     embedding_multiplier: Optional[float] = None
+    # mistral-style sliding-window attention: each token attends to the
+    # previous `sliding_window` tokens (inclusive of itself); None = full
+    sliding_window: Optional[int] = None
     # dropout (0 for RLHF)
     attn_pdrop: float = 0.0
     resid_pdrop: float = 0.0

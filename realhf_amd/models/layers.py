@@ -187,6 +187,8 @@ class ReaLModelBlock(nn.Module):
             and not cfg.qk_layernorm
             and self.hd in (64, 128)
             and _os.environ.get("REALHF_AMD_NO_FUSED_DECODE") != "1"
+            and (cfg.sliding_window is None
+                 or k_cache.shape[1] <= cfg.sliding_window)
         ):
             from realhf_amd import ops as _ops_pkg
 
@@ -258,7 +260,10 @@ class ReaLModelBlock(nn.Module):
             b_idx = torch.arange(bs, device=x.device)
             k_cache[b_idx, idx] = k.to(k_cache.dtype)
             v_cache[b_idx, idx] = v.to(v_cache.dtype)
-            attn_out = ops.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
+            attn_out = ops.attn_decode(
+                q, k_cache, v_cache, cache_seqlens, scale,
+                window=cfg.sliding_window,
+            )
         else:
             if k_cache is not None:
                 # prefill: write all tokens into the cache
@@ -271,7 +276,8 @@ class ReaLModelBlock(nn.Module):
                 k_cache[seq_id, positions] = k.detach().to(k_cache.dtype)
                 v_cache[seq_id, positions] = v.detach().to(v_cache.dtype)
             attn_out = ops.attn_varlen(
-                q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=scale
+                q, k, v, cu_seqlens, max_seqlen, causal=True,
+                softmax_scale=scale, window=cfg.sliding_window,
             )
         attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
         o = _linear(attn_out, self.p[f"{i}.attn.wo.weight"])

@@ -192,6 +192,7 @@ def attn_varlen_ref(
     cu_seqlens: torch.Tensor,  # [bs+1] int32
     causal: bool = True,
     softmax_scale: Optional[float] = None,
+    window: Optional[int] = None,  # sliding window incl. self (mistral)
 ) -> torch.Tensor:
     """Per-sequence SDPA in fp32 — the numerics oracle."""
     nq, hd = q.shape[1], q.shape[2]
@@ -215,13 +216,20 @@ def attn_varlen_ref(
                 torch.ones(L, L, dtype=torch.bool, device=q.device), diagonal=1
             )
             scores = scores.masked_fill(mask, float("-inf"))
+        if window is not None and L > window:
+            # visible keys for query i: (i - window, i]  (HF mistral mask)
+            wmask = torch.tril(
+                torch.ones(L, L, dtype=torch.bool, device=q.device),
+                diagonal=-window,
+            )
+            scores = scores.masked_fill(wmask, float("-inf"))
         probs = torch.softmax(scores, dim=-1)
         o = torch.matmul(probs, vi)  # [nq, L, hd]
         out[s:e] = o.transpose(0, 1).to(q.dtype)
     return out
 
 
-def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale):
+def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale, window=None):
     """Batched-padded torch implementation on GPU (used for backward
     recompute until the hand-written HIP backward lands): one set of
     batched rocBLAS GEMMs instead of a per-sequence python loop (the loop
@@ -259,6 +267,8 @@ def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale):
     mask = valid.unsqueeze(1).unsqueeze(2)  # [bs,1,1,L]
     if causal:
         cm = kpos.unsqueeze(0) <= kpos.unsqueeze(1)  # [Lq, Lk]
+        if window is not None:
+            cm = cm & (kpos.unsqueeze(0) > kpos.unsqueeze(1) - window)
         mask = mask & cm.unsqueeze(0).unsqueeze(0)
     scores = scores.masked_fill(~mask, float("-inf"))
     probs = torch.softmax(scores, dim=-1)
@@ -295,17 +305,27 @@ class _AttnVarlenFn(torch.autograd.Function):
         return dq, dk, dv, None, None, None, None
 
 
-def attn_varlen(q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=None):
+def attn_varlen(q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=None,
+                window=None):
     scale = softmax_scale or (1.0 / math.sqrt(q.shape[-1]))
-    if _ops.use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
+    if window is not None and window >= int(max_seqlen):
+        window = None  # window wider than any sequence: plain causal
+    if (
+        _ops.use_hip(q)
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] in (64, 128)
+        and window is None
+    ):
         return _AttnVarlenFn.apply(
             q.contiguous(), k.contiguous(), v.contiguous(),
             cu_seqlens, max_seqlen, causal, scale,
         )
     if q.is_cuda:
-        # odd head dims / dtypes: batched rocBLAS path (still GPU)
-        return _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale)
-    return attn_varlen_ref(q, k, v, cu_seqlens, causal, scale)
+        # odd head dims / dtypes / binding sliding window: batched rocBLAS
+        # path (still GPU)
+        return _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale,
+                                          window=window)
+    return attn_varlen_ref(q, k, v, cu_seqlens, causal, scale, window=window)
 
 
 # ---------------------------------------------------------------------------
@@ -317,6 +337,7 @@ def attn_decode_ref(
     v_cache: torch.Tensor,
     cache_seqlens: torch.Tensor,  # [bs] int32 — valid length INCLUDING new token
     softmax_scale: Optional[float] = None,
+    window: Optional[int] = None,
 ) -> torch.Tensor:
     bs, nq, hd = q.shape
     nkv = k_cache.shape[2]
@@ -325,8 +346,9 @@ def attn_decode_ref(
     out = torch.empty_like(q)
     for b in range(bs):
         L = int(cache_seqlens[b])
-        kb = k_cache[b, :L].transpose(0, 1).float()  # [nkv, L, hd]
-        vb = v_cache[b, :L].transpose(0, 1).float()
+        lo = max(0, L - window) if window is not None else 0
+        kb = k_cache[b, lo:L].transpose(0, 1).float()  # [nkv, L-lo, hd]
+        vb = v_cache[b, lo:L].transpose(0, 1).float()
         if rep > 1:
             kb = kb.repeat_interleave(rep, dim=0)
             vb = vb.repeat_interleave(rep, dim=0)
@@ -337,17 +359,22 @@ def attn_decode_ref(
     return out
 
 
-def attn_decode(q, k_cache, v_cache, cache_seqlens, softmax_scale=None):
+def attn_decode(q, k_cache, v_cache, cache_seqlens, softmax_scale=None,
+                window=None):
     scale = softmax_scale or (1.0 / math.sqrt(q.shape[-1]))
+    if window is not None and k_cache.shape[1] <= window:
+        window = None  # cache can never exceed the window: plain causal
     if (
         _ops.use_hip(q)
         and q.dtype == torch.bfloat16
         and q.shape[-1] in (64, 128)
         and q.shape[1] // k_cache.shape[2] in (1, 2, 4, 8)
+        and window is None
     ):
         C = _ops.require_hip()
         return C.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
-    return attn_decode_ref(q, k_cache, v_cache, cache_seqlens, scale)
+    return attn_decode_ref(q, k_cache, v_cache, cache_seqlens, scale,
+                           window=window)
 
 
 # ---------------------------------------------------------------------------

@@ -79,3 +79,32 @@ def test_generate_deterministic_with_generator():
     o1 = generate(model, packed, cu, gconfig, generator=torch.Generator().manual_seed(9))
     o2 = generate(model, packed, cu, gconfig, generator=torch.Generator().manual_seed(9))
     assert torch.equal(o1.gen_tokens, o2.gen_tokens)
+
+
+def test_mistral_sliding_window_generate_matches_hf():
+    """Greedy decode with a binding sliding window (cache grows past the
+    window) matches transformers mistral."""
+    fam = hf_reg.get_family("mistral")
+    cfg = fam.make_test_config()
+    cfg.dtype = "float32"
+    cfg.sliding_window = 6
+    hf_model = make_hf_model("mistral", cfg)
+    model = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    hf_reg.load_from_hf_state_dict(model, "mistral", hf_model.state_dict())
+
+    packed, cu, mx = packed_batch(cfg.vocab_size, bs=3, lo=3, hi=6, seed=11)
+    gconfig = GenerationHyperparameters(
+        max_new_tokens=10, greedy=True, use_hip_graph=False
+    )
+    out = generate(model, packed, cu, gconfig, eos_token_id=None)
+    for i in range(3):
+        s, e = int(cu[i]), int(cu[i + 1])
+        hf_out = hf_model.generate(
+            input_ids=packed[s:e].unsqueeze(0),
+            max_new_tokens=10,
+            do_sample=False,
+            use_cache=True,
+            pad_token_id=0,
+        )[0, e - s:]
+        got = out.gen_tokens[i, : int(out.gen_lengths[i])]
+        assert got.tolist() == hf_out.tolist(), (i, got.tolist(), hf_out.tolist())

@@ -60,6 +60,42 @@ def test_forward_matches_hf(family):
             )
 
 
+def test_mistral_sliding_window_matches_hf():
+    """Binding sliding window (window < seqlen) vs transformers mistral."""
+    fam = hf_reg.get_family("mistral")
+    cfg = fam.make_test_config()
+    cfg.dtype = "float32"
+    cfg.sliding_window = 6
+    hf_model = make_hf_model("mistral", cfg)
+    assert hf_model.config.sliding_window == 6
+
+    model = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    hf_reg.load_from_hf_state_dict(model, "mistral", hf_model.state_dict())
+
+    packed, cu, mx = packed_batch(cfg.vocab_size, bs=3, lo=10, hi=20, seed=7)
+    assert mx > cfg.sliding_window  # the window must actually bind
+    with torch.no_grad():
+        ours = model(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+        offs = cu.tolist()
+        for i in range(len(offs) - 1):
+            s, e = offs[i], offs[i + 1]
+            ref = hf_model(input_ids=packed[s:e].unsqueeze(0)).logits[0]
+            torch.testing.assert_close(
+                torch.log_softmax(ours[s:e].float(), -1),
+                torch.log_softmax(ref.float(), -1),
+                atol=2e-4,
+                rtol=2e-3,
+            )
+    # and the window must change the result vs full causal
+    cfg2 = fam.make_test_config()
+    cfg2.dtype = "float32"
+    model2 = ReaLModel(cfg2, device="cpu", dtype=torch.float32)
+    hf_reg.load_from_hf_state_dict(model2, "mistral", hf_model.state_dict())
+    with torch.no_grad():
+        full = model2(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+    assert not torch.allclose(ours, full)
+
+
 def test_logprob_gather_matches_hf_loss():
     fam = hf_reg.get_family("llama")
     cfg = fam.make_test_config()
