@@ -32,6 +32,9 @@ torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           torch::Tensor out32_ws, long splitk,
                           c10::optional<torch::Tensor> residual);
+torch::Tensor skinny_gemm2(torch::Tensor x, torch::Tensor w,
+                           torch::Tensor out32_ws, torch::Tensor sem,
+                           long splitk, c10::optional<torch::Tensor> residual);
 std::pair<std::vector<int64_t>, double> mcmc_search(
     int64_t n_gpus, std::vector<std::vector<std::vector<double>>> cand_rows,
     std::vector<std::vector<int64_t>> parents_in,
@@ -61,5 +64,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_qkv_decode", &rope_qkv_decode);
   m.def("grouped_gemm", &grouped_gemm);
   m.def("skinny_gemm", &skinny_gemm);
+  m.def("skinny_gemm2", &skinny_gemm2);
   m.def("mcmc_search", &mcmc_search);
 }

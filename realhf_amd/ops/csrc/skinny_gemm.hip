@@ -105,6 +105,140 @@ __global__ void sg_combine_kernel(const float* __restrict__ parts,
   }
 }
 
+// ---------------------------------------------------------------------------
+// v2: split-K combine fused into the GEMM kernel via self-resetting
+// semaphores — the LAST workgroup to finish an n-tile sums the fp32
+// partial slabs, folds the residual, converts to bf16 and resets the
+// tile's counter (so the sem buffer only needs zeroing once, at
+// allocation).  Removes the separate sg_combine launch (~5 us x 4 per
+// decode layer) and its extra fp32 pass.  256-deep k body keeps 8
+// 16-byte W loads in flight per lane (vs 4 in v1).
+__global__ __launch_bounds__(256, 4) void skinny_gemm2_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    float* __restrict__ out32, int* __restrict__ sem,
+    const bf16* __restrict__ resid, bf16* __restrict__ out,
+    int M, int N, int K, int kslice, int nks) {
+  const int ntile = blockIdx.x;
+  const int ks = blockIdx.y;
+  const int k_lo = ks * kslice;
+  const int k_hi = min(K, k_lo + kslice);
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int i16 = lane & 15;
+  const int g = lane >> 4;
+  const int n0 = ntile * SG_TN + wv * 16;
+
+  extern __shared__ __bf16 x_s[];  // [16][kslice] rows (zero-padded m >= M)
+  {
+    const int kn = k_hi - k_lo;
+    for (int idx = threadIdx.x * 8; idx < 16 * kslice; idx += 256 * 8) {
+      int m = idx / kslice;
+      int kk = idx % kslice;
+      bf16x8v v = {};
+      if (kk < kn && m < M)
+        v = *(const bf16x8v*)(x + (long)m * K + k_lo + kk);
+      *(bf16x8v*)(&x_s[(long)m * kslice + kk]) = v;
+    }
+  }
+  __syncthreads();
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+  const bf16* wrow = w + (long)(n0 + i16) * K;
+  int kk = k_lo;
+  for (; kk + 256 <= k_hi; kk += 256) {
+    bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
+    bf16x8v b1 = *(const bf16x8v*)(wrow + kk + 32 + g * 8);
+    bf16x8v b2 = *(const bf16x8v*)(wrow + kk + 64 + g * 8);
+    bf16x8v b3 = *(const bf16x8v*)(wrow + kk + 96 + g * 8);
+    bf16x8v b4 = *(const bf16x8v*)(wrow + kk + 128 + g * 8);
+    bf16x8v b5 = *(const bf16x8v*)(wrow + kk + 160 + g * 8);
+    bf16x8v b6 = *(const bf16x8v*)(wrow + kk + 192 + g * 8);
+    bf16x8v b7 = *(const bf16x8v*)(wrow + kk + 224 + g * 8);
+    const long xb = (long)i16 * kslice + (kk - k_lo) + g * 8;
+    bf16x8v a0 = *(const bf16x8v*)(&x_s[xb]);
+    bf16x8v a1 = *(const bf16x8v*)(&x_s[xb + 32]);
+    bf16x8v a2 = *(const bf16x8v*)(&x_s[xb + 64]);
+    bf16x8v a3 = *(const bf16x8v*)(&x_s[xb + 96]);
+    bf16x8v a4 = *(const bf16x8v*)(&x_s[xb + 128]);
+    bf16x8v a5 = *(const bf16x8v*)(&x_s[xb + 160]);
+    bf16x8v a6 = *(const bf16x8v*)(&x_s[xb + 192]);
+    bf16x8v a7 = *(const bf16x8v*)(&x_s[xb + 224]);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc1, 0, 0, 0);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b3, acc1, 0, 0, 0);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a4, b4, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a5, b5, acc1, 0, 0, 0);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a6, b6, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a7, b7, acc1, 0, 0, 0);
+  }
+  for (; kk + 32 <= k_hi; kk += 32) {
+    bf16x8v a0 = *(const bf16x8v*)(&x_s[(long)i16 * kslice + (kk - k_lo) + g * 8]);
+    bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
+  }
+  #pragma unroll
+  for (int r = 0; r < 4; r++) {
+    int m = g * 4 + r;
+    if (m < M)
+      out32[((long)ks * M + m) * N + n0 + i16] = acc0[r] + acc1[r];
+  }
+
+  // --- fused combine: last workgroup of this n-tile sums the slabs
+  __shared__ int is_last;
+  __threadfence();
+  __syncthreads();
+  if (threadIdx.x == 0)
+    is_last = (atomicAdd(&sem[ntile], 1) == nks - 1) ? 1 : 0;
+  __syncthreads();
+  if (!is_last) return;
+  __threadfence();  // acquire: see every slab written before the last inc
+  const int base_n = ntile * SG_TN;
+  for (int e = threadIdx.x; e < M * SG_TN; e += 256) {
+    int m = e >> 6, c = e & 63;
+    long off = (long)m * N + base_n + c;
+    float vsum = 0.f;
+    for (int s = 0; s < nks; s++) vsum += out32[(long)s * M * N + off];
+    if (resid) vsum += __bfloat162float(resid[off]);
+    short o = f2bf(vsum);
+    *((short*)out + off) = o;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) atomicExch(&sem[ntile], 0);  // self-reset
+}
+
+torch::Tensor skinny_gemm2(torch::Tensor x, torch::Tensor w,
+                           torch::Tensor out32_ws, torch::Tensor sem,
+                           long splitk, c10::optional<torch::Tensor> residual) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1);
+  TORCH_CHECK(w.dim() == 2 && w.stride(1) == 1);
+  int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(M <= 16 && K % 32 == 0 && N % SG_TN == 0);
+  TORCH_CHECK(x.stride(0) == K, "x must be contiguous");
+  int kslice = (K / (int)splitk + 63) / 64 * 64;
+  int nks = (K + kslice - 1) / kslice;
+  TORCH_CHECK(out32_ws.numel() >= (long)nks * M * N, "workspace too small");
+  TORCH_CHECK(sem.scalar_type() == torch::kInt32 && sem.numel() >= N / SG_TN,
+              "semaphore buffer too small (must be zero-initialized once)");
+  auto out = torch::empty({M, (long)N}, x.options());
+  const bf16* rptr = nullptr;
+  if (residual.has_value()) {
+    TORCH_CHECK(residual->is_contiguous() &&
+                residual->numel() == (long)M * N);
+    rptr = (const bf16*)residual->data_ptr();
+  }
+  dim3 grid(N / SG_TN, nks);
+  size_t lds = (size_t)16 * kslice * sizeof(short);
+  TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
+  hipLaunchKernelGGL(skinny_gemm2_kernel, grid, dim3(256), lds, cur_stream(),
+    (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+    out32_ws.data_ptr<float>(), sem.data_ptr<int>(), rptr,
+    (bf16*)out.data_ptr(), M, N, K, kslice, nks);
+  CHECK_CUDA_OK();
+  return out;
+}
+
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           torch::Tensor out32_ws, long splitk,
                           c10::optional<torch::Tensor> residual) {

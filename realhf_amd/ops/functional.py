@@ -506,7 +506,25 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
     )
     if residual is not None:
         residual = residual.contiguous()
-    return C.skinny_gemm(x, w, ws, splitk, residual)
+    if os.environ.get("REALHF_AMD_SKINNY_V1") == "1":
+        return C.skinny_gemm(x, w, ws, splitk, residual)
+    # v2: split-K combine fused into the GEMM kernel via self-resetting
+    # per-n-tile semaphores (saves a ~5us launch + an fp32 pass per call)
+    return C.skinny_gemm2(x, w, ws, _skinny_sem(x.device), splitk, residual)
+
+
+_SKINNY_SEM: dict = {}
+
+
+def _skinny_sem(device) -> torch.Tensor:
+    """Per-device split-K semaphore buffer (one int32 per n-tile; 1024
+    covers N up to 65536).  Zeroed ONCE at allocation; the kernel leaves
+    every counter back at zero."""
+    t = _SKINNY_SEM.get(device)
+    if t is None:
+        t = torch.zeros(1024, dtype=torch.int32, device=device)
+        _SKINNY_SEM[device] = t
+    return t
 
 
 # ---------------------------------------------------------------------------

@@ -323,3 +323,24 @@ def test_skinny_gemm_with_residual():
     x2 = x[:5]
     out2 = C.skinny_gemm(x2.contiguous(), w, ws, 8, None)
     torch.testing.assert_close(out2.float(), ref[:5], atol=0.3, rtol=3e-2)
+
+
+def test_skinny_gemm2_fused_combine():
+    """v2 (semaphore-fused split-K combine) vs fp32 reference, repeated
+    launches to exercise the self-resetting semaphores."""
+    torch.manual_seed(16)
+    sem = torch.zeros(1024, dtype=torch.int32, device="cuda")
+    for (M, K, N, sk) in [(16, 4096, 12288, 8), (16, 4096, 4096, 8),
+                          (16, 11008, 4096, 16), (5, 4096, 4096, 8)]:
+        x = (torch.randn(M, K, device="cuda") * 0.3).to(torch.bfloat16)
+        w = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)
+        res = (torch.randn(M, N, device="cuda") * 0.3).to(torch.bfloat16)
+        ws = torch.empty(32 * 16 * N, dtype=torch.float32, device="cuda")
+        ref = x.float() @ w.float().t()
+        for rep in range(3):  # reuse => semaphores must be back at zero
+            out = C.skinny_gemm2(x, w, ws, sem, sk, None)
+            torch.testing.assert_close(out.float(), ref, atol=0.3, rtol=3e-2)
+        out_r = C.skinny_gemm2(x, w, ws, sem, sk, res)
+        torch.testing.assert_close(out_r.float(), ref + res.float(),
+                                   atol=0.3, rtol=3e-2)
+        assert int(sem.abs().sum()) == 0  # all counters self-reset

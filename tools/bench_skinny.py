@@ -39,8 +39,10 @@ def bench(fn, iters=50):
 def main():
     M = 16
     ws = torch.empty(32 * M * 32000, dtype=torch.float32, device="cuda")
+    sem = torch.zeros(1024, dtype=torch.int32, device="cuda")
     print(f"{'shape':8} {'N':>6} {'K':>6} {'sk':>3} {'blaslt_us':>9} "
-          f"{'skinny_us':>9} {'roofline_us':>11} {'max_err':>8}")
+          f"{'skinny_us':>9} {'v2_us':>7} {'roofline_us':>11} "
+          f"{'max_err':>8} {'v2_err':>8}")
     for name, N, K in SHAPES:
         x = (torch.randn(M, K, device="cuda") * 0.3).to(torch.bfloat16)
         w = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)
@@ -52,9 +54,12 @@ def main():
                 continue
             out = C.skinny_gemm(x, w, ws, sk, None)
             err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+            out2 = C.skinny_gemm2(x, w, ws, sem, sk, None)
+            err2 = (out2.float() - ref).abs().max().item() / ref.abs().max().item()
             t_sk = bench(lambda: C.skinny_gemm(x, w, ws, sk, None))
+            t_v2 = bench(lambda: C.skinny_gemm2(x, w, ws, sem, sk, None))
             print(f"{name:8} {N:6d} {K:6d} {sk:3d} {t_blas:9.1f} {t_sk:9.1f} "
-                  f"{roof:11.1f} {err:8.4f}")
+                  f"{t_v2:7.1f} {roof:11.1f} {err:8.4f} {err2:8.4f}")
 
 
 if __name__ == "__main__":
