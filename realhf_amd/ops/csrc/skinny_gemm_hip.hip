@@ -119,8 +119,23 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm2_kernel(
     float* __restrict__ out32, int* __restrict__ sem,
     const bf16* __restrict__ resid, bf16* __restrict__ out,
     int M, int N, int K, int kslice, int nks) {
-  const int ntile = blockIdx.x;
-  const int ks = blockIdx.y;
+  // 1-D grid; when the tile count divides by 8, keep all K-slices of one
+  // n-tile on ONE XCD (the dispatcher places block b on XCD b%8) so the
+  // reducer reads same-XCD slabs — a pure speed choice, correctness comes
+  // from the agent-scope fences below (guide §6 G16 / split-K recipe).
+  const int nt_total = N / SG_TN;
+  int ntile, ks;
+  {
+    const int id = blockIdx.x;
+    if ((nt_total & 7) == 0) {
+      const int ix = id >> 3;
+      ntile = (id & 7) * (nt_total >> 3) + ix / nks;
+      ks = ix % nks;
+    } else {
+      ntile = id / nks;
+      ks = id % nks;
+    }
+  }
   const int k_lo = ks * kslice;
   const int k_hi = min(K, k_lo + kslice);
   const int lane = threadIdx.x & 63;
@@ -178,6 +193,7 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm2_kernel(
     bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
     acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
   }
+  // publish the fp32 slab with plain stores
   #pragma unroll
   for (int r = 0; r < 4; r++) {
     int m = g * 4 + r;
@@ -185,27 +201,59 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm2_kernel(
       out32[((long)ks * M + m) * N + n0 + i16] = acc0[r] + acc1[r];
   }
 
-  // --- fused combine: last workgroup of this n-tile sums the slabs
-  __shared__ int is_last;
-  __threadfence();
+  // --- in-launch split-K hand-off (guide split-K recipe, counter form):
+  // every wave drains its stores; lane 0 issues ONE agent-scope release
+  // (buffer_wbl2) with the post-fence wait restated (ROCm 7.2 drops it
+  // otherwise), THEN takes a relaxed ticket.  NEVER __threadfence() per
+  // block — that is a per-WG L2 writeback+invalidate, measured 4x slower
+  // end-to-end in the decode graph.
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
-  if (threadIdx.x == 0)
-    is_last = (atomicAdd(&sem[ntile], 1) == nks - 1) ? 1 : 0;
-  __syncthreads();
-  if (!is_last) return;
-  __threadfence();  // acquire: see every slab written before the last inc
-  const int base_n = ntile * SG_TN;
-  for (int e = threadIdx.x; e < M * SG_TN; e += 256) {
-    int m = e >> 6, c = e & 63;
-    long off = (long)m * N + base_n + c;
-    float vsum = 0.f;
-    for (int s = 0; s < nks; s++) vsum += out32[(long)s * M * N + off];
-    if (resid) vsum += __bfloat162float(resid[off]);
-    short o = f2bf(vsum);
-    *((short*)out + off) = o;
+  volatile int* flag = (volatile int*)x_s;  // reuse the ONE shared array
+  if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    int tick = __hip_atomic_fetch_add(&sem[ntile], 1, __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT);
+    flag[0] = (tick == nks - 1) ? 1 : 0;
   }
   __syncthreads();
-  if (threadIdx.x == 0) atomicExch(&sem[ntile], 0);  // self-reset
+  if (!flag[0]) return;
+
+  // --- last arriver reduces: one agent-scope acquire (drops this CU's
+  // L1), then plain 16-B slab loads.  Thread t owns row t/16, columns
+  // 4*(t%16)..+3 of the 16 x 64 tile.
+  if (threadIdx.x == 0)
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+  const int base_n = ntile * SG_TN;
+  const int m = threadIdx.x >> 4;
+  const int c4 = (threadIdx.x & 15) * 4;
+  if (m < M) {
+    const long off = (long)m * N + base_n + c4;
+    float4v vsum = *(const float4v*)(out32 + off);
+    for (int s = 1; s < nks; s++) {
+      float4v p = *(const float4v*)(out32 + (long)s * M * N + off);
+      #pragma unroll
+      for (int j = 0; j < 4; j++) vsum[j] += p[j];
+    }
+    short o4[4];
+    if (resid) {
+      short4v rv = *(const short4v*)((const short*)resid + off);
+      #pragma unroll
+      for (int j = 0; j < 4; j++)
+        o4[j] = f2bf(vsum[j] + __bfloat162float(((const bf16*)&rv)[j]));
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 4; j++) o4[j] = f2bf(vsum[j]);
+    }
+    *(short4v*)((short*)out + off) = *(short4v*)o4;
+  }
+  // self-reset so the persistent sem buffer is zero for the next launch
+  // (stream order makes this safe; the buffer is zeroed once at alloc)
+  if (threadIdx.x == 0)
+    __hip_atomic_store(&sem[ntile], 0, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
 }
 
 torch::Tensor skinny_gemm2(torch::Tensor x, torch::Tensor w,
@@ -229,7 +277,7 @@ torch::Tensor skinny_gemm2(torch::Tensor x, torch::Tensor w,
                 residual->numel() == (long)M * N);
     rptr = (const bf16*)residual->data_ptr();
   }
-  dim3 grid(N / SG_TN, nks);
+  dim3 grid((N / SG_TN) * nks);  // 1-D: kernel derives (tile, slice)
   size_t lds = (size_t)16 * kslice * sizeof(short);
   TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
   hipLaunchKernelGGL(skinny_gemm2_kernel, grid, dim3(256), lds, cur_stream(),

@@ -1,8 +1,11 @@
 #!/usr/bin/env python3
 """A/B the weight-streaming skinny GEMM vs torch.matmul (hipBLASLt) on the
 7B decode shapes.  Run on an MI355X."""
+import os
+import sys
 import time
 
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 
 import realhf_amd._C as C
