@@ -506,11 +506,13 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
     )
     if residual is not None:
         residual = residual.contiguous()
-    if os.environ.get("REALHF_AMD_SKINNY_V1") == "1":
-        return C.skinny_gemm(x, w, ws, splitk, residual)
-    # v2: split-K combine fused into the GEMM kernel via self-resetting
-    # per-n-tile semaphores (saves a ~5us launch + an fp32 pass per call)
-    return C.skinny_gemm2(x, w, ws, _skinny_sem(x.device), splitk, residual)
+    if os.environ.get("REALHF_AMD_SKINNY_V2") == "1":
+        # v2 (in-launch semaphore combine) measured SLOWER at every decode
+        # shape: the per-block agent-scope release (buffer_wbl2) does not
+        # amortize at 1.5k-2.7k blocks (qkv 67us vs 22.8us at splitk 8).
+        # Kept for reference / small-grid experiments only.
+        return C.skinny_gemm2(x, w, ws, _skinny_sem(x.device), splitk, residual)
+    return C.skinny_gemm(x, w, ws, splitk, residual)
 
 
 _SKINNY_SEM: dict = {}
