@@ -196,7 +196,11 @@ class ReaLModelBlock(nn.Module):
             qkv_dim = (self.nq + 2 * self.nkv) * self.hd
             merged = _maybe_merged(self.p, self._qkv_names, qkv_dim)
             if merged is not None:
-                qkv_raw = ops.maybe_skinny_linear(h, merged)
+                # launch-boundary reduce: hand the split-K fp32 slabs
+                # straight to rope_qkv_decode's prologue (no combine pass)
+                qkv_raw = ops.skinny_linear_nc(h, merged)
+                if qkv_raw is None:
+                    qkv_raw = ops.maybe_skinny_linear(h, merged)
                 if qkv_raw is None:
                     qkv_raw = _linear(h, merged)
             else:
@@ -219,6 +223,21 @@ class ReaLModelBlock(nn.Module):
             attn_out = ops.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
             attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
             wo = self.p[f"{i}.attn.wo.weight"]
+            if (
+                self.tp_size == 1
+                and f"{i}.attn.wo.bias" not in self.p
+                and cfg.norm_type == "rms"
+                and self.moe is None
+            ):
+                # launch-boundary reduce: o's split-K slabs are summed in
+                # the fused add+rmsnorm prologue (no combine, no bf16 o)
+                o_parts = ops.skinny_linear_nc(attn_out, wo)
+                if o_parts is not None:
+                    h2, x2 = C.add_rmsnorm_fwd(
+                        o_parts, x, self.p[f"{i}.mlp.ln.weight"],
+                        cfg.layer_norm_epsilon,
+                    )
+                    return self._mlp_body(h2, sp, residual=x2)
             o = ops.maybe_skinny_linear(attn_out, wo)
             if o is None:
                 o = _linear(attn_out, wo)
@@ -311,15 +330,26 @@ class ReaLModelBlock(nn.Module):
         if cfg.activation in ("silu", "geglu"):
             idim_local = self.p[f"{i}.mlp.gate.weight"].shape[0]
             merged = _maybe_merged(self.p, self._gu_names, 2 * idim_local)
+            act = None
             if merged is not None:
-                gu = ops.maybe_skinny_linear(h, merged)
-                if gu is None:
-                    gu = _linear(h, merged)
+                if cfg.activation == "silu" and self.tp_size == 1:
+                    # launch-boundary reduce: slabs summed in swiglu
+                    gu_parts = ops.skinny_linear_nc(h, merged)
+                    if gu_parts is not None:
+                        from realhf_amd import ops as _ops_pkg
+
+                        act = _ops_pkg.require_hip().swiglu_fwd(gu_parts)
+                if act is None:
+                    gu = ops.maybe_skinny_linear(h, merged)
+                    if gu is None:
+                        gu = _linear(h, merged)
             else:
                 gu = torch.cat(
                     [_linear(h, self.p[n]) for n in self._gu_names], dim=-1
                 )
-            if cfg.activation == "silu":
+            if act is not None:
+                pass
+            elif cfg.activation == "silu":
                 act = ops.swiglu(gu)
             else:  # geglu (gemma)
                 gate, up = gu.chunk(2, dim=-1)

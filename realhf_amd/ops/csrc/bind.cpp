@@ -32,6 +32,8 @@ torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           torch::Tensor out32_ws, long splitk,
                           c10::optional<torch::Tensor> residual);
+torch::Tensor skinny_gemm_nc(torch::Tensor x, torch::Tensor w,
+                             torch::Tensor out32_ws, long splitk);
 torch::Tensor skinny_gemm2(torch::Tensor x, torch::Tensor w,
                            torch::Tensor out32_ws, torch::Tensor sem,
                            long splitk, c10::optional<torch::Tensor> residual);
@@ -65,5 +67,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grouped_gemm", &grouped_gemm);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("skinny_gemm2", &skinny_gemm2);
+  m.def("skinny_gemm_nc", &skinny_gemm_nc);
   m.def("mcmc_search", &mcmc_search);
 }

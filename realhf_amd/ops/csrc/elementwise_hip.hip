@@ -101,6 +101,44 @@ __global__ void swiglu_fwd_kernel(const T* __restrict__ gu, T* __restrict__ out,
   }
 }
 
+// slab variant: gate_up comes in as fp32 split-K partial slabs
+// [nks, tokens, 2I] from skinny_gemm_nc (launch-boundary reduce)
+template <typename T>
+__global__ void swiglu_slab_kernel(const float* __restrict__ parts, int nks,
+                                   T* __restrict__ out, long tokens, int I) {
+  const long sstride = tokens * (2L * I);
+  long n = tokens * (I / 8);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    long t = i / (I / 8);
+    int d0 = (int)(i % (I / 8)) * 8;
+    const float* g = parts + t * (2L * I) + d0;
+    const float* u = g + I;
+    float gf[8], uf[8];
+    #pragma unroll
+    for (int j = 0; j < 8; j += 4) {
+      *(float4v*)(gf + j) = *(const float4v*)(g + j);
+      *(float4v*)(uf + j) = *(const float4v*)(u + j);
+    }
+    for (int s = 1; s < nks; s++) {
+      #pragma unroll
+      for (int j = 0; j < 8; j += 4) {
+        float4v pg = *(const float4v*)(g + (long)s * sstride + j);
+        float4v pu = *(const float4v*)(u + (long)s * sstride + j);
+        #pragma unroll
+        for (int q = 0; q < 4; q++) { gf[j + q] += pg[q]; uf[j + q] += pu[q]; }
+      }
+    }
+    short o[8];
+    #pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float sg = gf[j] / (1.f + __expf(-gf[j]));
+      ((T*)o)[j] = from_f32<T>(sg * uf[j]);
+    }
+    *(short8*)(out + t * (long)I + d0) = *(short8*)o;
+  }
+}
+
 template <typename T>
 __global__ void swiglu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ gu,
                                   T* __restrict__ dgu, long tokens, int I) {
@@ -131,9 +169,24 @@ __global__ void swiglu_bwd_kernel(const T* __restrict__ dy, const T* __restrict_
 }
 
 torch::Tensor swiglu_fwd(torch::Tensor gu) {
-  TORCH_CHECK(gu.is_contiguous() && gu.element_size() == 2);
+  TORCH_CHECK(gu.is_contiguous());
   int I2 = gu.size(-1);
   TORCH_CHECK(I2 % 16 == 0);
+  if (gu.dim() == 3 && gu.scalar_type() == torch::kFloat) {
+    // fp32 split-K partial slabs [nks, tokens, 2I] (launch-boundary reduce)
+    long tokens = gu.size(1);
+    int I = I2 / 2;
+    auto out = torch::empty({tokens, (long)I},
+                            gu.options().dtype(torch::kBFloat16));
+    long n = tokens * (I / 8);
+    int grid = (int)std::min<long>((n + 255) / 256, 8192);
+    hipLaunchKernelGGL((swiglu_slab_kernel<bf16>), dim3(grid), dim3(256), 0,
+      cur_stream(), gu.data_ptr<float>(), (int)gu.size(0),
+      (bf16*)out.data_ptr(), tokens, I);
+    CHECK_CUDA_OK();
+    return out;
+  }
+  TORCH_CHECK(gu.element_size() == 2);
   long tokens = gu.numel() / I2;
   int I = I2 / 2;
   auto out = torch::empty({tokens, I}, gu.options());

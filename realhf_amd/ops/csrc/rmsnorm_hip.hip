@@ -56,6 +56,70 @@ __global__ void rmsnorm_fwd_kernel(
   }
 }
 
+// slab variant of the fused add+rmsnorm: x comes in as fp32 split-K
+// partial slabs [nks, rows, H] from skinny_gemm_nc (launch-boundary
+// reduce); sum_out = sum_s(parts) + resid, out = rmsnorm(sum_out).
+template <typename T, int VEC>
+__global__ void add_rmsnorm_slab_kernel(
+    const float* __restrict__ parts, int nks, const T* __restrict__ resid,
+    const T* __restrict__ w, T* __restrict__ out, T* __restrict__ sum_out,
+    int rows, int H, float eps) {
+  __shared__ float red[8];
+  const long sstride = (long)rows * H;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    T* yr = out + (long)row * H;
+    float ss = 0.f;
+    for (int i = threadIdx.x * VEC; i < H; i += blockDim.x * VEC) {
+      float f[VEC];
+      const float* pr = parts + (long)row * H + i;
+      #pragma unroll
+      for (int j = 0; j < VEC; j += 4)
+        *(float4v*)(f + j) = *(const float4v*)(pr + j);
+      for (int s = 1; s < nks; s++) {
+        #pragma unroll
+        for (int j = 0; j < VEC; j += 4) {
+          float4v p = *(const float4v*)(pr + (long)s * sstride + j);
+          #pragma unroll
+          for (int q = 0; q < 4; q++) f[j + q] += p[q];
+        }
+      }
+      T r[VEC], v[VEC];
+      *(float4v*)r = *(const float4v*)(resid + (long)row * H + i);
+      #pragma unroll
+      for (int j = 0; j < VEC; j++) {
+        float fv = f[j] + to_f32<T>(r[j]);
+        v[j] = from_f32<T>(fv);
+        float fq = to_f32<T>(v[j]);
+        ss += fq * fq;
+      }
+      *(float4v*)(sum_out + (long)row * H + i) = *(float4v*)v;
+    }
+    ss = wave_sum(ss);
+    int wid = threadIdx.x / WAVE;
+    int nw = blockDim.x / WAVE;
+    if ((threadIdx.x & (WAVE - 1)) == 0) red[wid] = ss;
+    __syncthreads();
+    if (threadIdx.x < 8) {
+      float v2 = (threadIdx.x < nw) ? red[threadIdx.x] : 0.f;
+      for (int off = 4; off > 0; off >>= 1) v2 += __shfl_down(v2, off, 64);
+      if (threadIdx.x == 0) red[0] = v2;
+    }
+    __syncthreads();
+    float rs = rsqrtf(red[0] / H + eps);
+    const T* src2 = sum_out + (long)row * H;
+    for (int i = threadIdx.x * VEC; i < H; i += blockDim.x * VEC) {
+      T v[VEC], wv[VEC], o[VEC];
+      *(float4v*)v = *(const float4v*)(src2 + i);
+      *(float4v*)wv = *(const float4v*)(w + i);
+      #pragma unroll
+      for (int j = 0; j < VEC; j++)
+        o[j] = from_f32<T>(to_f32<T>(v[j]) * rs * to_f32<T>(wv[j]));
+      *(float4v*)(yr + i) = *(float4v*)o;
+    }
+    __syncthreads();
+  }
+}
+
 // backward: dx = rs * w * dy - rs^3/H * x * sum(dy * w * x)
 // dw: per-thread register partials over this block's rows (each thread owns
 // fixed columns), ONE atomicAdd per column per block at the end.
@@ -149,15 +213,28 @@ std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double 
 
 std::vector<torch::Tensor> add_rmsnorm_fwd(torch::Tensor x, torch::Tensor resid,
                                            torch::Tensor w, double eps) {
-  // out_norm = rmsnorm(x + resid); sum_out = x + resid (the new residual)
+  // out_norm = rmsnorm(x + resid); sum_out = x + resid (the new residual).
+  // x may also be fp32 split-K partial slabs [nks, rows, H] from
+  // skinny_gemm_nc — then x := sum over slabs (launch-boundary reduce).
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && resid.is_contiguous());
-  TORCH_CHECK(x.element_size() == 2);
   int H = x.size(-1);
-  long rows = x.numel() / H;
+  long rows = resid.numel() / H;
   TORCH_CHECK(H % 8 == 0);
-  auto out = torch::empty_like(x);
-  auto sum_out = torch::empty_like(x);
+  auto out = torch::empty_like(resid);
+  auto sum_out = torch::empty_like(resid);
   int grid = (int)std::min<long>(rows, 2048);
+  if (x.dim() == 3 && x.scalar_type() == torch::kFloat) {
+    TORCH_CHECK(resid.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(x.size(1) * x.size(2) == rows * H);
+    hipLaunchKernelGGL((add_rmsnorm_slab_kernel<bf16, 8>), dim3(grid),
+      dim3(256), 0, cur_stream(), x.data_ptr<float>(), (int)x.size(0),
+      (const bf16*)resid.data_ptr(), (const bf16*)w.data_ptr(),
+      (bf16*)out.data_ptr(), (bf16*)sum_out.data_ptr(), (int)rows, H,
+      (float)eps);
+    CHECK_CUDA_OK();
+    return {out, sum_out};
+  }
+  TORCH_CHECK(x.element_size() == 2);
   DISPATCH_BF16_FP16_FP32(x.scalar_type(), "add_rmsnorm", [&] {
     if constexpr (sizeof(scalar_t) == 2) {
       hipLaunchKernelGGL((rmsnorm_fwd_kernel<scalar_t, 8, true>), dim3(grid),

@@ -106,6 +106,35 @@ __global__ void sg_combine_kernel(const float* __restrict__ parts,
 }
 
 // ---------------------------------------------------------------------------
+// no-combine variant: run ONLY the split-K GEMM and hand the fp32 partial
+// slabs [nks, M, N] to the NEXT kernel's prologue (launch-boundary
+// reduce, guide split-K recipe: "combine in the NEXT kernel's prologue
+// ... costs nothing extra when that kernel exists anyway").  Consumers:
+// rope_qkv_decode / add_rmsnorm_fwd / swiglu_fwd slab modes.
+torch::Tensor skinny_gemm_nc(torch::Tensor x, torch::Tensor w,
+                             torch::Tensor out32_ws, long splitk) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1);
+  TORCH_CHECK(w.dim() == 2 && w.stride(1) == 1);
+  int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(M <= 16 && K % 32 == 0 && N % SG_TN == 0);
+  TORCH_CHECK(x.stride(0) == K, "x must be contiguous");
+  int kslice = (K / (int)splitk + 63) / 64 * 64;
+  int nks = (K + kslice - 1) / kslice;
+  TORCH_CHECK(out32_ws.numel() >= (long)nks * M * N, "workspace too small");
+  auto out32 = out32_ws.narrow(0, 0, (long)nks * M * N)
+                   .view({(long)nks, (long)M, (long)N});
+  dim3 grid(N / SG_TN, nks);
+  size_t lds = (size_t)16 * kslice * sizeof(short);
+  TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
+  hipLaunchKernelGGL(skinny_gemm_kernel, grid, dim3(256), lds, cur_stream(),
+    (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+    out32.data_ptr<float>(), M, N, K, kslice);
+  CHECK_CUDA_OK();
+  return out32;
+}
+
+// ---------------------------------------------------------------------------
 // v2: split-K combine fused into the GEMM kernel via self-resetting
 // semaphores — the LAST workgroup to finish an n-tile sums the fp32
 // partial slabs, folds the residual, converts to bf16 and resets the

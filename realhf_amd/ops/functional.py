@@ -515,6 +515,40 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
     return C.skinny_gemm(x, w, ws, splitk, residual)
 
 
+def skinny_linear_nc(x: torch.Tensor, w: torch.Tensor) -> Optional[torch.Tensor]:
+    """Like maybe_skinny_linear but WITHOUT the split-K combine: returns
+    the fp32 partial slabs [nks, M, N], to be summed in the CONSUMING
+    kernel's prologue (launch-boundary reduce: rope_qkv_decode /
+    add_rmsnorm_fwd / swiglu_fwd all accept slabs).  The slabs live in the
+    shared skinny workspace — consume them before the next skinny call."""
+    if not (
+        x.is_cuda
+        and x.dtype == torch.bfloat16
+        and not torch.is_grad_enabled()
+        and x.dim() == 2
+        and x.shape[0] <= 16
+        and x.stride(1) == 1
+        and x.stride(0) == x.shape[1]
+        and w.stride(1) == 1
+        and w.shape[0] % 64 == 0
+        and w.shape[1] % 32 == 0
+        and _ops.hip_available()
+    ):
+        return None
+    from realhf_amd.base.constants import get_global_memory_buffer
+
+    C = _ops.require_hip()
+    K = x.shape[1]
+    N = w.shape[0]
+    splitk = 8
+    while K // splitk > 1024:
+        splitk *= 2
+    ws = get_global_memory_buffer().get_tensor(
+        (2 * splitk * 16 * N,), torch.float32, "skinny_gemm_ws"
+    )
+    return C.skinny_gemm_nc(x, w, ws, splitk)
+
+
 _SKINNY_SEM: dict = {}
 
 

@@ -344,3 +344,58 @@ def test_skinny_gemm2_fused_combine():
         torch.testing.assert_close(out_r.float(), ref + res.float(),
                                    atol=0.3, rtol=3e-2)
         assert int(sem.abs().sum()) == 0  # all counters self-reset
+
+
+def test_slab_consumers_match_combined():
+    """skinny_gemm_nc slabs consumed in rope/add_rmsnorm/swiglu prologues
+    (launch-boundary reduce) match the combine-then-consume path."""
+    torch.manual_seed(17)
+    M, H = 16, 4096
+    ws = torch.empty(32 * 16 * 32000, dtype=torch.float32, device="cuda")
+
+    # --- add_rmsnorm slab mode
+    wo = (torch.randn(H, H, device="cuda") * 0.05).to(torch.bfloat16)
+    attn = (torch.randn(M, H, device="cuda") * 0.3).to(torch.bfloat16)
+    x = (torch.randn(M, H, device="cuda") * 0.3).to(torch.bfloat16)
+    g = torch.randn(H, device="cuda").to(torch.bfloat16)
+    o_bf = C.skinny_gemm(attn, wo, ws, 8, None)
+    ref_h2, ref_x2 = C.add_rmsnorm_fwd(o_bf, x, g, 1e-5)
+    parts = C.skinny_gemm_nc(attn, wo, ws, 8)
+    assert parts.shape == (8, M, H)
+    h2, x2 = C.add_rmsnorm_fwd(parts, x, g, 1e-5)
+    torch.testing.assert_close(x2.float(), ref_x2.float(), atol=0.1, rtol=5e-2)
+    torch.testing.assert_close(h2.float(), ref_h2.float(), atol=0.1, rtol=5e-2)
+
+    # --- swiglu slab mode
+    I = 11008
+    wgu = (torch.randn(2 * I, H, device="cuda") * 0.05).to(torch.bfloat16)
+    gu_bf = C.skinny_gemm(h2, wgu, ws, 8, None)
+    ref_act = C.swiglu_fwd(gu_bf)
+    gu_parts = C.skinny_gemm_nc(h2, wgu, ws, 8)
+    act = C.swiglu_fwd(gu_parts)
+    torch.testing.assert_close(act.float(), ref_act.float(), atol=0.1,
+                               rtol=5e-2)
+
+    # --- rope_qkv_decode slab mode (bs=16, llama-7B head geometry)
+    nq = nkv = 8
+    hd = 128
+    qkvd = (nq + 2 * nkv) * hd
+    wqkv = (torch.randn(qkvd, H, device="cuda") * 0.05).to(torch.bfloat16)
+    hin = (torch.randn(M, H, device="cuda") * 0.3).to(torch.bfloat16)
+    maxlen = 64
+    kc1 = torch.zeros(M, maxlen, nkv, hd, dtype=torch.bfloat16, device="cuda")
+    vc1 = torch.zeros_like(kc1)
+    kc2 = torch.zeros_like(kc1)
+    vc2 = torch.zeros_like(kc1)
+    lens = torch.randint(1, maxlen, (M,), dtype=torch.int32, device="cuda")
+    pos = torch.arange(maxlen, device="cuda", dtype=torch.float32)
+    inv = 1.0 / (10000.0 ** (torch.arange(0, hd, 2, device="cuda").float() / hd))
+    ang = pos[:, None] * inv[None, :]
+    cosb, sinb = ang.cos().contiguous(), ang.sin().contiguous()
+    qkv_bf = C.skinny_gemm(hin, wqkv, ws, 8, None)
+    q_ref = C.rope_qkv_decode(qkv_bf, None, kc1, vc1, lens, cosb, sinb, nq, True)
+    qkv_parts = C.skinny_gemm_nc(hin, wqkv, ws, 8)
+    q = C.rope_qkv_decode(qkv_parts, None, kc2, vc2, lens, cosb, sinb, nq, True)
+    torch.testing.assert_close(q.float(), q_ref.float(), atol=0.1, rtol=5e-2)
+    torch.testing.assert_close(kc2.float(), kc1.float(), atol=0.1, rtol=5e-2)
+    torch.testing.assert_close(vc2.float(), vc1.float(), atol=0.1, rtol=5e-2)
