@@ -16,6 +16,7 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 
 #define SG_TN 64  // n per workgroup (16 per wave)
 
+template <int NB>  // NB x 32-deep k body: NB 16-byte W loads in flight/lane
 __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w,
     float* __restrict__ out32, int M, int N, int K, int kslice) {
@@ -46,23 +47,25 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
   f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
   const bf16* wrow = w + (long)(n0 + i16) * K;
   int kk = k_lo;
-  // 128-deep body: issue all four W loads before the first MFMA so >=4
+  // NB*32-deep body: issue all NB W loads before the first MFMA so >= NB
   // 16B loads stay in flight per lane (one k-iter alone is latency-bound)
   // (nt loads measured 15-40% SLOWER here — keep plain loads)
-  for (; kk + 128 <= k_hi; kk += 128) {
-    bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
-    bf16x8v b1 = *(const bf16x8v*)(wrow + kk + 32 + g * 8);
-    bf16x8v b2 = *(const bf16x8v*)(wrow + kk + 64 + g * 8);
-    bf16x8v b3 = *(const bf16x8v*)(wrow + kk + 96 + g * 8);
+  for (; kk + NB * 32 <= k_hi; kk += NB * 32) {
+    bf16x8v b[NB], a[NB];
+    #pragma unroll
+    for (int t = 0; t < NB; t++)
+      b[t] = *(const bf16x8v*)(wrow + kk + t * 32 + g * 8);
     const long xb = (long)i16 * kslice + (kk - k_lo) + g * 8;
-    bf16x8v a0 = *(const bf16x8v*)(&x_s[xb]);
-    bf16x8v a1 = *(const bf16x8v*)(&x_s[xb + 32]);
-    bf16x8v a2 = *(const bf16x8v*)(&x_s[xb + 64]);
-    bf16x8v a3 = *(const bf16x8v*)(&x_s[xb + 96]);
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
-    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc1, 0, 0, 0);
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a2, b2, acc0, 0, 0, 0);
-    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a3, b3, acc1, 0, 0, 0);
+    #pragma unroll
+    for (int t = 0; t < NB; t++)
+      a[t] = *(const bf16x8v*)(&x_s[xb + t * 32]);
+    #pragma unroll
+    for (int t = 0; t < NB; t++) {
+      if (t & 1)
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[t], b[t], acc1, 0, 0, 0);
+      else
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[t], b[t], acc0, 0, 0, 0);
+    }
   }
   for (; kk + 32 <= k_hi; kk += 32) {
     bf16x8v a0 = *(const bf16x8v*)(&x_s[(long)i16 * kslice + (kk - k_lo) + g * 8]);
@@ -77,6 +80,15 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
     if (m < M)
       out32[((long)ks * M + m) * N + n0 + i16] = vsum;
   }
+}
+
+// pipeline depth: 4 (default) or 8 via REALHF_AMD_SKINNY_DEPTH=8
+static int sg_depth() {
+  static int d = [] {
+    const char* e = getenv("REALHF_AMD_SKINNY_DEPTH");
+    return (e && e[0] == '8') ? 8 : 4;
+  }();
+  return d;
 }
 
 // combine the nks fp32 partial slabs -> bf16 (+ optional residual add)
@@ -127,9 +139,14 @@ torch::Tensor skinny_gemm_nc(torch::Tensor x, torch::Tensor w,
   dim3 grid(N / SG_TN, nks);
   size_t lds = (size_t)16 * kslice * sizeof(short);
   TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
-  hipLaunchKernelGGL(skinny_gemm_kernel, grid, dim3(256), lds, cur_stream(),
-    (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
-    out32.data_ptr<float>(), M, N, K, kslice);
+  if (sg_depth() == 8)
+    hipLaunchKernelGGL(skinny_gemm_kernel<8>, grid, dim3(256), lds, cur_stream(),
+      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+      out32.data_ptr<float>(), M, N, K, kslice);
+  else
+    hipLaunchKernelGGL(skinny_gemm_kernel<4>, grid, dim3(256), lds, cur_stream(),
+      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+      out32.data_ptr<float>(), M, N, K, kslice);
   CHECK_CUDA_OK();
   return out32;
 }
@@ -336,9 +353,14 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
   dim3 grid(N / SG_TN, nks);
   size_t lds = (size_t)16 * kslice * sizeof(short);
   TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
-  hipLaunchKernelGGL(skinny_gemm_kernel, grid, dim3(256), lds, cur_stream(),
-    (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
-    out32.data_ptr<float>(), M, N, K, kslice);
+  if (sg_depth() == 8)
+    hipLaunchKernelGGL(skinny_gemm_kernel<8>, grid, dim3(256), lds, cur_stream(),
+      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+      out32.data_ptr<float>(), M, N, K, kslice);
+  else
+    hipLaunchKernelGGL(skinny_gemm_kernel<4>, grid, dim3(256), lds, cur_stream(),
+      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+      out32.data_ptr<float>(), M, N, K, kslice);
   auto out = torch::empty({M, (long)N}, x.options());
   long n = (long)M * N;
   int cgrid = (int)std::min<long>((n / 4 + 255) / 256, 2048);
