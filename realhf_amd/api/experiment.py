@@ -17,6 +17,18 @@ from realhf_amd.parallel.ddp import OptimizerConfig
 
 
 @dataclasses.dataclass
+class LoRAConfig:
+    """Low-rank adaptation of the attention projections
+    (reference: api/quickstart/model.py:56 LoRAConfig + the "lora"
+    ModelWrapperAbstraction wrapping c_attn/c_proj with a "squash" op).
+    delta_W = (scaling / dim) * B @ A; only A/B train."""
+
+    dim: int = 32
+    scaling: float = 32.0
+    squash_after_train: bool = False  # merge adapters into base on save
+
+
+@dataclasses.dataclass
 class ModelTrainEvalConfig:
     """One role's model (reference: api/quickstart/model.py:114)."""
 
@@ -31,6 +43,7 @@ class ModelTrainEvalConfig:
     parallel: ParallelismConfig = dataclasses.field(default_factory=ParallelismConfig)
     # gen-phase override strategy (enables param realloc between phases)
     gen_parallel: Optional[ParallelismConfig] = None
+    lora: Optional[LoRAConfig] = None  # train adapters only (base frozen)
 
 
 @dataclasses.dataclass

@@ -127,6 +127,14 @@ class ReaLModelBlock(nn.Module):
 
             self.moe = MoELayer(cfg, layer_idx, params, tp_size,
                                 ep_rank=ep_rank, ep_size=ep_size)
+        # LoRA adapters (injected by ReaLModel.attach_lora): short key
+        # ("wq"/"wk"/"wv"/"wo") -> (A [r, in], B [out, r]) views
+        self.lora: Dict[str, tuple] = {}
+        self.lora_scale = 0.0
+
+    def _lora_delta(self, x, short: str):
+        a, b = self.lora[short]
+        return ((x @ a.t()) @ b.t()) * self.lora_scale
 
     # -- attention --------------------------------------------------------
     def _qkv(self, x):
@@ -145,11 +153,18 @@ class ReaLModelBlock(nn.Module):
         q, k, v = qkv.split(
             [self.nq * self.hd, self.nkv * self.hd, self.nkv * self.hd], dim=-1
         )
+        if self.lora:
+            if "wq" in self.lora:
+                q = q + self._lora_delta(x, "wq")
+            if "wk" in self.lora:
+                k = k + self._lora_delta(x, "wk")
+            if "wv" in self.lora:
+                v = v + self._lora_delta(x, "wv")
         t = x.shape[0]
         return (
-            q.view(t, self.nq, self.hd),
-            k.view(t, self.nkv, self.hd),
-            v.view(t, self.nkv, self.hd),
+            q.reshape(t, self.nq, self.hd),
+            k.reshape(t, self.nkv, self.hd),
+            v.reshape(t, self.nkv, self.hd),
         )
 
     def forward(
@@ -189,6 +204,7 @@ class ReaLModelBlock(nn.Module):
             and _os.environ.get("REALHF_AMD_NO_FUSED_DECODE") != "1"
             and (cfg.sliding_window is None
                  or k_cache.shape[1] <= cfg.sliding_window)
+            and not self.lora  # adapters applied in the standard path
         ):
             from realhf_amd import ops as _ops_pkg
 
@@ -300,6 +316,8 @@ class ReaLModelBlock(nn.Module):
             )
         attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
         o = _linear(attn_out, self.p[f"{i}.attn.wo.weight"])
+        if "wo" in self.lora:
+            o = o + self._lora_delta(attn_out, "wo")
         if sp:
             o = mappings.reduce_scatter_to_sp_region(o)
         else:

@@ -120,6 +120,102 @@ class ReaLModel(nn.Module):
                 else:
                     t.zero_()
 
+    # ------------------------------------------------------------------ LoRA
+    # Reference: the "lora" ModelWrapperAbstraction (api/quickstart/
+    # model.py:56 LoRAConfig, :204 wrapper with lora_keys_to_replace
+    # ["c_attn.linear", "c_proj."] and the "squash" op).  Flat-param
+    # design: A/B adapters live in their OWN small contiguous buffer
+    # (`lora_flat`), so the LoRA optimizer is the same ZeRO-1 machinery
+    # pointed at that buffer while the base flat_param stays frozen.
+    def attach_lora(
+        self,
+        dim: int = 32,
+        scaling: float = 32.0,
+        keys=("attn.wq", "attn.wk", "attn.wv", "attn.wo"),
+    ):
+        assert getattr(self, "lora_flat", None) is None, "LoRA already attached"
+        cfg = self.config
+        specs: Dict[str, tuple] = {}  # name -> (start, end, shape)
+        off = 0
+
+        def _add(name, shape):
+            nonlocal off
+            n = shape[0] * shape[1]
+            specs[name] = (off, off + n, shape)
+            off += n
+
+        for idx in self.layer_indices:
+            if idx == 0 or idx == cfg.n_layers + 1:
+                continue
+            for k in keys:
+                wname = f"{idx}.{k}.weight"
+                if wname in self._params:
+                    out_d, in_d = self._params[wname].shape
+                    _add(f"{wname}.lora_A", (dim, in_d))
+                    _add(f"{wname}.lora_B", (out_d, dim))
+        self.lora_specs = specs
+        self.lora_dim = dim
+        self.lora_scale = scaling / dim
+        self.lora_flat = torch.zeros(off, dtype=self.dtype, device=self.device)
+        self.lora_grad: Optional[torch.Tensor] = None
+        self._map_lora()
+        with torch.no_grad():
+            for name, t in self.lora_params.items():
+                if name.endswith("lora_A"):
+                    # deterministic per-name seed: identical on every rank
+                    # (replicated-A TP shards must match across tp ranks)
+                    g = torch.Generator(device=t.device)
+                    g.manual_seed(abs(hash(name)) % (2**31))
+                    t.normal_(0.0, 1.0 / dim, generator=g)
+        self._inject_lora()
+
+    def _map_lora(self):
+        self.lora_params: Dict[str, torch.Tensor] = {}
+        for name, (s, e, shape) in self.lora_specs.items():
+            self.lora_params[name] = self.lora_flat[s:e].view(shape)
+
+    def _inject_lora(self):
+        for blk in self.layers:
+            if not isinstance(blk, ReaLModelBlock):
+                continue
+            i = blk.i
+            d = {}
+            for short in ("wq", "wk", "wv", "wo"):
+                a = self.lora_params.get(f"{i}.attn.{short}.weight.lora_A")
+                b = self.lora_params.get(f"{i}.attn.{short}.weight.lora_B")
+                if a is not None:
+                    d[short] = (a, b)
+            blk.lora = d
+            blk.lora_scale = self.lora_scale
+
+    def lora_grad_view(self, name: str) -> torch.Tensor:
+        s, e, shape = self.lora_specs[name]
+        return self.lora_grad[s:e].view(shape)
+
+    @torch.no_grad()
+    def squash_lora(self):
+        """Merge the adapters into the base weights and detach LoRA
+        (reference: lora_op_after_creation="squash")."""
+        for wname in {n.rsplit(".lora_", 1)[0] for n in self.lora_specs}:
+            a = self.lora_params[f"{wname}.lora_A"].float()
+            b = self.lora_params[f"{wname}.lora_B"].float()
+            w = self._params[wname]
+            w.add_((b @ a).to(w.dtype) * self.lora_scale)
+        self.lora_flat = None
+        self.lora_specs = {}
+        self.lora_params = {}
+        for blk in self.layers:
+            if isinstance(blk, ReaLModelBlock):
+                blk.lora = {}
+
+    def lora_state_dict(self) -> Dict[str, torch.Tensor]:
+        return {k: v.clone() for k, v in self.lora_params.items()}
+
+    def load_lora_state_dict(self, sd: Dict[str, torch.Tensor]):
+        with torch.no_grad():
+            for k, v in sd.items():
+                self.lora_params[k].copy_(v)
+
     # ------------------------------------------------------------------ fwd
     def forward(
         self,

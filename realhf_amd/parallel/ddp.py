@@ -48,6 +48,35 @@ class OptimizerConfig:
     offload: bool = False  # optimizer states in host memory (70B tier)
 
 
+class LoRAFacade:
+    """Adapter that lets ZeRO1Optimizer drive a ReaLModel's LoRA buffer:
+    exposes flat_param/flat_grad/_params/grad_view backed by the model's
+    lora_* attributes, so adapters get the same sharded AdamW + LR
+    schedule + clipping while the base flat_param stays frozen."""
+
+    def __init__(self, model):
+        self._m = model
+        assert getattr(model, "lora_flat", None) is not None, "attach_lora first"
+        self.flat_param = model.lora_flat
+        self.flat_grad = None
+
+    def _map_params(self):
+        self._m.lora_flat = self.flat_param
+        self._m._map_lora()
+        self._m._inject_lora()
+
+    def _build_modules(self):
+        pass
+
+    @property
+    def _params(self):
+        return self._m.lora_params
+
+    def grad_view(self, k):
+        self._m.lora_grad = self.flat_grad
+        return self._m.lora_grad_view(k)
+
+
 class ZeRO1Optimizer:
     def __init__(
         self,

@@ -146,8 +146,15 @@ class ZeRO1Backend(ModelBackend):
     def _initialize(self, model: Model, spec: FinetuneSpec) -> Model:
         if isinstance(self.optimizer, dict):
             self.optimizer = OptimizerConfig(**self.optimizer)
+        target = model.module
+        if getattr(model.module, "lora_flat", None) is not None:
+            # LoRA: same ZeRO-1 machinery over the small adapter buffer;
+            # the base flat_param stays frozen
+            from realhf_amd.parallel.ddp import LoRAFacade
+
+            target = LoRAFacade(model.module)
         opt = ZeRO1Optimizer(
-            model.module, self.optimizer,
+            target, self.optimizer,
             total_train_steps=spec.total_train_steps,
             bucket_size=self.bucket_size,
         )

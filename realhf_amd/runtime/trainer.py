@@ -426,6 +426,13 @@ class Trainer:
                     hf_reg.load_from_hf(m, mc.family, mc.path)
                 else:
                     m.random_init()
+                if getattr(mc, "lora", None) is not None and name in self.built.trainable:
+                    lc = mc.lora
+                    if isinstance(lc, dict):
+                        from realhf_amd.api.experiment import LoRAConfig
+
+                        lc = LoRAConfig(**lc)
+                    m.attach_lora(dim=lc.dim, scaling=lc.scaling)
                 model = Model(
                     name=name, module=m, tokenizer=self.tokenizer,
                     device=self.device, dtype=rcfg.torch_dtype,
