@@ -85,10 +85,25 @@ class PPOHyperparameters:
 
 
 @dataclasses.dataclass
+class SlurmConfig:
+    """Cluster options for mode=slurm (reference: scheduler/slurm/utils.py
+    SlurmLaunchInfo fields + cluster spec)."""
+
+    partition: Optional[str] = None
+    account: Optional[str] = None
+    time_limit: Optional[str] = None
+    container_image: Optional[str] = None
+    container_mounts: Optional[str] = None
+    gpus_per_node: int = 8
+    mem_per_node: Optional[str] = None
+
+
+@dataclasses.dataclass
 class CommonExperimentConfig:
     experiment_name: str = "exp"
     trial_name: str = "trial"
-    mode: str = "local"
+    mode: str = "local"  # local | slurm
+    slurm: SlurmConfig = dataclasses.field(default_factory=SlurmConfig)
     n_gpus: int = 1
     seed: int = 1
     allocation_mode: str = "global"  # global | manual | heuristic | d8t1p1-style

@@ -95,11 +95,26 @@ def main(argv=None):
             dist.destroy_process_group()
         return
 
-    # launcher mode: spawn local workers
+    # launcher mode: spawn workers via the configured scheduler
     from realhf_amd.base.testing import find_free_port
-    from realhf_amd.scheduler.local import LocalScheduler
 
     cmd = [sys.executable, "-m", "realhf_amd.apps.quickstart", exp_type] + list(argv[1:])
+    if cfg.mode == "slurm":
+        from realhf_amd.scheduler.slurm import SlurmScheduler
+
+        sc = cfg.slurm
+        sched = SlurmScheduler(
+            cfg.experiment_name, cfg.trial_name,
+            partition=sc.partition, account=sc.account,
+            time_limit=sc.time_limit, container_image=sc.container_image,
+            container_mounts=sc.container_mounts,
+            gpus_per_node=sc.gpus_per_node, mem_per_node=sc.mem_per_node,
+        )
+        sched.submit_array(cmd, cfg.n_gpus)
+        sched.wait()
+        return
+    from realhf_amd.scheduler.local import LocalScheduler
+
     sched = LocalScheduler(cfg.experiment_name, cfg.trial_name)
     sched.submit_array(cmd, cfg.n_gpus, master_port=find_free_port())
     sched.wait()
