@@ -34,6 +34,13 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> residual);
 torch::Tensor skinny_gemm_nc(torch::Tensor x, torch::Tensor w,
                              torch::Tensor out32_ws, long splitk);
+int64_t xgmi_create(int64_t rank, int64_t world, int64_t capacity);
+std::vector<py::bytes> xgmi_handles(int64_t h);
+void xgmi_connect(int64_t h, const std::vector<std::string>& data_handles,
+                  const std::vector<std::string>& sig_handles);
+torch::Tensor xgmi_all_reduce(int64_t h, torch::Tensor t);
+int64_t xgmi_status(int64_t h);
+void xgmi_destroy(int64_t h);
 torch::Tensor skinny_gemm2(torch::Tensor x, torch::Tensor w,
                            torch::Tensor out32_ws, torch::Tensor sem,
                            long splitk, c10::optional<torch::Tensor> residual);
@@ -68,5 +75,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm);
   m.def("skinny_gemm2", &skinny_gemm2);
   m.def("skinny_gemm_nc", &skinny_gemm_nc);
+  m.def("xgmi_create", &xgmi_create);
+  m.def("xgmi_handles", &xgmi_handles);
+  m.def("xgmi_connect", &xgmi_connect);
+  m.def("xgmi_all_reduce", &xgmi_all_reduce);
+  m.def("xgmi_status", &xgmi_status);
+  m.def("xgmi_destroy", &xgmi_destroy);
   m.def("mcmc_search", &mcmc_search);
 }

@@ -1,0 +1,99 @@
+"""Custom intra-node all-reduce over xGMI peer mappings.
+
+Reference counterpart: realhf/impl/model/parallel/custom_all_reduce.py +
+csrc/custom_all_reduce/ (vLLM-style CUDA-IPC allreduce; declared
+legacy/unused by the reference).  MI355X version: all 8 GPUs of a node
+are fully connected by xGMI, so a 1-stage direct-read sum is one hop of
+latency instead of a ring's 2(n-1) — a latency win for the small bf16
+TP all-reduces of decode (hidden-state messages are 32-128 KB at bs 16).
+
+Opt-in via REALHF_AMD_XGMI_AR=1 (RCCL remains the default; multi-GPU
+validation of this path on an 8-GPU node is scheduled for the next
+round).  Falls back to torch.distributed.all_reduce transparently for
+oversized messages or unsupported dtypes.
+"""
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from realhf_amd.base import logging
+
+logger = logging.getLogger("xgmi")
+
+_MAX_WORLD = 8
+
+
+class XgmiAllReduce:
+    """One IPC-shared buffer per rank + generation-counter barriers.
+
+    Handles are exchanged over the given (gloo or rccl) process group with
+    all_gather_object; every rank must call the constructor collectively.
+    """
+
+    def __init__(self, group=None, capacity_bytes: int = 16 << 20,
+                 device: Optional[torch.device] = None):
+        from realhf_amd import ops as ops_pkg
+
+        self._C = ops_pkg.require_hip()
+        self.group = group
+        self.world = dist.get_world_size(group)
+        self.rank = dist.get_rank(group)
+        assert 1 <= self.world <= _MAX_WORLD
+        self.capacity = capacity_bytes
+        if device is not None:
+            torch.cuda.set_device(device)
+        self.handle = self._C.xgmi_create(self.rank, self.world, capacity_bytes)
+        mine = [bytes(b) for b in self._C.xgmi_handles(self.handle)]
+        allh: list = [None] * self.world
+        dist.all_gather_object(allh, mine, group=group)
+        self._C.xgmi_connect(self.handle, [h[0] for h in allh],
+                             [h[1] for h in allh])
+        self._closed = False
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        """Sum across ranks; returns a NEW tensor (input unchanged)."""
+        if (
+            t.numel() * t.element_size() > self.capacity
+            or t.dtype not in (torch.bfloat16, torch.float32)
+            or t.numel() % 8 != 0
+        ):
+            out = t.clone()
+            dist.all_reduce(out, group=self.group)
+            return out
+        return self._C.xgmi_all_reduce(self.handle, t.contiguous())
+
+    def status_ok(self) -> bool:
+        """False if a barrier spin timed out (peer died mid-collective)."""
+        return int(self._C.xgmi_status(self.handle)) == 0
+
+    def close(self):
+        if not self._closed:
+            self._C.xgmi_destroy(self.handle)
+            self._closed = True
+
+
+_GLOBAL: Optional[XgmiAllReduce] = None
+
+
+def maybe_init_xgmi(group=None) -> Optional[XgmiAllReduce]:
+    """Create the process-global xGMI allreduce if opted in
+    (REALHF_AMD_XGMI_AR=1), CUDA is up, and the group fits one node."""
+    global _GLOBAL
+    if _GLOBAL is not None:
+        return _GLOBAL
+    if os.environ.get("REALHF_AMD_XGMI_AR") != "1":
+        return None
+    if not (dist.is_initialized() and torch.cuda.is_available()):
+        return None
+    if dist.get_world_size(group) > _MAX_WORLD:
+        return None
+    try:
+        _GLOBAL = XgmiAllReduce(group)
+        logger.info("xGMI custom all-reduce enabled (world=%d)",
+                    _GLOBAL.world)
+    except Exception as e:  # IPC unavailable etc. — RCCL fallback
+        logger.warning("xGMI allreduce unavailable (%s); using RCCL", e)
+        _GLOBAL = None
+    return _GLOBAL
