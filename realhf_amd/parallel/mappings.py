@@ -27,6 +27,14 @@ def _tp_group():
 def _all_reduce(x):
     if _tp_size() == 1:
         return x
+    if x.is_cuda and not torch.is_grad_enabled():
+        # opt-in xGMI direct-read allreduce for small decode-time TP
+        # messages (REALHF_AMD_XGMI_AR=1); returns None when disabled
+        from realhf_amd.parallel import xgmi
+
+        ar = xgmi.maybe_init_xgmi(_tp_group())
+        if ar is not None:
+            return ar.all_reduce(x.contiguous())
     dist.all_reduce(x.contiguous(), group=_tp_group())
     return x
 

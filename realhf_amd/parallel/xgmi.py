@@ -74,26 +74,29 @@ class XgmiAllReduce:
             self._closed = True
 
 
-_GLOBAL: Optional[XgmiAllReduce] = None
+_PER_GROUP: dict = {}
 
 
 def maybe_init_xgmi(group=None) -> Optional[XgmiAllReduce]:
-    """Create the process-global xGMI allreduce if opted in
-    (REALHF_AMD_XGMI_AR=1), CUDA is up, and the group fits one node."""
-    global _GLOBAL
-    if _GLOBAL is not None:
-        return _GLOBAL
+    """Create (once per process group) the xGMI allreduce if opted in
+    (REALHF_AMD_XGMI_AR=1), CUDA is up, and the group fits one node.
+    The constructor is collective: every rank of the group must reach the
+    first call together (guaranteed by the SPMD program order)."""
+    key = id(group)
+    if key in _PER_GROUP:
+        return _PER_GROUP[key]
     if os.environ.get("REALHF_AMD_XGMI_AR") != "1":
         return None
     if not (dist.is_initialized() and torch.cuda.is_available()):
         return None
     if dist.get_world_size(group) > _MAX_WORLD:
+        _PER_GROUP[key] = None
         return None
     try:
-        _GLOBAL = XgmiAllReduce(group)
-        logger.info("xGMI custom all-reduce enabled (world=%d)",
-                    _GLOBAL.world)
+        ar = XgmiAllReduce(group)
+        logger.info("xGMI custom all-reduce enabled (world=%d)", ar.world)
     except Exception as e:  # IPC unavailable etc. — RCCL fallback
         logger.warning("xGMI allreduce unavailable (%s); using RCCL", e)
-        _GLOBAL = None
-    return _GLOBAL
+        ar = None
+    _PER_GROUP[key] = ar
+    return ar

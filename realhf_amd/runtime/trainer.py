@@ -502,6 +502,7 @@ class Trainer:
                     )
                 if ctrl.save_freq_steps and self.global_step % ctrl.save_freq_steps == 0:
                     self.save()
+                    self.save_recover_ckpt(epoch, i + 1)
                 if ctrl.benchmark_steps:
                     if bench_t0 is None:
                         bench_t0 = time.time()
@@ -517,7 +518,7 @@ class Trainer:
                             )
                         self._save_recover_info(epoch, i)
                         return
-            self._save_recover_info(epoch + 1, 0)
+            self.save_recover_ckpt(epoch + 1, 0)
         if ctrl.save_freq_steps:
             self.save()
 
@@ -554,6 +555,41 @@ class Trainer:
             pickle.dump({"epoch": epoch, "step": step,
                          "global_step": self.global_step}, f)
 
+    def _recover_ckpt_path(self):
+        return os.path.join(
+            constants.RECOVER_ROOT(self.cfg.experiment_name, self.cfg.trial_name),
+            f"ckpt_rank{self.rank}.pt",
+        )
+
+    def save_recover_ckpt(self, epoch, step):
+        """Full resumable state: every trainable model's flat params +
+        its optimizer state (fp32 master/m/v shards), per rank.
+        Reference: the recover ckpt saved by model workers
+        (model_worker.py __save_model for recover + master's
+        recover_info)."""
+        if self.cfg.recover_mode == "disabled":
+            return
+        state = {"models": {}, "epoch": epoch, "step": step,
+                 "global_step": self.global_step}
+        for name in self.built.trainable:
+            if name not in self.models:
+                continue
+            eng = self.models[name].module
+            m = eng.module
+            opt = getattr(eng, "optimizer", None)
+            state["models"][str(name)] = {
+                "flat_param": m.flat_param.detach().cpu().clone(),
+                "lora_flat": (m.lora_flat.detach().cpu().clone()
+                              if getattr(m, "lora_flat", None) is not None
+                              else None),
+                "optimizer": ({k: (v.cpu() if torch.is_tensor(v) else v)
+                               for k, v in opt.state_dict().items()}
+                              if opt is not None else None),
+            }
+        os.makedirs(os.path.dirname(self._recover_ckpt_path()), exist_ok=True)
+        torch.save(state, self._recover_ckpt_path())
+        self._save_recover_info(epoch, step)
+
     def _maybe_load_recover(self):
         if self.cfg.recover_mode not in ("auto", "resume"):
             return None
@@ -563,5 +599,22 @@ class Trainer:
         with open(p, "rb") as f:
             info = pickle.load(f)
         self.global_step = info["global_step"]
+        ckpt = self._recover_ckpt_path()
+        if os.path.exists(ckpt):
+            state = torch.load(ckpt, map_location="cpu", weights_only=False)
+            for name in self.built.trainable:
+                st = state["models"].get(str(name))
+                if st is None or name not in self.models:
+                    continue
+                eng = self.models[name].module
+                m = eng.module
+                with torch.no_grad():
+                    m.flat_param.copy_(st["flat_param"].to(m.flat_param.device))
+                    if st["lora_flat"] is not None:
+                        m.lora_flat.copy_(st["lora_flat"].to(m.lora_flat.device))
+                opt = getattr(eng, "optimizer", None)
+                if opt is not None and st["optimizer"] is not None:
+                    opt.load_state_dict(st["optimizer"])
+            logger.info("restored model+optimizer state from %s", ckpt)
         logger.info("recovering from %s", info)
         return (info["epoch"], info["step"])

@@ -174,3 +174,42 @@ def test_ppo_realloc_two_ranks(tmp_path):
     data = str(tmp_path / "prompts.jsonl")
     _write_prompt_data(data)
     LocalMultiProcessTest(2, _ppo_realloc_worker, data, str(tmp_path / "root")).launch()
+
+
+def test_sft_recover_resumes_state(tmp_path):
+    """Kill-and-resume: second Trainer picks up global_step AND the saved
+    model+optimizer state (reference: recover_mode=auto resume path)."""
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "sft.jsonl")
+    _write_sft_data(data, n=16)
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+
+    def mkcfg():
+        cfg = SFTConfig(experiment_name="t-rec", trial_name="cpu", n_gpus=1)
+        cfg.model.dtype = "float32"
+        cfg.dataset.type_ = "prompt_answer"
+        cfg.dataset.path = data
+        cfg.dataset.train_bs_n_seqs = 4
+        cfg.exp_ctrl.total_train_epochs = 1
+        cfg.exp_ctrl.benchmark_steps = 2
+        cfg.exp_ctrl.save_freq_steps = 1  # recover ckpt every step
+        cfg.recover_mode = "auto"
+        return cfg
+
+    t1 = Trainer(mkcfg())
+    t1.run()
+    step1 = t1.global_step
+    name = t1.built.trainable[0]
+    flat_after = t1.models[name].module.module.flat_param.detach().clone()
+    opt_step = t1.models[name].module.optimizer.step_count
+    assert step1 >= 2 and opt_step >= 2
+
+    t2 = Trainer(mkcfg())
+    recov = t2._maybe_load_recover()
+    assert recov is not None
+    assert t2.global_step == step1
+    m2 = t2.models[name].module.module
+    torch.testing.assert_close(m2.flat_param, flat_after)
+    assert t2.models[name].module.optimizer.step_count == opt_step
