@@ -49,6 +49,7 @@ class ModelTrainEvalConfig:
 @dataclasses.dataclass
 class DatasetConfig:
     path: Optional[str] = None
+    valid_path: Optional[str] = None  # eval split (reference: valid_path)
     type_: str = "prompt"  # prompt | prompt_answer | rw_paired | synthetic_prompt
     max_prompt_len: int = 256
     max_seqlen: int = 1024

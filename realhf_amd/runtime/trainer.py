@@ -454,25 +454,57 @@ class Trainer:
                 real.async_offload(non_blocking=False)
         return models
 
-    def _build_dataloader(self):
+    def _build_dataloader(self, valid: bool = False):
         d = self.cfg.dataset
+        path = d.valid_path if valid else d.path
         ds_cfg = Abstraction(d.type_, dict(d.args))
         if d.type_ == "prompt":
-            ds_cfg.args.setdefault("path", d.path)
+            ds_cfg.args.setdefault("path", path)
             ds_cfg.args.setdefault("max_prompt_len", d.max_prompt_len)
         elif d.type_ == "prompt_answer":
-            ds_cfg.args.setdefault("path", d.path)
+            ds_cfg.args.setdefault("path", path)
             ds_cfg.args.setdefault("max_seqlen", d.max_seqlen)
         elif d.type_ == "rw_paired":
-            ds_cfg.args.setdefault("path", d.path)
+            ds_cfg.args.setdefault("path", path)
             ds_cfg.args.setdefault("max_seqlen", d.max_seqlen)
+        else:
+            ds_cfg.args["path"] = path
         # SPMD: every rank builds the identical dataset/loader (same seed)
         dataset = make_dataset(ds_cfg, seed=self.cfg.seed, dp_rank=0,
                                world_size=1, tokenizer=self.tokenizer)
         return PackedDataLoader(
-            dataset, batch_n_seqs=d.train_bs_n_seqs, shuffle=True,
+            dataset, batch_n_seqs=d.train_bs_n_seqs, shuffle=not valid,
             seed=self.cfg.seed,
         )
+
+    def evaluate(self) -> Dict[str, float]:
+        """Run each trainable model's interface.evaluate over the eval
+        split (reference: master worker eval step, ModelInterfaceType
+        .EVALUATE requests to model workers)."""
+        if not self.cfg.dataset.valid_path:
+            return {}
+        eval_dl = self._build_dataloader(valid=True)
+        out: Dict[str, float] = {}
+        for name in self.built.trainable:
+            if name not in self.models:
+                continue
+            for mfc in self.built.graph.mfcs:
+                if (mfc.model_name == name
+                        and mfc.interface_type == ModelInterfaceType.TRAIN_STEP):
+                    iface = self.built.interfaces[mfc.name]
+                    if not hasattr(iface, "evaluate"):
+                        break
+                    with constants.model_scope(str(name)):
+                        try:
+                            stats = iface.evaluate(self.models[name], eval_dl)
+                        except NotImplementedError:
+                            break
+                    out.update({f"{name.role}/{k}": v for k, v in stats.items()})
+                    break
+        if out and self.rank == 0:
+            logger.info("eval @ step %d: %s", self.global_step,
+                        {k: round(v, 4) for k, v in out.items()})
+        return out
 
     # ----------------------------------------------------------------- run
     def run(self):
@@ -503,6 +535,9 @@ class Trainer:
                 if ctrl.save_freq_steps and self.global_step % ctrl.save_freq_steps == 0:
                     self.save()
                     self.save_recover_ckpt(epoch, i + 1)
+                if (ctrl.eval_freq_steps
+                        and self.global_step % ctrl.eval_freq_steps == 0):
+                    self.evaluate()
                 if ctrl.benchmark_steps:
                     if bench_t0 is None:
                         bench_t0 = time.time()

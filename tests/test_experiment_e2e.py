@@ -213,3 +213,28 @@ def test_sft_recover_resumes_state(tmp_path):
     m2 = t2.models[name].module.module
     torch.testing.assert_close(m2.flat_param, flat_after)
     assert t2.models[name].module.optimizer.step_count == opt_step
+
+
+def test_sft_eval_split(tmp_path):
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "sft.jsonl")
+    valid = str(tmp_path / "valid.jsonl")
+    _write_sft_data(data, n=16)
+    _write_sft_data(valid, n=8)
+    cfg = SFTConfig(experiment_name="t-eval", trial_name="cpu", n_gpus=1)
+    cfg.model.dtype = "float32"
+    cfg.dataset.type_ = "prompt_answer"
+    cfg.dataset.path = data
+    cfg.dataset.valid_path = valid
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.exp_ctrl.total_train_epochs = 1
+    cfg.exp_ctrl.benchmark_steps = 2
+    cfg.exp_ctrl.eval_freq_steps = 1
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    t = Trainer(cfg)
+    t.run()
+    stats = t.evaluate()
+    assert "default/eval_loss" in stats
+    assert stats["default/eval_loss"] > 0
