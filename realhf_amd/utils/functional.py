@@ -108,3 +108,19 @@ def top_k_top_p_logits(
         scatter_mask = remove.scatter(1, sorted_idx, remove)
         logits.masked_fill_(scatter_mask, float("-inf"))
     return logits
+
+
+def apply_logits_mask(logits: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
+    """Mask forbidden vocab entries to -inf (reference:
+    utils/functional.py:214 apply_logits_mask — TP vocab-partition aware:
+    `logits` may be this tp rank's vocab slice [..., V/tp] while `mask`
+    (True = forbidden) covers the FULL vocab [..., V])."""
+    from realhf_amd.base import constants
+
+    v_local = logits.shape[-1]
+    if mask.shape[-1] != v_local:
+        tp = constants.tp_world_size() if constants.has_current() else 1
+        assert mask.shape[-1] == v_local * tp, (mask.shape, logits.shape, tp)
+        r = constants.tp_rank() if constants.has_current() else 0
+        mask = mask[..., r * v_local:(r + 1) * v_local]
+    return logits.masked_fill_(mask, float("-inf"))

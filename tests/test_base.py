@@ -88,3 +88,15 @@ def test_name_resolve_file(tmp_path):
     repo.clear_subtree("a")
     with pytest.raises(NameEntryNotFoundError):
         repo.get("a/b/c")
+
+
+def test_apply_logits_mask():
+    import torch
+
+    from realhf_amd.utils.functional import apply_logits_mask
+
+    logits = torch.zeros(3, 8)
+    mask = torch.zeros(3, 8, dtype=torch.bool)
+    mask[:, ::2] = True
+    out = apply_logits_mask(logits.clone(), mask)
+    assert torch.isinf(out[:, ::2]).all() and (out[:, 1::2] == 0).all()
