@@ -18,10 +18,13 @@
 //   Wave partials (m, s, acc) combine through LDS at the end.
 #include "common.h"
 
-#define DEC_WAVES 8
-
-template <int HD, int REP>
-__global__ __launch_bounds__(64 * DEC_WAVES) void attn_decode_kernel(
+// Wave count trade-off: the per-wave V tile is W*64*HD*2 bytes of LDS —
+// at HD=128, W=8 needs 128 KB (ONE workgroup per CU: the PV phase has
+// nothing co-resident to overlap with), W=4 needs 64 KB (TWO WGs per CU:
+// one streams KV while the other runs its softmax/PV phase).  Default 4;
+// REALHF_AMD_DEC_WAVES ∈ {2,4,8} for A/B.
+template <int HD, int REP, int W>
+__global__ __launch_bounds__(64 * W) void attn_decode_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
     const bf16* __restrict__ vc, const int* __restrict__ cache_seqlens,
     bf16* __restrict__ out, int bs, int nq, int nkv, long maxlen, float scale) {
@@ -33,9 +36,9 @@ __global__ __launch_bounds__(64 * DEC_WAVES) void attn_decode_kernel(
   const int w = threadIdx.x / WAVE;
 
   __shared__ float q_s[REP][HD];
-  __shared__ __bf16 v_s[DEC_WAVES][WAVE][HD];  // per-wave V chunk tile
-  __shared__ float m_s[DEC_WAVES], s_s[DEC_WAVES];
-  __shared__ float acc_s[DEC_WAVES][HD];
+  __shared__ __bf16 v_s[W][WAVE][HD];  // per-wave V chunk tile
+  __shared__ float m_s[W], s_s[W];
+  __shared__ float acc_s[W][HD];
 
   for (int i = threadIdx.x; i < REP * HD; i += blockDim.x) {
     int r = i / HD, d = i % HD;
@@ -56,7 +59,7 @@ __global__ __launch_bounds__(64 * DEC_WAVES) void attn_decode_kernel(
   const bf16* kb = kc + (long)b * maxlen * kv_stride + (long)kvh * HD;
   const bf16* vb = vc + (long)b * maxlen * kv_stride + (long)kvh * HD;
 
-  for (int base = w * WAVE; base < L; base += DEC_WAVES * WAVE) {
+  for (int base = w * WAVE; base < L; base += W * WAVE) {
     const int l = base + lane;
     const bool valid = l < L;
     float sc[REP];
@@ -97,9 +100,16 @@ __global__ __launch_bounds__(64 * DEC_WAVES) void attn_decode_kernel(
       // p broadcast by shfl (no cross-wave LDS -> in-wave visibility)
       for (int j = 0; j < nvalid; j++) {
         float pj = __shfl(p, j, 64);
-        #pragma unroll
-        for (int d = 0; d < DPL; d++)
-          acc_w[r][d] += pj * (float)v_s[w][j][lane * DPL + d];
+        if constexpr (DPL == 2) {
+          // one b32 LDS read for the lane's two dims
+          unsigned u = *(const unsigned*)(&v_s[w][j][lane * 2]);
+          acc_w[r][0] += pj * bf2f((short)(u & 0xffff));
+          acc_w[r][1] += pj * bf2f((short)(u >> 16));
+        } else {
+          #pragma unroll
+          for (int d = 0; d < DPL; d++)
+            acc_w[r][d] += pj * (float)v_s[w][j][lane * DPL + d];
+        }
       }
     }
   }
@@ -114,13 +124,13 @@ __global__ __launch_bounds__(64 * DEC_WAVES) void attn_decode_kernel(
     if (w == 0) {
       float M = -1e30f;
       #pragma unroll
-      for (int ww = 0; ww < DEC_WAVES; ww++) M = fmaxf(M, m_s[ww]);
+      for (int ww = 0; ww < W; ww++) M = fmaxf(M, m_s[ww]);
       float S = 0.f;
       float o[DPL];
       #pragma unroll
       for (int d = 0; d < DPL; d++) o[d] = 0.f;
       #pragma unroll
-      for (int ww = 0; ww < DEC_WAVES; ww++) {
+      for (int ww = 0; ww < W; ww++) {
         float f = __expf(m_s[ww] - M);
         S += f * s_s[ww];
         #pragma unroll
@@ -147,12 +157,22 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   auto cs = cache_seqlens.to(torch::kInt);
   auto out = torch::empty_like(q);
   dim3 grid(bs * nkv);
+  static int dec_waves = [] {
+    const char* e = getenv("REALHF_AMD_DEC_WAVES");
+    int v = e ? atoi(e) : 4;
+    return (v == 2 || v == 4 || v == 8) ? v : 4;
+  }();
   auto launch = [&](auto hd_c, auto rep_c) {
-    hipLaunchKernelGGL((attn_decode_kernel<hd_c.value, rep_c.value>), grid,
-      dim3(64 * DEC_WAVES), 0, cur_stream(), (const bf16*)q.data_ptr(),
-      (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
-      cs.data_ptr<int>(), (bf16*)out.data_ptr(), bs, nq, nkv, maxlen,
-      (float)scale);
+    auto go = [&](auto w_c) {
+      hipLaunchKernelGGL((attn_decode_kernel<hd_c.value, rep_c.value, w_c.value>),
+        grid, dim3(64 * w_c.value), 0, cur_stream(), (const bf16*)q.data_ptr(),
+        (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
+        cs.data_ptr<int>(), (bf16*)out.data_ptr(), bs, nq, nkv, maxlen,
+        (float)scale);
+    };
+    if (dec_waves == 2) go(std::integral_constant<int, 2>{});
+    else if (dec_waves == 8) go(std::integral_constant<int, 8>{});
+    else go(std::integral_constant<int, 4>{});
   };
   #define REP_SWITCH(HDV) \
     switch (rep) { \
