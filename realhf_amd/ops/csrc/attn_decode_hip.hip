@@ -18,6 +18,8 @@
 //   Wave partials (m, s, acc) combine through LDS at the end.
 #include "common.h"
 
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2q;
+
 // Wave count trade-off: the per-wave V tile is W*64*HD*2 bytes of LDS —
 // at HD=128, W=8 needs 128 KB (ONE workgroup per CU: the PV phase has
 // nothing co-resident to overlap with), W=4 needs 64 KB (TWO WGs per CU:
@@ -35,14 +37,14 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   const int w = threadIdx.x / WAVE;
 
-  __shared__ float q_s[REP][HD];
+  __shared__ __bf16 q_s[REP][HD];  // raw bf16: K-dot uses v_dot2 pairs
   __shared__ __bf16 v_s[W][WAVE][HD];  // per-wave V chunk tile
   __shared__ float m_s[W], s_s[W];
   __shared__ float acc_s[W][HD];
 
   for (int i = threadIdx.x; i < REP * HD; i += blockDim.x) {
     int r = i / HD, d = i % HD;
-    q_s[r][d] = __bfloat162float(q[((long)b * nq + kvh * REP + r) * HD + d]) * scale;
+    q_s[r][d] = (__bf16)q[((long)b * nq + kvh * REP + r) * HD + d];
   }
   __syncthreads();
 
@@ -73,16 +75,20 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
         short8 kv8 = *(const short8*)((const short*)krow + i * 8);
         short8 vv8 = *(const short8*)((const short*)vrow + i * 8);
         *(short8*)(&v_s[w][lane][i * 8]) = vv8;
-        float kf[8];
-        #pragma unroll
-        for (int j = 0; j < 8; j++) kf[j] = bf2f(kv8[j]);
+        // packed bf16 dots: 4 v_dot2 per 8 elems instead of 8 cvt + 8 fma
         #pragma unroll
         for (int r = 0; r < REP; r++) {
           #pragma unroll
-          for (int j = 0; j < 8; j++) sc[r] += q_s[r][i * 8 + j] * kf[j];
+          for (int j = 0; j < 4; j++) {
+            bf16x2q a = *(const bf16x2q*)(&q_s[r][i * 8 + j * 2]);
+            bf16x2q kk = *(const bf16x2q*)((const short*)&kv8 + j * 2);
+            sc[r] = __builtin_amdgcn_fdot2_f32_bf16(a, kk, sc[r], false);
+          }
         }
       }
     }
+    #pragma unroll
+    for (int r = 0; r < REP; r++) sc[r] *= scale;  // q kept raw bf16
     const int nvalid = min(WAVE, L - base);
     #pragma unroll
     for (int r = 0; r < REP; r++) {
