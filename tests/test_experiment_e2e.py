@@ -238,3 +238,43 @@ def test_sft_eval_split(tmp_path):
     stats = t.evaluate()
     assert "default/eval_loss" in stats
     assert stats["default/eval_loss"] > 0
+
+
+def _ppo_mixed4_worker(data, fileroot):
+    """4 ranks, mixed strategies per role: actor trains d2t2, generates
+    d1t2p2-like? -> keep pp out: gen d4t1, ref d2t2, rew d4 — exercises
+    realloc across tp<->dp remaps and partial-participation shards at a
+    world size above 2 (insurance for the 8-GPU scale run)."""
+    from realhf_amd.api.config import ParallelismConfig
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = PPOConfig(experiment_name="t-ppo4", trial_name="dist", n_gpus=4)
+    for mc in (cfg.actor, cfg.critic, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+    cfg.allocation_mode = "manual"
+    cfg.actor.parallel = ParallelismConfig(data_parallel_size=2,
+                                           tensor_parallel_size=2)
+    cfg.actor.gen_parallel = ParallelismConfig(data_parallel_size=4)
+    cfg.critic.parallel = ParallelismConfig(data_parallel_size=4)
+    cfg.ref.parallel = ParallelismConfig(data_parallel_size=2,
+                                         tensor_parallel_size=2)
+    cfg.rew.parallel = ParallelismConfig(data_parallel_size=4)
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 8
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 5
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    Trainer(cfg).run()
+
+
+@pytest.mark.distributed
+def test_ppo_mixed_strategies_four_ranks(tmp_path):
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data, n=16)
+    LocalMultiProcessTest(4, _ppo_mixed4_worker, data,
+                          str(tmp_path / "root")).launch()
