@@ -44,7 +44,9 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
   }
   __syncthreads();
 
-  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+  // 4 rotating accumulators relax the MFMA RAW chain (PMC: 35%
+  // SQ_WAIT_INST_ANY with 2 accs at the qkv shape)
+  f32x4 acc[4] = {};
   const bf16* wrow = w + (long)(n0 + i16) * K;
   int kk = k_lo;
   // NB*32-deep body: issue all NB W loads before the first MFMA so >= NB
@@ -60,23 +62,20 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
     for (int t = 0; t < NB; t++)
       a[t] = *(const bf16x8v*)(&x_s[xb + t * 32]);
     #pragma unroll
-    for (int t = 0; t < NB; t++) {
-      if (t & 1)
-        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[t], b[t], acc1, 0, 0, 0);
-      else
-        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[t], b[t], acc0, 0, 0, 0);
-    }
+    for (int t = 0; t < NB; t++)
+      acc[t & 3] =
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[t], b[t], acc[t & 3], 0, 0, 0);
   }
   for (; kk + 32 <= k_hi; kk += 32) {
     bf16x8v a0 = *(const bf16x8v*)(&x_s[(long)i16 * kslice + (kk - k_lo) + g * 8]);
     bf16x8v b0 = *(const bf16x8v*)(wrow + kk + g * 8);
-    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc0, 0, 0, 0);
+    acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[0], 0, 0, 0);
   }
   // non-atomic per-K-slice partial: out32[ks][m][n]
   #pragma unroll
   for (int r = 0; r < 4; r++) {
     int m = g * 4 + r;
-    float vsum = acc0[r] + acc1[r];
+    float vsum = acc[0][r] + acc[1][r] + acc[2][r] + acc[3][r];
     if (m < M)
       out32[((long)ks * M + m) * N + n0 + i16] = vsum;
   }
