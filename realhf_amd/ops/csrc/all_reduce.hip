@@ -105,6 +105,62 @@ __global__ void xar_1stage_kernel(XgmiComm c, T* __restrict__ out, long n,
   xar_barrier(c.flags, c.rank, c.world, gen + 1);
 }
 
+// 2-stage (reduce-scatter + all-gather) for LARGE messages: rank r sums
+// chunk r from every peer into its buffer's upper region, barrier, then
+// every rank gathers the reduced chunks.  Traffic ~2n per rank instead
+// of the 1-stage's world*n — wins once n is past the latency regime.
+// Layout: region A = [0, cap/2) holds the input copy; region B =
+// [cap/2, cap) holds this rank's reduced chunk.
+template <typename T>
+__global__ void xar_2stage_kernel(XgmiComm c, T* __restrict__ out, long n,
+                                  long boff_elems, unsigned gen) {
+  const long cn = n / c.world;  // chunk elems (host pads to world*8)
+  const long my0 = (long)c.rank * cn;
+  if (!xar_barrier(c.flags, c.rank, c.world, gen)) return;
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  // stage 1: reduce my chunk from all peers -> my region B
+  T* myB = (T*)c.data[c.rank] + boff_elems;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < cn;
+       i += stride) {
+    float acc[8] = {};
+    for (int p = 0; p < c.world; p++) {
+      const T* src = (const T*)c.data[p] + my0 + i;
+      if constexpr (sizeof(T) == 2) {
+        short8 v8 = *(const short8*)src;
+        #pragma unroll
+        for (int j = 0; j < 8; j++) acc[j] += to_f32<T>(((const T*)&v8)[j]);
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; j++) acc[j] += ((const float*)src)[j];
+      }
+    }
+    if constexpr (sizeof(T) == 2) {
+      short o[8];
+      #pragma unroll
+      for (int j = 0; j < 8; j++) ((T*)o)[j] = from_f32<T>(acc[j]);
+      *(short8*)(myB + i) = *(short8*)o;
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 8; j++) ((float*)myB)[i + j] = acc[j];
+    }
+  }
+  if (!xar_barrier(c.flags, c.rank, c.world, gen + 1)) return;
+  // stage 2: gather every rank's reduced chunk
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < n;
+       i += stride) {
+    int p = (int)(i / cn);
+    if (p >= c.world) p = c.world - 1;
+    const T* src = (const T*)c.data[p] + boff_elems + (i - (long)p * cn);
+    if constexpr (sizeof(T) == 2) {
+      *(short8*)(out + i) = *(const short8*)src;
+    } else {
+      #pragma unroll
+      for (int j = 0; j < 8; j++) ((float*)out)[i + j] = ((const float*)src)[j];
+    }
+  }
+  xar_barrier(c.flags, c.rank, c.world, gen + 2);
+}
+
 // ---------------------------------------------------------------------------
 // host side
 // ---------------------------------------------------------------------------
@@ -170,23 +226,43 @@ torch::Tensor xgmi_all_reduce(int64_t h, torch::Tensor t) {
   auto* c = (XgmiComm*)(intptr_t)h;
   TORCH_CHECK(t.is_cuda() && t.is_contiguous());
   long bytes = t.numel() * t.element_size();
-  TORCH_CHECK(bytes <= c->capacity, "tensor larger than xgmi buffer");
   TORCH_CHECK(t.numel() % 8 == 0, "numel must be a multiple of 8");
+  static long two_stage_min = [] {
+    const char* e = getenv("REALHF_AMD_XGMI_2STAGE_BYTES");
+    return e ? atol(e) : (long)(512 << 10);
+  }();
+  // 2-stage needs region B (upper half) + a world-divisible chunking
+  bool two_stage = c->world > 1 && bytes >= two_stage_min &&
+                   bytes <= c->capacity / 2 &&
+                   t.numel() % ((long)c->world * 8) == 0;
+  TORCH_CHECK(two_stage || bytes <= c->capacity,
+              "tensor larger than xgmi buffer");
   auto stream = cur_stream();
   xar_check(hipMemcpyAsync(c->data[c->rank], t.data_ptr(), bytes,
                            hipMemcpyDeviceToDevice, stream),
             "copy-in");
   auto out = torch::empty_like(t);
   unsigned gen = c->gen + 1;
-  c->gen += 2;  // this call consumes gen and gen+1
+  c->gen += two_stage ? 3 : 2;
   long vec = t.numel() / 8;
   int grid = (int)std::min<long>((vec + 255) / 256, XAR_MAX_BLOCKS);
+  long boff_elems = (c->capacity / 2) / t.element_size();
   if (t.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL((xar_1stage_kernel<bf16>), dim3(grid), dim3(256), 0,
-                       stream, *c, (bf16*)out.data_ptr(), t.numel(), gen);
+    if (two_stage)
+      hipLaunchKernelGGL((xar_2stage_kernel<bf16>), dim3(grid), dim3(256), 0,
+                         stream, *c, (bf16*)out.data_ptr(), t.numel(),
+                         boff_elems, gen);
+    else
+      hipLaunchKernelGGL((xar_1stage_kernel<bf16>), dim3(grid), dim3(256), 0,
+                         stream, *c, (bf16*)out.data_ptr(), t.numel(), gen);
   } else if (t.scalar_type() == torch::kFloat) {
-    hipLaunchKernelGGL((xar_1stage_kernel<float>), dim3(grid), dim3(256), 0,
-                       stream, *c, (float*)out.data_ptr(), t.numel(), gen);
+    if (two_stage)
+      hipLaunchKernelGGL((xar_2stage_kernel<float>), dim3(grid), dim3(256), 0,
+                         stream, *c, (float*)out.data_ptr(), t.numel(),
+                         boff_elems, gen);
+    else
+      hipLaunchKernelGGL((xar_1stage_kernel<float>), dim3(grid), dim3(256), 0,
+                         stream, *c, (float*)out.data_ptr(), t.numel(), gen);
   } else {
     TORCH_CHECK(false, "xgmi_all_reduce supports bf16/fp32");
   }

@@ -55,7 +55,7 @@ class XgmiAllReduce:
     def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         """Sum across ranks; returns a NEW tensor (input unchanged)."""
         if (
-            t.numel() * t.element_size() > self.capacity
+            t.numel() * t.element_size() > self.capacity // 2
             or t.dtype not in (torch.bfloat16, torch.float32)
             or t.numel() % 8 != 0
         ):

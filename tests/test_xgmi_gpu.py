@@ -39,6 +39,9 @@ def _xgmi_worker(rank, world, port, q):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    # force the 2-stage path for the "big" message below (the C++ reads
+    # this once, at the first all_reduce of the process)
+    os.environ["REALHF_AMD_XGMI_2STAGE_BYTES"] = "65536" 
     import torch
     import torch.distributed as dist
 
@@ -64,7 +67,13 @@ def _xgmi_worker(rank, world, port, q):
         ref = sum(ys)
         ok2 = bool(torch.allclose(out2.float().cpu(), ref, atol=0.1,
                                   rtol=0.05))
-        ok3 = ar.status_ok()
+        # large message -> 2-stage (reduce-scatter + all-gather) path
+        big = torch.full((65536,), float(rank + 1), device="cuda",
+                         dtype=torch.float32)
+        out3 = ar.all_reduce(big)
+        torch.cuda.synchronize()
+        ok2b = bool((out3 == expect).all().item())
+        ok3 = ar.status_ok() and ok2b
         ar.close()
         dist.barrier()
         q.put((rank, ok1 and ok2 and ok3, ""))
