@@ -45,7 +45,8 @@ __device__ __forceinline__ unsigned xar_load_sys(const unsigned* p) {
   return __hip_atomic_load(p, __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_SYSTEM);
 }
 
-// returns false on spin timeout (status word set); callers must exit
+// Arrival-only barrier (no data hand-off): returns false on spin timeout
+// (status word set); callers must exit.
 __device__ bool xar_barrier(unsigned* const* flags, int rank, int world,
                             unsigned gen) {
   const int b = blockIdx.x;
@@ -65,6 +66,44 @@ __device__ bool xar_barrier(unsigned* const* flags, int rank, int world,
   }
   __syncthreads();
   return true;
+}
+
+// Data-publishing barrier (guide §6 G16 recipe): EVERY wave drains its
+// stores (vmcnt is per-wave — lane 0's fence alone cannot see other
+// waves' in-flight writes), lane 0 issues a system-scope release with
+// the post-fence wait restated, THEN the arrival exchange; after the
+// spin, lane 0 does the system-scope acquire so this block's plain
+// loads see every peer's published data.
+__device__ bool xar_barrier_publish(unsigned* const* flags, int rank,
+                                    int world, unsigned gen) {
+  __shared__ int bar_fail;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    bar_fail = 0;
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+  __syncthreads();
+  const int b = blockIdx.x;
+  if (threadIdx.x < (unsigned)world) {
+    const int p = threadIdx.x;
+    xar_store_sys(&flags[p][b * XAR_MAX_WORLD + rank], gen);
+    unsigned long long spins = 0;
+    while (xar_load_sys(&flags[rank][b * XAR_MAX_WORLD + p]) < gen) {
+      if (++spins > XAR_SPIN_LIMIT) {
+        xar_store_sys(&flags[rank][XAR_STATUS_SLOT], 0xdead);
+        bar_fail = 1;
+        break;
+      }
+      __builtin_amdgcn_s_sleep(16);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x == 0)
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+  return bar_fail == 0;
 }
 
 template <typename T>
@@ -144,7 +183,7 @@ __global__ void xar_2stage_kernel(XgmiComm c, T* __restrict__ out, long n,
       for (int j = 0; j < 8; j++) ((float*)myB)[i + j] = acc[j];
     }
   }
-  if (!xar_barrier(c.flags, c.rank, c.world, gen + 1)) return;
+  if (!xar_barrier_publish(c.flags, c.rank, c.world, gen + 1)) return;
   // stage 2: gather every rank's reduced chunk
   for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < n;
        i += stride) {

@@ -73,10 +73,12 @@ def _xgmi_worker(rank, world, port, q):
         out3 = ar.all_reduce(big)
         torch.cuda.synchronize()
         ok2b = bool((out3 == expect).all().item())
-        ok3 = ar.status_ok() and ok2b
+        okst = ar.status_ok()
+        detail = (f"ok1={ok1} ok2={ok2} ok2b={ok2b} status={okst} "
+                  f"big_uniq={out3.unique()[:4].tolist()}")
         ar.close()
         dist.barrier()
-        q.put((rank, ok1 and ok2 and ok3, ""))
+        q.put((rank, ok1 and ok2 and ok2b and okst, detail))
     except Exception as e:  # pragma: no cover
         q.put((rank, False, repr(e)))
 
