@@ -35,19 +35,24 @@ logger = logging.getLogger("realloc")
 @dataclasses.dataclass(frozen=True)
 class ParallelStrategy:
     """Geometry + the global ranks hosting each (pp, dp, tp) coordinate.
-    rank_map[(pp, dp, tp)] = global rank."""
+    rank_map[(pp, dp, tp)] = global rank.  With expert parallelism,
+    `ep` divides dp and dp-coordinate d holds expert block d % ep
+    (matching the trainer's ep_rank = dp_rank % ep_size convention)."""
 
     pp: int
     dp: int
     tp: int
     rank_map: Tuple[Tuple[Tuple[int, int, int], int], ...]  # hashable
+    ep: int = 1
 
     @classmethod
-    def make(cls, pp: int, dp: int, tp: int, ranks: Optional[List[int]] = None):
+    def make(cls, pp: int, dp: int, tp: int, ranks: Optional[List[int]] = None,
+             ep: int = 1):
         n = pp * dp * tp
         if ranks is None:
             ranks = list(range(n))
         assert len(ranks) == n
+        assert dp % max(ep, 1) == 0
         rm = []
         i = 0
         for p in range(pp):
@@ -55,7 +60,7 @@ class ParallelStrategy:
                 for t in range(tp):
                     rm.append((((p, d, t)), ranks[i]))
                     i += 1
-        return cls(pp=pp, dp=dp, tp=tp, rank_map=tuple(rm))
+        return cls(pp=pp, dp=dp, tp=tp, rank_map=tuple(rm), ep=ep)
 
     def rank_of(self, p, d, t):
         return dict(self.rank_map)[(p, d, t)]
@@ -136,6 +141,14 @@ def _key_intervals_for_shard_pair(
     return si, di
 
 
+def _key_src_ep_rank(cfg: ReaLModelConfig, key: str, ep: int) -> Optional[int]:
+    """Which src ep block holds this key; None = present on every block."""
+    if ep <= 1 or ".experts." not in key:
+        return None
+    e = int(key.split(".experts.")[1].split(".")[0])
+    return e // (cfg.moe.num_experts // ep)
+
+
 _PLAN_CACHE: Dict = {}
 
 
@@ -155,20 +168,25 @@ def build_realloc_plan(
 
     src_parts = PL.partition_pipeline_layers(cfg, src.pp)
     dst_parts = PL.partition_pipeline_layers(cfg, dst.pp)
-    # layouts per (stage, tp shard) — same for every dp replica
+    # layouts per (stage, tp shard, ep block) — same for every dp replica
+    # within an ep block
     src_layouts = {
-        (p, t): PL.build_flat_layout(
-            cfg, list(range(*src_parts[p])), t, src.tp
+        (p, t, e): PL.build_flat_layout(
+            cfg, list(range(*src_parts[p])), t, src.tp,
+            ep_rank=e, ep_size=src.ep,
         )
         for p in range(src.pp)
         for t in range(src.tp)
+        for e in range(src.ep)
     }
     dst_layouts = {
-        (p, t): PL.build_flat_layout(
-            cfg, list(range(*dst_parts[p])), t, dst.tp
+        (p, t, e): PL.build_flat_layout(
+            cfg, list(range(*dst_parts[p])), t, dst.tp,
+            ep_rank=e, ep_size=dst.ep,
         )
         for p in range(dst.pp)
         for t in range(dst.tp)
+        for e in range(dst.ep)
     }
     layer_to_src_stage = {}
     for p, (lo, hi) in src_parts.items():
@@ -179,17 +197,26 @@ def build_realloc_plan(
     pair_src: Dict[Tuple[int, int], List[np.ndarray]] = {}
     pair_dst: Dict[Tuple[int, int], List[np.ndarray]] = {}
 
+    n_src_rep = src.dp // src.ep  # dp replicas per expert block
     for dp_ in range(dst.pp):
         d_lo, d_hi = dst_parts[dp_]
         for dd in range(dst.dp):
             for dt in range(dst.tp):
                 dst_rank = dst.rank_of(dp_, dd, dt)
-                dl = dst_layouts[(dp_, dt)]
+                dl = dst_layouts[(dp_, dt, dd % dst.ep)]
                 for layer in range(d_lo, d_hi):
                     sp_ = layer_to_src_stage[layer]
-                    # spread source load over src dp replicas
-                    sd = (dd + layer) % src.dp
                     for key in PL.keys_of_layer(cfg, layer):
+                        if key not in dl.specs:
+                            continue  # expert outside this dst ep block
+                        # expert keys exist only on src dp ranks whose ep
+                        # block owns the expert; others on every dp rank
+                        src_e = _key_src_ep_rank(cfg, key, src.ep)
+                        if src_e is None:
+                            sd = (dd + layer) % src.dp
+                        else:
+                            sd = src_e + src.ep * ((dd + layer) % n_src_rep)
+                        se = sd % src.ep
                         kind = PL.key_kind(key)
                         if kind == "head":
                             kind = PL.REPLICATED if cfg.is_critic else PL.VOCAB
@@ -200,7 +227,7 @@ def build_realloc_plan(
                             src_ts = range(src.tp)
                         for st in src_ts:
                             out = _key_intervals_for_shard_pair(
-                                cfg, key, src_layouts[(sp_, st)], dl,
+                                cfg, key, src_layouts[(sp_, st, se)], dl,
                                 st, src.tp, dt, dst.tp,
                             )
                             if out is None:

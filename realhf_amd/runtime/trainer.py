@@ -175,7 +175,12 @@ def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment
         name = ModelName(role, replica)
         rcfg = model_cfg_of(mc, name)
         par = _resolve_parallel(cfg, mc.parallel, world, rcfg.param_count())
-        model_strategies[name] = _strategy_for(par, world)
+        strat = _strategy_for(par, world)
+        if rcfg.moe is not None:
+            ep = min(rcfg.moe.expert_parallel_size, strat.dp)
+            if ep > 1 and strat.dp % ep == 0:
+                strat = dataclasses.replace(strat, ep=ep)
+        model_strategies[name] = strat
         model_cfgs[name] = rcfg
         model_roles[name] = mc
         return name, par
@@ -389,14 +394,12 @@ class Trainer:
             # register grid (group creation is collective — all ranks join)
             mc0 = self.built.model_roles[name]
             rcfg0 = self.built.model_cfgs[name]
-            ep = 1
-            if rcfg0.moe is not None:
-                ep = min(rcfg0.moe.expert_parallel_size, strat.dp)
+            ep = strat.ep
             topo = PipeDataTensorTopology(
                 num_pp=strat.pp, num_dp=strat.dp, num_tp=strat.tp,
                 sequence_parallel=mc0.parallel.sequence_parallel and strat.tp > 1,
                 gradient_checkpointing=mc0.gradient_checkpointing,
-                ep_size=ep if strat.dp % max(ep, 1) == 0 else 1,
+                ep_size=ep,
             )
             rank_mapping = {
                 topo.get_rank(pipe=p, data=d, tensor=t): r

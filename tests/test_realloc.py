@@ -185,3 +185,106 @@ def _ema_worker():
 def test_realloc_ema():
     """eta < 1 EMA-merges into the destination (ref-EMA)."""
     LocalMultiProcessTest(2, _ema_worker).launch()
+
+
+def _moe_cfg():
+    from realhf_amd.api.model import MoEConfig
+    cfg = make_test_config(n_layers=2, hidden_dim=32, n_heads=4, n_kv_heads=2,
+                           vocab_size=64)
+    cfg.moe = MoEConfig(num_experts=4, top_k=2, expert_parallel_size=2)
+    return cfg
+
+
+def test_plan_conservation_ep():
+    """EP: expert keys live only on their ep block; every dst element of
+    every (tp, ep) shard is written exactly once."""
+    cfg = _moe_cfg()
+    for (s_geom, s_ep), (d_geom, d_ep) in [
+        (((1, 2, 1), 2), ((1, 2, 1), 1)),   # ep2 -> replicated
+        (((1, 2, 1), 1), ((1, 2, 1), 2)),   # replicated -> ep2
+        (((1, 4, 1), 2), ((1, 2, 2), 2)),   # ep2 dp4 -> ep2 dp2 tp2
+        (((1, 4, 1), 4), ((1, 2, 1), 2)),   # ep4 -> ep2
+    ]:
+        src = ParallelStrategy.make(*s_geom, ep=s_ep)
+        dst = ParallelStrategy.make(*d_geom, ep=d_ep)
+        plan = build_realloc_plan(cfg, src, dst)
+        dst_parts = PL.partition_pipeline_layers(cfg, dst.pp)
+        for p in range(dst.pp):
+            for d in range(dst.dp):
+                for t in range(dst.tp):
+                    r = dst.rank_of(p, d, t)
+                    layout = PL.build_flat_layout(
+                        cfg, list(range(*dst_parts[p])), t, dst.tp,
+                        ep_rank=d % dst.ep, ep_size=dst.ep,
+                    )
+                    expect = sum(layout.specs[k].numel for k in layout.keys)
+                    ivs = np.concatenate(
+                        [tr.dst_intervals for tr in plan.transfers
+                         if tr.dst_rank == r], axis=0)
+                    ivs = ivs[np.argsort(ivs[:, 0])]
+                    total = int((ivs[:, 1] - ivs[:, 0]).sum())
+                    assert total == expect, (s_geom, s_ep, d_geom, d_ep,
+                                             p, d, t, total, expect)
+                    assert (ivs[1:, 0] >= ivs[:-1, 1]).all()
+
+
+def _fill_model_from_full_ep(model, cfg, sd):
+    with torch.no_grad():
+        for k in model.layout.keys:
+            shard = PL.tp_partition(cfg, k, sd[k], model.tp_rank, model.tp_size)
+            model.param_view(k).copy_(shard)
+
+
+def _realloc_ep_worker(s_ep, d_ep):
+    import torch.distributed as dist
+
+    from realhf_amd.models.real_model import ReaLModel
+
+    cfg = _moe_cfg()
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg)
+    rank = dist.get_rank()
+    src = ParallelStrategy.make(1, 4, 1, ep=s_ep)
+    dst = ParallelStrategy.make(1, 2, 2, ep=d_ep)
+
+    src_model = None
+    for (p, d, t), r in src.rank_map:
+        if r == rank:
+            src_model = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                                  tp_rank=t, tp_size=1, pp_rank=p, pp_size=1,
+                                  ep_rank=d % s_ep, ep_size=s_ep)
+            _fill_model_from_full_ep(src_model, cfg, sd)
+    dst_model = None
+    for (p, d, t), r in dst.rank_map:
+        if r == rank:
+            dst_model = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                                  tp_rank=t, tp_size=2, pp_rank=p, pp_size=1,
+                                  ep_rank=d % d_ep, ep_size=d_ep)
+            with torch.no_grad():
+                dst_model.flat_param.zero_()
+    plan = build_realloc_plan(cfg, src, dst)
+    execute_realloc(
+        plan,
+        src_model.flat_param if src_model is not None else None,
+        dst_model.flat_param if dst_model is not None else None,
+    )
+    if dst_model is not None:
+        _check_model_vs_full(dst_model, cfg, sd)
+    dist.barrier()
+    plan2 = build_realloc_plan(cfg, dst, src)
+    if src_model is not None:
+        with torch.no_grad():
+            src_model.flat_param.zero_()
+    execute_realloc(
+        plan2,
+        dst_model.flat_param if dst_model is not None else None,
+        src_model.flat_param if src_model is not None else None,
+    )
+    if src_model is not None:
+        _check_model_vs_full(src_model, cfg, sd)
+
+
+@pytest.mark.distributed
+@pytest.mark.parametrize("s_ep,d_ep", [(2, 1), (4, 2), (2, 2)])
+def test_realloc_roundtrip_ep(s_ep, d_ep):
+    LocalMultiProcessTest(4, _realloc_ep_worker, s_ep, d_ep).launch()
