@@ -278,3 +278,62 @@ def test_ppo_mixed_strategies_four_ranks(tmp_path):
     _write_prompt_data(data, n=16)
     LocalMultiProcessTest(4, _ppo_mixed4_worker, data,
                           str(tmp_path / "root")).launch()
+
+
+def test_dpo_experiment_single_process(tmp_path):
+    from realhf_amd.api.experiment import DPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "rw.jsonl")
+    _write_rw_data(data)
+    cfg = DPOConfig(experiment_name="t-dpo", trial_name="cpu", n_gpus=1)
+    cfg.actor.dtype = "float32"
+    cfg.ref.dtype = "float32"
+    cfg.dataset.type_ = "rw_paired"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.exp_ctrl.benchmark_steps = 2
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
+
+
+def test_grpo_experiment_single_process(tmp_path):
+    from realhf_amd.api.experiment import GRPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data)
+    cfg = GRPOConfig(experiment_name="t-grpo", trial_name="cpu", n_gpus=1)
+    for mc in (cfg.actor, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.group_size = 2
+    cfg.ppo.gen.max_new_tokens = 6
+    cfg.ppo.gen.min_new_tokens = 2
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
+
+
+def test_gen_experiment_single_process(tmp_path):
+    from realhf_amd.api.experiment import GenerationConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data)
+    cfg = GenerationConfig(experiment_name="t-gen", trial_name="cpu", n_gpus=1)
+    cfg.model.dtype = "float32"
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.gen.max_new_tokens = 6
+    cfg.gen.use_hip_graph = False
+    cfg.exp_ctrl.benchmark_steps = 1
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
