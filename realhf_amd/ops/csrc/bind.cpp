@@ -34,6 +34,10 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> residual);
 torch::Tensor skinny_gemm_nc(torch::Tensor x, torch::Tensor w,
                              torch::Tensor out32_ws, long splitk);
+std::vector<torch::Tensor> attn_varlen_bwd(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor dout,
+    torch::Tensor lse, torch::Tensor Dsum, torch::Tensor cu_seqlens,
+    bool causal, double scale);
 int64_t xgmi_create(int64_t rank, int64_t world, int64_t capacity);
 std::vector<py::bytes> xgmi_handles(int64_t h);
 void xgmi_connect(int64_t h, const std::vector<std::string>& data_handles,
@@ -75,6 +79,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm);
   m.def("skinny_gemm2", &skinny_gemm2);
   m.def("skinny_gemm_nc", &skinny_gemm_nc);
+  m.def("attn_varlen_bwd", &attn_varlen_bwd);
   m.def("xgmi_create", &xgmi_create);
   m.def("xgmi_handles", &xgmi_handles);
   m.def("xgmi_connect", &xgmi_connect);

@@ -289,11 +289,25 @@ class _AttnVarlenFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out):
-        # Backward by blocked recompute with rocBLAS GEMMs (fp32 math).
-        # Hand-written HIP bwd is the planned upgrade; attention bwd is a
-        # small share of train-step FLOPs at the bench seqlens (SURVEY §7
-        # hard-parts note), so this path is correct-first.
         q, k, v, out, lse, cu_seqlens = ctx.saved_tensors
+        if os.environ.get("REALHF_AMD_HIP_ATTN_BWD") == "1":
+            # hand-written MFMA backward (attn_bwd.hip): kv-stationary,
+            # dQ via fp32 atomics, per-q-head dK/dV reduced here for GQA
+            C = _ops.require_hip()
+            dout = grad_out.contiguous()
+            dsum = (dout.float() * out.float()).sum(-1)  # [total, nq]
+            dq32, dk32, dv32 = C.attn_varlen_bwd(
+                q, k, v, dout, lse, dsum, cu_seqlens, ctx.causal, ctx.scale
+            )
+            nq, nkv = q.shape[1], k.shape[1]
+            rep = nq // nkv
+            if rep > 1:
+                t = dk32.shape[0]
+                dk32 = dk32.view(t, nkv, rep, -1).sum(2)
+                dv32 = dv32.view(t, nkv, rep, -1).sum(2)
+            return (dq32.to(q.dtype), dk32.to(k.dtype), dv32.to(v.dtype),
+                    None, None, None, None)
+        # Fallback: blocked recompute with rocBLAS GEMMs (fp32 softmax).
         with torch.enable_grad():
             qg = q.detach().requires_grad_(True)
             kg = k.detach().requires_grad_(True)

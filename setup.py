@@ -25,6 +25,7 @@ SRC = [
     "realhf_amd/ops/csrc/skinny_gemm.hip",
     "realhf_amd/ops/csrc/attn_varlen.hip",
     "realhf_amd/ops/csrc/all_reduce.hip",
+    "realhf_amd/ops/csrc/attn_bwd.hip",
 ]
 
 setup(
