@@ -184,17 +184,24 @@ __global__ void xar_2stage_kernel(XgmiComm c, T* __restrict__ out, long n,
     }
   }
   if (!xar_barrier_publish(c.flags, c.rank, c.world, gen + 1)) return;
-  // stage 2: gather every rank's reduced chunk
-  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < n;
-       i += stride) {
-    int p = (int)(i / cn);
-    if (p >= c.world) p = c.world - 1;
-    const T* src = (const T*)c.data[p] + boff_elems + (i - (long)p * cn);
-    if constexpr (sizeof(T) == 2) {
-      *(short8*)(out + i) = *(const short8*)src;
-    } else {
-      #pragma unroll
-      for (int j = 0; j < 8; j++) ((float*)out)[i + j] = ((const float*)src)[j];
+  // stage 2: gather every rank's reduced chunk.  CRITICAL: the barrier is
+  // pairwise per-BLOCK (block b syncs with block b of each peer), so
+  // block b may only read intra-chunk offsets that block b itself wrote
+  // in stage 1 — the loops below mirror stage 1's offset mapping exactly
+  // (a full-chunk read pattern raced with peers' other blocks: measured
+  // as zeroed stretches of the peer chunk under load).
+  for (int p = 0; p < c.world; p++) {
+    const T* srcB = (const T*)c.data[p] + boff_elems;
+    T* outp = out + (long)p * cn;
+    for (long o = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; o < cn;
+         o += stride) {
+      if constexpr (sizeof(T) == 2) {
+        *(short8*)(outp + o) = *(const short8*)(srcB + o);
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; j++)
+          ((float*)outp)[o + j] = ((const float*)srcB)[o + j];
+      }
     }
   }
   xar_barrier(c.flags, c.rank, c.world, gen + 2);
