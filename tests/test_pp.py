@@ -175,3 +175,50 @@ def _pp_gen_worker():
 @pytest.mark.distributed
 def test_pp2_generate_matches_single():
     LocalMultiProcessTest(2, _pp_gen_worker).launch()
+
+
+def _pp_tp_forward_worker():
+    """Combined tp2 x pp2 on 4 ranks: forward matches single-process."""
+    import torch.distributed as dist
+
+    from realhf_amd.api.config import Abstraction, ModelName
+    from realhf_amd.api.model import FinetuneSpec, Model, make_backend
+    import realhf_amd.runtime.engine  # noqa: F401
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.runtime.engine import sample_to_packed
+
+    cfg = make_test_config(n_layers=4, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=41)
+    init_global_constants(num_dp=1, num_tp=2, num_pp=2, model_name="m")
+    g = constants.grid_of("m")
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                  tp_rank=g.tp_rank, tp_size=2,
+                  pp_rank=g.pp_rank, pp_size=2)
+    _fill_model_from_full(m, cfg, sd)
+    model = Model(ModelName("m", 0), m, None, torch.device("cpu"),
+                  torch.float32)
+    model = make_backend(Abstraction("inference")).initialize(
+        model, FinetuneSpec(1, 64, 4))
+    batch = _mk_batch(cfg)
+    with constants.model_scope("m"):
+        out = model.module.forward(batch, n_mbs=2)
+    if m.is_last_stage:
+        single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+        _fill_model_from_full(single, cfg, sd)
+        ids, cu, mx = sample_to_packed(batch)
+        with torch.no_grad():
+            ref = single(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=mx)
+        # last stage returns this tp rank's vocab-parallel logit shard
+        vshard = cfg.vocab_size // 2
+        ref_shard = ref[:, g.tp_rank * vshard:(g.tp_rank + 1) * vshard]
+        torch.testing.assert_close(out, ref_shard, atol=2e-4, rtol=2e-4)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_pp2_tp2_forward():
+    LocalMultiProcessTest(4, _pp_tp_forward_worker).launch()
