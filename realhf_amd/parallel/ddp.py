@@ -22,7 +22,8 @@ mixed-precision optimizer.
 """
 import dataclasses
 import math
-from typing import Dict
+import os
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist
@@ -96,6 +97,24 @@ class ZeRO1Optimizer:
         self.dp_group = g.dp_group() if g is not None else None
         self.dp_size = g.dp_size if g is not None else 1
         self.dp_rank = g.dp_rank if g is not None else 0
+        # SP shards tokens over the tp group, so REPLICATED params'
+        # grads (layernorms, critic head) are per-tp-rank partials and
+        # must sum over tp before the DP reduce-scatter.
+        self._sp_repl_ivs = []
+        self.tp_group = None
+        if (g is not None and g.tp_size > 1 and constants.has_current()
+                and constants.sequence_parallel() and hasattr(model, "layout")):
+            from realhf_amd.models import param_layout as PL
+
+            self.tp_group = g.tp_group()
+            for k in model.layout.keys:
+                kind = PL.key_kind(k)
+                if kind == "head":
+                    kind = (PL.REPLICATED if model.config.is_critic
+                            else PL.VOCAB)
+                if kind == PL.REPLICATED:
+                    sp_ = model.layout.specs[k]
+                    self._sp_repl_ivs.append((sp_.start, sp_.end))
 
         n = model.flat_param.numel()
         # pad so every DP shard is 256-element aligned
@@ -119,15 +138,59 @@ class ZeRO1Optimizer:
         s1 = s0 + self.shard_size
         self.shard_bounds = (s0, s1)
         state_dev = "cpu" if cfg.offload else dev
-        self.master = self._param_padded[s0:s1].to(torch.float32).to(state_dev)
+
+        # ---- bucketed comm/compute overlap (opt-in, dp>1) --------------
+        # With REALHF_AMD_ZERO_OVERLAP=1, grads reduce-scatter per BUCKET
+        # as soon as the backward has produced every grad overlapping the
+        # bucket (post-accumulate-grad hooks, armed for the LAST
+        # minibatch only).  Ownership becomes per-bucket (rank r owns
+        # slice r of each bucket) so RS outputs land contiguously in
+        # grad_shard; master/exp_avg use the same per-bucket layout and
+        # params all-gather back per bucket.
+        self.bucket_size = bucket_size
+        self.overlap_comm = (
+            os.environ.get("REALHF_AMD_ZERO_OVERLAP") == "1"
+            and self.dp_size > 1
+            and hasattr(model, "layout")
+            and not self._sp_repl_ivs  # SP tp-reduce must precede RS
+        )
+        if self.overlap_comm:
+            align = 256 * self.dp_size
+            self._bsz = max(align, (bucket_size // align) * align)
+            self.buckets = []  # (b0, b1, shard_off)
+            off, soff = 0, 0
+            while off < self.n_pad:
+                b1 = min(self.n_pad, off + self._bsz)
+                self.buckets.append((off, b1, soff))
+                soff += (b1 - off) // self.dp_size
+                off = b1
+            pieces = []
+            for b0, b1, _ in self.buckets:
+                l = (b1 - b0) // self.dp_size
+                pieces.append(
+                    self._param_padded[b0 + self.dp_rank * l:
+                                       b0 + (self.dp_rank + 1) * l])
+            self.master = torch.cat(pieces).to(torch.float32).to(state_dev)
+            self._shard_bf16 = torch.empty(
+                self.shard_size, dtype=model.flat_param.dtype, device=dev)
+            # param -> overlapped bucket ids
+            self._param_buckets = {}
+            for k in model.layout.keys:
+                sp = model.layout.specs[k]
+                self._param_buckets[k] = list(
+                    range(sp.start // self._bsz, (sp.end - 1) // self._bsz + 1))
+            self._bucket_left: List[int] = []
+            self._reduced: List[bool] = []
+            self._works: list = []
+            self._armed = False
+        else:
+            self.master = self._param_padded[s0:s1].to(torch.float32).to(state_dev)
         self.exp_avg = torch.zeros_like(self.master)
         self.exp_avg_sq = torch.zeros_like(self.master)
         self.grad_shard = torch.empty(
             self.shard_size, dtype=model.flat_param.dtype, device=dev
         )
 
-        # gradient hooks for bucketed overlap
-        self.bucket_size = bucket_size
         self.overlap = overlap_reduce_scatter and self.dp_size > 1
         self._comm_stream = (
             torch.cuda.Stream() if (self.overlap and dev.type == "cuda") else None
@@ -142,7 +205,47 @@ class ZeRO1Optimizer:
             if not p.requires_grad:
                 p.requires_grad_(True)
             p.grad = self.model.grad_view(k)
+            if self.overlap_comm:
+                bids = self._param_buckets.get(k)
+                if bids is not None:
+                    p.register_post_accumulate_grad_hook(
+                        self._make_bucket_hook(bids))
         self._grad_views_attached = True
+
+    # ---------------- bucketed overlap machinery ----------------------
+    def _make_bucket_hook(self, bids):
+        def hook(_p):
+            if not self._armed:
+                return
+            for b in bids:
+                self._bucket_left[b] -= 1
+                if self._bucket_left[b] == 0:
+                    self._flush_bucket(b)
+        return hook
+
+    def arm_overlap(self):
+        """Call right before the LAST minibatch's backward: buckets
+        reduce-scatter as soon as their grads are complete."""
+        if not self.overlap_comm:
+            return
+        counts = [0] * len(self.buckets)
+        for k in self.model._params:
+            for b in self._param_buckets.get(k, ()):
+                counts[b] += 1
+        self._bucket_left = counts
+        self._reduced = [False] * len(self.buckets)
+        self._works = []
+        self._armed = True
+
+    def _flush_bucket(self, b):
+        b0, b1, soff = self.buckets[b]
+        l = (b1 - b0) // self.dp_size
+        w = dist.reduce_scatter_tensor(
+            self.grad_shard[soff:soff + l], self.grad_padded[b0:b1],
+            op=dist.ReduceOp.AVG, group=self.dp_group, async_op=True,
+        )
+        self._works.append(w)
+        self._reduced[b] = True
 
     def zero_grad(self):
         self.grad_padded.zero_()
@@ -170,8 +273,23 @@ class ZeRO1Optimizer:
         cfg = self.cfg
         dev = self.grad_padded.device
 
+        # 0. SP: sum replicated params' grads over the tp group
+        if self._sp_repl_ivs:
+            for a, b in self._sp_repl_ivs:
+                dist.all_reduce(self.grad_padded[a:b], group=self.tp_group)
+
         # 1. reduce-scatter grads over DP (average)
-        if self.dp_size > 1:
+        if self.overlap_comm:
+            # flush buckets the hooks missed (unused params / not armed)
+            for b in range(len(self.buckets)):
+                if not self._reduced[b]:
+                    self._flush_bucket(b)
+            for w in self._works:
+                w.wait()
+            self._works = []
+            self._armed = False
+            gshard = self.grad_shard
+        elif self.dp_size > 1:
             dist.reduce_scatter_tensor(
                 self.grad_shard, self.grad_padded, op=dist.ReduceOp.AVG,
                 group=self.dp_group,
@@ -201,7 +319,8 @@ class ZeRO1Optimizer:
         # 3. AdamW on this rank's shard (bf16 grads read directly)
         lr = self._lr()
         s0, s1 = self.shard_bounds
-        param_shard = self._param_padded[s0:s1]
+        param_shard = (self._shard_bf16 if self.overlap_comm
+                       else self._param_padded[s0:s1])
         g_in = gshard.float().cpu() if cfg.offload else gshard
         ops.fused_adamw(
             self.master, g_in, self.exp_avg, self.exp_avg_sq,
@@ -214,7 +333,16 @@ class ZeRO1Optimizer:
             param_shard.copy_(self.master.to(param_shard.dtype))
 
         # 4. all-gather updated params
-        if self.dp_size > 1:
+        if self.overlap_comm:
+            works = []
+            for b0, b1, soff in self.buckets:
+                l = (b1 - b0) // self.dp_size
+                works.append(dist.all_gather_into_tensor(
+                    self._param_padded[b0:b1], self._shard_bf16[soff:soff + l],
+                    group=self.dp_group, async_op=True))
+            for w in works:
+                w.wait()
+        elif self.dp_size > 1:
             dist.all_gather_into_tensor(
                 self._param_padded, param_shard.contiguous(), group=self.dp_group
             )

@@ -61,10 +61,13 @@ class PipelinableTrainEngine(PipelinableEngine):
         self.optimizer.zero_grad()
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 and input_.bs >= n_mbs else [input_]
         stats: Dict[str, float] = {}
-        for mb in mbs:
+        for i, mb in enumerate(mbs):
             ids, cu, mx = sample_to_packed(mb)
             out = self.model(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=mx)
             loss, st = loss_fn(out, mb)
+            if i == len(mbs) - 1 and hasattr(self.optimizer, "arm_overlap"):
+                # bucketed reduce-scatter overlaps this (last) backward
+                self.optimizer.arm_overlap()
             (loss / len(mbs)).backward()
             for k, v in st.items():
                 stats[k] = stats.get(k, 0.0) + float(v) / len(mbs)
