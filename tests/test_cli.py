@@ -45,3 +45,19 @@ def test_cli_help():
     )
     assert out.returncode == 0
     assert "ppo" in out.stdout
+
+
+def test_examples_run(tmp_path):
+    """The example scripts are runnable user documentation — keep them
+    working (reference parity: examples/{load_and_eval_rw,new_algorithms,
+    customized_exp})."""
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for script in ("examples/load_and_eval_rw.py",
+                   "examples/new_algorithms/reinforce.py",
+                   "examples/customized_exp/ppo_ref_ema.py"):
+        r = subprocess.run([sys.executable, os.path.join(root, script)],
+                           capture_output=True, text=True, timeout=300)
+        assert r.returncode == 0, (script, r.stdout[-800:], r.stderr[-800:])
