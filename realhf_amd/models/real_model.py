@@ -163,9 +163,13 @@ class ReaLModel(nn.Module):
             for name, t in self.lora_params.items():
                 if name.endswith("lora_A"):
                     # deterministic per-name seed: identical on every rank
-                    # (replicated-A TP shards must match across tp ranks)
+                    # (replicated-A TP shards must match across tp ranks).
+                    # NOTE zlib.crc32, not hash() — python str hashes are
+                    # salted per process.
+                    import zlib
+
                     g = torch.Generator(device=t.device)
-                    g.manual_seed(abs(hash(name)) % (2**31))
+                    g.manual_seed(zlib.crc32(name.encode()) % (2**31))
                     t.normal_(0.0, 1.0 / dim, generator=g)
         self._inject_lora()
 
