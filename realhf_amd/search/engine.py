@@ -34,6 +34,34 @@ GEN_BW_EFF = 0.65
 GEN_TOKEN_OVERHEAD_S = 2.0e-3 / 512  # residual per-token fixed cost
 
 
+def load_cost_table(path: str):
+    """Calibrate the analytic cost model from measured numbers
+    (tools/profile_layers.py output; reference counterpart: the search
+    engine's ProfileLayers-produced op-cost tables, search_engine/
+    layers.py:56).  Recognized keys: bf16_tf, hbm_gbps, xgmi_link_gbps,
+    train_eff, inf_eff, gen_bw_eff."""
+    import json
+
+    global BF16_PEAK_TF, HBM_GBPS, XGMI_LINK_GBPS
+    global TRAIN_EFF, INF_EFF, GEN_BW_EFF
+    with open(path) as f:
+        t = json.load(f)
+    BF16_PEAK_TF = t.get("bf16_tf", BF16_PEAK_TF)
+    HBM_GBPS = t.get("hbm_gbps", HBM_GBPS)
+    XGMI_LINK_GBPS = t.get("xgmi_link_gbps", XGMI_LINK_GBPS)
+    TRAIN_EFF = t.get("train_eff", TRAIN_EFF)
+    INF_EFF = t.get("inf_eff", INF_EFF)
+    GEN_BW_EFF = t.get("gen_bw_eff", GEN_BW_EFF)
+    logger.info("cost model calibrated from %s: tf=%.0f hbm=%.0f", path,
+                BF16_PEAK_TF, HBM_GBPS)
+
+
+import os as _os  # noqa: E402
+
+if _os.environ.get("REALHF_AMD_COST_TABLE"):
+    load_cost_table(_os.environ["REALHF_AMD_COST_TABLE"])
+
+
 @dataclasses.dataclass
 class MFCSpec:
     name: str

@@ -67,3 +67,16 @@ def test_search_allocation_mode_in_experiment():
     built = build_experiment(cfg, 8)
     assert cfg.allocation_mode == "manual"
     assert built.allocations["actor_train"].strategy.world >= 4
+
+
+def test_cost_table_calibration(tmp_path):
+    import json
+
+    from realhf_amd.search import engine as E
+
+    old = (E.BF16_PEAK_TF, E.HBM_GBPS)
+    p = tmp_path / "ct.json"
+    p.write_text(json.dumps({"bf16_tf": 1000.0, "hbm_gbps": 5000.0}))
+    E.load_cost_table(str(p))
+    assert E.BF16_PEAK_TF == 1000.0 and E.HBM_GBPS == 5000.0
+    E.BF16_PEAK_TF, E.HBM_GBPS = old
