@@ -60,6 +60,12 @@ import os as _os  # noqa: E402
 
 if _os.environ.get("REALHF_AMD_COST_TABLE"):
     load_cost_table(_os.environ["REALHF_AMD_COST_TABLE"])
+else:
+    # measured-on-MI355X defaults shipped with the package
+    _default = _os.path.join(_os.path.dirname(__file__), "..", "data",
+                             "cost_table_gfx950.json")
+    if _os.path.exists(_default):
+        load_cost_table(_default)
 
 
 @dataclasses.dataclass
