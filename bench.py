@@ -131,7 +131,11 @@ def main():
         with scope(n):
             m = ReaLModel(cfg, device=device,
                           dtype=torch.bfloat16 if use_cuda else torch.float32)
-            torch.manual_seed(4242 + hash(n) % 1000)  # same weights on all ranks
+            # zlib.crc32, NOT hash(): python str hashes are salted per
+            # process, which would give each DP rank different weights
+            import zlib
+
+            torch.manual_seed(4242 + zlib.crc32(n.encode()) % 1000)
             m.random_init()
             model = Model(
                 name=ModelName(n, 0), module=m, tokenizer=None, device=device,
