@@ -154,6 +154,19 @@ def main():
     mem = torch.cuda.memory_allocated() / 2**30 if use_cuda else 0.0
     log(f"models built in {time.time() - t0:.1f}s; mem={mem:.1f} GiB")
 
+    # DP replicas must start from identical weights (weak-scaling
+    # validity): verify with a cheap checksum all-reduce
+    if world > 1:
+        for n in names:
+            with scope(n):
+                mm = models[n].module.module
+                cs = mm.flat_param.float().sum()
+                lo, hi = cs.clone(), cs.clone()
+                dist.all_reduce(lo, op=dist.ReduceOp.MIN)
+                dist.all_reduce(hi, op=dist.ReduceOp.MAX)
+                assert torch.allclose(lo, hi, rtol=1e-5), (
+                    f"{n}: DP replicas differ at init ({lo} vs {hi})")
+
     gconfig = dict(
         max_new_tokens=args.gen_len, min_new_tokens=args.gen_len,
         greedy=False, top_k=1000, top_p=0.9, temperature=1.0,
