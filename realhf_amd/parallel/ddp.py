@@ -9,9 +9,9 @@ Mechanics (per train step, DP size d):
   1. backward accumulates into the padded flat grad buffer (param.grad
      views);
   2. one reduce-scatter of the flat grad over DP; global grad-norm clip
-     (DP-reduced).  (Bucketed overlap with backward is the planned
-     refinement; the single large reduce-scatter is already the
-     xGMI-friendly shape — few large collectives.)
+     (DP-reduced).  With REALHF_AMD_ZERO_OVERLAP=1 the reduce-scatter
+     runs per BUCKET as the backward completes each bucket's grads
+     (per-bucket ownership; see the overlap block in __init__).
   3. fused AdamW (HIP kernel) updates the fp32 master shard of this rank
      and writes bf16 back into the param shard;
   4. all-gather of param shards into the flat param buffer (one RCCL
@@ -85,7 +85,6 @@ class ZeRO1Optimizer:
         cfg: OptimizerConfig,
         total_train_steps: int = 1000,
         bucket_size: int = 40_000_000,  # 40M elements ≈ 80 MB bf16 buckets
-        overlap_reduce_scatter: bool = True,
     ):
         self.model = model
         self.cfg = cfg
@@ -191,10 +190,6 @@ class ZeRO1Optimizer:
             self.shard_size, dtype=model.flat_param.dtype, device=dev
         )
 
-        self.overlap = overlap_reduce_scatter and self.dp_size > 1
-        self._comm_stream = (
-            torch.cuda.Stream() if (self.overlap and dev.type == "cuda") else None
-        )
         self._grad_views_attached = False
 
     # ------------------------------------------------------------------
