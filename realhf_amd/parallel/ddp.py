@@ -244,6 +244,9 @@ class ZeRO1Optimizer:
 
     def zero_grad(self):
         self.grad_padded.zero_()
+        if self.overlap_comm:
+            self._armed = False  # a failed backward must not leave stale state
+            self._works = []
         if not self._grad_views_attached:
             self.attach_grads()
 
