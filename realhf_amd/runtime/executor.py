@@ -19,6 +19,7 @@ Per MFC and step:
   4. run post-hooks (realloc back / offload).
 """
 import dataclasses
+import os
 import time
 from typing import Any, Dict, List, Optional
 
@@ -161,6 +162,14 @@ class DFGExecutor:
                         "%s on rank %d took %.3fs", mfc.name, self.rank,
                         time.time() - t0,
                     )
+                    if os.environ.get("REALHF_AMD_LOG_MEM") == "1":
+                        # per-MFC memory table (reference: model_worker
+                        # __log_gpu_stats:999)
+                        from realhf_amd.base.monitor import gpu_memory_stats
+
+                        logger.info("%s mem: %s", mfc.name,
+                                    {k: round(v, 2) for k, v in
+                                     gpu_memory_stats().items()})
                     if isinstance(res, SequenceSample):
                         if mfc.output_key_remap:
                             res.remap_keys_(mfc.output_key_remap)
