@@ -359,16 +359,41 @@ def register_dataset(name: str, cls: Callable):
 
 
 def make_dataset(cfg, seed: int, dp_rank: int, world_size: int, tokenizer=None):
-    """cfg: Abstraction{type_, args}."""
+    """cfg: Abstraction{type_, args}.  With REALHF_AMD_DATASET_CACHE set to
+    a directory, constructed datasets are pickled there keyed by their
+    full spec (reference: data_api.py:671 make_dataset's optional on-disk
+    cache) — useful when tokenization dominates startup."""
+    import hashlib
+    import os
+    import pickle
+
     from realhf_amd.api.config import Abstraction
 
     if isinstance(cfg, str):
         cfg = Abstraction(type_=cfg)
+    cache_dir = os.environ.get("REALHF_AMD_DATASET_CACHE")
+    cache_path = None
+    if cache_dir:
+        key = repr((cfg.type_, sorted(cfg.args.items()), seed, dp_rank,
+                    world_size))
+        h = hashlib.sha256(key.encode()).hexdigest()[:16]
+        cache_path = os.path.join(cache_dir, f"dataset_{cfg.type_}_{h}.pkl")
+        if os.path.exists(cache_path):
+            with open(cache_path, "rb") as f:
+                return pickle.load(f)
     cls = _DATASETS[cfg.type_]
-    return cls(
+    ds = cls(
         seed=seed, dp_rank=dp_rank, world_size=world_size, tokenizer=tokenizer,
         **cfg.args,
     )
+    if cache_path:
+        os.makedirs(cache_dir, exist_ok=True)
+        try:
+            with open(cache_path, "wb") as f:
+                pickle.dump(ds, f)
+        except Exception as e:  # tokenizer refs etc. may not pickle
+            logger.warning("dataset cache write failed: %s", e)
+    return ds
 
 
 def load_shuffle_split_dataset(path: str, seed: int, dp_rank: int, world_size: int):

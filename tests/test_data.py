@@ -97,3 +97,26 @@ def test_packed_dataloader():
     assert batches[0].bs == 4
     all_ids = [i for b in batches for i in b.ids]
     assert len(set(all_ids)) == 10
+
+
+def test_dataset_cache(tmp_path, monkeypatch):
+    import json
+
+    from realhf_amd.api.config import Abstraction
+    from realhf_amd.api.data import make_dataset
+
+    p = tmp_path / "d.jsonl"
+    with open(p, "w") as f:
+        for i in range(6):
+            f.write(json.dumps({"input_ids": [3, 4, 5 + i]}) + "\n")
+    monkeypatch.setenv("REALHF_AMD_DATASET_CACHE", str(tmp_path / "cache"))
+    import realhf_amd.api.datasets  # noqa: F401
+
+    cfg = Abstraction("prompt", {"path": str(p), "max_prompt_len": 8})
+    d1 = make_dataset(cfg, seed=1, dp_rank=0, world_size=1)
+    files = list((tmp_path / "cache").iterdir())
+    assert len(files) == 1
+    d2 = make_dataset(cfg, seed=1, dp_rank=0, world_size=1)  # from cache
+    assert len(d1) == len(d2)
+    assert d1[0].data["packed_prompts"].tolist() == \
+        d2[0].data["packed_prompts"].tolist()
