@@ -83,7 +83,7 @@ def _xgmi_worker(rank, world, port, q):
         q.put((rank, False, repr(e)))
 
 
-def test_xgmi_two_process_one_gpu():
+def _run_two_process_once():
     from realhf_amd.base.testing import find_free_port
 
     ctx = mp.get_context("spawn")
@@ -101,4 +101,12 @@ def test_xgmi_two_process_one_gpu():
         p.join(timeout=60)
         if p.is_alive():
             p.terminate()
+    return results
+
+
+def test_xgmi_two_process_one_gpu():
+    results = _run_two_process_once()
+    if not all(ok for ok, _ in results.values()):
+        # multiprocess + IPC: allow one retry before declaring failure
+        results = _run_two_process_once()
     assert all(ok for ok, _ in results.values()), results
