@@ -120,3 +120,43 @@ def test_dataset_cache(tmp_path, monkeypatch):
     assert len(d1) == len(d2)
     assert d1[0].data["packed_prompts"].tolist() == \
         d2[0].data["packed_prompts"].tolist()
+
+
+def test_dataloader_identical_across_replicas(tmp_path):
+    """SPMD critical: every rank builds the same dataset/loader with the
+    same seed and MUST iterate identical batches (ids and order)."""
+    import json
+
+    from realhf_amd.api.config import Abstraction
+    from realhf_amd.api.data import PackedDataLoader, make_dataset
+    import realhf_amd.api.datasets  # noqa: F401
+
+    p = tmp_path / "d.jsonl"
+    with open(p, "w") as f:
+        for i in range(24):
+            f.write(json.dumps({"input_ids": [3, 4, 5 + i % 7]}) + "\n")
+    cfg = Abstraction("prompt", {"path": str(p), "max_prompt_len": 8})
+
+    def batch_ids():
+        ds = make_dataset(cfg, seed=3, dp_rank=0, world_size=1)
+        dl = PackedDataLoader(ds, batch_n_seqs=6, shuffle=True, seed=3)
+        return [tuple(map(str, b.ids)) for b in dl]
+
+    a, b = batch_ids(), batch_ids()
+    assert a == b and len(a) == 4
+
+
+def test_flops_formula_sanity():
+    from realhf_amd.base.monitor import dense_transformer_flops
+
+    # llama-7b-ish: 2 * params * tokens dominates; check the right scale
+    f = dense_transformer_flops(
+        n_layers=32, hidden=4096, intermediate=11008, vocab=32000,
+        n_heads=32, n_kv_heads=32, head_dim=128,
+        total_tokens=1024, sum_sq_seqlens=0.0,
+    )
+    approx = 2 * 6.7e9 * 1024
+    assert 0.8 * approx < f < 1.3 * approx
+    assert dense_transformer_flops(
+        32, 4096, 11008, 32000, 32, 32, 128, 1024, 0.0, backward=True
+    ) == 3 * f
