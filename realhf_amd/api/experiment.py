@@ -114,6 +114,8 @@ class CommonExperimentConfig:
     dataset: DatasetConfig = dataclasses.field(default_factory=DatasetConfig)
     tokenizer_path: Optional[str] = None
     recover_mode: str = "disabled"  # disabled | auto | resume
+    # gang restarts on worker failure (pairs with recover_mode=auto)
+    max_restarts: int = 0
 
 
 @dataclasses.dataclass

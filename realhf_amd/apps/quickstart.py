@@ -115,7 +115,8 @@ def main(argv=None):
         return
     from realhf_amd.scheduler.local import LocalScheduler
 
-    sched = LocalScheduler(cfg.experiment_name, cfg.trial_name)
+    sched = LocalScheduler(cfg.experiment_name, cfg.trial_name,
+                           max_restarts=cfg.max_restarts)
     sched.submit_array(cmd, cfg.n_gpus, master_port=find_free_port())
     sched.wait()
 

@@ -61,3 +61,41 @@ def test_examples_run(tmp_path):
         r = subprocess.run([sys.executable, os.path.join(root, script)],
                            capture_output=True, text=True, timeout=300)
         assert r.returncode == 0, (script, r.stdout[-800:], r.stderr[-800:])
+
+
+def test_scheduler_gang_restart(tmp_path):
+    """A worker that crashes once is gang-restarted and the job completes
+    (elastic recovery; pairs with recover_mode=auto checkpoints)."""
+    import sys
+
+    from realhf_amd.scheduler.local import LocalScheduler
+
+    marker = tmp_path / "crashed_once"
+    script = tmp_path / "w.py"
+    script.write_text(
+        "import os, sys\n"
+        f"m = {str(marker)!r}\n"
+        "if not os.path.exists(m):\n"
+        "    open(m, 'w').close()\n"
+        "    sys.exit(3)\n"  # first attempt: rank crashes
+        "print('ok rank', os.environ.get('RANK'))\n"
+    )
+    sched = LocalScheduler("exp", "t0", max_restarts=2)
+    sched.submit_array([sys.executable, str(script)], n_procs=2)
+    assert sched.wait(timeout=60) == 0
+    assert marker.exists()
+
+
+def test_scheduler_restart_exhausted(tmp_path):
+    import sys
+
+    import pytest as _pytest
+
+    from realhf_amd.scheduler.local import JobException, LocalScheduler
+
+    script = tmp_path / "w.py"
+    script.write_text("import sys; sys.exit(5)\n")
+    sched = LocalScheduler("exp", "t1", max_restarts=1)
+    sched.submit_array([sys.executable, str(script)], n_procs=2)
+    with _pytest.raises(JobException):
+        sched.wait(timeout=60)
