@@ -54,7 +54,7 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
     const bf16* __restrict__ v, const int* __restrict__ cu_seqlens,
     const int* __restrict__ blk_seq, const int* __restrict__ blk_qstart,
     bf16* __restrict__ out, float* __restrict__ lse,
-    int nq, int nkv, float scale, bool causal) {
+    int nq, int nkv, float scale, bool causal, int window) {
   constexpr int HDCH = HD / 32;  // mfma K-chunks over head_dim
   const int blk = blockIdx.x;
   const int qh = blockIdx.y;
@@ -100,8 +100,15 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
 
   // causal: keys needed up to q0_local + QBLK - 1 (inclusive); else all L
   const int kv_end = causal ? min(L, q0_local + QBLK) : L;
+  // sliding window (mistral): query qrow sees keys (qrow-window, qrow];
+  // the workgroup's earliest needed key is q0_local - window + 1
+  int kv_begin = 0;
+  if (window > 0) {
+    kv_begin = q0_local - window + 1;
+    kv_begin = (kv_begin > 0) ? (kv_begin / KVBLK) * KVBLK : 0;
+  }
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
+  for (int kv0 = kv_begin; kv0 < kv_end; kv0 += KVBLK) {
     const int kchunk = min(KVBLK, kv_end - kv0);
     // ---- stage K and V(T) tiles ------------------------------------
     __syncthreads();
@@ -143,7 +150,8 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
       #pragma unroll
       for (int ks = 0; ks < 4; ks++) {
         int kidx = kv0 + ks * 16 + i16;
-        bool ok = (kidx < kv_end) && (!causal || kidx <= qrow) && (qrow < L);
+        bool ok = (kidx < kv_end) && (!causal || kidx <= qrow) && (qrow < L)
+                  && (window <= 0 || kidx > qrow - window);
         float sv = ok ? s_sub[ks][r] * scale : -1e30f;
         s_sub[ks][r] = sv;
         mx = fmaxf(mx, sv);
@@ -217,7 +225,8 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
 
 std::vector<torch::Tensor> attn_varlen_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
-    torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale) {
+    torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale,
+    long window) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16,
               "attn_varlen_fwd: bf16 only");
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
@@ -249,13 +258,13 @@ std::vector<torch::Tensor> attn_varlen_fwd(
       cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
       (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
       bseq_t.data_ptr<int>(), bqs_t.data_ptr<int>(), (bf16*)out.data_ptr(),
-      lse.data_ptr<float>(), nq, nkv, (float)scale, causal);
+      lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
   } else if (hd == 64) {
     hipLaunchKernelGGL((attn_varlen_fwd_kernel<64>), grid, dim3(64 * AV_WAVES), 0,
       cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
       (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
       bseq_t.data_ptr<int>(), bqs_t.data_ptr<int>(), (bf16*)out.data_ptr(),
-      lse.data_ptr<float>(), nq, nkv, (float)scale, causal);
+      lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
   } else {
     TORCH_CHECK(false, "unsupported head_dim ", hd);
   }

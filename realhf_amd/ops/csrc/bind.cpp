@@ -24,7 +24,8 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
                           torch::Tensor cache_seqlens, double scale);
 std::vector<torch::Tensor> attn_varlen_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
-    torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale);
+    torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale,
+    long window);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor tr16_probe(torch::Tensor addr_elem);
 torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,

@@ -280,17 +280,21 @@ def _attn_varlen_blocked_torch(q, k, v, cu_seqlens, causal, scale, window=None):
 
 class _AttnVarlenFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, cu_seqlens, max_seqlen, causal, scale):
+    def forward(ctx, q, k, v, cu_seqlens, max_seqlen, causal, scale,
+                window=None):
         C = _ops.require_hip()
-        out, lse = C.attn_varlen_fwd(q, k, v, cu_seqlens, int(max_seqlen), causal, scale)
+        out, lse = C.attn_varlen_fwd(q, k, v, cu_seqlens, int(max_seqlen),
+                                     causal, scale, int(window or 0))
         ctx.save_for_backward(q, k, v, out, lse, cu_seqlens)
         ctx.causal, ctx.scale, ctx.max_seqlen = causal, scale, max_seqlen
+        ctx.window = window
         return out
 
     @staticmethod
     def backward(ctx, grad_out):
         q, k, v, out, lse, cu_seqlens = ctx.saved_tensors
-        if os.environ.get("REALHF_AMD_NO_HIP_ATTN_BWD") != "1":
+        if (os.environ.get("REALHF_AMD_NO_HIP_ATTN_BWD") != "1"
+                and ctx.window is None):
             # hand-written MFMA backward (attn_bwd.hip): kv-stationary,
             # dQ via fp32 atomics, per-q-head dK/dV reduced here for GQA.
             # In-context A/B: 3.51 vs 3.19 samples/s over the padded
@@ -308,17 +312,18 @@ class _AttnVarlenFn(torch.autograd.Function):
                 dk32 = dk32.view(t, nkv, rep, -1).sum(2)
                 dv32 = dv32.view(t, nkv, rep, -1).sum(2)
             return (dq32.to(q.dtype), dk32.to(k.dtype), dv32.to(v.dtype),
-                    None, None, None, None)
+                    None, None, None, None, None)
         # Fallback: blocked recompute with rocBLAS GEMMs (fp32 softmax).
         with torch.enable_grad():
             qg = q.detach().requires_grad_(True)
             kg = k.detach().requires_grad_(True)
             vg = v.detach().requires_grad_(True)
             ref = _attn_varlen_blocked_torch(
-                qg, kg, vg, cu_seqlens, ctx.causal, ctx.scale
+                qg, kg, vg, cu_seqlens, ctx.causal, ctx.scale,
+                window=ctx.window,
             )
             dq, dk, dv = torch.autograd.grad(ref, (qg, kg, vg), grad_out)
-        return dq, dk, dv, None, None, None, None
+        return dq, dk, dv, None, None, None, None, None
 
 
 def attn_varlen(q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=None,
@@ -330,11 +335,11 @@ def attn_varlen(q, k, v, cu_seqlens, max_seqlen, causal=True, softmax_scale=None
         _ops.use_hip(q)
         and q.dtype == torch.bfloat16
         and q.shape[-1] in (64, 128)
-        and window is None
+        and (window is None or causal)
     ):
         return _AttnVarlenFn.apply(
             q.contiguous(), k.contiguous(), v.contiguous(),
-            cu_seqlens, max_seqlen, causal, scale,
+            cu_seqlens, max_seqlen, causal, scale, window,
         )
     if q.is_cuda:
         # odd head dims / dtypes / binding sliding window: batched rocBLAS

@@ -147,7 +147,7 @@ def test_attn_varlen_fwd(nq, nkv, lens):
     k = torch.randn(total, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
     v = torch.randn(total, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
     scale = 1.0 / np.sqrt(hd)
-    out, lse = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale)
+    out, lse = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale, 0)
     ref = F.attn_varlen_ref(
         q.float().cpu(), k.float().cpu(), v.float().cpu(), cu.cpu(), True, scale
     )
@@ -428,7 +428,7 @@ def test_attn_varlen_bwd_hip_matches_ref():
             ref, (qr, kr, vr), dout.float())
 
         # HIP fwd for lse, then HIP bwd
-        out, lse = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale)
+        out, lse = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale, 0)
         dsum = (dout.float() * out.float()).sum(-1)
         dq32, dk32, dv32 = C.attn_varlen_bwd(q, k, v, dout.contiguous(),
                                              lse, dsum, cu, True, scale)
@@ -439,3 +439,30 @@ def test_attn_varlen_bwd_hip_matches_ref():
         torch.testing.assert_close(dv32, ref_dv, atol=0.15, rtol=5e-2)
         torch.testing.assert_close(dk32, ref_dk, atol=0.15, rtol=5e-2)
         torch.testing.assert_close(dq32, ref_dq, atol=0.15, rtol=5e-2)
+
+
+def test_attn_varlen_fwd_sliding_window():
+    """MFMA forward with a binding sliding window vs the fp32 oracle."""
+    from realhf_amd.ops import functional as F
+
+    torch.manual_seed(23)
+    nq = nkv = 8
+    hd = 128
+    for window in (32, 100):
+        lens = [96, 64, 200, 40]
+        cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                          dtype=torch.int32, device="cuda")
+        total = sum(lens)
+        q = (torch.randn(total, nq, hd, device="cuda") * 0.3).to(torch.bfloat16)
+        k = (torch.randn(total, nkv, hd, device="cuda") * 0.3).to(torch.bfloat16)
+        v = (torch.randn(total, nkv, hd, device="cuda") * 0.3).to(torch.bfloat16)
+        scale = hd ** -0.5
+        out, _ = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale, window)
+        ref = F.attn_varlen_ref(q.float().cpu(), k.float().cpu(),
+                                v.float().cpu(), cu.cpu(), True, scale,
+                                window=window)
+        torch.testing.assert_close(out.float().cpu(), ref, atol=0.05,
+                                   rtol=3e-2)
+        # and the window must differ from full causal
+        out_full, _ = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale, 0)
+        assert not torch.allclose(out, out_full)

@@ -30,7 +30,7 @@ def run(shape_name, lens, nq, nkv, hd=128, iters=30):
     k = (torch.randn(total, nkv, hd, device="cuda") * 0.5).to(torch.bfloat16)
     v = (torch.randn(total, nkv, hd, device="cuda") * 0.5).to(torch.bfloat16)
     scale = hd ** -0.5
-    us = bench(lambda: C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale),
+    us = bench(lambda: C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale, 0),
                iters)
     flops = sum(2 * 2 * (l * l / 2) * nq * hd for l in lens)
     print(f"{shape_name:24} {us:9.1f} us  {flops / us / 1e6:7.1f} TF/s")
