@@ -337,3 +337,29 @@ def test_gen_experiment_single_process(tmp_path):
     cfg.exp_ctrl.benchmark_steps = 1
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
     Trainer(cfg).run()
+
+
+def test_sft_training_deterministic(tmp_path):
+    """Two identical runs produce identical losses — the SPMD design
+    depends on every rank computing bit-identical programs."""
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "sft.jsonl")
+    _write_sft_data(data, n=16)
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+
+    def run(trial):
+        cfg = SFTConfig(experiment_name="t-det", trial_name=trial, n_gpus=1)
+        cfg.model.dtype = "float32"
+        cfg.dataset.type_ = "prompt_answer"
+        cfg.dataset.path = data
+        cfg.dataset.train_bs_n_seqs = 4
+        cfg.exp_ctrl.benchmark_steps = 2
+        t = Trainer(cfg)
+        t.run()
+        name = t.built.trainable[0]
+        return t.models[name].module.module.flat_param.clone()
+
+    a, b = run("a"), run("b")
+    assert torch.equal(a, b)
