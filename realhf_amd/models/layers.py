@@ -202,8 +202,6 @@ class ReaLModelBlock(nn.Module):
             and not cfg.qk_layernorm
             and self.hd in (64, 128)
             and _os.environ.get("REALHF_AMD_NO_FUSED_DECODE") != "1"
-            and (cfg.sliding_window is None
-                 or k_cache.shape[1] <= cfg.sliding_window)
             and not self.lora  # adapters applied in the standard path
         ):
             from realhf_amd import ops as _ops_pkg
@@ -236,7 +234,8 @@ class ReaLModelBlock(nn.Module):
                 qkv_raw, bias, k_cache, v_cache,
                 cache_seqlens, cos, sin, self.nq, cfg.apply_rotary,
             )
-            attn_out = ops.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
+            attn_out = ops.attn_decode(q, k_cache, v_cache, cache_seqlens,
+                                       scale, window=cfg.sliding_window)
             attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
             wo = self.p[f"{i}.attn.wo.weight"]
             if (

@@ -29,7 +29,8 @@ template <int HD, int REP, int W>
 __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
     const bf16* __restrict__ vc, const int* __restrict__ cache_seqlens,
-    bf16* __restrict__ out, int bs, int nq, int nkv, long maxlen, float scale) {
+    bf16* __restrict__ out, int bs, int nq, int nkv, long maxlen, float scale,
+    int window) {
   constexpr int DPL = HD / WAVE;  // dims per lane (2 for hd=128)
   const int b = blockIdx.x / nkv;
   const int kvh = blockIdx.x % nkv;
@@ -61,9 +62,13 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
   const bf16* kb = kc + (long)b * maxlen * kv_stride + (long)kvh * HD;
   const bf16* vb = vc + (long)b * maxlen * kv_stride + (long)kvh * HD;
 
-  for (int base = w * WAVE; base < L; base += W * WAVE) {
+  // sliding window (mistral): only keys in [L - window, L) attend
+  const int lo = (window > 0 && L > window) ? (L - window) : 0;
+  const int origin = (lo / (W * WAVE)) * (W * WAVE);
+
+  for (int base = origin + w * WAVE; base < L; base += W * WAVE) {
     const int l = base + lane;
-    const bool valid = l < L;
+    const bool valid = l < L && l >= lo;
     float sc[REP];
     #pragma unroll
     for (int r = 0; r < REP; r++) sc[r] = 0.f;
@@ -153,7 +158,8 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
 }
 
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
-                          torch::Tensor cache_seqlens, double scale) {
+                          torch::Tensor cache_seqlens, double scale,
+                          long window) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.is_contiguous() && kc.is_contiguous() && vc.is_contiguous());
   int bs = q.size(0), nq = q.size(1), hd = q.size(2);
@@ -174,7 +180,7 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
         grid, dim3(64 * w_c.value), 0, cur_stream(), (const bf16*)q.data_ptr(),
         (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
         cs.data_ptr<int>(), (bf16*)out.data_ptr(), bs, nq, nkv, maxlen,
-        (float)scale);
+        (float)scale, (int)window);
     };
     if (dec_waves == 2) go(std::integral_constant<int, 2>{});
     else if (dec_waves == 8) go(std::integral_constant<int, 8>{});

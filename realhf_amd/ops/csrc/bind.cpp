@@ -21,7 +21,8 @@ void fused_adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                  double b2, double eps, double wd, long step, double gscale,
                  bool write_bf16);
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
-                          torch::Tensor cache_seqlens, double scale);
+                          torch::Tensor cache_seqlens, double scale,
+                          long window);
 std::vector<torch::Tensor> attn_varlen_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale,

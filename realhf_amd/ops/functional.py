@@ -390,10 +390,10 @@ def attn_decode(q, k_cache, v_cache, cache_seqlens, softmax_scale=None,
         and q.dtype == torch.bfloat16
         and q.shape[-1] in (64, 128)
         and q.shape[1] // k_cache.shape[2] in (1, 2, 4, 8)
-        and window is None
     ):
         C = _ops.require_hip()
-        return C.attn_decode(q, k_cache, v_cache, cache_seqlens, scale)
+        return C.attn_decode(q, k_cache, v_cache, cache_seqlens, scale,
+                             int(window or 0))
     return attn_decode_ref(q, k_cache, v_cache, cache_seqlens, scale,
                            window=window)
 

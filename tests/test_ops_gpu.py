@@ -164,7 +164,7 @@ def test_attn_decode(nq, nkv, hd):
     kc = torch.randn(bs, maxlen, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
     vc = torch.randn(bs, maxlen, nkv, hd, dtype=torch.bfloat16, device="cuda") * 0.5
     scale = 1.0 / np.sqrt(hd)
-    out = C.attn_decode(q, kc, vc, lens, scale)
+    out = C.attn_decode(q, kc, vc, lens, scale, 0)
     ref = F.attn_decode_ref(
         q.float().cpu(), kc.float().cpu(), vc.float().cpu(), lens.cpu(), scale
     )
@@ -466,3 +466,24 @@ def test_attn_varlen_fwd_sliding_window():
         # and the window must differ from full causal
         out_full, _ = C.attn_varlen_fwd(q, k, v, cu, max(lens), True, scale, 0)
         assert not torch.allclose(out, out_full)
+
+
+def test_attn_decode_sliding_window():
+    from realhf_amd.ops import functional as F
+
+    torch.manual_seed(25)
+    bs, nq, nkv, hd, maxlen = 4, 8, 8, 128, 96
+    q = (torch.randn(bs, nq, hd, device="cuda") * 0.3).to(torch.bfloat16)
+    kc = (torch.randn(bs, maxlen, nkv, hd, device="cuda") * 0.3).to(torch.bfloat16)
+    vc = (torch.randn(bs, maxlen, nkv, hd, device="cuda") * 0.3).to(torch.bfloat16)
+    lens = torch.tensor([96, 40, 7, 64], dtype=torch.int32, device="cuda")
+    scale = hd ** -0.5
+    for window in (16, 48):
+        out = C.attn_decode(q, kc, vc, lens, scale, window)
+        ref = F.attn_decode_ref(q.float().cpu(), kc.float().cpu(),
+                                vc.float().cpu(), lens.cpu(), scale,
+                                window=window)
+        torch.testing.assert_close(out.float().cpu(), ref, atol=0.05,
+                                   rtol=3e-2)
+    full = C.attn_decode(q, kc, vc, lens, scale, 0)
+    assert not torch.allclose(C.attn_decode(q, kc, vc, lens, scale, 16), full)

@@ -33,7 +33,7 @@ def main():
     print(f"{'ctx':>5} {'us':>8} {'TB/s':>6}")
     for ctx in (128, 256, 384, 512, 640):
         cs = torch.full((bs,), ctx, dtype=torch.int32, device="cuda")
-        t = bench(lambda: C.attn_decode(q, kc, vc, cs, scale))
+        t = bench(lambda: C.attn_decode(q, kc, vc, cs, scale, 0))
         bytes_ = bs * nkv * ctx * hd * 2 * 2
         print(f"{ctx:5d} {t:8.1f} {bytes_ / t / 1e6:6.2f}")
 
