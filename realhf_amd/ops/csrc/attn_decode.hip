@@ -71,6 +71,13 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
     float sc[REP];
     #pragma unroll
     for (int r = 0; r < REP; r++) sc[r] = 0.f;
+    if (!valid) {
+      // windowed: front-invalid lanes are INSIDE the PV j-range, so
+      // their V tile rows must be zeros (0 * uninitialized LDS = NaN)
+      #pragma unroll
+      for (int i = 0; i < HD / 8; i++)
+        *(short8*)(&v_s[w][lane][i * 8]) = short8{};
+    }
     if (valid) {
       const bf16* krow = kb + (long)l * kv_stride;
       const bf16* vrow = vb + (long)l * kv_stride;
