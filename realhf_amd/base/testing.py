@@ -65,18 +65,38 @@ class LocalMultiProcessTest:
         self.world_size = world_size
         self.backend = backend
         self.timeout = timeout_secs
+        self._spawn = (fn, args, kwargs)
+        self._make_procs()
+
+    def _make_procs(self):
+        fn, args, kwargs = self._spawn
         ctx = mp.get_context("spawn")
         self.errq = ctx.Queue()
         port = find_free_port()
         self.procs = [
             ctx.Process(
                 target=_proc_entry,
-                args=(r, world_size, port, backend, fn, args, kwargs, self.errq),
+                args=(r, self.world_size, port, self.backend, fn, args,
+                      kwargs, self.errq),
             )
-            for r in range(world_size)
+            for r in range(self.world_size)
         ]
 
     def launch(self):
+        errs = self._launch_once()
+        if errs and all(
+            "rendezvous" in tb or "EADDRINUSE" in tb
+            or "Address already in use" in tb or "Connection" in tb
+            for _, tb in errs
+        ):
+            # one retry on a fresh port: rendezvous races with TIME_WAIT
+            # sockets of the previous test are a known flake source
+            self._make_procs()
+            errs = self._launch_once()
+        if errs:
+            raise RuntimeError("\n".join(f"[rank {r}]\n{tb}" for r, tb in errs))
+
+    def _launch_once(self):
         for p in self.procs:
             p.start()
         for p in self.procs:
@@ -88,8 +108,7 @@ class LocalMultiProcessTest:
             if p.is_alive():
                 p.terminate()
                 errs.append((p.name, "timeout"))
-        if errs:
-            raise RuntimeError("\n".join(f"[rank {r}]\n{tb}" for r, tb in errs))
+        return errs
 
 
 def find_free_port() -> int:
