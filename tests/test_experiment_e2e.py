@@ -340,10 +340,15 @@ def test_gen_experiment_single_process(tmp_path):
 
 
 def test_sft_training_deterministic(tmp_path):
-    """Two identical runs produce identical losses — the SPMD design
-    depends on every rank computing bit-identical programs."""
+    """Two identical runs produce identical params — the SPMD design
+    depends on every rank computing identical programs.  Single-threaded:
+    multi-threaded CPU reductions are not bitwise-stable under load
+    (dynamic OMP scheduling), which is orthogonal to what this guards."""
     from realhf_amd.api.experiment import SFTConfig
     from realhf_amd.runtime.trainer import Trainer
+
+    n_threads = torch.get_num_threads()
+    torch.set_num_threads(1)
 
     data = str(tmp_path / "sft.jsonl")
     _write_sft_data(data, n=16)
@@ -361,5 +366,8 @@ def test_sft_training_deterministic(tmp_path):
         name = t.built.trainable[0]
         return t.models[name].module.module.flat_param.clone()
 
-    a, b = run("a"), run("b")
+    try:
+        a, b = run("a"), run("b")
+    finally:
+        torch.set_num_threads(n_threads)
     assert torch.equal(a, b)
