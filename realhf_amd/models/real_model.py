@@ -75,7 +75,11 @@ class ReaLModel(nn.Module):
         return self.config.n_layers + 2
 
     def instantiate(self):
-        self.flat_param = torch.empty(
+        # zeros, not empty: the layout's 64-element alignment gaps are
+        # never written by loads/inits, and garbage there makes whole-
+        # buffer checksums/comparisons (bench DP check, realloc tests,
+        # determinism tests) allocator-history-dependent
+        self.flat_param = torch.zeros(
             self.layout.total_numel, dtype=self.dtype, device=self.device
         )
         self._map_params()
