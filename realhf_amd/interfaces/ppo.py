@@ -274,6 +274,9 @@ class PPOActorInterface(ModelInterface):
 class PPOCriticInterface(ModelInterface):
     n_minibatches: int = 4
     kl_ctl: float = 0.1
+    adaptive_kl_ctl: bool = False
+    adaptive_kl_target: float = 6.0
+    adaptive_kl_horizon: float = 10000.0
     value_eps_clip: float = 0.2
     max_reward_clip: float = 20.0
     discount: float = 1.0
@@ -283,7 +286,15 @@ class PPOCriticInterface(ModelInterface):
 
     def __post_init__(self):
         self._rms = ppo_math.ExponentialRunningMeanStd() if self.value_norm else None
-        self._kl_ctl = ppo_math.FixedKLController(self.kl_ctl)
+        # The critic must adapt its KL coefficient in lockstep with the actor
+        # (reference ppo_interface.py:658) — otherwise value targets are
+        # computed from a stale kl_ctl while the actor's rewards drift.
+        if self.adaptive_kl_ctl:
+            self._kl_ctl = ppo_math.AdaptiveKLController(
+                self.kl_ctl, self.adaptive_kl_target, self.adaptive_kl_horizon
+            )
+        else:
+            self._kl_ctl = ppo_math.FixedKLController(self.kl_ctl)
 
     @torch.no_grad()
     def inference(self, model: Model, data: SequenceSample, n_mbs=None):
@@ -345,6 +356,13 @@ class PPOCriticInterface(ModelInterface):
             for k, v in stats.items():
                 all_stats[k] = all_stats.get(k, 0.0) + v / len(mbs)
         model.inc_version()
+        with torch.no_grad():
+            m = prep["loss_mask"].float()
+            n = m.sum().clamp(min=1)
+            kl = float(
+                (-prep["kl_rewards"] * m).sum() / n / max(self._kl_ctl.value, 1e-8)
+            )
+        self._kl_ctl.update(kl, n_steps=int(data.bs))
         all_stats["returns_mean"] = float(returns.mean())
         return all_stats
 

@@ -88,15 +88,21 @@ def get_packed_rewards(
     logprobs: torch.Tensor,  # [total - bs]
     ref_logprobs: torch.Tensor,  # [total - bs]
     short_cu_seqlens: torch.Tensor,  # [bs+1] over the (len-1) reward positions
-    seq_no_eos_mask: torch.Tensor,  # [bs] — True if no EOS (score not applied... still applied at end)
+    seq_no_eos_mask: torch.Tensor,  # [bs] — True if the sequence was truncated (no EOS)
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """KL-penalty token rewards + clipped score at the last position
-    (reference: ppo_functional.py:291)."""
+    (reference: ppo_functional.py:291).
+
+    Truncated (no-EOS) sequences get NO score at the end: they already
+    bootstrap from the last value in GAE, so adding the score would
+    double-count reward and bias the policy against emitting EOS
+    (reference: ppo_functional.py:306 torch.where(seq_no_eos_mask, 0, score)).
+    """
     kl_rewards = -kl_ctl * (logprobs - ref_logprobs)
     rewards = kl_rewards.clone()
     ends = short_cu_seqlens[1:].long() - 1
     sc = score.clamp(-clip_reward_value, clip_reward_value)
-    rewards[ends] += sc
+    rewards[ends] += torch.where(seq_no_eos_mask.bool(), torch.zeros_like(sc), sc)
     return kl_rewards, rewards
 
 

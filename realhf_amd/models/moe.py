@@ -14,6 +14,7 @@ Expert GEMMs run through the hand-written MFMA grouped-GEMM kernel
 path uses per-expert GEMMs (rocBLAS) so autograd handles the backward
 (hand-written grouped bwd is a noted follow-up).
 """
+import contextlib
 from typing import Dict, List
 
 import torch
@@ -24,8 +25,29 @@ from realhf_amd.api.model import ReaLModelConfig
 from realhf_amd.base import constants
 from realhf_amd.parallel import mappings
 
-# aux losses collected during forward; training interfaces drain this.
+# Aux losses collected during the TRAINING forward only; the training
+# interface's loss_fn drains this.  Collection is explicitly scoped (engine
+# wraps the train forward in aux_loss_collection()) so that no-grad
+# inference/generation and gradient-checkpoint RECOMPUTE (which re-runs the
+# forward during backward) never pollute the list for the next minibatch.
 _AUX_LOSSES: List[torch.Tensor] = []
+_COLLECT_AUX = False
+
+
+@contextlib.contextmanager
+def aux_loss_collection():
+    global _COLLECT_AUX, _AUX_LOSSES
+    prev = _COLLECT_AUX
+    _AUX_LOSSES = []
+    _COLLECT_AUX = True
+    try:
+        yield
+    finally:
+        _COLLECT_AUX = prev
+
+
+def _collecting() -> bool:
+    return _COLLECT_AUX and torch.is_grad_enabled()
 
 
 def pop_aux_losses() -> List[torch.Tensor]:
@@ -81,13 +103,13 @@ class TopKRouter(torch.nn.Module):
             scores, idx = torch.topk(probs, moe.top_k, dim=-1)
             if moe.norm_topk_prob:
                 scores = scores / scores.sum(dim=-1, keepdim=True)
-            if moe.routing_type == "aux_loss" and self.training:
+            if moe.routing_type == "aux_loss" and _collecting():
                 counts = torch.bincount(
                     idx.flatten(), minlength=moe.num_experts
                 )
                 aux = switch_load_balancing_loss(probs, counts, moe.top_k)
                 _AUX_LOSSES.append(moe.aux_loss_coef * aux)
-            if moe.z_loss_coef > 0 and self.training:
+            if moe.z_loss_coef > 0 and _collecting():
                 z = torch.logsumexp(logits, dim=-1).square().mean()
                 _AUX_LOSSES.append(moe.z_loss_coef * z)
         return scores, idx

@@ -19,6 +19,7 @@ import torch.distributed as dist
 from realhf_amd.api.data import SequenceSample
 from realhf_amd.api.model import GenerationHyperparameters, PipelinableEngine
 from realhf_amd.base import constants, logging, seeding
+from realhf_amd.models import moe as moe_mod
 from realhf_amd.runtime.engine import sample_to_packed
 
 logger = logging.getLogger("pp")
@@ -90,6 +91,7 @@ class PipelinedEngine(PipelinableEngine):
     def forward(self, input_: SequenceSample, n_mbs: Optional[int] = None,
                 post_hook: Optional[Callable] = None,
                 aggregate_fn: Callable = None):
+        self.model.eval()
         g = self._grid()
         n_mbs = n_mbs or g.pp_size
         n_mbs = min(n_mbs, input_.bs)
@@ -122,6 +124,7 @@ class PipelinedEngine(PipelinableEngine):
     def train_batch(self, input_: SequenceSample, loss_fn: Callable,
                     version_steps: int = 0, n_mbs: Optional[int] = None) -> Dict:
         g = self._grid()
+        self.model.train()
         self.optimizer.zero_grad()
         n_mbs = n_mbs or self.n_mbs_train or (2 * g.pp_size)
         n_mbs = max(min(n_mbs, input_.bs), 1)
@@ -145,22 +148,31 @@ class PipelinedEngine(PipelinableEngine):
             if not self.model.is_first_stage:
                 hidden = _recv((total, h_dim), dtype, g.pp_prev_global_rank(), dev)
                 hidden.requires_grad_(True)
-            with torch.enable_grad():
+            with torch.enable_grad(), moe_mod.aux_loss_collection():
                 out = self._stage_fwd(mb, hidden)
+                aux = moe_mod.pop_aux_losses()
             if not self.model.is_last_stage:
                 _send(out.detach(), g.pp_next_global_rank())
-            fwd_state[i] = (hidden, out, mb)
+            fwd_state[i] = (hidden, out, mb, aux)
 
         def do_bwd(i):
-            hidden, out, mb = fwd_state.pop(i)
+            hidden, out, mb, aux = fwd_state.pop(i)
             if self.model.is_last_stage:
                 loss, st = loss_fn(out, mb)
+                for a in aux:
+                    loss = loss + a
                 (loss / n_mbs).backward()
                 for k, v in st.items():
                     stats[k] = stats.get(k, 0.0) + float(v) / n_mbs
             else:
                 gout = _recv(tuple(out.shape), dtype, g.pp_next_global_rank(), dev)
-                torch.autograd.backward(out, grad_tensors=gout)
+                if aux:
+                    # mid-stage MoE aux losses backprop locally alongside
+                    # the received activation grads
+                    s = torch.stack(aux).sum() / n_mbs
+                    torch.autograd.backward([out, s], grad_tensors=[gout, None])
+                else:
+                    torch.autograd.backward(out, grad_tensors=gout)
             if hidden is not None:
                 _send(hidden.grad, g.pp_prev_global_rank())
 
@@ -200,6 +212,7 @@ class PipelinedEngine(PipelinableEngine):
         from realhf_amd.models import generation as genmod
         from realhf_amd.parallel import mappings
 
+        self.model.eval()
         g = self._grid()
         gconfig = gconfig or GenerationHyperparameters()
         cfg = self.model.config

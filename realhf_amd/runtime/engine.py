@@ -24,6 +24,7 @@ from realhf_amd.api.model import (
 )
 from realhf_amd.base import constants, logging, seeding
 from realhf_amd.models import generation as genmod
+from realhf_amd.models import moe as moe_mod
 from realhf_amd.parallel.ddp import OptimizerConfig, ZeRO1Optimizer
 
 logger = logging.getLogger("engine")
@@ -58,13 +59,22 @@ class PipelinableTrainEngine(PipelinableEngine):
 
     def train_batch(self, input_: SequenceSample, loss_fn: Callable,
                     version_steps: int = 0, n_mbs: Optional[int] = None) -> Dict:
+        self.model.train()
         self.optimizer.zero_grad()
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 and input_.bs >= n_mbs else [input_]
         stats: Dict[str, float] = {}
         for i, mb in enumerate(mbs):
             ids, cu, mx = sample_to_packed(mb)
-            out = self.model(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=mx)
-            loss, st = loss_fn(out, mb)
+            # MoE aux losses are collected only inside this scope — never
+            # during inference or gradient-checkpoint recompute in backward.
+            with moe_mod.aux_loss_collection():
+                out = self.model(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=mx)
+                loss, st = loss_fn(out, mb)
+                # interfaces that know about MoE pop aux losses themselves
+                # inside loss_fn; anything left over is added here so plain
+                # interfaces (SFT/DPO/...) still train the router
+                for aux in moe_mod.pop_aux_losses():
+                    loss = loss + aux
             if i == len(mbs) - 1 and hasattr(self.optimizer, "arm_overlap"):
                 # bucketed reduce-scatter overlaps this (last) backward
                 self.optimizer.arm_overlap()
@@ -77,6 +87,7 @@ class PipelinableTrainEngine(PipelinableEngine):
 
     @torch.no_grad()
     def eval_batch(self, input_: SequenceSample, loss_fn: Callable, n_mbs=None):
+        self.model.eval()
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 else [input_]
         stats: Dict[str, float] = {}
         for mb in mbs:
@@ -91,6 +102,7 @@ class PipelinableTrainEngine(PipelinableEngine):
     def forward(self, input_: SequenceSample, n_mbs: Optional[int] = None,
                 post_hook: Optional[Callable] = None,
                 aggregate_fn: Callable = _default_aggregate):
+        self.model.eval()
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 and input_.bs >= n_mbs else [input_]
         outs = []
         for mb in mbs:
@@ -105,6 +117,7 @@ class PipelinableTrainEngine(PipelinableEngine):
     def generate(self, input_: SequenceSample, tokenizer=None,
                  gconfig: Optional[GenerationHyperparameters] = None,
                  n_mbs: Optional[int] = None, **gen_kw):
+        self.model.eval()
         gconfig = gconfig or GenerationHyperparameters()
         key = "packed_prompts" if "packed_prompts" in input_.keys else "packed_input_ids"
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 and input_.bs >= n_mbs else [input_]

@@ -267,6 +267,7 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
     critic_iface = Abstraction("ppo_critic", {
         "n_minibatches": ppo.ppo_n_minibatches,
         "value_eps_clip": ppo.value_eps_clip, "kl_ctl": ppo.kl_ctl,
+        "adaptive_kl_ctl": ppo.use_adaptive_kl_ctl,
         "max_reward_clip": ppo.max_reward_clip, "discount": ppo.discount,
         "gae_lambda": ppo.gae_lambda, "value_norm": ppo.value_norm,
     })
@@ -360,6 +361,9 @@ class Trainer:
             "cuda", int(os.environ.get("LOCAL_RANK", 0))
         ) if torch.cuda.is_available() else torch.device("cpu")
         self.tokenizer = self._load_tokenizer()
+        # The dataloader is built BEFORE the models so FinetuneSpec sees the
+        # real dataset length (LR warmup/decay horizon depends on it).
+        self.train_dl = self._build_dataloader()
         self.models = self._build_models()
         self.executor = DFGExecutor(
             self.built.graph, self.built.allocations, self.models,
@@ -442,7 +446,7 @@ class Trainer:
                 )
                 spec = FinetuneSpec(
                     self.cfg.exp_ctrl.total_train_epochs,
-                    10_000,
+                    len(self.train_dl.dataset),
                     self.cfg.dataset.train_bs_n_seqs,
                 )
                 if name in self.built.trainable:
@@ -512,7 +516,7 @@ class Trainer:
     # ----------------------------------------------------------------- run
     def run(self):
         cfg = self.cfg
-        dl = self._build_dataloader()
+        dl = self.train_dl
         ctrl = cfg.exp_ctrl
         bench_t0 = None
         recover = self._maybe_load_recover()
@@ -608,7 +612,8 @@ class Trainer:
         if self.cfg.recover_mode == "disabled":
             return
         state = {"models": {}, "epoch": epoch, "step": step,
-                 "global_step": self.global_step}
+                 "global_step": self.global_step,
+                 "interfaces": self._interface_states()}
         for name in self.built.trainable:
             if name not in self.models:
                 continue
@@ -627,6 +632,35 @@ class Trainer:
         os.makedirs(os.path.dirname(self._recover_ckpt_path()), exist_ok=True)
         torch.save(state, self._recover_ckpt_path())
         self._save_recover_info(epoch, step)
+
+    def _interface_states(self):
+        """Algorithm state living on interfaces (value-norm running stats,
+        adaptive KL controller value) — without it, elastic restart silently
+        resets value normalization to mean 0/std 1 and kl_ctl to init."""
+        out = {}
+        for mfc_name, iface in self.built.interfaces.items():
+            st = {}
+            kl = getattr(iface, "_kl_ctl", None)
+            if kl is not None:
+                st["kl_ctl_value"] = kl.value
+            rms = getattr(iface, "_rms", None)
+            if rms is not None:
+                st["value_norm"] = rms.state_dict()
+            if st:
+                out[mfc_name] = st
+        return out
+
+    def _restore_interface_states(self, states):
+        for mfc_name, st in (states or {}).items():
+            iface = self.built.interfaces.get(mfc_name)
+            if iface is None:
+                continue
+            kl = getattr(iface, "_kl_ctl", None)
+            if kl is not None and "kl_ctl_value" in st:
+                kl.value = st["kl_ctl_value"]
+            rms = getattr(iface, "_rms", None)
+            if rms is not None and "value_norm" in st:
+                rms.load_state_dict(st["value_norm"])
 
     def _maybe_load_recover(self):
         if self.cfg.recover_mode not in ("auto", "resume"):
@@ -653,6 +687,7 @@ class Trainer:
                 opt = getattr(eng, "optimizer", None)
                 if opt is not None and st["optimizer"] is not None:
                     opt.load_state_dict(st["optimizer"])
+            self._restore_interface_states(state.get("interfaces"))
             logger.info("restored model+optimizer state from %s", ckpt)
         logger.info("recovering from %s", info)
         return (info["epoch"], info["step"])
