@@ -59,6 +59,10 @@ class MFCDef:
     n_mbs: Optional[int] = None
     balanced_dp: bool = False
     log_return_value: bool = False
+    # scheduling hint: lower runs earlier among ready MFCs — used to place
+    # whole-mesh MFCs before disjoint-mesh concurrent groups so sub-mesh
+    # ranks reach their own MFC without waiting (executor program order)
+    priority: int = 0
     # hooks attached by the experiment planner
     pre_hooks: List[Any] = dataclasses.field(default_factory=list)
     post_hooks: List[Any] = dataclasses.field(default_factory=list)
@@ -103,13 +107,14 @@ class DFG:
                 indeg[c.name] += 1
         from collections import deque
 
-        q = deque(sorted([m for m in self.mfcs if indeg[m.name] == 0], key=lambda x: x.name))
+        q = deque(sorted([m for m in self.mfcs if indeg[m.name] == 0],
+                         key=lambda x: (x.priority, x.name)))
         out = []
         by_name = {m.name: m for m in self.mfcs}
         while q:
             m = q.popleft()
             out.append(m)
-            for c in sorted(m.children, key=lambda x: x.name):
+            for c in sorted(m.children, key=lambda x: (x.priority, x.name)):
                 indeg[c.name] -= 1
                 if indeg[c.name] == 0:
                     q.append(by_name[c.name])

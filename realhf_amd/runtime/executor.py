@@ -1,4 +1,4 @@
-"""SPMD dataflow-graph executor.
+"""SPMD dataflow-graph executor with CONCURRENT mesh-scoped MFCs.
 
 Replaces the reference's master/worker runtime (realhf/system/
 master_worker.py + model_worker.py + request_reply_stream.py) for the
@@ -7,21 +7,37 @@ workers with a req->syn->ack total-order protocol, EVERY rank runs the
 same deterministic program over the DFG — per-MFC participation, DP
 splits, data merges and reallocation plans are all pure functions of
 (config, step), so no control messages exist in the hot path at all.
-The req->syn->ack protocol's purpose (identical request order on every
-worker, master_worker.py:74-95) is inherited by construction.
+
+Concurrency model (reference: one asyncio request+reply coroutine pair
+per MFC, master_worker.py:455,602): independent MFCs allocated to
+DISJOINT device meshes execute simultaneously in wall-clock because no
+collective in the per-MFC path spans more than (producer mesh | consumer
+mesh).  A rank simply skips MFCs it does not participate in and proceeds
+to its next MFC; the static ownership plan (below) guarantees every
+collective is entered by exactly its group's members in a single global
+program order, so the schedule is deadlock-free by construction.
+
+Data ownership: after an MFC completes, its output keys are held
+(full-batch, DP-merged) by every rank of its mesh.  The `_static_plan`
+computed at init simulates one step and records, per MFC, which keys
+must be broadcast from a holder to the consumer mesh — the analogue of
+the reference's data-transfer plan + send/recv caches
+(impl/model/comm/data_transfer.py:123-323), but derived offline.
 
 Per MFC and step:
-  1. run pre-hooks (parameter realloc in / reload from offload),
-  2. participating ranks take their balanced DP shard of the input keys
+  1. broadcast missing input keys to this MFC's mesh (plan-driven),
+  2. run pre-hooks (parameter realloc in / reload from offload),
+  3. participating ranks take their balanced DP shard of the input keys
      and call the interface (generate / inference / train_step),
-  3. DP-head outputs are merged into the replicated batch store
-     (runtime/data_transfer.py),
-  4. run post-hooks (realloc back / offload).
+  4. DP-head outputs are merged across the MESH (device-native gather),
+  5. run post-hooks (realloc back / offload).
+Train stats are merged once per step at the step boundary (the only
+world-scoped communication in the loop).
 """
 import dataclasses
 import os
 import time
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List, Optional, Set, Tuple
 
 import torch
 import torch.distributed as dist
@@ -31,12 +47,21 @@ from realhf_amd.api.data import SequenceSample
 from realhf_amd.api.dfg import DFG, MFCDef, OffloadHook, ParamReallocHook
 from realhf_amd.api.model import Model, ModelInterface
 from realhf_amd.base import constants, logging
+from realhf_amd.base.topology import (
+    ParallelGrid,
+    PipeDataTensorTopology,
+    new_or_get_group,
+)
 from realhf_amd.parallel.realloc import (
     ParallelStrategy,
     build_realloc_plan,
     execute_realloc,
 )
-from realhf_amd.runtime.data_transfer import dp_shard, gather_across_dp
+from realhf_amd.runtime.data_transfer import (
+    broadcast_sample,
+    dp_shard,
+    gather_across_dp,
+)
 
 logger = logging.getLogger("executor")
 
@@ -51,6 +76,20 @@ class MFCAllocation:
     @property
     def ranks(self):
         return sorted(r for _, r in self.strategy.rank_map)
+
+
+@dataclasses.dataclass
+class _TransferOp:
+    key: str
+    src: int  # global rank holding the key
+    group_ranks: Tuple[int, ...]  # {src} | missing consumer ranks
+
+
+@dataclasses.dataclass
+class _MFCPlan:
+    mesh: Tuple[int, ...]
+    transfers: List[_TransferOp]
+    scope: str  # constants scope name carrying the executing grid
 
 
 class DFGExecutor:
@@ -68,7 +107,106 @@ class DFGExecutor:
         self.interfaces = interfaces
         self.model_strategies = model_strategies
         self.rank = dist.get_rank() if dist.is_initialized() else 0
+        self.world = dist.get_world_size() if dist.is_initialized() else 1
         self._step = 0
+        self._topo_order = self.graph.topological_order()
+        self._register_mfc_grids()
+        self._plan = self._static_plan()
+        self._make_groups()
+        self._mfc_wall: Dict[str, float] = {}
+
+    # ----------------------------------------------------- static planning
+    def _register_mfc_grids(self):
+        """An MFC may execute on a SUB-MESH of its model's mesh with its
+        own DP layout (valid whenever both layouts are pure-DP: every
+        mesh rank holds full replica weights, so reallocation is FREE —
+        the 288 GB HBM3E design point).  Register a grid per such MFC so
+        dp_rank/dp_group inside the interface reflect the executing mesh."""
+        self._mfc_scopes: Dict[str, str] = {}
+        for mfc in self._topo_order:
+            alloc = self.allocations[mfc.name]
+            mstrat = self.model_strategies[mfc.model_name]
+            if alloc.strategy == mstrat:
+                self._mfc_scopes[mfc.name] = str(mfc.model_name)
+                continue
+            s = alloc.strategy
+            assert s.tp == mstrat.tp and s.pp == mstrat.pp, (
+                f"{mfc.name}: sub-mesh execution requires identical tp/pp "
+                f"(got mfc {s.tp}x{s.pp} vs model {mstrat.tp}x{mstrat.pp}); "
+                "differing shard layouts need a separate replica + realloc"
+            )
+            assert set(alloc.ranks) <= {r for _, r in mstrat.rank_map}, (
+                f"{mfc.name}: executing mesh must be a subset of the model mesh"
+            )
+            scope = f"{mfc.model_name}@{mfc.name}"
+            topo = PipeDataTensorTopology(
+                num_pp=s.pp, num_dp=s.dp, num_tp=s.tp,
+                sequence_parallel=alloc.sequence_parallel,
+                gradient_checkpointing=alloc.gradient_checkpointing,
+                ep_size=s.ep,
+            )
+            rank_mapping = {
+                topo.get_rank(pipe=p, data=d, tensor=t): r
+                for (p, d, t), r in s.rank_map
+            }
+            if dist.is_initialized():
+                grid = ParallelGrid(topo, rank_mapping)
+            else:
+                from realhf_amd.base.topology import FakeGrid
+
+                grid = FakeGrid(0, topo)
+            constants.set_grid(scope, grid)
+            self._mfc_scopes[mfc.name] = scope
+
+    def _static_plan(self) -> Dict[str, _MFCPlan]:
+        """Simulate one step to derive per-MFC input transfers.  Holder
+        sets evolve deterministically, so the plan is identical on every
+        rank and valid for all steps."""
+        world_ranks = frozenset(range(self.world))
+        holders: Dict[str, Set[int]] = {}
+        plan: Dict[str, _MFCPlan] = {}
+        for mfc in self._topo_order:
+            alloc = self.allocations[mfc.name]
+            mesh = tuple(alloc.ranks)
+            ops: List[_TransferOp] = []
+            for k in mfc.input_keys:
+                h = holders.get(k)
+                if h is None:
+                    h = set(world_ranks)  # dataset key: replicated batch
+                    holders[k] = h
+                need = set(mesh) - h
+                if need:
+                    src = min(h)
+                    ops.append(_TransferOp(
+                        key=k, src=src,
+                        group_ranks=tuple(sorted({src} | need)),
+                    ))
+                    h |= need
+            for k in mfc.output_keys:
+                holders[k] = set(mesh)
+            plan[mfc.name] = _MFCPlan(
+                mesh=mesh, transfers=ops, scope=self._mfc_scopes[mfc.name]
+            )
+        return plan
+
+    def _make_groups(self):
+        """Pre-create every process group the step will use (group
+        creation is collective over the WORLD — it must happen here, in
+        identical order on all ranks, never inside the concurrent loop)."""
+        if not dist.is_initialized():
+            return
+        for mfc in self._topo_order:
+            p = self._plan[mfc.name]
+            if 1 < len(p.mesh) < self.world:
+                new_or_get_group(list(p.mesh))
+            for t in p.transfers:
+                if len(t.group_ranks) < self.world:
+                    new_or_get_group(list(t.group_ranks))
+
+    def _mesh_group(self, mesh: Tuple[int, ...]):
+        if not dist.is_initialized() or len(mesh) == self.world:
+            return None  # default group
+        return new_or_get_group(list(mesh))
 
     # ------------------------------------------------------------- hooks
     def _run_hook(self, hook, mfc: MFCDef, pre: bool):
@@ -119,89 +257,171 @@ class DFGExecutor:
     # -------------------------------------------------------------- step
     def run_step(self, batch: SequenceSample) -> Dict[str, Any]:
         """Run one full DFG iteration on `batch` (the replicated global
-        batch).  Returns merged train stats."""
-        store = batch
-        all_stats: Dict[str, Any] = {}
-        for mfc in self.graph.topological_order():
+        batch).  Returns merged train stats (on every rank)."""
+        # store: key -> single-key full-batch sample (this rank's copy;
+        # present only if this rank holds the key)
+        store: Dict[str, SequenceSample] = {
+            k: batch.select_keys([k]) for k in batch.keys
+        }
+        local_stats: Dict[str, dict] = {}
+        device = None
+        for m in self.models.values():
+            device = m.device
+            break
+        dev = device if (device is not None and device.type == "cuda") else None
+
+        for mfc in self._topo_order:
+            plan = self._plan[mfc.name]
             alloc = self.allocations[mfc.name]
+            mesh = plan.mesh
+            on_mesh = self.rank in mesh
+
+            # 1. plan-driven input transfers (only src/missing ranks engage)
+            for t in plan.transfers:
+                if self.rank not in t.group_ranks:
+                    continue
+                grp = (new_or_get_group(list(t.group_ranks))
+                       if dist.is_initialized() and len(t.group_ranks) < self.world
+                       else None)
+                sample = store.get(t.key) if self.rank == t.src else None
+                received = broadcast_sample(sample, t.src, grp, device=dev)
+                if self.rank != t.src:
+                    store[t.key] = received
+
+            # 2. pre-hooks
             for h in mfc.pre_hooks:
                 self._run_hook(h, mfc, pre=True)
 
-            participating = self.rank in alloc.ranks
             local_out = None
-            local_stats = None
-            if participating:
-                name = str(mfc.model_name)
-                with constants.model_scope(name):
-                    g = constants.grid()
-                    model = self.models[mfc.model_name]
-                    real = (
-                        model.module.model
-                        if hasattr(model.module, "model") else model.module
-                    )
-                    if getattr(real, "_offloaded", False):
-                        real.reload_from_offload()
-                    inp = store.select_keys(
-                        [k for k in mfc.input_keys if k in store.keys]
-                    )
-                    if mfc.input_key_remap:
-                        inp.remap_keys_(mfc.input_key_remap)
-                    shard = dp_shard(inp, g.dp_rank, g.dp_size)
-                    shard = shard.to_device(model.device)
-                    iface = self.interfaces[mfc.name]
-                    t0 = time.time()
-                    if mfc.interface_type == ModelInterfaceType.GENERATE:
-                        res = iface.generate(model, shard, n_mbs=alloc.n_mbs)
-                    elif mfc.interface_type == ModelInterfaceType.INFERENCE:
-                        res = iface.inference(model, shard, n_mbs=alloc.n_mbs)
-                    elif mfc.interface_type == ModelInterfaceType.TRAIN_STEP:
-                        res = iface.train_step(model, shard, n_mbs=alloc.n_mbs)
-                    else:
-                        res = iface.evaluate(model, [shard])
-                    logger.debug(
-                        "%s on rank %d took %.3fs", mfc.name, self.rank,
-                        time.time() - t0,
-                    )
-                    if os.environ.get("REALHF_AMD_LOG_MEM") == "1":
-                        # per-MFC memory table (reference: model_worker
-                        # __log_gpu_stats:999)
-                        from realhf_amd.base.monitor import gpu_memory_stats
+            if on_mesh:
+                t0 = time.time()
+                with constants.model_scope(plan.scope):
+                    local_out = self._exec_mfc(mfc, alloc, store, local_stats)
+                self._mfc_wall[mfc.name] = time.time() - t0
 
-                        logger.info("%s mem: %s", mfc.name,
-                                    {k: round(v, 2) for k, v in
-                                     gpu_memory_stats().items()})
-                    if isinstance(res, SequenceSample):
-                        if mfc.output_key_remap:
-                            res.remap_keys_(mfc.output_key_remap)
-                        # one DP head per shard reports (tp 0, last pp stage)
-                        if g.tp_rank == 0 and g.pp_rank == g.pp_size - 1:
-                            local_out = res
-                    elif isinstance(res, dict):
-                        if g.tp_rank == 0 and g.pp_rank == g.pp_size - 1:
-                            local_stats = res
-
-            if mfc.interface_type == ModelInterfaceType.TRAIN_STEP:
-                merged_stats = gather_across_dp_stats(local_stats)
-                if merged_stats:
-                    for k, v in merged_stats.items():
-                        all_stats[f"{mfc.name}/{k}"] = v
-            else:
-                merged = gather_across_dp(local_out)
+            # 4. merge outputs across this mesh's DP heads
+            if mfc.interface_type != ModelInterfaceType.TRAIN_STEP and on_mesh:
+                if len(mesh) == 1:
+                    merged = local_out
+                else:
+                    grp = self._mesh_group(mesh)
+                    merged = gather_across_dp(local_out, group=grp, device=dev)
                 if merged is not None:
-                    order = _order_by_ids(merged.ids, store.ids)
-                    if len(order) == merged.bs == store.bs:
-                        merged = merged.select_idx(order)
-                        store.update_(merged)
-                    else:
-                        # the producer re-keyed the batch (e.g. GRPO group
-                        # expansion: each prompt becomes group_size
-                        # responses with fresh ids) — it becomes the store
-                        store = merged
+                    self._merge_outputs(mfc, merged, store)
 
+            # 5. post-hooks
             for h in mfc.post_hooks:
                 self._run_hook(h, mfc, pre=False)
+
         self._step += 1
-        return all_stats
+        return self._merge_step_stats(local_stats)
+
+    def _exec_mfc(self, mfc: MFCDef, alloc: MFCAllocation,
+                  store: Dict[str, SequenceSample],
+                  local_stats: Dict[str, dict]):
+        g = constants.grid()
+        model = self.models[mfc.model_name]
+        real = (
+            model.module.model
+            if hasattr(model.module, "model") else model.module
+        )
+        if getattr(real, "_offloaded", False):
+            real.reload_from_offload()
+        inp = self._assemble_input(mfc, store)
+        if mfc.input_key_remap:
+            inp.remap_keys_(mfc.input_key_remap)
+        shard = dp_shard(inp, g.dp_rank, g.dp_size)
+        shard = shard.to_device(model.device)
+        iface = self.interfaces[mfc.name]
+        if mfc.interface_type == ModelInterfaceType.GENERATE:
+            res = iface.generate(model, shard, n_mbs=alloc.n_mbs)
+        elif mfc.interface_type == ModelInterfaceType.INFERENCE:
+            res = iface.inference(model, shard, n_mbs=alloc.n_mbs)
+        elif mfc.interface_type == ModelInterfaceType.TRAIN_STEP:
+            res = iface.train_step(model, shard, n_mbs=alloc.n_mbs)
+        else:
+            res = iface.evaluate(model, [shard])
+        if os.environ.get("REALHF_AMD_LOG_MEM") == "1":
+            # per-MFC memory table (reference: model_worker __log_gpu_stats:999)
+            from realhf_amd.base.monitor import gpu_memory_stats
+
+            logger.info("%s mem: %s", mfc.name,
+                        {k: round(v, 2) for k, v in gpu_memory_stats().items()})
+        local_out = None
+        if isinstance(res, SequenceSample):
+            if mfc.output_key_remap:
+                res.remap_keys_(mfc.output_key_remap)
+            # one DP head per shard reports (tp 0, last pp stage)
+            if g.tp_rank == 0 and g.pp_rank == g.pp_size - 1:
+                local_out = res
+        elif isinstance(res, dict):
+            # every DP head contributes; step-end merge averages over them
+            if g.tp_rank == 0 and g.pp_rank == g.pp_size - 1:
+                local_stats[mfc.name] = res
+        return local_out
+
+    def _assemble_input(self, mfc: MFCDef,
+                        store: Dict[str, SequenceSample]) -> SequenceSample:
+        keys = [k for k in mfc.input_keys if k in store]
+        assert keys, (mfc.name, mfc.input_keys, list(store))
+        base = store[keys[0]]
+        out = base
+        for k in keys[1:]:
+            s = store[k]
+            if list(map(str, s.ids)) != list(map(str, out.ids)):
+                order = _order_by_ids(s.ids, out.ids)
+                assert len(order) == out.bs, (
+                    f"{mfc.name}: id mismatch joining key {k}"
+                )
+                s = s.select_idx(order)
+            if out is base:
+                out = SequenceSample(
+                    keys=tuple(base.keys), ids=list(base.ids),
+                    seqlens=dict(base.seqlens), data=dict(base.data),
+                )
+            out.update_(s)
+        return out
+
+    def _merge_outputs(self, mfc: MFCDef, merged: SequenceSample,
+                       store: Dict[str, SequenceSample]):
+        ref_key = next(iter(store), None)
+        same_batch = (
+            ref_key is not None
+            and sorted(map(str, merged.ids)) == sorted(map(str, store[ref_key].ids))
+        )
+        if same_batch:
+            order = _order_by_ids(merged.ids, store[ref_key].ids)
+            merged = merged.select_idx(order)
+        else:
+            # the producer re-keyed the batch (e.g. GRPO group expansion:
+            # each prompt becomes group_size responses with fresh ids) —
+            # its output id-space becomes the store's
+            for k in list(store):
+                del store[k]
+        for k in merged.keys:
+            store[k] = merged.select_keys([k])
+
+    def _merge_step_stats(self, local_stats: Dict[str, dict]) -> Dict[str, Any]:
+        """One world-scoped object gather per STEP (not per MFC) — the
+        step boundary is a natural sync point; this also drains any rank
+        skew before the next dataloader batch."""
+        flat = {
+            f"{mfc}/{k}": float(v)
+            for mfc, st in local_stats.items()
+            for k, v in st.items() if isinstance(v, (int, float))
+        }
+        for mfc, w in self._mfc_wall.items():
+            flat[f"{mfc}/wall_s"] = w
+        if not dist.is_initialized():
+            return flat
+        bucket: List = [None] * self.world
+        dist.all_gather_object(bucket, flat)
+        vals: Dict[str, List[float]] = {}
+        for b in bucket:
+            for k, v in (b or {}).items():
+                vals.setdefault(k, []).append(v)
+        out = {k: sum(v) / len(v) for k, v in vals.items()}
+        return out
 
 
 def _order_by_ids(ids: List, target_ids: List) -> List[int]:
@@ -210,6 +430,7 @@ def _order_by_ids(ids: List, target_ids: List) -> List[int]:
 
 
 def gather_across_dp_stats(local: Optional[dict]) -> Optional[dict]:
+    """Legacy helper (kept for tests): world-mean of per-rank stat dicts."""
     if not dist.is_initialized():
         return local
     world = dist.get_world_size()

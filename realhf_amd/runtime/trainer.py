@@ -171,11 +171,20 @@ def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment
     allocations: Dict[str, MFCAllocation] = {}
     trainable: List[ModelName] = []
 
-    def add_model(role, mc, replica=0):
+    def add_model(role, mc, replica=0, ranks=None):
         name = ModelName(role, replica)
         rcfg = model_cfg_of(mc, name)
         par = _resolve_parallel(cfg, mc.parallel, world, rcfg.param_count())
-        strat = _strategy_for(par, world)
+        if (ranks is not None and par.tensor_parallel_size == 1
+                and par.pipeline_parallel_size == 1):
+            # asymmetric heuristic: model lives only on a rank subset,
+            # pure DP over it (288 GB replication design point).  Models
+            # whose resolved layout needs TP/PP keep the full mesh.
+            ranks = list(ranks)
+            par = ParallelismConfig(data_parallel_size=len(ranks))
+            strat = ParallelStrategy.make(1, len(ranks), 1, ranks=ranks)
+        else:
+            strat = _strategy_for(par, world)
         if rcfg.moe is not None:
             ep = min(rcfg.moe.expert_parallel_size, strat.dp)
             if ep > 1 and strat.dp % ep == 0:
@@ -186,17 +195,18 @@ def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment
         return name, par
 
     def add_mfc(name, model_name, itype, iface_cfg, inp, out, mc, par,
-                n_mbs=None):
+                n_mbs=None, alloc_strategy=None, priority=0):
         mfcs.append(
             MFCDef(
                 name=name, model_name=model_name, interface_type=itype,
                 interface_impl=iface_cfg, input_keys=tuple(inp),
                 output_keys=tuple(out), n_seqs=cfg.dataset.train_bs_n_seqs,
+                priority=priority,
             )
         )
         interfaces[name] = make_interface(iface_cfg)
         allocations[name] = MFCAllocation(
-            strategy=model_strategies[model_name],
+            strategy=alloc_strategy or model_strategies[model_name],
             sequence_parallel=par.sequence_parallel,
             gradient_checkpointing=mc.gradient_checkpointing,
             n_mbs=n_mbs,
@@ -246,14 +256,45 @@ def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment
     )
 
 
+def _asym_inference_meshes(cfg, world: int):
+    """MI355X heuristic (reference counterpart: ppo_exp.py:419-613,
+    recalibrated for 288 GB HBM3E): with ≤~30B-param models every role
+    fits fully replicated, so all layouts are pure DP and sub-mesh
+    execution needs NO parameter reallocation (every rank already holds
+    full weights).  The asymmetric lever that remains profitable is
+    CONCURRENCY of the independent inference MFCs: critic_inf runs on
+    the first half of the node while rew_inf runs on the second half,
+    simultaneously (ref_inf keeps the whole node and runs first).  At
+    80 GB the reference instead must shard/offload — its level-2/3/4
+    split — which on this hardware would only add transfer cost."""
+    if world < 2 or cfg.allocation_mode != "heuristic":
+        return None
+    half = world // 2
+    return {
+        "first_half": list(range(half)),
+        "second_half": list(range(half, world)),
+    }
+
+
 def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
     T = ModelInterfaceType
     ppo = cfg.ppo
+    asym = _asym_inference_meshes(cfg, world)
     actor, apar = add_model("actor", cfg.actor)
     critic, cpar = add_model("critic", cfg.critic)
     ref, refpar = add_model("ref", cfg.ref)
-    rew, rewpar = add_model("rew", cfg.rew)
+    # the reward model only ever serves rew_inf: under the asymmetric
+    # heuristic it is instantiated on HALF the node only (memory saving +
+    # concurrency with critic_inf on the other half)
+    rew, rewpar = add_model("rew", cfg.rew,
+                            ranks=asym["second_half"] if asym else None)
     trainable += [actor, critic]
+    critic_inf_strategy = None
+    if (asym and cpar.tensor_parallel_size == 1
+            and cpar.pipeline_parallel_size == 1):
+        critic_inf_strategy = ParallelStrategy.make(
+            1, len(asym["first_half"]), 1, ranks=asym["first_half"]
+        )
 
     gen_cfg = dataclasses.asdict(ppo.gen)
     actor_iface = Abstraction("ppo_actor", {
@@ -294,12 +335,17 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
             ParamReallocHook(source=actor, target=gen_model)
         )
         gen_mfc.post_hooks.append(OffloadHook())
+    # execution order under the asymmetric heuristic: ref_inf first on the
+    # WHOLE node, then critic_inf (first half) and rew_inf (second half)
+    # CONCURRENTLY — each sub-mesh rank skips the other MFC and proceeds
+    add_mfc("ref_inf", ref, T.INFERENCE, actor_iface,
+            ["packed_input_ids"], ["packed_ref_logprobs"], cfg.ref, refpar,
+            priority=-1)
+    add_mfc("critic_inf", critic, T.INFERENCE, critic_iface,
+            ["packed_input_ids"], ["values"], cfg.critic, cpar,
+            alloc_strategy=critic_inf_strategy)
     add_mfc("rew_inf", rew, T.INFERENCE, rw_iface,
             ["packed_input_ids"], ["rewards"], cfg.rew, rewpar)
-    add_mfc("ref_inf", ref, T.INFERENCE, actor_iface,
-            ["packed_input_ids"], ["packed_ref_logprobs"], cfg.ref, refpar)
-    add_mfc("critic_inf", critic, T.INFERENCE, critic_iface,
-            ["packed_input_ids"], ["values"], cfg.critic, cpar)
     train_keys = ["packed_input_ids", "packed_logprobs", "packed_ref_logprobs",
                   "rewards", "values", "prompt_mask", "seq_no_eos_mask"]
     add_mfc("actor_train", actor, T.TRAIN_STEP, actor_iface, train_keys, [],
@@ -317,9 +363,14 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
 def _build_grpo(cfg: GRPOConfig, world, add_model, add_mfc, mfcs, trainable):
     T = ModelInterfaceType
     ppo = cfg.ppo
+    asym = _asym_inference_meshes(cfg, world)
     actor, apar = add_model("actor", cfg.actor)
-    ref, refpar = add_model("ref", cfg.ref)
-    rew, rewpar = add_model("rew", cfg.rew)
+    # GRPO has no critic: ref_inf and rew_inf are the whole inference
+    # level — give each half the node and run them concurrently
+    ref, refpar = add_model("ref", cfg.ref,
+                            ranks=asym["first_half"] if asym else None)
+    rew, rewpar = add_model("rew", cfg.rew,
+                            ranks=asym["second_half"] if asym else None)
     trainable.append(actor)
     iface = Abstraction("grpo", {
         "n_minibatches": ppo.ppo_n_minibatches,
