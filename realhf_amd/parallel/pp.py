@@ -51,6 +51,70 @@ def _recv(shape, dtype, src_global: int, device):
     return buf
 
 
+class _PPDecodeGraph:
+    """Per-(stage, microbatch) hipGraph of the decode forward inside the
+    pipeline (reference: per-microbatch CUDA graphs in PipeGenInstrSet,
+    pipe_runner.py:422-475).  Static buffers: token ids (first stage) or
+    received hidden states (later stages) + cache_seqlens; the KV caches
+    are already static.  The p2p recv writes into the static input
+    buffer, then one graph launch replaces ~32 kernel launches/layer."""
+
+    def __init__(self, model, kv_caches, bs, is_first, device, hidden_dim,
+                 dtype):
+        self.model = model
+        self.kv = kv_caches
+        self.is_first = is_first
+        self.graph = None
+        self.failed = False
+        self.in_tokens = torch.zeros(bs, dtype=torch.long, device=device)
+        self.in_hidden = (None if is_first else
+                          torch.zeros(bs, hidden_dim, dtype=dtype,
+                                      device=device))
+        self.in_cache_seqlens = torch.zeros(bs, dtype=torch.int32,
+                                            device=device)
+        self.out = None
+
+    def _eager(self):
+        return self.model(
+            packed_input_ids=self.in_tokens if self.is_first else None,
+            hidden_states=self.in_hidden,
+            kv_caches=self.kv,
+            cache_seqlens=self.in_cache_seqlens,
+            decode=True,
+        )
+
+    def _capture(self):
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._eager()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.out = self._eager()
+
+    def step(self, tokens, hidden, cache_seqlens):
+        if self.is_first:
+            self.in_tokens.copy_(tokens)
+        else:
+            self.in_hidden.copy_(hidden)
+        self.in_cache_seqlens.copy_(cache_seqlens)
+        if self.graph is None and not self.failed:
+            try:
+                self._capture()
+            except Exception as e:
+                logger.warning("PP decode hipGraph capture failed (%s); "
+                               "eager decode", e)
+                self.failed = True
+                torch.cuda.synchronize()
+        if self.graph is None:
+            return self._eager()
+        self.graph.replay()
+        return self.out
+
+
 class PipelinedEngine(PipelinableEngine):
     """pp > 1 engine (train + forward + generate).  The module is a
     ReaLModel holding only this stage's layers."""
@@ -204,11 +268,22 @@ class PipelinedEngine(PipelinableEngine):
                  gconfig: Optional[GenerationHyperparameters] = None,
                  n_mbs: Optional[int] = None, return_prompt_logprobs=False,
                  **kw):
-        """Token-level pipelined generation: prefill passes through all
-        stages; each decode step is a short pipeline pass, with the last
-        stage sampling and sending next tokens to the first stage
-        (reference: GenerateSchedule + Send/RecvNextTokens).
-        Correctness-first: one microbatch (no token-interleaving yet)."""
+        """Token-level PIPELINED generation (reference: GenerateSchedule
+        static_schedule.py:195-306 + pipe_runner.py:179-256, 422-475).
+
+        n_mbs microbatches (default = pp stages) are interleaved
+        round-robin: while microbatch i's token is being sampled on the
+        last stage, microbatch i+1 occupies the earlier stages, so the
+        per-token pipeline bubble shrinks from (S-1)/S to ~0 once
+        n_mbs >= S.  After sampling, the last stage broadcasts
+        (tokens, done) over the PP group — one tiny collective that both
+        feeds the first stage's next forward AND gives every stage the
+        SAME termination view, replacing the reference's burn-out
+        signal protocol: a microbatch whose sequences have all hit EOS
+        (checked every 8 tokens to amortize the host sync) is retired by
+        all stages at the same round, so short batches exit early.  Each
+        (stage, microbatch) decode forward is hipGraph-captured with
+        static recv/token buffers (_PPDecodeGraph)."""
         from realhf_amd.models import generation as genmod
         from realhf_amd.parallel import mappings
 
@@ -222,106 +297,176 @@ class PipelinedEngine(PipelinableEngine):
         pad = getattr(tokenizer, "pad_token_id", 0) if tokenizer else 0
         if pad is None:
             pad = 0
+        max_new = gconfig.max_new_tokens
+        S = g.pp_size
+        first, last = self.model.is_first_stage, self.model.is_last_stage
+        pp_group = g.pp_group()
 
         key = "packed_prompts" if "packed_prompts" in input_.keys else "packed_input_ids"
-        prompts, cu, mx = sample_to_packed(input_, key)
-        bs = cu.shape[0] - 1
-        prompt_lens = (cu[1:] - cu[:-1]).to(dev)
-        max_new = gconfig.max_new_tokens
-        cache_len = int(prompt_lens.max()) + max_new
-        nkv_local = max(cfg.n_kv_heads // self.model.tp_size, 1)
-        n_blocks = sum(
-            1 for i in self.model.layer_indices if 1 <= i <= cfg.n_layers
-        )
-        kv_caches = [
-            (
-                torch.zeros(bs, cache_len, nkv_local, cfg.head_dim, dtype=dtype, device=dev),
-                torch.zeros(bs, cache_len, nkv_local, cfg.head_dim, dtype=dtype, device=dev),
-            )
-            for _ in range(n_blocks)
-        ]
+        M = n_mbs or S
+        M = max(1, min(M, input_.bs))
+        mbs = input_.split(M) if M > 1 else [input_]
+        M = len(mbs)
         gen = torch.Generator(device=dev)
         gen.manual_seed(seeding.base_seed() + 97 * g.dp_rank)
+        nkv_local = max(cfg.n_kv_heads // self.model.tp_size, 1)
+        n_blocks = sum(1 for i in self.model.layer_indices
+                       if 1 <= i <= cfg.n_layers)
 
-        # ---- prefill --------------------------------------------------
-        hidden = None
-        total = prompts.shape[0]
-        if not self.model.is_first_stage:
-            hidden = _recv((total, cfg.hidden_dim), dtype, g.pp_prev_global_rank(), dev)
-        out = self.model(
-            packed_input_ids=prompts if self.model.is_first_stage else None,
-            cu_seqlens=cu, max_seqlen=mx, hidden_states=hidden,
-            kv_caches=kv_caches,
+        class _MB:
+            pass
+
+        st: List[_MB] = []
+        for mb in mbs:
+            s = _MB()
+            s.prompts, s.cu, s.mx = sample_to_packed(mb, key)
+            s.bs = s.cu.shape[0] - 1
+            s.prompt_lens = (s.cu[1:] - s.cu[:-1]).to(dev)
+            cache_len = int(s.prompt_lens.max()) + max_new
+            s.kv = [
+                (torch.zeros(s.bs, cache_len, nkv_local, cfg.head_dim,
+                             dtype=dtype, device=dev),
+                 torch.zeros(s.bs, cache_len, nkv_local, cfg.head_dim,
+                             dtype=dtype, device=dev))
+                for _ in range(n_blocks)
+            ]
+            s.cache_seqlens = s.prompt_lens.to(torch.int32).clone()
+            s.r = 0  # pipeline round: r produces token r (0 = prefill)
+            s.active = True
+            s.prompt_logprobs = None
+            s.graph = None
+            # broadcast payload: [tokens(int64) | done(int64)] as one row pair
+            s.bc = torch.zeros(2, s.bs, dtype=torch.long, device=dev)
+            s.done = torch.zeros(s.bs, dtype=torch.bool, device=dev)
+            if last:
+                s.gen_tokens = torch.full((s.bs, max_new), pad,
+                                          dtype=torch.long, device=dev)
+                s.gen_logprobs = torch.zeros(s.bs, max_new,
+                                             dtype=torch.float32, device=dev)
+                s.gen_lengths = torch.zeros(s.bs, dtype=torch.long, device=dev)
+            st.append(s)
+
+        use_graph = (
+            gconfig.use_hip_graph and dev.type == "cuda"
+            and dtype == torch.bfloat16 and cfg.head_dim in (64, 128)
         )
-        prompt_logprobs = None
-        if self.model.is_last_stage:
-            if return_prompt_logprobs:
-                from realhf_amd.parallel.tp import packed_shifted_logprobs
 
-                prompt_logprobs = packed_shifted_logprobs(out, cu, prompts)
-            last = out[(cu[1:].long() - 1)]
-            cur_logits = mappings.gather_from_tp_region(last).float()
-        else:
-            _send(out, g.pp_next_global_rank())
+        last_rank = g.global_rank_of(S - 1, g.dp_rank, g.tp_rank)
 
-        gen_tokens = torch.full((bs, max_new), pad, dtype=torch.long, device=dev)
-        gen_logprobs = torch.zeros(bs, max_new, dtype=torch.float32, device=dev)
-        done = torch.zeros(bs, dtype=torch.bool, device=dev)
-        gen_lengths = torch.zeros(bs, dtype=torch.long, device=dev)
-        cache_seqlens = prompt_lens.to(torch.int32).clone()
-        first_rank = g.global_rank_of(0, g.dp_rank, g.tp_rank)
-        last_rank = g.global_rank_of(g.pp_size - 1, g.dp_rank, g.tp_rank)
+        def sample_and_bcast(s, cur_logits):
+            """LAST STAGE ONLY: sample token r, update bookkeeping,
+            broadcast (tokens, done) to the pipeline.  Non-last stages
+            receive this broadcast at the START of the microbatch's NEXT
+            slot (not here) — otherwise an early stage would block on the
+            sample of mb i before computing mb i+1 and the pipeline would
+            serialize."""
+            r = s.r
+            if eos is not None and r < gconfig.min_new_tokens:
+                cur_logits[:, eos] = float("-inf")
+            tokens, logp = genmod._sample_from_logits(cur_logits, gconfig, gen)
+            tokens = torch.where(s.done, torch.full_like(tokens, pad), tokens)
+            s.gen_tokens[:, r] = tokens
+            s.gen_logprobs[:, r] = torch.where(
+                s.done, torch.zeros_like(logp), logp)
+            s.gen_lengths += (~s.done).long()
+            if eos is not None:
+                s.done = s.done | (tokens == eos)
+            s.bc[0] = tokens
+            s.bc[1] = s.done.long() if eos is not None else 0
+            if S > 1:
+                dist.broadcast(s.bc, src=last_rank, group=pp_group)
 
-        tokens = None
-        for t in range(max_new):
-            # sampling on last stage; tokens broadcast last -> first
-            if self.model.is_last_stage:
-                if eos is not None and t < gconfig.min_new_tokens:
-                    cur_logits[:, eos] = float("-inf")
-                tokens, logp = genmod._sample_from_logits(cur_logits, gconfig, gen)
-                tokens = torch.where(done, torch.full_like(tokens, pad), tokens)
-                gen_tokens[:, t] = tokens
-                gen_logprobs[:, t] = torch.where(done, torch.zeros_like(logp), logp)
-                gen_lengths += (~done).long()
-                if eos is not None:
-                    done = done | (tokens == eos)
-                if not self.model.is_first_stage and t < max_new - 1:
-                    _send(tokens, first_rank)
-                    _send(done.to(torch.uint8), first_rank)
-            if t == max_new - 1:
-                break
-            if self.model.is_first_stage and not self.model.is_last_stage:
-                tokens = _recv((bs,), torch.long, last_rank, dev)
-                done_u8 = _recv((bs,), torch.uint8, last_rank, dev)
-                if bool(done_u8.all()):
-                    # termination: propagate via the normal flow below
-                    pass
-            cache_seqlens += 1
+        def should_retire(s) -> bool:
+            """Identical decision on every stage at slot r from the
+            (mb, r-1) broadcast; host sync amortized over 8 tokens."""
+            if s.r >= max_new:
+                return True
+            return (eos is not None and (s.r & 7) == 0
+                    and bool(s.bc[1].all()))
+
+        # ---- prefill (fill-drain over microbatches, round r = 0) ----------
+        for s in st:
             hidden = None
-            if not self.model.is_first_stage:
-                hidden = _recv((bs, cfg.hidden_dim), dtype, g.pp_prev_global_rank(), dev)
+            if not first:
+                hidden = _recv((s.prompts.shape[0], cfg.hidden_dim), dtype,
+                               g.pp_prev_global_rank(), dev)
             out = self.model(
-                packed_input_ids=tokens if self.model.is_first_stage else None,
-                hidden_states=hidden, kv_caches=kv_caches,
-                cache_seqlens=cache_seqlens, decode=True,
+                packed_input_ids=s.prompts if first else None,
+                cu_seqlens=s.cu, max_seqlen=s.mx, hidden_states=hidden,
+                kv_caches=s.kv,
             )
-            if not self.model.is_last_stage:
-                _send(out, g.pp_next_global_rank())
+            if last:
+                if return_prompt_logprobs:
+                    from realhf_amd.parallel.tp import packed_shifted_logprobs
+
+                    s.prompt_logprobs = packed_shifted_logprobs(
+                        out, s.cu, s.prompts)
+                cur_logits = mappings.gather_from_tp_region(
+                    out[(s.cu[1:].long() - 1)]).float()
+                sample_and_bcast(s, cur_logits)  # token 0
             else:
-                cur_logits = mappings.gather_from_tp_region(out).float()
+                _send(out, g.pp_next_global_rank())
+            s.r = 1
+
+        # ---- interleaved decode rounds (r >= 1) ---------------------------
+        # All stages walk microbatches in the same (r-major, mb-minor)
+        # order, so every pp-group collective matches by program order.
+        self._gen_decode_slots = 0
+        self._gen_trace = []  # per-rank event log: ("bc"|"fwd", mb, r)
+        while any(s.active for s in st):
+            for i, s in enumerate(st):
+                if not s.active:
+                    continue
+                if S > 1 and not last:
+                    # receive (tokens, done) of round r-1 sampled on the
+                    # last stage while this stage was busy with other mbs
+                    dist.broadcast(s.bc, src=last_rank, group=pp_group)
+                    self._gen_trace.append(("bc", i, s.r - 1))
+                if should_retire(s):
+                    s.active = False
+                    continue
+                self._gen_decode_slots += 1
+                self._gen_trace.append(("fwd", i, s.r))
+                s.cache_seqlens += 1
+                tokens = s.bc[0]
+                if s.graph is None and use_graph:
+                    s.graph = _PPDecodeGraph(self.model, s.kv, s.bs, first, dev,
+                                             cfg.hidden_dim, dtype)
+                if not first:
+                    hidden = _recv((s.bs, cfg.hidden_dim), dtype,
+                                   g.pp_prev_global_rank(), dev)
+                else:
+                    hidden = None
+                if s.graph is not None:
+                    out = s.graph.step(tokens, hidden, s.cache_seqlens)
+                else:
+                    out = self.model(
+                        packed_input_ids=tokens if first else None,
+                        hidden_states=hidden, kv_caches=s.kv,
+                        cache_seqlens=s.cache_seqlens, decode=True,
+                    )
+                if last:
+                    cur_logits = mappings.gather_from_tp_region(out).float()
+                    sample_and_bcast(s, cur_logits)
+                else:
+                    _send(out, g.pp_next_global_rank())
+                s.r += 1
 
         _flush_sends()
-        if not self.model.is_last_stage:
+        if not last:
             return None
-        max_len = int(gen_lengths.max())
-        return [(
-            genmod.GenerationOutput(
-                gen_tokens=gen_tokens[:, :max_len],
-                gen_logprobs=gen_logprobs[:, :max_len],
-                gen_lengths=gen_lengths,
-                no_eos_mask=~done,
-                prompt_logprobs=prompt_logprobs,
-            ),
-            prompts,
-            cu,
-        )]
+        outs = []
+        for s, mb in zip(st, mbs):
+            max_len = int(s.gen_lengths.max())
+            outs.append((
+                genmod.GenerationOutput(
+                    gen_tokens=s.gen_tokens[:, :max_len],
+                    gen_logprobs=s.gen_logprobs[:, :max_len],
+                    gen_lengths=s.gen_lengths,
+                    no_eos_mask=~s.done,
+                    prompt_logprobs=s.prompt_logprobs,
+                ),
+                s.prompts,
+                s.cu,
+            ))
+        return outs

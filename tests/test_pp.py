@@ -160,15 +160,28 @@ def _pp_gen_worker():
     gconfig = GenerationHyperparameters(max_new_tokens=6, greedy=True,
                                         use_hip_graph=False)
     with constants.model_scope("m"):
-        outs = engine.generate(batch, gconfig=gconfig)
+        # n_mbs=2: token-interleaved pipelined decode over 2 microbatches
+        outs = engine.generate(batch, gconfig=gconfig, n_mbs=2)
     if m.is_last_stage:
-        (gen_out, prompts, cu) = outs[0]
+        assert len(outs) == 2  # one GenerationOutput per microbatch
         single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
         _fill_model_from_full(single, cfg, sd)
         cu_t = torch.tensor([0] + list(np.cumsum(lens)), dtype=torch.int32)
         ref = gen_single(single, toks, cu_t, gconfig)
-        assert torch.equal(gen_out.gen_tokens, ref.gen_tokens), (
-            gen_out.gen_tokens, ref.gen_tokens)
+        # microbatches may reorder sequences (balanced split): match each
+        # generated row back to the reference row by prompt identity
+        prompt_of = {}
+        for j in range(len(lens)):
+            prompt_of[tuple(toks[cu_t[j]:cu_t[j+1]].tolist())] = j
+        n_checked = 0
+        for gen_out, prompts, cu in outs:
+            for i in range(cu.shape[0] - 1):
+                p = tuple(prompts[cu[i]:cu[i+1]].tolist())
+                j = prompt_of[p]
+                assert torch.equal(gen_out.gen_tokens[i], ref.gen_tokens[j]), (
+                    gen_out.gen_tokens[i], ref.gen_tokens[j])
+                n_checked += 1
+        assert n_checked == len(lens)
     dist.barrier()
 
 
@@ -222,3 +235,114 @@ def _pp_tp_forward_worker():
 @pytest.mark.distributed
 def test_pp2_tp2_forward():
     LocalMultiProcessTest(4, _pp_tp_forward_worker).launch()
+
+
+def _pp_gen_early_term_worker():
+    """A batch whose sequences hit EOS at token 1 must retire its
+    microbatch early: decode slots executed << max_new_tokens
+    (reference: GenerateSchedule termination protocol + burn-out,
+    pipe_runner.py:179-256)."""
+    import torch.distributed as dist
+
+    from realhf_amd.api.data import SequenceSample
+    from realhf_amd.api.model import GenerationHyperparameters
+    from realhf_amd.base import constants
+    from realhf_amd.models.generation import generate as gen_single
+    from realhf_amd.models.real_model import ReaLModel
+
+    cfg = make_test_config(n_layers=4, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=51)
+    engine, m = _make_pp_engine(cfg, sd, 2)
+    rng = np.random.RandomState(9)
+    lens = [5]
+    toks = torch.from_numpy(rng.randint(0, 128, size=sum(lens))).long()
+    # learn what greedy decoding emits at step 1 -> use it as EOS
+    single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    _fill_model_from_full(single, cfg, sd)
+    cu_t = torch.tensor([0, lens[0]], dtype=torch.int32)
+    probe = gen_single(single, toks, cu_t,
+                       GenerationHyperparameters(max_new_tokens=4, greedy=True,
+                                                 use_hip_graph=False))
+    eos = int(probe.gen_tokens[0, 1])
+
+    class _Tok:
+        eos_token_id = eos
+        pad_token_id = 0
+
+    batch = SequenceSample(
+        keys=("packed_prompts",), ids=["e0"],
+        seqlens={"packed_prompts": [[lens[0]]]},
+        data={"packed_prompts": toks},
+    )
+    max_new = 64
+    gconfig = GenerationHyperparameters(max_new_tokens=max_new, greedy=True,
+                                        use_hip_graph=False)
+    with constants.model_scope("m"):
+        outs = engine.generate(batch, tokenizer=_Tok(), gconfig=gconfig,
+                               n_mbs=1)
+    # EVERY stage must retire early (slots counted locally per rank)
+    assert engine._gen_decode_slots <= 10, engine._gen_decode_slots
+    if m.is_last_stage:
+        (gen_out, _, _) = outs[0]
+        assert int(gen_out.gen_lengths[0]) == 2, gen_out.gen_lengths
+        assert not bool(gen_out.no_eos_mask[0])
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_pp2_generate_early_termination():
+    LocalMultiProcessTest(2, _pp_gen_early_term_worker).launch()
+
+
+def _pp_gen_interleave_worker():
+    """>= 2 microbatches in flight per token round: the FIRST stage must
+    issue microbatch i+1's decode forward BEFORE it receives microbatch
+    i's next-token broadcast — i.e. while mb i is still traversing the
+    later stages (reference: GenerateSchedule interleaving,
+    static_schedule.py:215-306)."""
+    import torch.distributed as dist
+
+    from realhf_amd.api.data import SequenceSample
+    from realhf_amd.api.model import GenerationHyperparameters
+    from realhf_amd.base import constants
+
+    cfg = make_test_config(n_layers=4, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=51)
+    engine, m = _make_pp_engine(cfg, sd, 2)
+    rng = np.random.RandomState(9)
+    lens = [5, 8, 6, 7]
+    toks = torch.from_numpy(rng.randint(0, 128, size=sum(lens))).long()
+    batch = SequenceSample(
+        keys=("packed_prompts",), ids=[f"i{j}" for j in range(4)],
+        seqlens={"packed_prompts": [[l] for l in lens]},
+        data={"packed_prompts": toks},
+    )
+    gconfig = GenerationHyperparameters(max_new_tokens=5, greedy=True,
+                                        use_hip_graph=False)
+    with constants.model_scope("m"):
+        engine.generate(batch, gconfig=gconfig, n_mbs=2)
+    if m.is_first_stage:
+        tr = engine._gen_trace
+        # find a round r with: fwd(mb0,r) < fwd(mb1,r) < bc(mb0,r)
+        ok = False
+        for r in range(1, 4):
+            try:
+                f0 = tr.index(("fwd", 0, r))
+                f1 = tr.index(("fwd", 1, r))
+                b0 = tr.index(("bc", 0, r))
+            except ValueError:
+                continue
+            if f0 < f1 < b0:
+                ok = True
+                break
+        assert ok, f"no interleaved round found in trace: {tr}"
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_pp2_generate_microbatch_interleaving():
+    LocalMultiProcessTest(2, _pp_gen_interleave_worker).launch()
