@@ -9,10 +9,12 @@ all experts and stubs the expert group to self (megatron.py:108, SURVEY.md
 §2.3 row EP).  Here experts shard over an EP group (a block of DP ranks)
 with token all-to-all over xGMI; EP degree = cfg.moe.expert_parallel_size.
 
-Expert GEMMs run through the hand-written MFMA grouped-GEMM kernel
-(ops/csrc/grouped_gemm.hip) on the rollout/inference path; the training
-path uses per-expert GEMMs (rocBLAS) so autograd handles the backward
-(hand-written grouped bwd is a noted follow-up).
+Expert GEMMs run through the hand-written MFMA grouped-GEMM kernels for
+BOTH inference and training (fwd: ops/csrc/grouped_gemm.hip; bwd dX/dW:
+ops/csrc/grouped_gemm_bwd.hip, via the _GroupedGemm autograd function) —
+matching the reference's grouped_gemm.ops.gmm training path
+(experts.py:194-207).  The per-expert rocBLAS loop remains as the
+CPU/odd-shape fallback (REALHF_AMD_MOE_LOOP=1 forces it, for tests).
 """
 import contextlib
 from typing import Dict, List
@@ -115,6 +117,34 @@ class TopKRouter(torch.nn.Module):
         return scores, idx
 
 
+class _GroupedGemm(torch.autograd.Function):
+    """out[seg_e] = x[seg_e] @ W[e]^T through the MFMA grouped-GEMM
+    kernels; backward stays native (dX = dOut @ W[e], dW = dOut^T x)."""
+
+    @staticmethod
+    def forward(ctx, x, w, counts_cpu):
+        from realhf_amd import ops as _ops_pkg
+
+        C = _ops_pkg.require_hip()
+        ctx.save_for_backward(x, w, counts_cpu)
+        return C.grouped_gemm(x.contiguous(), w, counts_cpu)
+
+    @staticmethod
+    def backward(ctx, dout):
+        from realhf_amd import ops as _ops_pkg
+
+        C = _ops_pkg.require_hip()
+        x, w, counts_cpu = ctx.saved_tensors
+        dout = dout.contiguous()
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = C.grouped_gemm_dx(dout, w, counts_cpu)
+        if ctx.needs_input_grad[1]:
+            dw = C.grouped_gemm_dw(dout, x.contiguous(), counts_cpu,
+                                   w.shape[0])
+        return dx, dw, None
+
+
 class _AllToAll(torch.autograd.Function):
     """Autograd-wrapped all_to_all_single over the EP group: backward is
     the reverse exchange (xGMI all-pairs — the natural fit)."""
@@ -188,29 +218,36 @@ class MoELayer(torch.nn.Module):
         )
 
     def _experts_forward(self, x_sorted: torch.Tensor, counts_cpu: torch.Tensor):
-        """x_sorted: tokens grouped by local expert; counts_cpu [n_local]."""
+        """x_sorted: tokens grouped by local expert; counts_cpu [n_local].
+
+        Training AND inference run through the hand-written MFMA grouped
+        GEMM (fwd: grouped_gemm.hip, bwd dX/dW: grouped_gemm_bwd.hip) —
+        the reference trains through grouped_gemm.ops.gmm
+        (experts.py:194-207); the per-expert rocBLAS loop below is only
+        the CPU/odd-shape fallback."""
+        import os
+
         use_grouped = (
             x_sorted.is_cuda
             and x_sorted.dtype == torch.bfloat16
-            and not torch.is_grad_enabled()
+            and os.environ.get("REALHF_AMD_MOE_LOOP") != "1"
         )
         if use_grouped:
-            from realhf_amd import ops as _ops_pkg
-
-            C = _ops_pkg.require_hip()
             wg = self._expert_weight_stack("gate")
             wu = self._expert_weight_stack("up")
             wd = self._expert_weight_stack("down")
             idim = wg.shape[1] if wg is not None else 0
+            hid = x_sorted.shape[1]
             if (
                 wg is not None and wu is not None and wd is not None
-                and x_sorted.shape[1] % 32 == 0 and idim % 64 == 0
-                and x_sorted.shape[1] % 64 == 0
+                # fwd needs K%32, N%64; bwd dX needs K%64, dW needs N%64 —
+                # both dims multiple of 64 covers every direction
+                and hid % 64 == 0 and idim % 64 == 0
             ):
-                gate = C.grouped_gemm(x_sorted, wg, counts_cpu)
-                up = C.grouped_gemm(x_sorted, wu, counts_cpu)
+                gate = _GroupedGemm.apply(x_sorted, wg, counts_cpu)
+                up = _GroupedGemm.apply(x_sorted, wu, counts_cpu)
                 act = (F.silu(gate.float()) * up.float()).to(x_sorted.dtype)
-                return C.grouped_gemm(act, wd, counts_cpu)
+                return _GroupedGemm.apply(act, wd, counts_cpu)
         outs = torch.empty(
             x_sorted.shape[0],
             self.p[f"{self.i}.mlp.experts.{self.local_e0}.down.weight"].shape[0],

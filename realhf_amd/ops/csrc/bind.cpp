@@ -31,6 +31,10 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor tr16_probe(torch::Tensor addr_elem);
 torch::Tensor grouped_gemm(torch::Tensor x, torch::Tensor w,
                            torch::Tensor seg_lens_cpu);
+torch::Tensor grouped_gemm_dx(torch::Tensor dout, torch::Tensor w,
+                              torch::Tensor seg_lens_cpu);
+torch::Tensor grouped_gemm_dw(torch::Tensor dout, torch::Tensor x,
+                              torch::Tensor seg_lens_cpu, long E);
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           torch::Tensor out32_ws, long splitk,
                           c10::optional<torch::Tensor> residual);
@@ -78,6 +82,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tr16_probe", &tr16_probe);
   m.def("rope_qkv_decode", &rope_qkv_decode);
   m.def("grouped_gemm", &grouped_gemm);
+  m.def("grouped_gemm_dx", &grouped_gemm_dx);
+  m.def("grouped_gemm_dw", &grouped_gemm_dw);
   m.def("skinny_gemm", &skinny_gemm);
   m.def("skinny_gemm2", &skinny_gemm2);
   m.def("skinny_gemm_nc", &skinny_gemm_nc);

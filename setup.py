@@ -22,6 +22,7 @@ SRC = [
     "realhf_amd/ops/csrc/attn_decode.hip",
     "realhf_amd/ops/csrc/rope_decode.hip",
     "realhf_amd/ops/csrc/grouped_gemm.hip",
+    "realhf_amd/ops/csrc/grouped_gemm_bwd.hip",
     "realhf_amd/ops/csrc/skinny_gemm.hip",
     "realhf_amd/ops/csrc/attn_varlen.hip",
     "realhf_amd/ops/csrc/all_reduce.hip",

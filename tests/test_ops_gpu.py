@@ -284,11 +284,11 @@ def test_moe_grouped_path_matches_loop():
     cu = torch.tensor([0, 48, 96], dtype=torch.int32, device="cuda")
     with torch.no_grad():
         out_grouped = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=48)
-        with torch.enable_grad():  # forces the per-expert loop path
-            m.allocate_grad_buffer()
-            for k, p in m._params.items():
-                p.requires_grad_(True)
+        os.environ["REALHF_AMD_MOE_LOOP"] = "1"  # force per-expert loop ref
+        try:
             out_loop = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=48)
+        finally:
+            del os.environ["REALHF_AMD_MOE_LOOP"]
     torch.testing.assert_close(
         out_grouped.float(), out_loop.float(), atol=5e-2, rtol=5e-2
     )
@@ -487,3 +487,68 @@ def test_attn_decode_sliding_window():
                                    rtol=3e-2)
     full = C.attn_decode(q, kc, vc, lens, scale, 0)
     assert not torch.allclose(C.attn_decode(q, kc, vc, lens, scale, 16), full)
+
+
+def test_grouped_gemm_backward_kernels():
+    """dX/dW MFMA kernels vs fp32 torch references (per expert)."""
+    torch.manual_seed(14)
+    E, N, K = 4, 256, 192
+    lens = torch.tensor([70, 0, 129, 33], dtype=torch.int32)
+    total = int(lens.sum())
+    x = (torch.randn(total, K, device="cuda") * 0.5).to(torch.bfloat16)
+    w = (torch.randn(E, N, K, device="cuda") * 0.5).to(torch.bfloat16)
+    dout = (torch.randn(total, N, device="cuda") * 0.5).to(torch.bfloat16)
+    dx = C.grouped_gemm_dx(dout, w, lens)
+    dw = C.grouped_gemm_dw(dout, x, lens, E)
+    off = 0
+    for e in range(E):
+        n = int(lens[e])
+        dw_ref = (dout[off:off + n].float().t() @ x[off:off + n].float()
+                  if n else torch.zeros(N, K, device="cuda"))
+        torch.testing.assert_close(dw[e].float(), dw_ref, atol=0.5, rtol=3e-2)
+        if n:
+            dx_ref = dout[off:off + n].float() @ w[e].float()
+            torch.testing.assert_close(dx[off:off + n].float(), dx_ref,
+                                       atol=0.5, rtol=3e-2)
+        off += n
+
+
+def test_moe_training_grouped_backward_matches_loop():
+    """A full MoE train forward+backward on the native grouped path must
+    produce the same flat-buffer gradients as the per-expert loop path
+    (VERDICT round-1 item #4: Mixtral GRPO trains off the native kernel
+    without this)."""
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.models.real_model import ReaLModel
+
+    fam = hf_reg.get_family("mixtral")
+    cfg = fam.make_test_config(
+        n_layers=1, hidden_dim=64, n_heads=1, n_kv_heads=1, vocab_size=128,
+        intermediate_dim=128, head_dim=64,
+    )
+
+    def run(force_loop):
+        torch.manual_seed(15)
+        m = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
+        m.random_init()
+        m.train()
+        m.allocate_grad_buffer()
+        for k, p in m._params.items():
+            p.requires_grad_(True)
+            p.grad = m.grad_view(k)  # autograd accumulates into flat_grad
+        toks = torch.randint(0, 128, (96,), device="cuda")
+        cu = torch.tensor([0, 48, 96], dtype=torch.int32, device="cuda")
+        if force_loop:
+            os.environ["REALHF_AMD_MOE_LOOP"] = "1"
+        try:
+            out = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=48)
+            out.float().square().mean().backward()
+        finally:
+            os.environ.pop("REALHF_AMD_MOE_LOOP", None)
+        return out.detach().float(), m.flat_grad.detach().float().clone()
+
+    out_g, grad_g = run(force_loop=False)
+    out_l, grad_l = run(force_loop=True)
+    torch.testing.assert_close(out_g, out_l, atol=5e-2, rtol=5e-2)
+    # bf16 kernels vs bf16 loop: small elementwise tolerance on grads
+    torch.testing.assert_close(grad_g, grad_l, atol=8e-2, rtol=8e-2)
