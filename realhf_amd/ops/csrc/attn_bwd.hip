@@ -60,7 +60,7 @@ __global__ __launch_bounds__(64 * BW_WAVES, 2) void attn_varlen_bwd_kernel(
     const int* __restrict__ cu_seqlens, const int* __restrict__ blk_seq,
     const int* __restrict__ blk_kstart, float* __restrict__ dq32,
     float* __restrict__ dk32, float* __restrict__ dv32,
-    int nq, int nkv, float scale, bool causal) {
+    int nq, int nkv, float scale, bool causal, int window) {
   constexpr int HDCH = HD / 32;
   const int blk = blockIdx.x;
   const int qh = blockIdx.y;
@@ -113,7 +113,11 @@ __global__ __launch_bounds__(64 * BW_WAVES, 2) void attn_varlen_bwd_kernel(
   }
 
   const int q_begin = causal ? (k0 / BW_QT) * BW_QT : 0;
-  for (int q0 = q_begin; q0 < L; q0 += BW_QT) {
+  // sliding window (mistral, fwd-matching semantics: query q sees keys
+  // (q-window, q]): key kidx is consumed only by queries < kidx + window,
+  // so this kv block's last relevant query is k0 + BW_KV - 1 + window - 1
+  const int q_end = (window > 0) ? min(L, k0 + BW_KV + window - 1) : L;
+  for (int q0 = q_begin; q0 < q_end; q0 += BW_QT) {
     const int qn = min(BW_QT, L - q0);
     // ---- stage Q/dO tile + lse + D --------------------------------
     __syncthreads();
@@ -159,7 +163,8 @@ __global__ __launch_bounds__(64 * BW_WAVES, 2) void attn_varlen_bwd_kernel(
       #pragma unroll
       for (int r = 0; r < 4; r++) {
         const int kidx = k0 + w * 16 + g * 4 + r;
-        bool ok = (kidx < L) && (qidx < L) && (!causal || kidx <= qidx);
+        bool ok = (kidx < L) && (qidx < L) && (!causal || kidx <= qidx)
+                  && (window <= 0 || kidx > qidx - window);
         float p = ok ? __expf(st[qs][r] * scale - lse_s[qs * 16 + i16]) : 0.f;
         float ds = ok ? p * (dpt[qs][r] - d_s[qs * 16 + i16]) * scale : 0.f;
         pt_s[w][g * 4 + r][qs * 16 + i16] = (__bf16)p;
@@ -235,7 +240,7 @@ __global__ __launch_bounds__(64 * BW_WAVES, 2) void attn_varlen_bwd_kernel(
 std::vector<torch::Tensor> attn_varlen_bwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor dout,
     torch::Tensor lse, torch::Tensor Dsum, torch::Tensor cu_seqlens,
-    bool causal, double scale) {
+    bool causal, double scale, long window) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous() &&
               dout.is_contiguous());
@@ -271,7 +276,7 @@ std::vector<torch::Tensor> attn_varlen_bwd(
       Dsum.data_ptr<float>(), cu_dev.data_ptr<int>(),
       bseq_t.data_ptr<int>(), bks_t.data_ptr<int>(),
       dq32.data_ptr<float>(), dk32.data_ptr<float>(), dv32.data_ptr<float>(),
-      nq, nkv, (float)scale, causal);
+      nq, nkv, (float)scale, causal, (int)window);
   } else {
     hipLaunchKernelGGL((attn_varlen_bwd_kernel<64>), grid,
       dim3(64 * BW_WAVES), 0, cur_stream(), (const bf16*)q.data_ptr(),
@@ -280,7 +285,7 @@ std::vector<torch::Tensor> attn_varlen_bwd(
       Dsum.data_ptr<float>(), cu_dev.data_ptr<int>(),
       bseq_t.data_ptr<int>(), bks_t.data_ptr<int>(),
       dq32.data_ptr<float>(), dk32.data_ptr<float>(), dv32.data_ptr<float>(),
-      nq, nkv, (float)scale, causal);
+      nq, nkv, (float)scale, causal, (int)window);
   }
   CHECK_CUDA_OK();
   return {dq32, dk32, dv32};

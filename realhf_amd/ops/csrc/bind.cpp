@@ -43,7 +43,7 @@ torch::Tensor skinny_gemm_nc(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> attn_varlen_bwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor dout,
     torch::Tensor lse, torch::Tensor Dsum, torch::Tensor cu_seqlens,
-    bool causal, double scale);
+    bool causal, double scale, long window);
 int64_t xgmi_create(int64_t rank, int64_t world, int64_t capacity);
 std::vector<py::bytes> xgmi_handles(int64_t h);
 void xgmi_connect(int64_t h, const std::vector<std::string>& data_handles,

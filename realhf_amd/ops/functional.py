@@ -293,17 +293,18 @@ class _AttnVarlenFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out):
         q, k, v, out, lse, cu_seqlens = ctx.saved_tensors
-        if (os.environ.get("REALHF_AMD_NO_HIP_ATTN_BWD") != "1"
-                and ctx.window is None):
+        if os.environ.get("REALHF_AMD_NO_HIP_ATTN_BWD") != "1":
             # hand-written MFMA backward (attn_bwd.hip): kv-stationary,
-            # dQ via fp32 atomics, per-q-head dK/dV reduced here for GQA.
+            # dQ via fp32 atomics, per-q-head dK/dV reduced here for GQA;
+            # sliding-window via key-range restriction like the forward.
             # In-context A/B: 3.51 vs 3.19 samples/s over the padded
             # recompute fallback (REALHF_AMD_NO_HIP_ATTN_BWD=1).
             C = _ops.require_hip()
             dout = grad_out.contiguous()
             dsum = (dout.float() * out.float()).sum(-1)  # [total, nq]
             dq32, dk32, dv32 = C.attn_varlen_bwd(
-                q, k, v, dout, lse, dsum, cu_seqlens, ctx.causal, ctx.scale
+                q, k, v, dout, lse, dsum, cu_seqlens, ctx.causal, ctx.scale,
+                int(ctx.window or 0),
             )
             nq, nkv = q.shape[1], k.shape[1]
             rep = nq // nkv
