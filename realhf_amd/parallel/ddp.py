@@ -184,11 +184,42 @@ class ZeRO1Optimizer:
             self._armed = False
         else:
             self.master = self._param_padded[s0:s1].to(torch.float32).to(state_dev)
+        if cfg.offload and dev.type == "cuda":
+            # pinned host states: the offloaded step streams chunks
+            # through the GPU (H2D -> fused AdamW -> D2H) on a side
+            # stream, double-buffered — PCIe-bound instead of CPU-bound
+            # (reference counterpart: DeepSpeed ZeRO-offload,
+            # deepspeed.py:276-359; round-1 blocked on a sync .cpu()).
+            try:
+                self.master = self.master.pin_memory()
+            except RuntimeError:
+                logger.warning("pin_memory failed; offload uses pageable host mem")
         self.exp_avg = torch.zeros_like(self.master)
         self.exp_avg_sq = torch.zeros_like(self.master)
+        if cfg.offload and dev.type == "cuda" and self.master.is_pinned():
+            try:
+                self.exp_avg = self.exp_avg.pin_memory()
+                self.exp_avg_sq = self.exp_avg_sq.pin_memory()
+            except RuntimeError:
+                pass
         self.grad_shard = torch.empty(
             self.shard_size, dtype=model.flat_param.dtype, device=dev
         )
+        if cfg.offload and dev.type == "cuda":
+            self._off_chunk = int(os.environ.get(
+                "REALHF_AMD_OFFLOAD_CHUNK", 32 * 1024 * 1024
+            ))  # fp32 elems: 128 MB/tensor default
+            self._off_stream = torch.cuda.Stream()
+            self._off_stage = [
+                {
+                    "master": torch.empty(self._off_chunk, dtype=torch.float32, device=dev),
+                    "m": torch.empty(self._off_chunk, dtype=torch.float32, device=dev),
+                    "v": torch.empty(self._off_chunk, dtype=torch.float32, device=dev),
+                    "h2d": torch.cuda.Event(),
+                    "done": torch.cuda.Event(),
+                }
+                for _ in range(2)
+            ]
 
         self._grad_views_attached = False
 
@@ -319,16 +350,25 @@ class ZeRO1Optimizer:
         s0, s1 = self.shard_bounds
         param_shard = (self._shard_bf16 if self.overlap_comm
                        else self._param_padded[s0:s1])
-        g_in = gshard.float().cpu() if cfg.offload else gshard
-        ops.fused_adamw(
-            self.master, g_in, self.exp_avg, self.exp_avg_sq,
-            lr=lr, beta1=cfg.beta1, beta2=cfg.beta2, eps=cfg.eps,
-            weight_decay=cfg.weight_decay, step=self.step_count,
-            bf16_out=None if cfg.offload else param_shard,
-            grad_scale=gscale,
-        )
-        if cfg.offload:
+        if cfg.offload and dev.type == "cuda":
+            self._adamw_offloaded(gshard, param_shard, lr, gscale)
+        elif cfg.offload:
+            # CPU path (tests / no GPU): states live on host, torch math
+            g_in = gshard.float().cpu()
+            ops.fused_adamw(
+                self.master, g_in, self.exp_avg, self.exp_avg_sq,
+                lr=lr, beta1=cfg.beta1, beta2=cfg.beta2, eps=cfg.eps,
+                weight_decay=cfg.weight_decay, step=self.step_count,
+                bf16_out=None, grad_scale=gscale,
+            )
             param_shard.copy_(self.master.to(param_shard.dtype))
+        else:
+            ops.fused_adamw(
+                self.master, gshard, self.exp_avg, self.exp_avg_sq,
+                lr=lr, beta1=cfg.beta1, beta2=cfg.beta2, eps=cfg.eps,
+                weight_decay=cfg.weight_decay, step=self.step_count,
+                bf16_out=param_shard, grad_scale=gscale,
+            )
 
         # 4. all-gather updated params
         if self.overlap_comm:
@@ -348,6 +388,47 @@ class ZeRO1Optimizer:
         if grad_norm is not None:
             out["grad_norm"] = grad_norm
         return out
+
+    @torch.no_grad()
+    def _adamw_offloaded(self, gshard, param_shard, lr, gscale):
+        """Pipelined offloaded AdamW: host-resident fp32 states stream
+        through the GPU in double-buffered chunks.  Per chunk i:
+        H2D(master,m,v) on the side stream -> fused AdamW on the default
+        stream (writes the bf16 param-shard chunk in place) -> D2H of the
+        updated states back to pinned host memory.  Chunk i+1's uploads
+        overlap chunk i's compute+downloads, so the step runs at PCIe
+        bandwidth instead of blocking on a whole-shard sync copy."""
+        cfg = self.cfg
+        n = self.shard_size
+        cur = torch.cuda.current_stream()
+        cs = self._off_stream
+        cs.wait_stream(cur)  # grads must be final before chunks start
+        chunks = list(range(0, n, self._off_chunk))
+        for idx, c0 in enumerate(chunks):
+            c1 = min(n, c0 + self._off_chunk)
+            l = c1 - c0
+            st = self._off_stage[idx % 2]
+            with torch.cuda.stream(cs):
+                # buffer reuse is ordered by the side stream itself: the
+                # previous D2H of this buffer was issued on cs earlier
+                st["master"][:l].copy_(self.master[c0:c1], non_blocking=True)
+                st["m"][:l].copy_(self.exp_avg[c0:c1], non_blocking=True)
+                st["v"][:l].copy_(self.exp_avg_sq[c0:c1], non_blocking=True)
+                st["h2d"].record(cs)
+            cur.wait_event(st["h2d"])
+            ops.fused_adamw(
+                st["master"][:l], gshard[c0:c1], st["m"][:l], st["v"][:l],
+                lr=lr, beta1=cfg.beta1, beta2=cfg.beta2, eps=cfg.eps,
+                weight_decay=cfg.weight_decay, step=self.step_count,
+                bf16_out=param_shard[c0:c1], grad_scale=gscale,
+            )
+            st["done"].record(cur)
+            with torch.cuda.stream(cs):
+                cs.wait_event(st["done"])
+                self.master[c0:c1].copy_(st["master"][:l], non_blocking=True)
+                self.exp_avg[c0:c1].copy_(st["m"][:l], non_blocking=True)
+                self.exp_avg_sq[c0:c1].copy_(st["v"][:l], non_blocking=True)
+        cur.wait_stream(cs)  # states must be home before checkpoint/next step
 
     # ------------------------------------------------------------------
     def state_dict(self):

@@ -600,3 +600,44 @@ def test_attn_varlen_bwd_sliding_window():
         dq_f, _, _ = C.attn_varlen_bwd(q, k, v, dout.contiguous(), lse,
                                        dsum, cu, True, scale, 0)
         assert not torch.allclose(dq32, dq_f)
+
+
+def test_offloaded_adamw_matches_resident():
+    """Pipelined host-offloaded AdamW (chunked H2D/compute/D2H on a side
+    stream) must produce the same parameters as the GPU-resident
+    optimizer (70B-tier machinery, VERDICT round-1 item #6)."""
+    from realhf_amd.models.hf.llama import make_test_config
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel.ddp import OptimizerConfig, ZeRO1Optimizer
+
+    os.environ["REALHF_AMD_OFFLOAD_CHUNK"] = "100000"  # force many chunks
+
+    def run(offload):
+        torch.manual_seed(33)
+        cfg = make_test_config(n_layers=2, hidden_dim=128, n_heads=2,
+                               n_kv_heads=2, head_dim=64,
+                               intermediate_dim=256, vocab_size=256)
+        cfg.family = "llama"
+        m = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
+        m.random_init()
+        opt = ZeRO1Optimizer(
+            m, OptimizerConfig(lr=1e-2, warmup_steps_proportion=0.0,
+                               offload=offload),
+            total_train_steps=10,
+        )
+        for _ in range(3):  # multi-step: state round-trips must persist
+            opt.zero_grad()
+            toks = torch.randint(0, 256, (64,), device="cuda")
+            cu = torch.tensor([0, 64], dtype=torch.int32, device="cuda")
+            out = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=64)
+            out.float().square().mean().backward()
+            opt.step()
+        torch.cuda.synchronize()
+        return m.flat_param.detach().clone()
+
+    try:
+        p_resident = run(offload=False)
+        p_offload = run(offload=True)
+    finally:
+        del os.environ["REALHF_AMD_OFFLOAD_CHUNK"]
+    torch.testing.assert_close(p_offload, p_resident, atol=0, rtol=0)
