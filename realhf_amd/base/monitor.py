@@ -137,6 +137,48 @@ def gpu_memory_stats() -> Dict[str, float]:
     }
 
 
+def estimate_mfc_flops(itype, cfg, seqlens, out_seqlens=None) -> float:
+    """Model FLOPs of one MFC execution over a packed shard (reference:
+    InterfaceDataAmount + caculate_llama_*_flops, master_worker.py:233 /
+    monitor.py:277).  `seqlens`: input per-seq lengths; `out_seqlens`:
+    full prompt+generation lengths for GENERATE MFCs."""
+    tokens = float(sum(seqlens))
+    ssq = float(sum(l * l for l in seqlens))
+    kw = dict(
+        n_layers=cfg.n_layers, hidden=cfg.hidden_dim,
+        intermediate=cfg.intermediate_dim, vocab=cfg.vocab_size,
+        n_heads=cfg.n_heads, n_kv_heads=cfg.n_kv_heads,
+        head_dim=cfg.head_dim,
+    )
+    name = getattr(itype, "name", str(itype)).upper()
+    if name == "TRAIN_STEP":
+        return dense_transformer_flops(total_tokens=tokens,
+                                       sum_sq_seqlens=ssq, backward=True, **kw)
+    if name == "GENERATE" and out_seqlens is not None:
+        new_total = float(sum(o - i for o, i in zip(out_seqlens, seqlens)))
+        # decode attention reads ctx t for each generated position t
+        sum_ctx = float(sum((o * o - i * i) / 2.0
+                            for o, i in zip(out_seqlens, seqlens)))
+
+        class _C:
+            pass
+
+        c = _C()
+        for k, v in (("n_layers", cfg.n_layers), ("hidden_dim", cfg.hidden_dim),
+                     ("intermediate_dim", cfg.intermediate_dim),
+                     ("vocab_size", cfg.vocab_size), ("n_heads", cfg.n_heads),
+                     ("n_kv_heads", cfg.n_kv_heads), ("head_dim", cfg.head_dim)):
+            setattr(c, k, v)
+        prefill_flops = dense_transformer_flops(
+            total_tokens=tokens, sum_sq_seqlens=ssq, **kw)
+        decode_flops = dense_transformer_flops(
+            total_tokens=new_total, sum_sq_seqlens=0.0, **kw)
+        return prefill_flops + decode_flops + (
+            4 * cfg.n_heads * cfg.head_dim * sum_ctx * cfg.n_layers)
+    return dense_transformer_flops(total_tokens=tokens, sum_sq_seqlens=ssq,
+                                   **kw)
+
+
 def log_tflops(name: str, flops: float, seconds: float, n_gpus: int = 1):
     tf = flops / seconds / 1e12 / n_gpus
     util = tf / MI355X_BF16_DENSE_TFLOPS * 100

@@ -114,6 +114,7 @@ class DFGExecutor:
         self._plan = self._static_plan()
         self._make_groups()
         self._mfc_wall: Dict[str, float] = {}
+        self._mfc_flops: Dict[str, float] = {}
 
     # ----------------------------------------------------- static planning
     def _register_mfc_grids(self):
@@ -297,6 +298,8 @@ class DFGExecutor:
                 t0 = time.time()
                 with constants.model_scope(plan.scope):
                     local_out = self._exec_mfc(mfc, alloc, store, local_stats)
+                if torch.cuda.is_available():
+                    torch.cuda.synchronize()
                 self._mfc_wall[mfc.name] = time.time() - t0
 
             # 4. merge outputs across this mesh's DP heads
@@ -358,7 +361,32 @@ class DFGExecutor:
             # every DP head contributes; step-end merge averages over them
             if g.tp_rank == 0 and g.pp_rank == g.pp_size - 1:
                 local_stats[mfc.name] = res
+        self._record_flops(mfc, getattr(real, "config", None), shard, res, g)
         return local_out
+
+    def _record_flops(self, mfc, cfg, shard, res, g):
+        """Per-MFC model-FLOPs estimate (reference: master_worker's
+        TFLOP/s line, master_worker.py:1461-1488) — merged into the step
+        stats as <mfc>/tflops_per_gpu at the step boundary."""
+        if cfg is None:
+            return
+        try:
+            from realhf_amd.base.monitor import estimate_mfc_flops
+
+            key = ("packed_input_ids" if "packed_input_ids" in shard.keys
+                   else next(iter(shard.keys)))
+            seqlens = [sum(x) for x in shard.seqlens[key]]
+            out_seqlens = None
+            if (mfc.interface_type == ModelInterfaceType.GENERATE
+                    and isinstance(res, SequenceSample)
+                    and "packed_input_ids" in res.keys):
+                out_seqlens = [sum(x) for x in
+                               res.seqlens["packed_input_ids"]]
+            fl = estimate_mfc_flops(mfc.interface_type, cfg, seqlens,
+                                    out_seqlens)
+            self._mfc_flops[mfc.name] = fl / max(1, g.tp_size * g.pp_size)
+        except Exception:  # FLOPs logging must never break the step
+            self._mfc_flops.pop(mfc.name, None)
 
     def _assemble_input(self, mfc: MFCDef,
                         store: Dict[str, SequenceSample]) -> SequenceSample:
@@ -412,6 +440,9 @@ class DFGExecutor:
         }
         for mfc, w in self._mfc_wall.items():
             flat[f"{mfc}/wall_s"] = w
+            fl = self._mfc_flops.get(mfc)
+            if fl and w > 0:
+                flat[f"{mfc}/tflops_per_gpu"] = fl / w / 1e12
         if not dist.is_initialized():
             return flat
         bucket: List = [None] * self.world

@@ -582,14 +582,29 @@ class Trainer:
                 self.global_step += 1
                 if self.rank == 0:
                     toks = sum(batch.main_seqlens())
+                    perf = {k: v for k, v in stats.items()
+                            if k.endswith(("/wall_s", "/tflops_per_gpu"))}
                     logger.info(
                         "epoch %d step %d (global %d): %.2fs (%.0f tok/s, "
                         "%.2f samples/s) %s",
                         epoch, i, self.global_step, dt, toks / dt,
                         batch.bs / dt,
                         {k: round(v, 4) for k, v in stats.items()
-                         if isinstance(v, float)},
+                         if isinstance(v, float) and k not in perf},
                     )
+                    if perf:
+                        # reference-style per-MFC perf line
+                        # (master_worker.py:1461-1488)
+                        logger.info(
+                            "perf: %s",
+                            " | ".join(
+                                f"{m}: {stats.get(m + '/wall_s', 0):.2f}s"
+                                + (f" {stats[m + '/tflops_per_gpu']:.0f} TF/s"
+                                   if m + "/tflops_per_gpu" in stats else "")
+                                for m in sorted({k.split("/")[0]
+                                                 for k in perf})
+                            ),
+                        )
                 if ctrl.save_freq_steps and self.global_step % ctrl.save_freq_steps == 0:
                     self.save()
                     self.save_recover_ckpt(epoch, i + 1)
