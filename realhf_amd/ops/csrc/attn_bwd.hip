@@ -34,6 +34,29 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
 typedef __attribute__((address_space(3))) bf16x4v lds_b64_t;
 
+static __global__ void bw_build_blk_offsets(const int* __restrict__ cu,
+                                             int bs, int blk,
+                                             int* __restrict__ out) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    int acc = 0;
+    for (int i = 0; i < bs; i++) {
+      out[i] = acc;
+      int L = cu[i + 1] - cu[i];
+      acc += (L + blk - 1) / blk;
+    }
+    out[bs] = acc;
+  }
+}
+
+DEVINL int bw_blk_lookup(const int* __restrict__ off, int bs, int blk_id) {
+  int lo = 0, hi = bs;  // largest i with off[i] <= blk_id
+  while (lo + 1 < hi) {
+    int mid = (lo + hi) >> 1;
+    if (off[mid] <= blk_id) lo = mid; else hi = mid;
+  }
+  return lo;
+}
+
 #define BW_WAVES 4
 #define BW_KV 64      // keys per workgroup (16 per wave)
 #define BW_QT 32      // queries per tile
@@ -57,16 +80,17 @@ __global__ __launch_bounds__(64 * BW_WAVES, 2) void attn_varlen_bwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const bf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ Dsum,
-    const int* __restrict__ cu_seqlens, const int* __restrict__ blk_seq,
-    const int* __restrict__ blk_kstart, float* __restrict__ dq32,
+    const int* __restrict__ cu_seqlens, const int* __restrict__ blk_offsets,
+    int n_seqs, float* __restrict__ dq32,
     float* __restrict__ dk32, float* __restrict__ dv32,
     int nq, int nkv, float scale, bool causal, int window) {
   constexpr int HDCH = HD / 32;
   const int blk = blockIdx.x;
+  if (blk >= blk_offsets[n_seqs]) return;  // over-provisioned grid tail
   const int qh = blockIdx.y;
   const int kvh = qh / (nq / nkv);
-  const int seq = blk_seq[blk];
-  const int k0 = blk_kstart[blk];
+  const int seq = bw_blk_lookup(blk_offsets, n_seqs, blk);
+  const int k0 = (blk - blk_offsets[seq]) * BW_KV;
   const int s0 = cu_seqlens[seq], s1 = cu_seqlens[seq + 1];
   const int L = s1 - s0;
   const int lane = threadIdx.x & 63;
@@ -246,27 +270,17 @@ std::vector<torch::Tensor> attn_varlen_bwd(
               dout.is_contiguous());
   int total = q.size(0), nq = q.size(1), hd = q.size(2);
   int nkv = k.size(1);
-  auto cu_cpu = cu_seqlens.to(torch::kInt).cpu();
   auto cu_dev = cu_seqlens.to(torch::kInt).to(q.device());
-  int bs = cu_cpu.numel() - 1;
-  const int* cu = cu_cpu.data_ptr<int>();
-  std::vector<int> bseq, bks;
-  for (int i = 0; i < bs; i++) {
-    int L = cu[i + 1] - cu[i];
-    for (int ks = 0; ks < L; ks += BW_KV) {
-      bseq.push_back(i);
-      bks.push_back(ks);
-    }
-  }
-  auto bseq_t = torch::from_blob(bseq.data(), {(long)bseq.size()},
-                                 torch::kInt).to(q.device());
-  auto bks_t = torch::from_blob(bks.data(), {(long)bks.size()},
-                                torch::kInt).to(q.device());
+  int bs = cu_dev.numel() - 1;
+  auto iopts = torch::TensorOptions().dtype(torch::kInt).device(q.device());
+  auto blk_off = torch::empty({(long)bs + 1}, iopts);
+  hipLaunchKernelGGL(bw_build_blk_offsets, dim3(1), dim3(64), 0,
+    cur_stream(), cu_dev.data_ptr<int>(), bs, BW_KV, blk_off.data_ptr<int>());
   auto f32 = q.options().dtype(torch::kFloat);
   auto dq32 = torch::zeros({(long)total, (long)nq, (long)hd}, f32);
   auto dk32 = torch::empty({(long)total, (long)nq, (long)hd}, f32);
   auto dv32 = torch::empty({(long)total, (long)nq, (long)hd}, f32);
-  dim3 grid((unsigned)bseq.size(), nq);
+  dim3 grid((unsigned)(total / BW_KV + bs), nq);
   TORCH_CHECK(hd == 128 || hd == 64, "attn_varlen_bwd: hd 64/128 only");
   if (hd == 128) {
     hipLaunchKernelGGL((attn_varlen_bwd_kernel<128>), grid,
@@ -274,7 +288,7 @@ std::vector<torch::Tensor> attn_varlen_bwd(
       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
       (const bf16*)dout.data_ptr(), lse.data_ptr<float>(),
       Dsum.data_ptr<float>(), cu_dev.data_ptr<int>(),
-      bseq_t.data_ptr<int>(), bks_t.data_ptr<int>(),
+      blk_off.data_ptr<int>(), bs,
       dq32.data_ptr<float>(), dk32.data_ptr<float>(), dv32.data_ptr<float>(),
       nq, nkv, (float)scale, causal, (int)window);
   } else {
@@ -283,7 +297,7 @@ std::vector<torch::Tensor> attn_varlen_bwd(
       (const bf16*)k.data_ptr(), (const bf16*)v.data_ptr(),
       (const bf16*)dout.data_ptr(), lse.data_ptr<float>(),
       Dsum.data_ptr<float>(), cu_dev.data_ptr<int>(),
-      bseq_t.data_ptr<int>(), bks_t.data_ptr<int>(),
+      blk_off.data_ptr<int>(), bs,
       dq32.data_ptr<float>(), dk32.data_ptr<float>(), dv32.data_ptr<float>(),
       nq, nkv, (float)scale, causal, (int)window);
   }

@@ -1,0 +1,81 @@
+"""Multi-rank RCCL validation on ONE physical GPU (VERDICT round-1 item
+#2): two processes share cuda:0 over RCCL and run a bench-shaped PPO
+step through the concurrent executor — ZeRO-1 with bucketed overlap ON,
+the asymmetric heuristic (critic_inf rank 0 | rew_inf rank 1, reward
+model instantiated on rank 1 only), device-native sample
+gather/broadcast, and a DP-replica weight checksum after the optimizer
+step.  This is the closest 1-GPU proxy for the driver's 8-GPU scale run."""
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from realhf_amd.base.testing import LocalMultiProcessTest
+
+pytestmark = [pytest.mark.gpu, pytest.mark.distributed]
+
+if not torch.cuda.is_available():
+    pytest.skip("needs GPU", allow_module_level=True)
+
+
+def _worker(data, fileroot):
+    import torch.distributed as dist
+
+    from realhf_amd.api.config import ModelName
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    torch.cuda.set_device(0)  # both ranks share the one physical GPU
+    os.environ["LOCAL_RANK"] = "0"
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    os.environ["REALHF_AMD_ZERO_OVERLAP"] = "1"
+    try:
+        cfg = PPOConfig(experiment_name="t-mr", trial_name="gpu", n_gpus=2)
+        cfg.allocation_mode = "heuristic"
+        cfg.dataset.type_ = "prompt"
+        cfg.dataset.path = data
+        cfg.dataset.train_bs_n_seqs = 8
+        cfg.dataset.max_prompt_len = 8
+        cfg.ppo.gen.max_new_tokens = 8
+        cfg.ppo.gen.min_new_tokens = 2
+        cfg.ppo.gen.use_hip_graph = False
+        cfg.ppo.ppo_n_minibatches = 2
+        cfg.exp_ctrl.benchmark_steps = 2
+        t = Trainer(cfg)
+        # asymmetric shape engaged
+        assert t.executor._plan["critic_inf"].mesh == (0,)
+        assert t.executor._plan["rew_inf"].mesh == (1,)
+        rew = ModelName("rew", 0)
+        assert (rew in t.models) == (dist.get_rank() == 1)
+        # ZeRO overlap active on the trainable engines
+        for name in t.built.trainable:
+            if name in t.models:
+                assert t.models[name].module.optimizer.overlap_comm
+        t.run()
+        # DP replicas must agree bit-for-bit after optimizer steps
+        for name in t.built.trainable:
+            if name not in t.models:
+                continue
+            fp = t.models[name].module.module.flat_param
+            cs = fp.float().sum()
+            lo, hi = cs.clone(), cs.clone()
+            dist.all_reduce(lo, op=dist.ReduceOp.MIN)
+            dist.all_reduce(hi, op=dist.ReduceOp.MAX)
+            assert torch.allclose(lo, hi, rtol=0, atol=0), (
+                f"{name}: DP replicas diverged ({lo} vs {hi})")
+        dist.barrier()
+    finally:
+        os.environ.pop("REALHF_AMD_ZERO_OVERLAP", None)
+
+
+def test_two_rank_rccl_ppo_heuristic_overlap(tmp_path):
+    rng = np.random.RandomState(5)
+    data = str(tmp_path / "p.jsonl")
+    with open(data, "w") as f:
+        for _ in range(16):
+            f.write(json.dumps(
+                {"input_ids": rng.randint(3, 60, size=8).tolist()}) + "\n")
+    LocalMultiProcessTest(2, _worker, data, str(tmp_path / "root"),
+                          backend="nccl", timeout_secs=600).launch()
