@@ -95,54 +95,74 @@ class BuiltExperiment:
 def _apply_search_allocation(cfg, world: int):
     """allocation_mode=search: pick per-MFC strategies with the MCMC
     search engine, then write them back into the model configs (reference:
-    apps/main.py experiment._search() -> search_rpc_allocations)."""
-    from realhf_amd.models import hf as _hf
+    apps/main.py experiment._search() -> search_rpc_allocations).
+    Handles any experiment type whose roles expose .parallel
+    (PPO, GRPO, DPO, ...)."""
     from realhf_amd.search.engine import MFCSpec, search_allocations
 
-    if not isinstance(cfg, PPOConfig):
-        logger.warning("allocation_mode=search currently tunes PPO only; "
-                       "using heuristic for %s", type(cfg).__name__)
+    role_cfgs = {
+        name: getattr(cfg, name)
+        for name in ("actor", "critic", "ref", "rew", "model")
+        if hasattr(cfg, name)
+    }
+    if not role_cfgs:
+        logger.warning("allocation_mode=search: no role configs on %s; "
+                       "using heuristic", type(cfg).__name__)
         cfg.allocation_mode = "heuristic"
         return
-
-    def spec_of(name, role, mc, itype, gen_tokens=0, avg_seqlen=None):
-        if mc.path:
-            rcfg = _hf.config_from_hf_path(mc.family, mc.path)
-        else:
-            rcfg = _hf.get_family(mc.family).make_test_config()
-        p = rcfg.param_count()
-        return MFCSpec(
-            name=name, role=role, interface_type=itype,
-            n_seqs=cfg.dataset.train_bs_n_seqs,
-            avg_seqlen=avg_seqlen or (cfg.dataset.max_prompt_len
-                                      + cfg.ppo.gen.max_new_tokens),
-            gen_tokens=gen_tokens, param_bytes=p * 2.0,
-            flops_per_token=2.0 * p,
-        )
 
     T = ModelInterfaceType
     tmp = build_experiment(
         dataclasses.replace(cfg, allocation_mode="heuristic"), world
     )
+    gen_max = (cfg.ppo.gen.max_new_tokens if hasattr(cfg, "ppo")
+               else getattr(getattr(cfg, "gen", None), "max_new_tokens", 0))
+    prompt_len = getattr(cfg.dataset, "max_prompt_len", None) or 512
+    full_len = prompt_len + gen_max if gen_max else (
+        getattr(cfg.dataset, "max_seqlen", None) or 1024)
     specs = {}
     for m in tmp.graph.mfcs:
-        role = m.model_name.role
-        mc = {"actor": cfg.actor, "critic": cfg.critic, "ref": cfg.ref,
-              "rew": cfg.rew}[role]
-        gen_toks = cfg.ppo.gen.max_new_tokens if m.interface_type == T.GENERATE else 0
-        avg = cfg.dataset.max_prompt_len if m.interface_type == T.GENERATE else None
-        specs[m.name] = spec_of(m.name, role, mc, m.interface_type,
-                                gen_tokens=gen_toks, avg_seqlen=avg)
+        rcfg = tmp.model_cfgs[m.model_name]
+        p = rcfg.param_count()
+        is_gen = m.interface_type == T.GENERATE
+        specs[m.name] = MFCSpec(
+            name=m.name, role=m.model_name.role,
+            interface_type=m.interface_type,
+            n_seqs=cfg.dataset.train_bs_n_seqs,
+            avg_seqlen=prompt_len if is_gen else full_len,
+            gen_tokens=gen_max if is_gen else 0,
+            param_bytes=p * 2.0, flops_per_token=2.0 * p,
+            n_layers=rcfg.n_layers, hidden_dim=rcfg.hidden_dim,
+            n_kv_heads=rcfg.n_kv_heads, head_dim=rcfg.head_dim,
+            gradient_checkpointing=getattr(
+                tmp.model_roles[m.model_name], "gradient_checkpointing",
+                False),
+            offload_optimizer=getattr(
+                tmp.model_roles[m.model_name].optimizer, "offload", False),
+            n_minibatches=getattr(getattr(cfg, "ppo", None),
+                                  "ppo_n_minibatches", 1),
+        )
+    trainable_roles = sorted({n.role for n in tmp.trainable})
     alloc, cost = search_allocations(
-        tmp.graph, specs, trainable_roles=["actor", "critic"], n_gpus=world,
+        tmp.graph, specs, trainable_roles=trainable_roles, n_gpus=world,
     )
     cfg.allocation_mode = "manual"
-    cfg.actor.parallel = alloc["actor_train"]
-    cfg.critic.parallel = alloc["critic_train"]
-    cfg.ref.parallel = alloc["ref_inf"]
-    cfg.rew.parallel = alloc["rew_inf"]
-    if alloc["actor_gen"] != alloc["actor_train"]:
-        cfg.actor.gen_parallel = alloc["actor_gen"]
+    # write back: a role's layout = its TRAIN_STEP mfc if trainable, else
+    # its first mfc; a GENERATE mfc with a different layout becomes the
+    # gen replica (realloc hooks engage in build_experiment)
+    by_role: Dict[str, List] = {}
+    for m in tmp.graph.mfcs:
+        by_role.setdefault(m.model_name.role, []).append(m)
+    for role, ms in by_role.items():
+        mc = role_cfgs.get(role) or role_cfgs.get("model")
+        if mc is None:
+            continue
+        train = [m for m in ms if m.interface_type == T.TRAIN_STEP]
+        main = train[0] if train else ms[0]
+        mc.parallel = alloc[main.name]
+        gen = [m for m in ms if m.interface_type == T.GENERATE]
+        if gen and alloc[gen[0].name] != alloc[main.name]:
+            mc.gen_parallel = alloc[gen[0].name]
     logger.info("search allocation (est %.3fs/step): %s", cost,
                 {k: str(v) for k, v in alloc.items()})
 

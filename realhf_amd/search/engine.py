@@ -79,6 +79,15 @@ class MFCSpec:
     gen_tokens: int = 0  # for GENERATE
     param_bytes: float = 0.0  # full model, bf16
     flops_per_token: float = 0.0  # fwd
+    # memory-model inputs (reference: estimate_rpc_memory_cost,
+    # estimate.py:387-450)
+    n_layers: int = 32
+    hidden_dim: int = 4096
+    n_kv_heads: int = 8
+    head_dim: int = 128
+    gradient_checkpointing: bool = True
+    offload_optimizer: bool = False
+    n_minibatches: int = 1
 
 
 def enumerate_strategies(n_gpus: int) -> List[ParallelismConfig]:
@@ -116,11 +125,36 @@ def estimate_time_s(m: MFCSpec, par: ParallelismConfig) -> float:
 
 
 def estimate_mem_bytes(m: MFCSpec, par: ParallelismConfig, trainable: bool) -> float:
-    shard = m.param_bytes / (par.tensor_parallel_size * par.pipeline_parallel_size)
+    """Peak device bytes of one MFC execution — params/grads/optimizer
+    (ZeRO-1 over dp, optionally host-offloaded), activations (with the
+    grad-checkpoint discount), and the decode KV cache for GENERATE
+    (reference: estimate_rpc_memory_cost, estimate.py:387-450 — extended
+    with the KV-cache/offload terms the reference marks TODO)."""
+    tp, pp, dp = (par.tensor_parallel_size, par.pipeline_parallel_size,
+                  par.data_parallel_size)
+    shard = m.param_bytes / (tp * pp)
+    seqs_per_dp = max(1, m.n_seqs // dp)
+    mem = shard  # bf16 params
     if trainable:
-        # bf16 params + bf16 grads + fp32 master/m/v sharded over dp
-        return shard * 2 + shard * 6 / par.data_parallel_size
-    return shard
+        mem += shard  # bf16 grads
+        if not m.offload_optimizer:
+            mem += shard * 6 / dp  # fp32 master + m + v over ZeRO dp
+        # activations of one microbatch: checkpointing keeps ~2 tensors
+        # per layer boundary, else ~14 per layer (attn+mlp intermediates)
+        mb_tokens = seqs_per_dp * m.avg_seqlen / max(1, m.n_minibatches)
+        per_tok = m.hidden_dim * 2 * (2 * pp if m.gradient_checkpointing
+                                      else 14 * m.n_layers / pp)
+        mem += mb_tokens * per_tok / tp
+    elif m.interface_type == ModelInterfaceType.GENERATE:
+        # contiguous KV caches [bs/dp, prompt+gen, nkv/tp, hd] x 2 x layers
+        cache_len = m.avg_seqlen + m.gen_tokens
+        kv = (seqs_per_dp * cache_len * max(1, m.n_kv_heads // tp)
+              * m.head_dim * 2 * 2 * m.n_layers / pp)
+        mem += kv
+        mem += seqs_per_dp * m.hidden_dim * 2 * 4  # decode activations
+    else:
+        mem += seqs_per_dp * m.avg_seqlen * m.hidden_dim * 2 * 4 / tp
+    return mem
 
 
 def search_allocations(

@@ -80,3 +80,55 @@ def test_cost_table_calibration(tmp_path):
     E.load_cost_table(str(p))
     assert E.BF16_PEAK_TF == 1000.0 and E.HBM_GBPS == 5000.0
     E.BF16_PEAK_TF, E.HBM_GBPS = old
+
+
+def test_memory_model_terms():
+    """KV cache, activation, and optimizer-offload terms of the memory
+    model (reference: estimate_rpc_memory_cost, estimate.py:387-450)."""
+    from realhf_amd.api.config import ModelInterfaceType, ParallelismConfig
+    from realhf_amd.search.engine import MFCSpec, estimate_mem_bytes
+
+    base = dict(name="g", role="actor", n_seqs=128, avg_seqlen=128,
+                param_bytes=14e9, flops_per_token=14e9,
+                n_layers=32, hidden_dim=4096, n_kv_heads=8, head_dim=128)
+    par = ParallelismConfig(data_parallel_size=1)
+    gen_short = MFCSpec(interface_type=ModelInterfaceType.GENERATE,
+                        gen_tokens=128, **base)
+    gen_long = MFCSpec(interface_type=ModelInterfaceType.GENERATE,
+                       gen_tokens=1024, **base)
+    m_s = estimate_mem_bytes(gen_short, par, trainable=False)
+    m_l = estimate_mem_bytes(gen_long, par, trainable=False)
+    assert m_l > m_s  # KV cache grows with generation length
+    # exact KV delta: bs * dtok * nkv * hd * 2 * 2 * L
+    expect = 128 * (1024 - 128) * 8 * 128 * 2 * 2 * 32
+    assert abs((m_l - m_s) - expect) / expect < 1e-6
+
+    tr = MFCSpec(interface_type=ModelInterfaceType.TRAIN_STEP, **base)
+    tr_off = MFCSpec(interface_type=ModelInterfaceType.TRAIN_STEP,
+                     offload_optimizer=True, **base)
+    m_tr = estimate_mem_bytes(tr, par, trainable=True)
+    m_off = estimate_mem_bytes(tr_off, par, trainable=True)
+    assert m_tr - m_off == pytest.approx(14e9 * 6)  # fp32 states offloaded
+
+    ckpt = MFCSpec(interface_type=ModelInterfaceType.TRAIN_STEP,
+                   gradient_checkpointing=True, **base)
+    nockpt = MFCSpec(interface_type=ModelInterfaceType.TRAIN_STEP,
+                     gradient_checkpointing=False, **base)
+    assert (estimate_mem_bytes(nockpt, par, True)
+            > estimate_mem_bytes(ckpt, par, True))
+
+
+def test_search_allocation_grpo(tmp_path):
+    """allocation_mode=search now extends past PPO (round-1 gap)."""
+    from realhf_amd.api.experiment import GRPOConfig
+    from realhf_amd.runtime.trainer import _apply_search_allocation
+
+    cfg = GRPOConfig(experiment_name="s-grpo", trial_name="t", n_gpus=8)
+    cfg.allocation_mode = "search"
+    cfg.dataset.train_bs_n_seqs = 32
+    cfg.dataset.max_prompt_len = 64
+    cfg.ppo.gen.max_new_tokens = 64
+    _apply_search_allocation(cfg, 8)
+    assert cfg.allocation_mode == "manual"
+    assert cfg.actor.parallel.world_size <= 8
+    assert cfg.ref.parallel.world_size <= 8
