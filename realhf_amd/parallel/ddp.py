@@ -147,12 +147,17 @@ class ZeRO1Optimizer:
         # grad_shard; master/exp_avg use the same per-bucket layout and
         # params all-gather back per bucket.
         self.bucket_size = bucket_size
+        # DEFAULT ON since round 2 (REALHF_AMD_ZERO_OVERLAP=0 opts out):
+        # the dp>1 CPU multiprocess tests run this path on every PR, and
+        # it is the configuration the 8-GPU scale bench runs with.
         self.overlap_comm = (
-            os.environ.get("REALHF_AMD_ZERO_OVERLAP") == "1"
+            os.environ.get("REALHF_AMD_ZERO_OVERLAP", "1") != "0"
             and self.dp_size > 1
             and hasattr(model, "layout")
             and not self._sp_repl_ivs  # SP tp-reduce must precede RS
         )
+        if self.overlap_comm and self._gloo_cuda():
+            self.overlap_comm = False  # async RS unavailable on gloo+CUDA
         if self.overlap_comm:
             align = 256 * self.dp_size
             self._bsz = max(align, (bucket_size // align) * align)
@@ -281,6 +286,13 @@ class ZeRO1Optimizer:
         if not self._grad_views_attached:
             self.attach_grads()
 
+    def _gloo_cuda(self) -> bool:
+        try:
+            return (self.model.flat_param.is_cuda
+                    and dist.get_backend(self.dp_group) == "gloo")
+        except Exception:
+            return False
+
     # ------------------------------------------------------------------
     def _lr(self) -> float:
         c = self.cfg
@@ -319,10 +331,17 @@ class ZeRO1Optimizer:
             self._armed = False
             gshard = self.grad_shard
         elif self.dp_size > 1:
-            dist.reduce_scatter_tensor(
-                self.grad_shard, self.grad_padded, op=dist.ReduceOp.AVG,
-                group=self.dp_group,
-            )
+            if self._gloo_cuda():
+                # gloo lacks CUDA reduce-scatter: all-reduce + local slice
+                dist.all_reduce(self.grad_padded, group=self.dp_group)
+                self.grad_padded /= self.dp_size
+                s0_, s1_ = self.shard_bounds
+                self.grad_shard.copy_(self.grad_padded[s0_:s1_])
+            else:
+                dist.reduce_scatter_tensor(
+                    self.grad_shard, self.grad_padded, op=dist.ReduceOp.AVG,
+                    group=self.dp_group,
+                )
             gshard = self.grad_shard
         else:
             gshard = self.grad_padded  # no copy needed at dp==1
@@ -381,9 +400,21 @@ class ZeRO1Optimizer:
             for w in works:
                 w.wait()
         elif self.dp_size > 1:
-            dist.all_gather_into_tensor(
-                self._param_padded, param_shard.contiguous(), group=self.dp_group
-            )
+            if self._gloo_cuda():
+                # gloo lacks CUDA all-gather: one broadcast per dp rank
+                for r in range(self.dp_size):
+                    piece = self._param_padded[r * self.shard_size:
+                                               (r + 1) * self.shard_size]
+                    if r == self.dp_rank:
+                        piece.copy_(param_shard)
+                    src_g = dist.get_process_group_ranks(self.dp_group)[r] \
+                        if self.dp_group is not None else r
+                    dist.broadcast(piece, src=src_g, group=self.dp_group)
+            else:
+                dist.all_gather_into_tensor(
+                    self._param_padded, param_shard.contiguous(),
+                    group=self.dp_group
+                )
         out = {"lr": lr}
         if grad_norm is not None:
             out["grad_norm"] = grad_norm
