@@ -282,6 +282,20 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t)
 
+    # DP replicas must still agree after the timed steps (guards the
+    # bucketed ZeRO overlap path: a mis-ordered reduce-scatter would
+    # silently diverge the replicas and invalidate the number)
+    if world > 1:
+        for n in ("actor", "critic"):
+            with scope(n):
+                mm = models[n].module.module
+                cs = mm.flat_param.float().sum()
+                lo, hi = cs.clone(), cs.clone()
+                dist.all_reduce(lo, op=dist.ReduceOp.MIN)
+                dist.all_reduce(hi, op=dist.ReduceOp.MAX)
+                assert torch.allclose(lo, hi, rtol=1e-6, atol=0), (
+                    f"{n}: DP replicas diverged after training ({lo} vs {hi})")
+
     n_gpus = world
     total_samples = args.seqs_per_gpu * n_gpus * args.steps
     samples_per_sec = total_samples / elapsed
