@@ -10,7 +10,11 @@ grouped_gemm_dx_kernel / grouped_gemm_dw_kernel among the top entries:
       python tools/bench_moe_train.py --steps 5
 """
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
@@ -28,8 +32,6 @@ def main():
                    help="force the per-expert rocBLAS loop (A/B baseline)")
     args = p.parse_args()
     if args.loop:
-        import os
-
         os.environ["REALHF_AMD_MOE_LOOP"] = "1"
 
     import realhf_amd.models.hf as hf_reg
