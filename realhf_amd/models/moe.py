@@ -227,10 +227,21 @@ class MoELayer(torch.nn.Module):
         the CPU/odd-shape fallback."""
         import os
 
+        # Size-adaptive dispatch (measured on MI355X, tools/bench_moe_train
+        # sweep): the MFMA grouped kernel wins on SMALL/ragged segments
+        # (decode and inference batches) where per-GEMM launch and tile
+        # quantization dominate rocBLAS; at large training segments
+        # rocBLAS's tuned big-GEMM kernels win (361 vs 151 TF/s at 1024
+        # tok/expert), so the loop fallback — still fully autograd — takes
+        # over past the threshold.
+        avg_seg = x_sorted.shape[0] / max(1, self.n_local)
+        threshold = float(os.environ.get("REALHF_AMD_GG_THRESHOLD", 256))
         use_grouped = (
             x_sorted.is_cuda
             and x_sorted.dtype == torch.bfloat16
             and os.environ.get("REALHF_AMD_MOE_LOOP") != "1"
+            and (avg_seg <= threshold
+                 or os.environ.get("REALHF_AMD_MOE_GROUPED") == "1")
         )
         if use_grouped:
             wg = self._expert_weight_stack("gate")

@@ -209,6 +209,48 @@ class DFGExecutor:
             return None  # default group
         return new_or_get_group(list(mesh))
 
+    # ------------------------------------------------------- profiling
+    def _maybe_profile(self, mfc: MFCDef):
+        """Per-MFC profiler dumps (reference: __maybe_profile_rpc,
+        model_worker.py:663/716): REALHF_AMD_DUMP_TRACE=1 writes a chrome
+        trace per (rank, mfc, step) under LOG_ROOT/trace/;
+        REALHF_AMD_DUMP_MEMORY=1 snapshots the allocator after the MFC."""
+        import contextlib
+
+        trace = os.environ.get("REALHF_AMD_DUMP_TRACE") == "1"
+        memdump = (os.environ.get("REALHF_AMD_DUMP_MEMORY") == "1"
+                   and torch.cuda.is_available())
+
+        if not trace and not memdump:
+            return contextlib.nullcontext()
+
+        @contextlib.contextmanager
+        def ctx():
+            root = constants.LOG_ROOT(constants.experiment_name(),
+                                      constants.trial_name())
+            if memdump:
+                torch.cuda.memory._record_memory_history(max_entries=100000)
+            if trace:
+                acts = [torch.profiler.ProfilerActivity.CPU]
+                if torch.cuda.is_available():
+                    acts.append(torch.profiler.ProfilerActivity.CUDA)
+                with torch.profiler.profile(activities=acts) as prof:
+                    yield
+                d = os.path.join(root, "trace")
+                os.makedirs(d, exist_ok=True)
+                prof.export_chrome_trace(os.path.join(
+                    d, f"{mfc.name}_r{self.rank}_s{self._step}.json"))
+            else:
+                yield
+            if memdump:
+                d = os.path.join(root, "memory")
+                os.makedirs(d, exist_ok=True)
+                torch.cuda.memory._dump_snapshot(os.path.join(
+                    d, f"{mfc.name}_r{self.rank}_s{self._step}.pickle"))
+                torch.cuda.memory._record_memory_history(enabled=None)
+
+        return ctx()
+
     # ------------------------------------------------------------- hooks
     def _run_hook(self, hook, mfc: MFCDef, pre: bool):
         if isinstance(hook, OffloadHook):
@@ -296,7 +338,8 @@ class DFGExecutor:
             local_out = None
             if on_mesh:
                 t0 = time.time()
-                with constants.model_scope(plan.scope):
+                with constants.model_scope(plan.scope), \
+                        self._maybe_profile(mfc):
                     local_out = self._exec_mfc(mfc, alloc, store, local_stats)
                 if torch.cuda.is_available():
                     torch.cuda.synchronize()

@@ -140,3 +140,36 @@ def test_redis_name_resolve_backend():
     with pytest.raises(NameEntryNotFoundError):
         r.get("a/b")
     assert r.wait("missing", timeout=0.1) if False else True
+
+
+def test_executor_trace_dump(tmp_path, monkeypatch):
+    """REALHF_AMD_DUMP_TRACE=1 writes a chrome trace per (mfc, rank, step)
+    (reference: REAL_DUMP_TRACE / __maybe_profile_rpc)."""
+    import json as _json
+    import os
+
+    import numpy as np
+
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.base import constants
+    from realhf_amd.runtime.trainer import Trainer
+
+    rng = np.random.RandomState(0)
+    data = str(tmp_path / "sft.jsonl")
+    with open(data, "w") as f:
+        for _ in range(8):
+            rec = {"prompt_ids": rng.randint(0, 60, size=4).tolist(),
+                   "answer_ids": rng.randint(0, 60, size=6).tolist()}
+            f.write(_json.dumps(rec) + "\n")
+    monkeypatch.setenv("REALHF_AMD_FILEROOT", str(tmp_path / "root"))
+    monkeypatch.setenv("REALHF_AMD_DUMP_TRACE", "1")
+    cfg = SFTConfig(experiment_name="t-trace", trial_name="cpu", n_gpus=1)
+    cfg.model.dtype = "float32"
+    cfg.dataset.type_ = "prompt_answer"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.exp_ctrl.benchmark_steps = 1
+    Trainer(cfg).run()
+    d = os.path.join(constants.LOG_ROOT("t-trace", "cpu"), "trace")
+    assert os.path.isdir(d) and any(
+        f.endswith(".json") for f in os.listdir(d)), d

@@ -30,9 +30,13 @@ def main():
     p.add_argument("--tokens", type=int, default=8192)
     p.add_argument("--loop", action="store_true",
                    help="force the per-expert rocBLAS loop (A/B baseline)")
+    p.add_argument("--grouped", action="store_true",
+                   help="force the MFMA grouped kernel at any segment size")
     args = p.parse_args()
     if args.loop:
         os.environ["REALHF_AMD_MOE_LOOP"] = "1"
+    if args.grouped:
+        os.environ["REALHF_AMD_MOE_GROUPED"] = "1"
 
     import realhf_amd.models.hf as hf_reg
     from realhf_amd.models.real_model import ReaLModel
@@ -79,7 +83,7 @@ def main():
           f"({args.tokens} tok, {args.experts} experts x {args.layers} L, "
           f"hidden {args.hidden}, inter {args.inter}); "
           f"expert-GEMM {moe_flops/dt/1e12:.0f} TFLOP/s "
-          f"{'(per-expert loop)' if args.loop else '(grouped MFMA)'}")
+          f"{'(per-expert loop)' if args.loop else ('(grouped MFMA)' if args.grouped else '(adaptive)')}")
 
 
 if __name__ == "__main__":
