@@ -137,10 +137,19 @@ class ReaLModelBlock(nn.Module):
         return ((x @ a.t()) @ b.t()) * self.lora_scale
 
     # -- attention --------------------------------------------------------
-    def _qkv(self, x):
+    def _qkv(self, x, fused_sp=None):
         qkv_dim = (self.nq + 2 * self.nkv) * self.hd
         merged = _maybe_merged(self.p, self._qkv_names, qkv_dim)
-        if merged is not None:
+        if fused_sp is not None and merged is not None:
+            # x is PRE-mapping: the fused fn applies (copy|SP-gather) in
+            # forward and overlaps the bwd grad collective with the
+            # weight-grad GEMM (mappings._ColumnParallelLinear)
+            qkv = mappings.column_parallel_linear(x, merged, fused_sp)
+        elif fused_sp is not None:
+            x = (mappings.gather_from_sp_region(x) if fused_sp
+                 else mappings.copy_to_tp_region(x))
+            qkv = torch.cat([_linear(x, self.p[n]) for n in self._qkv_names], dim=-1)
+        elif merged is not None:
             qkv = _linear(x, merged)
         else:
             qkv = torch.cat([_linear(x, self.p[n]) for n in self._qkv_names], dim=-1)
@@ -160,7 +169,7 @@ class ReaLModelBlock(nn.Module):
                 k = k + self._lora_delta(x, "wk")
             if "wv" in self.lora:
                 v = v + self._lora_delta(x, "wv")
-        t = x.shape[0]
+        t = qkv.shape[0]  # with fused SP, x is the pre-gather shard
         return (
             q.reshape(t, self.nq, self.hd),
             k.reshape(t, self.nkv, self.hd),
@@ -182,10 +191,15 @@ class ReaLModelBlock(nn.Module):
         cfg = self.cfg
         h = _norm(cfg, x, self.p[f"{i}.attn.ln.weight"], self.p.get(f"{i}.attn.ln.bias"))
         sp = constants.has_current() and constants.sequence_parallel()
-        if sp:
-            h = mappings.gather_from_sp_region(h)
-        else:
-            h = mappings.copy_to_tp_region(h)
+        # fused column linear (async bwd comm) takes the PRE-mapping h;
+        # decode/no-grad/LoRA keep the explicit mapping route
+        fused_col = (self.tp_size > 1 and torch.is_grad_enabled()
+                     and not self.lora and not decode)
+        if not fused_col:
+            if sp:
+                h = mappings.gather_from_sp_region(h)
+            else:
+                h = mappings.copy_to_tp_region(h)
 
         scale = 1.0 / math.sqrt(self.hd)
         if cfg.scale_attn_by_inverse_layer_idx:
@@ -269,7 +283,7 @@ class ReaLModelBlock(nn.Module):
             x = x + o
             return self._mlp(x, sp)
 
-        q, k, v = self._qkv(h)
+        q, k, v = self._qkv(h, fused_sp=(sp if fused_col else None))
         if cfg.apply_rotary:
             # graph-capture-safe length bound: never read positions back
             if decode:
@@ -340,15 +354,26 @@ class ReaLModelBlock(nn.Module):
         value is the full residual-stream output."""
         cfg = self.cfg
         i = self.i
-        if sp:
-            h = mappings.gather_from_sp_region(h)
-        else:
-            h = mappings.copy_to_tp_region(h)
+        fused_col = self.tp_size > 1 and torch.is_grad_enabled()
+        if fused_col and cfg.activation in ("silu", "geglu"):
+            idim_local = self.p[f"{i}.mlp.gate.weight"].shape[0]
+            merged_f = _maybe_merged(self.p, self._gu_names, 2 * idim_local)
+            if merged_f is None:
+                fused_col = False
+        elif fused_col:
+            fused_col = False  # single-up MLP: keep the mapping route
+        if not fused_col:
+            if sp:
+                h = mappings.gather_from_sp_region(h)
+            else:
+                h = mappings.copy_to_tp_region(h)
         if cfg.activation in ("silu", "geglu"):
             idim_local = self.p[f"{i}.mlp.gate.weight"].shape[0]
             merged = _maybe_merged(self.p, self._gu_names, 2 * idim_local)
             act = None
-            if merged is not None:
+            if fused_col:
+                gu = mappings.column_parallel_linear(h, merged, sp)
+            elif merged is not None:
                 if cfg.activation == "silu" and self.tp_size == 1:
                     # launch-boundary reduce: slabs summed in swiglu
                     gu_parts = ops.skinny_linear_nc(h, merged)
@@ -360,7 +385,7 @@ class ReaLModelBlock(nn.Module):
                     gu = ops.maybe_skinny_linear(h, merged)
                     if gu is None:
                         gu = _linear(h, merged)
-            else:
+            elif not fused_col:
                 gu = torch.cat(
                     [_linear(h, self.p[n]) for n in self._gu_names], dim=-1
                 )

@@ -228,14 +228,13 @@ class MoELayer(torch.nn.Module):
         import os
 
         # Size-adaptive dispatch (measured on MI355X, tools/bench_moe_train
-        # sweep): the MFMA grouped kernel wins on SMALL/ragged segments
-        # (decode and inference batches) where per-GEMM launch and tile
-        # quantization dominate rocBLAS; at large training segments
-        # rocBLAS's tuned big-GEMM kernels win (361 vs 151 TF/s at 1024
-        # tok/expert), so the loop fallback — still fully autograd — takes
-        # over past the threshold.
+        # sweep, fwd+bwd): at 64 routed tokens/expert the MFMA grouped
+        # kernel wins 1.4x over the per-expert rocBLAS loop (14.5 vs
+        # 20.6 ms); by 256/expert the loop edges ahead (19.7 vs 23.4) and
+        # at 1024/expert it wins 2x (tuned big-GEMM kernels).  Crossover
+        # ~128 tokens/expert; both paths are fully autograd.
         avg_seg = x_sorted.shape[0] / max(1, self.n_local)
-        threshold = float(os.environ.get("REALHF_AMD_GG_THRESHOLD", 256))
+        threshold = float(os.environ.get("REALHF_AMD_GG_THRESHOLD", 128))
         use_grouped = (
             x_sorted.is_cuda
             and x_sorted.dtype == torch.bfloat16

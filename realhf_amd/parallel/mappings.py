@@ -161,6 +161,56 @@ class _GatherFromTPRegion(torch.autograd.Function):
         return _split_along_last_dim(g)
 
 
+class _ColumnParallelLinear(torch.autograd.Function):
+    """Fused (copy-to-TP | SP-gather) + GEMM with ASYNC backward comm
+    (reference: LinearWithGradAccumulationAndAsyncCommunication,
+    model_parallel/modules.py:232-340): the input-grad all-reduce (or
+    SP reduce-scatter) is launched async and OVERLAPS the weight-grad
+    GEMM, hiding the TP collective behind compute.  With SP, only the
+    sequence shard of x is saved; the backward re-gathers it (memory
+    for one extra all-gather, as the reference does)."""
+
+    @staticmethod
+    def forward(ctx, x, w, sp):
+        ctx.sp = sp
+        total_x = _gather_along_first_dim(x) if sp else x
+        ctx.save_for_backward(x, w)
+        return torch.nn.functional.linear(total_x, w)
+
+    @staticmethod
+    def backward(ctx, gy):
+        x, w = ctx.saved_tensors
+        sp = ctx.sp
+        tp = _tp_size()
+        gy2 = gy.reshape(-1, gy.shape[-1]).contiguous()
+        total_x = _gather_along_first_dim(x) if sp else x
+        gx_total = (gy2 @ w).view(*gy.shape[:-1], w.shape[1])
+        handle = None
+        if tp > 1 and sp:
+            shape = list(gx_total.shape)
+            shape[0] //= tp
+            gx = torch.empty(shape, dtype=gx_total.dtype,
+                             device=gx_total.device)
+            handle = dist.reduce_scatter_tensor(
+                gx, gx_total.contiguous(), group=_tp_group(), async_op=True)
+        elif tp > 1:
+            gx = gx_total.contiguous()
+            handle = dist.all_reduce(gx, group=_tp_group(), async_op=True)
+        else:
+            gx = gx_total
+        # the weight-grad GEMM runs while the collective is in flight
+        gw = gy2.t() @ total_x.reshape(-1, total_x.shape[-1])
+        if handle is not None:
+            handle.wait()
+        return gx, gw, None
+
+
+def column_parallel_linear(x, w, sp: bool):
+    """y = x @ w^T with the TP input mapping fused in; `x` must be the
+    PRE-mapping tensor (SP shard when sp=True, full tokens otherwise)."""
+    return _ColumnParallelLinear.apply(x, w, sp)
+
+
 def copy_to_tp_region(x):
     return _CopyToTPRegion.apply(x)
 

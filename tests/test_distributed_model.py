@@ -237,3 +237,61 @@ def _dp_overlap_worker():
 def test_dp2_zero1_overlap_matches_single():
     """Bucketed reduce-scatter overlapped with backward == single."""
     LocalMultiProcessTest(2, _dp_overlap_worker).launch()
+
+
+def _fused_col_linear_worker():
+    """Fused column-parallel linear (async bwd comm) must match the
+    explicit mapping route bit-for-bit in fwd and grads (tp=2, SP on and
+    off)."""
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.parallel import mappings
+
+    for sp in (False, True):
+        torch.manual_seed(11)
+        tokens = 8  # per rank when sp
+        in_dim, out_dim = 16, 12
+        with constants.model_scope("m"):
+            tp = constants.tp_world_size()
+            x = torch.randn(tokens, in_dim, requires_grad=True)
+            w = torch.randn(out_dim, in_dim, requires_grad=True)
+            # broadcast so every tp rank has identical x/w
+            dist.broadcast(x.data, src=0)
+            dist.broadcast(w.data, src=0)
+            if sp:
+                r = constants.tp_rank()
+                xs = x[r * (tokens // tp):(r + 1) * (tokens // tp)]
+                xs = xs.detach().requires_grad_(True)
+            else:
+                xs = x
+
+            y1 = mappings.column_parallel_linear(xs, w, sp)
+            g = torch.randn_like(y1)
+            dist.broadcast(g, src=0)
+            y1.backward(g)
+            gx1, gw1 = xs.grad.clone(), w.grad.clone()
+
+            xs2 = xs.detach().requires_grad_(True)
+            w2 = w.detach().requires_grad_(True)
+            h = (mappings.gather_from_sp_region(xs2) if sp
+                 else mappings.copy_to_tp_region(xs2))
+            y2 = torch.nn.functional.linear(h, w2)
+            y2.backward(g)
+
+            torch.testing.assert_close(y1, y2)
+            torch.testing.assert_close(gx1, xs2.grad)
+            torch.testing.assert_close(gw1, w2.grad)
+    dist.barrier()
+
+
+def _fused_col_entry():
+    from realhf_amd.base.testing import init_global_constants
+
+    init_global_constants(num_dp=1, num_tp=2, num_pp=1, model_name="m")
+    _fused_col_linear_worker()
+
+
+@pytest.mark.distributed
+def test_fused_column_parallel_linear_tp2():
+    LocalMultiProcessTest(2, _fused_col_entry).launch()
