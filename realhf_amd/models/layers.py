@@ -445,5 +445,9 @@ class OutputHead(nn.Module):
         w = self.tied_w if self.tied_w is not None else self.p.get(f"{self.i}.head.weight")
         if cfg.is_critic:
             return _linear(h.float(), w.float())  # [total, 1] fp32
+        tp = constants.tp_world_size() if constants.has_current() else 1
+        if tp > 1 and torch.is_grad_enabled():
+            # fused copy-to-TP + GEMM: bwd all-reduce overlaps dW
+            return mappings.column_parallel_linear(h, w, False)
         h = mappings.copy_to_tp_region(h)
         return _linear(h, w)  # [total, vocab/tp] — vocab-parallel logits
