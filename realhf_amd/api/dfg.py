@@ -21,7 +21,10 @@ from realhf_amd.api.config import (
 
 @dataclasses.dataclass
 class OffloadHook:
-    """Offload model weights to pinned host memory after the MFC."""
+    """Release model weights after the MFC: to pinned host memory
+    (to="cpu", PCIe reload) or ZeRO-3-style sharded across the DP group
+    (to="dp_shard", one xGMI all-gather to restore — faster and
+    host-RAM-free; the 288 GB-native choice for frozen 70B ref/RM)."""
 
     to: str = "cpu"
 

@@ -257,7 +257,11 @@ class DFGExecutor:
             m = self.models.get(mfc.model_name)
             if m is not None:
                 real = m.module.model if hasattr(m.module, "model") else m.module
-                real.async_offload()
+                if hook.to == "dp_shard":
+                    with constants.model_scope(str(mfc.model_name)):
+                        real.shard_to_dp()
+                else:
+                    real.async_offload()
             return
         if isinstance(hook, ParamReallocHook):
             src_name = hook.source if hook.source is not None else mfc.model_name
@@ -373,6 +377,9 @@ class DFGExecutor:
         )
         if getattr(real, "_offloaded", False):
             real.reload_from_offload()
+        if getattr(real, "_dp_sharded", False):
+            with constants.model_scope(str(mfc.model_name)):
+                real.gather_from_dp()
         inp = self._assemble_input(mfc, store)
         if mfc.input_key_remap:
             inp.remap_keys_(mfc.input_key_remap)

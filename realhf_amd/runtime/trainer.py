@@ -375,9 +375,13 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
 
     for m in mfcs:
         if m.name == "rew_inf" and cfg.rew.offload:
-            m.post_hooks.append(OffloadHook())
+            m.post_hooks.append(OffloadHook(to=_offload_style(cfg.rew)))
         if m.name == "ref_inf" and cfg.ref.offload:
-            m.post_hooks.append(OffloadHook())
+            m.post_hooks.append(OffloadHook(to=_offload_style(cfg.ref)))
+
+
+def _offload_style(mc) -> str:
+    return mc.offload if isinstance(mc.offload, str) else "cpu"
 
 
 def _build_grpo(cfg: GRPOConfig, world, add_model, add_mfc, mfcs, trainable):
@@ -414,9 +418,9 @@ def _build_grpo(cfg: GRPOConfig, world, add_model, add_mfc, mfcs, trainable):
              "rewards", "prompt_mask"], [], cfg.actor, apar)
     for m in mfcs:
         if m.name == "rew_inf" and cfg.rew.offload:
-            m.post_hooks.append(OffloadHook())
+            m.post_hooks.append(OffloadHook(to=_offload_style(cfg.rew)))
         if m.name == "ref_inf" and cfg.ref.offload:
-            m.post_hooks.append(OffloadHook())
+            m.post_hooks.append(OffloadHook(to=_offload_style(cfg.ref)))
 
 
 # ---------------------------------------------------------------------------
