@@ -187,6 +187,16 @@ class ZeRO1Optimizer:
             self._reduced: List[bool] = []
             self._works: list = []
             self._armed = False
+        elif cfg.offload:
+            # chunked fp32 conversion: a whole-shard .float() on device
+            # would spike 2x shard bytes (275 GB for 70B at dp1)
+            self.master = torch.empty(self.shard_size, dtype=torch.float32,
+                                      device="cpu")
+            ch = 64 * 1024 * 1024
+            for c0 in range(0, self.shard_size, ch):
+                c1 = min(self.shard_size, c0 + ch)
+                self.master[c0:c1].copy_(
+                    self._param_padded[s0 + c0:s0 + c1].float())
         else:
             self.master = self._param_padded[s0:s1].to(torch.float32).to(state_dev)
         if cfg.offload and dev.type == "cuda":
