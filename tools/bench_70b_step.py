@@ -38,13 +38,20 @@ def main():
 
     if args.model == "70b":
         cfg = llama70b_config()
-    else:  # ~34B (CodeLlama-34b geometry)
+    elif args.model == "34b":  # CodeLlama-34b geometry
         cfg = llama70b_config()
         cfg.n_layers = 48
         cfg.hidden_dim = 8192
         cfg.intermediate_dim = 22016
         cfg.n_heads = 64
         cfg.n_kv_heads = 8
+    else:  # 13b geometry (safe-margin offload-tier measurement)
+        cfg = llama70b_config()
+        cfg.n_layers = 40
+        cfg.hidden_dim = 5120
+        cfg.intermediate_dim = 13824
+        cfg.n_heads = 40
+        cfg.n_kv_heads = 40
     print(f"model: {cfg.n_layers}L hidden {cfg.hidden_dim} "
           f"({cfg.param_count()/1e9:.1f}B params)", flush=True)
 
