@@ -96,13 +96,18 @@ class ZeRO1Optimizer:
         self.dp_group = g.dp_group() if g is not None else None
         self.dp_size = g.dp_size if g is not None else 1
         self.dp_rank = g.dp_rank if g is not None else 0
-        # SP shards tokens over the tp group, so REPLICATED params'
-        # grads (layernorms, critic head) are per-tp-rank partials and
-        # must sum over tp before the DP reduce-scatter.
-        self._sp_repl_ivs = []
+        # REPLICATED params (layernorms, critic head) exist identically
+        # on every tp rank.  Two uses: (a) under SP their grads are
+        # per-tp-rank partials and must sum over tp BEFORE the DP
+        # reduce-scatter; (b) the global grad norm must count them ONCE,
+        # not tp x (they are discounted by 1/tp before the model-group
+        # norm reduction below).
+        self._repl_ivs = []  # flat intervals of replicated params
+        self._sp_repl_ivs = []  # same, but only when SP needs the tp-sum
         self.tp_group = None
-        if (g is not None and g.tp_size > 1 and constants.has_current()
-                and constants.sequence_parallel() and hasattr(model, "layout")):
+        self.tp_size = g.tp_size if g is not None else 1
+        self.model_group = g.model_group() if g is not None else None
+        if (g is not None and g.tp_size > 1 and hasattr(model, "layout")):
             from realhf_amd.models import param_layout as PL
 
             self.tp_group = g.tp_group()
@@ -113,7 +118,9 @@ class ZeRO1Optimizer:
                             else PL.VOCAB)
                 if kind == PL.REPLICATED:
                     sp_ = model.layout.specs[k]
-                    self._sp_repl_ivs.append((sp_.start, sp_.end))
+                    self._repl_ivs.append((sp_.start, sp_.end))
+            if constants.has_current() and constants.sequence_parallel():
+                self._sp_repl_ivs = list(self._repl_ivs)
 
         n = model.flat_param.numel()
         # pad so every DP shard is 256-element aligned
@@ -154,7 +161,8 @@ class ZeRO1Optimizer:
             os.environ.get("REALHF_AMD_ZERO_OVERLAP", "1") != "0"
             and self.dp_size > 1
             and hasattr(model, "layout")
-            and not self._sp_repl_ivs  # SP tp-reduce must precede RS
+            and self.tp_size == 1  # tp: SP tp-reduce + norm discount
+                                   # need the contiguous shard layout
         )
         if self.overlap_comm and self._gloo_cuda():
             self.overlap_comm = False  # async RS unavailable on gloo+CUDA
@@ -359,14 +367,25 @@ class ZeRO1Optimizer:
         # 2. grad-norm clip — computed WITHOUT materializing an fp32 copy
         #    (a 7B fp32 grad copy is a 28 GiB spike).  The clip factor is
         #    folded into the AdamW kernel's grad_scale.
-        #    NOTE with TP>1 replicated-param grads are counted tp× —
-        #    conservative overestimate, acceptable for clipping.
         grad_norm = None
         gscale = 1.0
         if cfg.gradient_clipping and cfg.gradient_clipping > 0:
             sq = torch.linalg.vector_norm(gshard, dtype=torch.float32) ** 2
-            if self.dp_size > 1:
-                dist.all_reduce(sq, group=self.dp_group)
+            if self.tp_size > 1 and not self.overlap_comm:
+                # replicated params appear on every tp rank: count once
+                s0_, s1_ = self.shard_bounds
+                for a, b in self._repl_ivs:
+                    lo, hi = max(a, s0_), min(b, s1_)
+                    if lo < hi:
+                        part = torch.linalg.vector_norm(
+                            gshard[lo - s0_:hi - s0_],
+                            dtype=torch.float32) ** 2
+                        sq -= part * (self.tp_size - 1) / self.tp_size
+            if self.model_group is not None and (
+                    self.dp_size > 1 or self.tp_size > 1):
+                # model-group reduction (dp x tp x pp): one global norm,
+                # identical clip factor on every rank
+                dist.all_reduce(sq, group=self.model_group)
             grad_norm = float(sq.sqrt())
             if not math.isfinite(grad_norm):
                 logger.warning("non-finite grad norm %s — skipping step", grad_norm)
