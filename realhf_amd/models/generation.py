@@ -95,6 +95,13 @@ class DecodeGraph:
         )
 
     def capture(self, tokens, cache_seqlens):
+        import os
+
+        if os.environ.get("REALHF_AMD_FORCE_GRAPH_FAIL") == "1":
+            # test hook: exercise the eager fallback path (a capture can
+            # fail for real when the forward contains a graph-unsafe op,
+            # e.g. an in-graph RCCL collective on some topologies)
+            raise RuntimeError("forced capture failure (test hook)")
         self.in_tokens = tokens.clone()
         self.in_cache_seqlens = cache_seqlens.clone()
         torch.cuda.synchronize()
