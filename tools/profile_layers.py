@@ -62,6 +62,65 @@ def main():
         table["xgmi_link_gbps"] = round(bus / 7, 1)  # 7 links per GPU
         dist.destroy_process_group()
 
+    # end-to-end efficiencies on a real (small) llama: these drive the
+    # search engine's per-MFC time estimates (reference counterpart:
+    # profiled per-op tables, search_engine/estimate.py:135-450)
+    from realhf_amd.api.model import GenerationHyperparameters
+    from realhf_amd.models.generation import generate
+    from realhf_amd.models.hf.llama import make_test_config
+    from realhf_amd.models.real_model import ReaLModel
+
+    cfg = make_test_config(
+        n_layers=16, hidden_dim=2048, n_heads=16, n_kv_heads=16,
+        head_dim=128, intermediate_dim=5632, vocab_size=32000,
+    )
+    cfg.family = "llama"
+    model = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
+    model.random_init()
+    p = cfg.param_count()
+    tokens = 4096
+    toks = torch.randint(0, 32000, (tokens,), device="cuda")
+    cu = torch.arange(0, tokens + 1, 512, dtype=torch.int32, device="cuda")
+    fwd_flops = 2.0 * p * tokens
+
+    with torch.no_grad():
+        t = _time(lambda: model(packed_input_ids=toks, cu_seqlens=cu,
+                                max_seqlen=512), iters=5, warmup=2)
+    table["inf_eff"] = round(fwd_flops / t / 1e12 / table["bf16_tf"], 3)
+
+    model.allocate_grad_buffer()
+    for kk, pm in model._params.items():
+        pm.requires_grad_(True)
+        pm.grad = model.grad_view(kk)
+
+    def _train():
+        model.flat_grad.zero_()
+        out = model(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=512)
+        out.float().square().mean().backward()
+
+    t = _time(_train, iters=5, warmup=2)
+    table["train_eff"] = round(3 * fwd_flops / t / 1e12 / table["bf16_tf"], 3)
+
+    # decode: per-token time -> fraction of HBM weight-streaming bound
+    for kk, pm in model._params.items():
+        pm.requires_grad_(False)
+        pm.grad = None
+    bs = 64
+    ptoks = torch.randint(0, 32000, (bs * 64,), device="cuda")
+    pcu = torch.arange(0, bs * 64 + 1, 64, dtype=torch.int32, device="cuda")
+    g = GenerationHyperparameters(max_new_tokens=64, min_new_tokens=64,
+                                  greedy=True, use_hip_graph=True)
+    with torch.no_grad():
+        generate(model, ptoks, pcu, g)  # warm session + graph
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        generate(model, ptoks, pcu, g)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+    per_tok = dt / 64
+    ideal = (p * 2) / (table["hbm_gbps"] * 1e9)
+    table["gen_bw_eff"] = round(ideal / per_tok, 3)
+
     with open(args.out, "w") as f:
         json.dump(table, f, indent=1)
     print(json.dumps(table))
