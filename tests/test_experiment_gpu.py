@@ -128,3 +128,34 @@ def test_tiny_ppo_with_rccl_groups(tmp_path):
 
         constants.clear_grids()
         clear_group_cache()
+
+
+def test_tiny_grpo_mixtral_gpu(tmp_path):
+    """BASELINE config #5 end-to-end ON THE HIP PATH: Mixtral-style GRPO
+    (bf16, grouped-GEMM experts, native attention/decode kernels) for
+    two steps on one GPU."""
+    from realhf_amd.api.experiment import GRPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    rng = np.random.RandomState(7)
+    data = str(tmp_path / "p.jsonl")
+    with open(data, "w") as f:
+        for _ in range(8):
+            f.write(json.dumps(
+                {"input_ids": rng.randint(3, 60, size=8).tolist()}) + "\n")
+    cfg = GRPOConfig(experiment_name="t-grpo-gpu", trial_name="g", n_gpus=1)
+    for mc in (cfg.actor, cfg.ref):
+        mc.family = "mixtral"
+    cfg.rew.family = "llama"
+    cfg.rew.is_critic = True
+    cfg.group_size = 2
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 8
+    cfg.ppo.gen.min_new_tokens = 2
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 2
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
