@@ -59,10 +59,14 @@ def main():
                    help="per-phase sync+timing (serializes side-stream overlap)")
     p.add_argument("--no-offload-frozen", dest="offload_frozen",
                    action="store_false", default=True)
-    p.add_argument("--offload-set", type=str, default="ref,rew",
-                   help="comma list of frozen roles to offload "
-                        "(profiling showed reload copies are ~15%% of GPU "
-                        "time; fewer offloads trade memory for copies)")
+    p.add_argument("--offload-set", type=str, default="",
+                   help="comma list of frozen roles to offload.  Default "
+                        "NONE since round 2: 4x7B fits 265.6 GiB resident "
+                        "on 288 GB, and the A/B measured 3.67 samples/s "
+                        "resident vs 3.50 offloaded (reload copyBuffer "
+                        "traffic was 15%% of GPU time in rocprof) — "
+                        "offload pays only when memory forces it (70B "
+                        "tier, or --offload-set ref,rew)")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
