@@ -24,15 +24,24 @@ typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2q;
 // nothing co-resident to overlap with), W=4 needs 64 KB (TWO WGs per CU:
 // one streams KV while the other runs its softmax/PV phase).  Default 4;
 // REALHF_AMD_DEC_WAVES ∈ {2,4,8} for A/B.
+// Split-K over the key range (flash-decoding): at decode batch sizes
+// bs*nkv < 256 a one-WG-per-(batch, kv-head) grid leaves most of the
+// 256 CUs idle; gridDim.y splits each sequence's keys into `nsplit`
+// ranges whose unnormalized partials (acc, m, s) land in `ws` and are
+// softmax-combined by attn_decode_combine_kernel.  nsplit is chosen
+// from bs*nkv only (static for hipGraph capture); per-batch ranges are
+// derived in-kernel from the dynamic cache_seqlens.
 template <int HD, int REP, int W>
 __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ kc,
     const bf16* __restrict__ vc, const int* __restrict__ cache_seqlens,
-    bf16* __restrict__ out, int bs, int nq, int nkv, long maxlen, float scale,
-    int window) {
+    bf16* __restrict__ out, float* __restrict__ ws, int bs, int nq, int nkv,
+    long maxlen, float scale, int window) {
   constexpr int DPL = HD / WAVE;  // dims per lane (2 for hd=128)
   const int b = blockIdx.x / nkv;
   const int kvh = blockIdx.x % nkv;
+  const int nsplit = gridDim.y;
+  const int sp = blockIdx.y;
   const int L = cache_seqlens[b];
   const int lane = threadIdx.x & (WAVE - 1);
   const int w = threadIdx.x / WAVE;
@@ -64,8 +73,14 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
   // sliding window (mistral): only keys in [L - window, L) attend
   const int lo = (window > 0 && L > window) ? (L - window) : 0;
   const int origin = (lo / (W * WAVE)) * (W * WAVE);
+  // this split's aligned key range [r0, r1)
+  const int chunk = W * WAVE;
+  const int nchunks = (L - origin + chunk - 1) / chunk;
+  const int per_sp = (nchunks + nsplit - 1) / nsplit;
+  const int r0 = origin + sp * per_sp * chunk;
+  const int r1 = min((long)L, (long)origin + (long)(sp + 1) * per_sp * chunk);
 
-  for (int base = origin + w * WAVE; base < L; base += W * WAVE) {
+  for (int base = r0 + w * WAVE; base < r1; base += W * WAVE) {
     const int l = base + lane;
     const bool valid = l < L && l >= lo;
     float sc[REP];
@@ -100,7 +115,7 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
     }
     #pragma unroll
     for (int r = 0; r < REP; r++) sc[r] *= scale;  // q kept raw bf16
-    const int nvalid = min(WAVE, L - base);
+    const int nvalid = min(WAVE, r1 - base);
     #pragma unroll
     for (int r = 0; r < REP; r++) {
       if (!valid) sc[r] = -1e30f;
@@ -148,19 +163,54 @@ __global__ __launch_bounds__(64 * W) void attn_decode_kernel(
       for (int d = 0; d < DPL; d++) o[d] = 0.f;
       #pragma unroll
       for (int ww = 0; ww < W; ww++) {
-        float f = __expf(m_s[ww] - M);
+        float f = (s_s[ww] > 0.f) ? __expf(m_s[ww] - M) : 0.f;
         S += f * s_s[ww];
         #pragma unroll
         for (int d = 0; d < DPL; d++) o[d] += f * acc_s[ww][lane * DPL + d];
       }
-      float inv = (S > 0.f) ? 1.f / S : 0.f;
-      #pragma unroll
-      for (int d = 0; d < DPL; d++)
-        out[((long)b * nq + kvh * REP + r) * HD + lane * DPL + d] =
-            __float2bfloat16(o[d] * inv);
+      const long head = (long)b * nq + kvh * REP + r;
+      if (ws == nullptr) {
+        float inv = (S > 0.f) ? 1.f / S : 0.f;
+        #pragma unroll
+        for (int d = 0; d < DPL; d++)
+          out[head * HD + lane * DPL + d] = __float2bfloat16(o[d] * inv);
+      } else {
+        // unnormalized split partial: [head][split][HD dims | m | s]
+        float* row = ws + (head * nsplit + sp) * (HD + 2);
+        #pragma unroll
+        for (int d = 0; d < DPL; d++) row[lane * DPL + d] = o[d];
+        if (lane == 0) { row[HD] = M; row[HD + 1] = S; }
+      }
     }
     __syncthreads();
   }
+}
+
+template <int HD>
+__global__ __launch_bounds__(WAVE) void attn_decode_combine_kernel(
+    const float* __restrict__ ws, bf16* __restrict__ out, int nsplit) {
+  constexpr int DPL = HD / WAVE;
+  const long head = blockIdx.x;
+  const int lane = threadIdx.x;
+  const float* base = ws + head * (long)nsplit * (HD + 2);
+  float M = -1e30f;
+  for (int s = 0; s < nsplit; s++) M = fmaxf(M, base[s * (HD + 2) + HD]);
+  float S = 0.f;
+  float o[DPL];
+  #pragma unroll
+  for (int d = 0; d < DPL; d++) o[d] = 0.f;
+  for (int s = 0; s < nsplit; s++) {
+    const float* row = base + s * (HD + 2);
+    float ss = row[HD + 1];
+    float f = (ss > 0.f) ? __expf(row[HD] - M) : 0.f;
+    S += f * ss;
+    #pragma unroll
+    for (int d = 0; d < DPL; d++) o[d] += f * row[lane * DPL + d];
+  }
+  float inv = (S > 0.f) ? 1.f / S : 0.f;
+  #pragma unroll
+  for (int d = 0; d < DPL; d++)
+    out[head * HD + lane * DPL + d] = __float2bfloat16(o[d] * inv);
 }
 
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
@@ -174,19 +224,46 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   int rep = nq / nkv;
   auto cs = cache_seqlens.to(torch::kInt);
   auto out = torch::empty_like(q);
-  dim3 grid(bs * nkv);
   static int dec_waves = [] {
     const char* e = getenv("REALHF_AMD_DEC_WAVES");
     int v = e ? atoi(e) : 4;
     return (v == 2 || v == 4 || v == 8) ? v : 4;
   }();
+  // split-K factor: fill the 256-CU chip when bs*nkv is small (decode
+  // batches).  Static per shape -> hipGraph-safe; REALHF_AMD_DEC_SPLIT
+  // overrides for A/B.
+  static int dec_split_env = [] {
+    const char* e = getenv("REALHF_AMD_DEC_SPLIT");
+    return e ? atoi(e) : 0;
+  }();
+  int nsplit = dec_split_env > 0
+      ? dec_split_env
+      : std::min<long>(8, std::max<long>(1, (2 * 256) / std::max(1, bs * nkv)));
+  torch::Tensor ws;
+  float* ws_ptr = nullptr;
+  if (nsplit > 1) {
+    ws = torch::empty({(long)bs * nq * nsplit * (hd + 2)},
+                      q.options().dtype(torch::kFloat));
+    ws_ptr = ws.data_ptr<float>();
+  }
+  dim3 grid(bs * nkv, nsplit);
   auto launch = [&](auto hd_c, auto rep_c) {
     auto go = [&](auto w_c) {
       hipLaunchKernelGGL((attn_decode_kernel<hd_c.value, rep_c.value, w_c.value>),
         grid, dim3(64 * w_c.value), 0, cur_stream(), (const bf16*)q.data_ptr(),
         (const bf16*)kc.data_ptr(), (const bf16*)vc.data_ptr(),
-        cs.data_ptr<int>(), (bf16*)out.data_ptr(), bs, nq, nkv, maxlen,
+        cs.data_ptr<int>(), (bf16*)out.data_ptr(), ws_ptr, bs, nq, nkv, maxlen,
         (float)scale, (int)window);
+      if (ws_ptr) {
+        if (hd_c.value == 128)
+          hipLaunchKernelGGL((attn_decode_combine_kernel<128>),
+            dim3((unsigned)(bs * nq)), dim3(WAVE), 0, cur_stream(),
+            ws_ptr, (bf16*)out.data_ptr(), nsplit);
+        else
+          hipLaunchKernelGGL((attn_decode_combine_kernel<64>),
+            dim3((unsigned)(bs * nq)), dim3(WAVE), 0, cur_stream(),
+            ws_ptr, (bf16*)out.data_ptr(), nsplit);
+      }
     };
     if (dec_waves == 2) go(std::integral_constant<int, 2>{});
     else if (dec_waves == 8) go(std::integral_constant<int, 8>{});
