@@ -346,3 +346,73 @@ def _pp_gen_interleave_worker():
 @pytest.mark.distributed
 def test_pp2_generate_microbatch_interleaving():
     LocalMultiProcessTest(2, _pp_gen_interleave_worker).launch()
+
+
+def _pp_tp_gen_worker():
+    """tp2 x pp2 pipelined generation matches the single-process
+    engine (greedy): exercises the TP logits gather + shared-seed
+    sampling inside the interleaved PP decode schedule."""
+    import torch.distributed as dist
+
+    from realhf_amd.api.config import Abstraction, ModelName
+    from realhf_amd.api.data import SequenceSample
+    from realhf_amd.api.model import (
+        FinetuneSpec,
+        GenerationHyperparameters,
+        Model,
+        make_backend,
+    )
+    import realhf_amd.runtime.engine  # noqa: F401
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models.generation import generate as gen_single
+    from realhf_amd.models.real_model import ReaLModel
+
+    cfg = make_test_config(n_layers=4, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=43)
+    init_global_constants(num_dp=1, num_tp=2, num_pp=2, model_name="m")
+    g = constants.grid_of("m")
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                  tp_rank=g.tp_rank, tp_size=2,
+                  pp_rank=g.pp_rank, pp_size=2)
+    _fill_model_from_full(m, cfg, sd)
+    model = Model(ModelName("m", 0), m, None, torch.device("cpu"),
+                  torch.float32)
+    model = make_backend(Abstraction("inference")).initialize(
+        model, FinetuneSpec(1, 64, 4))
+    rng = np.random.RandomState(19)
+    lens = [5, 8, 6]
+    toks = torch.from_numpy(rng.randint(0, 128, size=sum(lens))).long()
+    batch = SequenceSample(
+        keys=("packed_prompts",), ids=[f"t{j}" for j in range(3)],
+        seqlens={"packed_prompts": [[l] for l in lens]},
+        data={"packed_prompts": toks},
+    )
+    gconfig = GenerationHyperparameters(max_new_tokens=6, greedy=True,
+                                        use_hip_graph=False)
+    with constants.model_scope("m"):
+        outs = model.module.generate(batch, gconfig=gconfig, n_mbs=2)
+    if m.is_last_stage:
+        single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+        _fill_model_from_full(single, cfg, sd)
+        cu_t = torch.tensor([0] + list(np.cumsum(lens)), dtype=torch.int32)
+        ref = gen_single(single, toks, cu_t, gconfig)
+        prompt_of = {}
+        for j in range(len(lens)):
+            prompt_of[tuple(toks[cu_t[j]:cu_t[j+1]].tolist())] = j
+        n = 0
+        for gen_out, prompts, cu in outs:
+            for i in range(cu.shape[0] - 1):
+                p = tuple(prompts[cu[i]:cu[i+1]].tolist())
+                j = prompt_of[p]
+                assert torch.equal(gen_out.gen_tokens[i], ref.gen_tokens[j])
+                n += 1
+        assert n == len(lens)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_pp2_tp2_generate():
+    LocalMultiProcessTest(4, _pp_tp_gen_worker).launch()
