@@ -289,7 +289,14 @@ class ReaLModelBlock(nn.Module):
             if decode:
                 rot_len = int(k_cache.shape[1])
             else:
-                rot_len = max(int(max_seqlen), cfg.max_position_embeddings)
+                ms = max_seqlen
+                if ms is None:  # CP: token shard with explicit positions
+                    from realhf_amd.parallel import cp as cp_mod
+
+                    c = cp_mod.current()
+                    ms = (c.info.full_max if c is not None
+                          else cfg.max_position_embeddings)
+                rot_len = max(int(ms), cfg.max_position_embeddings)
             cos, sin = ops.rotary_cache.get(
                 self.hd,
                 rot_len,
@@ -313,20 +320,44 @@ class ReaLModelBlock(nn.Module):
                 window=cfg.sliding_window,
             )
         else:
-            if k_cache is not None:
-                # prefill: write all tokens into the cache
-                bs = k_cache.shape[0]
-                seq_id = torch.bucketize(
-                    torch.arange(x.shape[0], device=x.device),
-                    cu_seqlens[1:].long(),
-                    right=True,
+            from realhf_amd.parallel import cp as cp_mod
+
+            cpctx = cp_mod.current()
+            if cpctx is not None:
+                # Ulysses context parallelism: tokens are sharded over
+                # the CP group; one all-to-all gives this rank ALL
+                # tokens for 1/cp of the heads, the varlen kernel runs
+                # on full sequences, and a second all-to-all restores
+                # the (token-shard, all-heads) layout (parallel/cp.py).
+                assert k_cache is None, "CP: no KV-cache prefill/generation"
+                assert q.shape[1] % cpctx.size == 0, (
+                    f"q heads {q.shape[1]} not divisible by cp {cpctx.size}")
+                assert k.shape[1] % cpctx.size == 0, (
+                    f"kv heads {k.shape[1]} not divisible by cp {cpctx.size}")
+                q = cp_mod.seq_gather_head_scatter(q)
+                k = cp_mod.seq_gather_head_scatter(k)
+                v = cp_mod.seq_gather_head_scatter(v)
+                attn_out = ops.attn_varlen(
+                    q, k, v, cpctx.info.full_cu, cpctx.info.full_max,
+                    causal=True, softmax_scale=scale,
+                    window=cfg.sliding_window,
                 )
-                k_cache[seq_id, positions] = k.detach().to(k_cache.dtype)
-                v_cache[seq_id, positions] = v.detach().to(v_cache.dtype)
-            attn_out = ops.attn_varlen(
-                q, k, v, cu_seqlens, max_seqlen, causal=True,
-                softmax_scale=scale, window=cfg.sliding_window,
-            )
+                attn_out = cp_mod.head_gather_seq_scatter(attn_out)
+            else:
+                if k_cache is not None:
+                    # prefill: write all tokens into the cache
+                    bs = k_cache.shape[0]
+                    seq_id = torch.bucketize(
+                        torch.arange(x.shape[0], device=x.device),
+                        cu_seqlens[1:].long(),
+                        right=True,
+                    )
+                    k_cache[seq_id, positions] = k.detach().to(k_cache.dtype)
+                    v_cache[seq_id, positions] = v.detach().to(v_cache.dtype)
+                attn_out = ops.attn_varlen(
+                    q, k, v, cu_seqlens, max_seqlen, causal=True,
+                    softmax_scale=scale, window=cfg.sliding_window,
+                )
         attn_out = attn_out.reshape(attn_out.shape[0], self.nq * self.hd)
         o = _linear(attn_out, self.p[f"{i}.attn.wo.weight"])
         if "wo" in self.lora:
