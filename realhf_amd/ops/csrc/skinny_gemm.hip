@@ -16,8 +16,12 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 
 #define SG_TN 64  // n per workgroup (16 per wave)
 
-template <int NB>  // NB x 32-deep k body: NB 16-byte W loads in flight/lane
-__global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
+// NB x 32-deep k body: NB 16-byte W loads in flight/lane.  WVS = waves
+// per workgroup (n-tile = WVS*16): 4 (=SG_TN 64) default; 2 halves the
+// tile for TAIL-BOUND small-N shapes (o-proj N=4096: 64 tiles left 88%
+// of wave cycles parked in PMC — 128 tiles x half work fills the tail).
+template <int NB, int WVS = 4>
+__global__ __launch_bounds__(64 * WVS, 4) void skinny_gemm_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w,
     float* __restrict__ out32, int M, int N, int K, int kslice) {
   const int ntile = blockIdx.x;
@@ -28,12 +32,12 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
   const int wv = threadIdx.x >> 6;
   const int i16 = lane & 15;
   const int g = lane >> 4;
-  const int n0 = ntile * SG_TN + wv * 16;
+  const int n0 = ntile * (WVS * 16) + wv * 16;
 
   extern __shared__ __bf16 x_s[];  // [16][kslice] rows (zero-padded m >= M)
   {
     const int kn = k_hi - k_lo;
-    for (int idx = threadIdx.x * 8; idx < 16 * kslice; idx += 256 * 8) {
+    for (int idx = threadIdx.x * 8; idx < 16 * kslice; idx += 64 * WVS * 8) {
       int m = idx / kslice;
       int kk = idx % kslice;
       bf16x8v v = {};
@@ -90,6 +94,24 @@ static int sg_depth() {
   return d;
 }
 
+// n-tile width threshold: shapes with N <= this use the 32-wide tile
+// (2 waves) to double the workgroup count; REALHF_AMD_SG_TN32_MAXN=0
+// disables, default 4096 (o-proj) pending A/B
+static int sg_tn32_maxn() {
+  static int v = [] {
+    const char* e = getenv("REALHF_AMD_SG_TN32_MAXN");
+    return e ? atoi(e) : 4096;
+  }();
+  return v;
+}
+
+template <typename F4, typename F8, typename F4w2, typename F8w2>
+static void sg_dispatch(int N, F4 f4, F8 f8, F4w2 f4w2, F8w2 f8w2) {
+  bool tn32 = N <= sg_tn32_maxn() && (N % 32) == 0;
+  if (sg_depth() == 8) { if (tn32) f8w2(); else f8(); }
+  else { if (tn32) f4w2(); else f4(); }
+}
+
 // combine the nks fp32 partial slabs -> bf16 (+ optional residual add)
 __global__ void sg_combine_kernel(const float* __restrict__ parts,
                                   const bf16* __restrict__ resid,
@@ -135,17 +157,20 @@ torch::Tensor skinny_gemm_nc(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(out32_ws.numel() >= (long)nks * M * N, "workspace too small");
   auto out32 = out32_ws.narrow(0, 0, (long)nks * M * N)
                    .view({(long)nks, (long)M, (long)N});
-  dim3 grid(N / SG_TN, nks);
   size_t lds = (size_t)16 * kslice * sizeof(short);
   TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
-  if (sg_depth() == 8)
-    hipLaunchKernelGGL(skinny_gemm_kernel<8>, grid, dim3(256), lds, cur_stream(),
+  auto launch = [&](auto nb, auto wvs) {
+    dim3 grid(N / (16 * wvs.value), nks);
+    hipLaunchKernelGGL((skinny_gemm_kernel<nb.value, wvs.value>), grid,
+      dim3(64 * wvs.value), lds, cur_stream(),
       (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
       out32.data_ptr<float>(), M, N, K, kslice);
-  else
-    hipLaunchKernelGGL(skinny_gemm_kernel<4>, grid, dim3(256), lds, cur_stream(),
-      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
-      out32.data_ptr<float>(), M, N, K, kslice);
+  };
+  using c4 = std::integral_constant<int, 4>;
+  using c8 = std::integral_constant<int, 8>;
+  using c2 = std::integral_constant<int, 2>;
+  sg_dispatch(N, [&]{ launch(c4{}, c4{}); }, [&]{ launch(c8{}, c4{}); },
+              [&]{ launch(c4{}, c2{}); }, [&]{ launch(c8{}, c2{}); });
   CHECK_CUDA_OK();
   return out32;
 }
@@ -349,17 +374,20 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
   int nks = (K + kslice - 1) / kslice;
   TORCH_CHECK(out32_ws.numel() >= (long)nks * M * N, "workspace too small");
   auto out32 = out32_ws.narrow(0, 0, (long)nks * M * N);
-  dim3 grid(N / SG_TN, nks);
   size_t lds = (size_t)16 * kslice * sizeof(short);
   TORCH_CHECK(lds <= 160 * 1024, "kslice too large for LDS");
-  if (sg_depth() == 8)
-    hipLaunchKernelGGL(skinny_gemm_kernel<8>, grid, dim3(256), lds, cur_stream(),
+  auto launch = [&](auto nb, auto wvs) {
+    dim3 grid(N / (16 * wvs.value), nks);
+    hipLaunchKernelGGL((skinny_gemm_kernel<nb.value, wvs.value>), grid,
+      dim3(64 * wvs.value), lds, cur_stream(),
       (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
       out32.data_ptr<float>(), M, N, K, kslice);
-  else
-    hipLaunchKernelGGL(skinny_gemm_kernel<4>, grid, dim3(256), lds, cur_stream(),
-      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
-      out32.data_ptr<float>(), M, N, K, kslice);
+  };
+  using c4 = std::integral_constant<int, 4>;
+  using c8 = std::integral_constant<int, 8>;
+  using c2 = std::integral_constant<int, 2>;
+  sg_dispatch(N, [&]{ launch(c4{}, c4{}); }, [&]{ launch(c8{}, c4{}); },
+              [&]{ launch(c4{}, c2{}); }, [&]{ launch(c8{}, c2{}); });
   auto out = torch::empty({M, (long)N}, x.options());
   long n = (long)M * N;
   int cgrid = (int)std::min<long>((n / 4 + 255) / 256, 2048);
