@@ -95,12 +95,14 @@ static int sg_depth() {
 }
 
 // n-tile width threshold: shapes with N <= this use the 32-wide tile
-// (2 waves) to double the workgroup count; REALHF_AMD_SG_TN32_MAXN=0
-// disables, default 4096 (o-proj) pending A/B
+// (2 waves).  Default 0 = DISABLED: the in-context A/B measured 3.55 vs
+// 3.66 samples/s with it on at 4096 — doubling the workgroups also
+// doubles the per-WG x-LDS stage and the tail moves, it does not
+// shrink.  Kept env-gated (REALHF_AMD_SG_TN32_MAXN) for other shapes.
 static int sg_tn32_maxn() {
   static int v = [] {
     const char* e = getenv("REALHF_AMD_SG_TN32_MAXN");
-    return e ? atoi(e) : 4096;
+    return e ? atoi(e) : 0;
   }();
   return v;
 }
