@@ -288,3 +288,71 @@ def _realloc_ep_worker(s_ep, d_ep):
 @pytest.mark.parametrize("s_ep,d_ep", [(2, 1), (4, 2), (2, 2)])
 def test_realloc_roundtrip_ep(s_ep, d_ep):
     LocalMultiProcessTest(4, _realloc_ep_worker, s_ep, d_ep).launch()
+
+
+def _realloc_worker_ex(src_geom, dst_geom, src_ranks, dst_ranks, family,
+                       is_critic):
+    """Extended matrix: disjoint rank sets (the replica/asym flow), other
+    model families, and critic head shapes (reference crown-jewel style:
+    tests/comm/test_param_realloc.py parametrization)."""
+    import torch.distributed as dist
+
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.models.real_model import ReaLModel
+
+    fam = hf_reg.get_family(family)
+    kw = dict(n_layers=4, hidden_dim=64, n_heads=8, vocab_size=128)
+    if family != "gpt2":  # gpt2 is MHA: kv heads fixed = n_heads
+        kw["n_kv_heads"] = 4
+    cfg = fam.make_test_config(**kw)
+    cfg.is_critic = is_critic
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=101)
+    rank = dist.get_rank()
+    sp, sdp, st = src_geom
+    dp, ddp, dt = dst_geom
+    src = ParallelStrategy.make(sp, sdp, st, ranks=src_ranks)
+    dst = ParallelStrategy.make(dp, ddp, dt, ranks=dst_ranks)
+
+    src_model = None
+    for (p, d, t), r in src.rank_map:
+        if r == rank:
+            src_model = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                                  tp_rank=t, tp_size=st, pp_rank=p, pp_size=sp)
+            _fill_model_from_full(src_model, cfg, sd)
+    dst_model = None
+    for (p, d, t), r in dst.rank_map:
+        if r == rank:
+            dst_model = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                                  tp_rank=t, tp_size=dt, pp_rank=p, pp_size=dp)
+            with torch.no_grad():
+                dst_model.flat_param.zero_()
+    plan = build_realloc_plan(cfg, src, dst)
+    execute_realloc(
+        plan,
+        src_model.flat_param if src_model is not None else None,
+        dst_model.flat_param if dst_model is not None else None,
+    )
+    if dst_model is not None:
+        _check_model_vs_full(dst_model, cfg, sd)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+@pytest.mark.parametrize(
+    "src_geom,dst_geom,src_ranks,dst_ranks,family,is_critic",
+    [
+        # DISJOINT rank sets: train mesh [0,1] -> gen mesh [2,3]
+        ((1, 2, 1), (1, 2, 1), [0, 1], [2, 3], "llama", False),
+        ((1, 1, 2), (2, 1, 1), [0, 1], [2, 3], "llama", False),
+        # critic head (replicated 1-dim output) across tp<->pp remaps
+        ((1, 2, 2), (2, 2, 1), None, None, "llama", True),
+        # second family
+        ((1, 1, 4), (4, 1, 1), None, None, "gpt2", False),
+        ((2, 2, 1), (1, 1, 4), None, None, "gpt2", True),
+    ],
+)
+def test_realloc_matrix_extended(src_geom, dst_geom, src_ranks, dst_ranks,
+                                 family, is_critic):
+    LocalMultiProcessTest(4, _realloc_worker_ex, src_geom, dst_geom,
+                          src_ranks, dst_ranks, family, is_critic).launch()
