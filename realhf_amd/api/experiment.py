@@ -9,7 +9,7 @@ same idea: structured configs translated into (DFG, allocations, worker
 state).
 """
 import dataclasses
-from typing import Any, Dict, Optional
+from typing import Union, Any, Dict, Optional
 
 from realhf_amd.api.config import ParallelismConfig
 from realhf_amd.api.model import GenerationHyperparameters
@@ -38,7 +38,9 @@ class ModelTrainEvalConfig:
     init_critic_from_actor: bool = False
     dtype: str = "bfloat16"
     gradient_checkpointing: bool = False
-    offload: bool = False  # offload params when idle
+    # False | True ("cpu": pinned-host offload) | "dp_shard" (ZeRO-3-style
+    # sharding over the DP group, restored by one xGMI all-gather)
+    offload: Union[bool, str] = False
     optimizer: OptimizerConfig = dataclasses.field(default_factory=OptimizerConfig)
     parallel: ParallelismConfig = dataclasses.field(default_factory=ParallelismConfig)
     # gen-phase override strategy (enables param realloc between phases)

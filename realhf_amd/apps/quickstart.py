@@ -46,7 +46,11 @@ def _coerce(raw: str, t, current):
     if isinstance(t, str):  # string annotations
         t = {"int": int, "float": float, "bool": bool, "str": str}.get(t, str)
     if t is bool or isinstance(current, bool):
-        return raw.lower() in ("1", "true", "yes")
+        if raw.lower() in ("1", "true", "yes"):
+            return True
+        if raw.lower() in ("0", "false", "no"):
+            return False
+        return raw  # bool|str fields, e.g. offload=dp_shard
     if t is int or isinstance(current, int) and not isinstance(current, bool):
         return int(raw)
     if t is float or isinstance(current, float):
