@@ -188,6 +188,23 @@ class GenerationConfig(CommonExperimentConfig):
     )
 
 
+@dataclasses.dataclass
+class ProfileConfig(CommonExperimentConfig):
+    """Time interfaces across parallel strategies with mock data
+    (reference: ProfileConfig, experiments/benchmark/profile_exp.py:61)."""
+
+    model: ModelTrainEvalConfig = dataclasses.field(
+        default_factory=ModelTrainEvalConfig
+    )
+    interfaces: str = "inference,train_step"  # comma list (+generate)
+    strategies: str = "d1"  # ';'-separated d{dp}t{tp}p{pp} specs
+    n_seqs: int = 16
+    seq_len: int = 512
+    gen_tokens: int = 128
+    n_steps: int = 3
+    warmup: int = 1
+
+
 EXPERIMENT_TYPES = {
     "sft": SFTConfig,
     "rw": RWConfig,
@@ -195,4 +212,5 @@ EXPERIMENT_TYPES = {
     "ppo": PPOConfig,
     "grpo": GRPOConfig,
     "gen": GenerationConfig,
+    "profile": ProfileConfig,
 }

@@ -91,9 +91,16 @@ def main(argv=None):
             dist.init_process_group(backend)
             if torch.cuda.is_available():
                 torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
-        from realhf_amd.runtime.trainer import Trainer
+        from realhf_amd.api.experiment import ProfileConfig
 
-        Trainer(cfg).run()
+        if isinstance(cfg, ProfileConfig):
+            from realhf_amd.runtime.profiler import run_profile
+
+            run_profile(cfg)
+        else:
+            from realhf_amd.runtime.trainer import Trainer
+
+            Trainer(cfg).run()
         if dist.is_initialized():
             dist.barrier()
             dist.destroy_process_group()

@@ -99,3 +99,30 @@ def test_scheduler_restart_exhausted(tmp_path):
     sched.submit_array([sys.executable, str(script)], n_procs=2)
     with _pytest.raises(JobException):
         sched.wait(timeout=60)
+
+
+def test_profile_experiment(tmp_path, monkeypatch):
+    """Profile experiment: times interfaces across strategies with mock
+    data (reference: ProfileConfig, profile_exp.py:61)."""
+    import json
+    import os
+
+    from realhf_amd.apps.quickstart import parse_cli
+    from realhf_amd.base import constants
+    from realhf_amd.runtime.profiler import run_profile
+
+    monkeypatch.setenv("REALHF_AMD_FILEROOT", str(tmp_path / "root"))
+    exp, cfg = parse_cli([
+        "profile", "experiment_name=t-prof", "trial_name=x",
+        "model.family=llama", "model.dtype=float32", "n_gpus=1",
+        "strategies=d1t1p1", "interfaces=inference,train_step,generate",
+        "n_seqs=2", "seq_len=16", "gen_tokens=4", "n_steps=1", "warmup=0",
+    ])
+    assert exp == "profile"
+    res = run_profile(cfg)
+    assert "d1t1p1" in res
+    for k in ("inference", "train_step", "generate"):
+        assert res["d1t1p1"][k]["seconds"] > 0
+    out = os.path.join(constants.LOG_ROOT("t-prof", "x"),
+                       "profile_result.json")
+    assert json.load(open(out))["d1t1p1"]
