@@ -41,3 +41,7 @@ if __name__ == "__main__":
     run("bench-mb 4x640 h32", [640] * 4, 32, 32, iters=iters)
     run("bench-full 16x640 h32", [640] * 16, 32, 32, iters=iters)
     run("long 16x2048 h32gqa8", [2048] * 16, 32, 8, iters=iters)
+    # long-context tier (device-built block lists; grad-ckpt training
+    # shapes for the CP/packing long-context story)
+    run("16k 2x16384 h32gqa8", [16384] * 2, 32, 8, iters=max(3, iters // 6))
+    run("32k 1x32768 h32gqa8", [32768], 32, 8, iters=max(3, iters // 10))
