@@ -1,11 +1,14 @@
 #!/usr/bin/env python3
 """Standalone A/B + profiling target for the flash-attention varlen
 forward kernel."""
+import os
 import sys
 import time
 
 import numpy as np
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import realhf_amd._C as C
 
