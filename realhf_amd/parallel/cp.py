@@ -107,11 +107,12 @@ class _SeqHeadAllToAll(torch.autograd.Function):
     def forward(ctx, x, gather_seq: bool):
         st = _STATE
         ctx.gather_seq = gather_seq
+        ctx.st = st  # captured: backward may run after the context exits
         return _a2a(x, st, gather_seq)
 
     @staticmethod
     def backward(ctx, g):
-        return _a2a(g.contiguous(), _STATE, not ctx.gather_seq), None
+        return _a2a(g.contiguous(), ctx.st, not ctx.gather_seq), None
 
 
 def _a2a(x, st, gather_seq):

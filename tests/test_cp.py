@@ -55,7 +55,9 @@ def _cp_worker():
     with cp.context(None, rank, world, info):
         out = m(packed_input_ids=local_ids, positions=local_pos,
                 cu_seqlens=None, max_seqlen=None)
-        (out.float().square().sum() / info.total).backward()
+    # backward OUTSIDE the context: the autograd fn must have captured
+    # the cp state at forward time
+    (out.float().square().sum() / info.total).backward()
     g = m.flat_grad.clone()
     dist.all_reduce(g)  # weights replicated: CP grads sum like DP
 
