@@ -187,7 +187,13 @@ def main():
         })
     )
     critic_iface = make_interface(
-        Abstraction("ppo_critic", {"n_minibatches": args.n_minibatches})
+        Abstraction("ppo_critic", {
+            "n_minibatches": args.n_minibatches,
+            # critic_train is the LAST phase of the step: its final
+            # param all-gather hides under the NEXT step's generation
+            # (engines wait it before any param use)
+            "defer_final_allgather": world > 1,
+        })
     )
     rew_iface = make_interface(Abstraction("paired_rw"))
 
@@ -298,6 +304,7 @@ def main():
     # bucketed ZeRO overlap path: a mis-ordered reduce-scatter would
     # silently diverge the replicas and invalidate the number)
     if world > 1:
+        models["critic"].module.optimizer.finish_allgather()
         for n in ("actor", "critic"):
             with scope(n):
                 mm = models[n].module.module

@@ -109,6 +109,10 @@ class PPOActorInterface(ModelInterface):
     gae_lambda: float = 1.0
     adv_norm: bool = True
     early_stop_imp_ratio: Optional[float] = None
+    # leave the LAST minibatch's ZeRO param all-gather in flight; it is
+    # waited at the model's next engine call (e.g. the critic's AG hides
+    # under the following 2.7 s generation phase at dp8)
+    defer_final_allgather: bool = False
 
     def __post_init__(self):
         if isinstance(self.gconfig, dict):
@@ -217,10 +221,12 @@ class PPOActorInterface(ModelInterface):
         all_stats: Dict[str, float] = {}
         n_mb = min(self.n_minibatches, train_sample.bs)
         mbs = train_sample.split(n_mb) if n_mb > 1 else [train_sample]
-        for mb in mbs:
+        for i_mb, mb in enumerate(mbs):
             stats = model.module.train_batch(
                 mb, self._loss_fn, version_steps=model.version.global_step,
                 n_mbs=n_mbs,
+                defer_allgather=(self.defer_final_allgather
+                                 and i_mb == len(mbs) - 1),
             )
             if (
                 self.early_stop_imp_ratio is not None
@@ -283,6 +289,7 @@ class PPOCriticInterface(ModelInterface):
     gae_lambda: float = 1.0
     value_norm: bool = True
     value_loss_type: str = "huber"
+    defer_final_allgather: bool = False
 
     def __post_init__(self):
         self._rms = ppo_math.ExponentialRunningMeanStd() if self.value_norm else None
@@ -348,10 +355,12 @@ class PPOCriticInterface(ModelInterface):
         all_stats: Dict[str, float] = {}
         n_mb = min(self.n_minibatches, train_sample.bs)
         mbs = train_sample.split(n_mb) if n_mb > 1 else [train_sample]
-        for mb in mbs:
+        for i_mb, mb in enumerate(mbs):
             stats = model.module.train_batch(
                 mb, self._loss_fn, version_steps=model.version.global_step,
                 n_mbs=n_mbs,
+                defer_allgather=(self.defer_final_allgather
+                                 and i_mb == len(mbs) - 1),
             )
             for k, v in stats.items():
                 all_stats[k] = all_stats.get(k, 0.0) + v / len(mbs)

@@ -247,6 +247,11 @@ class ZeRO1Optimizer:
             ]
 
         self._grad_views_attached = False
+        # async param all-gathers left in flight by step(defer_allgather=
+        # True) — waited by finish_allgather() (engines call it before
+        # any param use).  Lets the LAST PPO minibatch's all-gather hide
+        # under the next MFC (e.g. critic AG under the 2.7 s generation).
+        self._pending_ag: List = []
 
     # ------------------------------------------------------------------
     def attach_grads(self):
@@ -298,7 +303,13 @@ class ZeRO1Optimizer:
         self._works.append(w)
         self._reduced[b] = True
 
+    def finish_allgather(self):
+        for w in self._pending_ag:
+            w.wait()
+        self._pending_ag = []
+
     def zero_grad(self):
+        self.finish_allgather()
         self.grad_padded.zero_()
         if self.overlap_comm:
             self._armed = False  # a failed backward must not leave stale state
@@ -329,7 +340,7 @@ class ZeRO1Optimizer:
         return c.lr
 
     @torch.no_grad()
-    def step(self) -> Dict[str, float]:
+    def step(self, defer_allgather: bool = False) -> Dict[str, float]:
         self.step_count += 1
         cfg = self.cfg
         dev = self.grad_padded.device
@@ -428,8 +439,11 @@ class ZeRO1Optimizer:
                 works.append(dist.all_gather_into_tensor(
                     self._param_padded[b0:b1], self._shard_bf16[soff:soff + l],
                     group=self.dp_group, async_op=True))
-            for w in works:
-                w.wait()
+            if defer_allgather:
+                self._pending_ag = works
+            else:
+                for w in works:
+                    w.wait()
         elif self.dp_size > 1:
             if self._gloo_cuda():
                 # gloo lacks CUDA all-gather: one broadcast per dp rank
@@ -442,10 +456,14 @@ class ZeRO1Optimizer:
                         if self.dp_group is not None else r
                     dist.broadcast(piece, src=src_g, group=self.dp_group)
             else:
-                dist.all_gather_into_tensor(
+                w = dist.all_gather_into_tensor(
                     self._param_padded, param_shard.contiguous(),
-                    group=self.dp_group
+                    group=self.dp_group, async_op=True
                 )
+                if defer_allgather:
+                    self._pending_ag = [w]
+                else:
+                    w.wait()
         out = {"lr": lr}
         if grad_norm is not None:
             out["grad_norm"] = grad_norm

@@ -186,7 +186,8 @@ class PipelinedEngine(PipelinableEngine):
 
     # -------------------------------------------------------------- train
     def train_batch(self, input_: SequenceSample, loss_fn: Callable,
-                    version_steps: int = 0, n_mbs: Optional[int] = None) -> Dict:
+                    version_steps: int = 0, n_mbs: Optional[int] = None,
+                    defer_allgather: bool = False) -> Dict:
         g = self._grid()
         self.model.train()
         self.optimizer.zero_grad()
@@ -254,7 +255,7 @@ class PipelinedEngine(PipelinableEngine):
             bi += 1
 
         _flush_sends()
-        opt_stats = self.optimizer.step()
+        opt_stats = self.optimizer.step(defer_allgather=defer_allgather)
         stats.update(opt_stats)
         return stats
 

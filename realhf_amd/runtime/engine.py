@@ -58,9 +58,10 @@ class PipelinableTrainEngine(PipelinableEngine):
         return self.model
 
     def train_batch(self, input_: SequenceSample, loss_fn: Callable,
-                    version_steps: int = 0, n_mbs: Optional[int] = None) -> Dict:
+                    version_steps: int = 0, n_mbs: Optional[int] = None,
+                    defer_allgather: bool = False) -> Dict:
         self.model.train()
-        self.optimizer.zero_grad()
+        self.optimizer.zero_grad()  # also waits any deferred all-gather
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 and input_.bs >= n_mbs else [input_]
         stats: Dict[str, float] = {}
         for i, mb in enumerate(mbs):
@@ -81,13 +82,19 @@ class PipelinableTrainEngine(PipelinableEngine):
             (loss / len(mbs)).backward()
             for k, v in st.items():
                 stats[k] = stats.get(k, 0.0) + float(v) / len(mbs)
-        opt_stats = self.optimizer.step()
+        opt_stats = self.optimizer.step(defer_allgather=defer_allgather)
         stats.update(opt_stats)
         return stats
+
+    def _finish_ag(self):
+        opt = getattr(self, "optimizer", None)
+        if opt is not None and hasattr(opt, "finish_allgather"):
+            opt.finish_allgather()
 
     @torch.no_grad()
     def eval_batch(self, input_: SequenceSample, loss_fn: Callable, n_mbs=None):
         self.model.eval()
+        self._finish_ag()
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 else [input_]
         stats: Dict[str, float] = {}
         for mb in mbs:
@@ -103,6 +110,7 @@ class PipelinableTrainEngine(PipelinableEngine):
                 post_hook: Optional[Callable] = None,
                 aggregate_fn: Callable = _default_aggregate):
         self.model.eval()
+        self._finish_ag()
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 and input_.bs >= n_mbs else [input_]
         outs = []
         for mb in mbs:
@@ -118,6 +126,7 @@ class PipelinableTrainEngine(PipelinableEngine):
                  gconfig: Optional[GenerationHyperparameters] = None,
                  n_mbs: Optional[int] = None, **gen_kw):
         self.model.eval()
+        self._finish_ag()
         gconfig = gconfig or GenerationHyperparameters()
         key = "packed_prompts" if "packed_prompts" in input_.keys else "packed_input_ids"
         mbs = input_.split(n_mbs) if n_mbs and n_mbs > 1 and input_.bs >= n_mbs else [input_]

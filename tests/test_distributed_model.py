@@ -334,3 +334,50 @@ def _dp_shard_worker():
 @pytest.mark.distributed
 def test_dp_shard_round_trip():
     LocalMultiProcessTest(2, _dp_shard_worker).launch()
+
+
+def _defer_ag_worker():
+    """step(defer_allgather=True) leaves the param all-gather in flight;
+    finish_allgather() (called by every engine entry) completes it to
+    the same parameters as the blocking step."""
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel.ddp import OptimizerConfig, ZeRO1Optimizer
+
+    init_global_constants(num_dp=2, num_tp=1, num_pp=1, model_name="m")
+    cfg = make_test_config(n_layers=2, hidden_dim=32, n_heads=4, n_kv_heads=4,
+                           vocab_size=96)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=111)
+
+    def run(defer):
+        with constants.model_scope("m"):
+            torch.manual_seed(7)
+            m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+            _fill_model_from_full(m, cfg, sd)
+            opt = ZeRO1Optimizer(
+                m, OptimizerConfig(lr=1e-2, warmup_steps_proportion=0.0))
+            for _ in range(2):
+                opt.zero_grad()
+                toks = torch.randint(0, 96, (12,))
+                cu = torch.tensor([0, 12], dtype=torch.int32)
+                if opt.overlap_comm:
+                    opt.arm_overlap()
+                out = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=12)
+                out.float().square().mean().backward()
+                opt.step(defer_allgather=defer)
+            opt.finish_allgather()
+            return m.flat_param.detach().clone()
+
+    p_block = run(defer=False)
+    p_defer = run(defer=True)
+    torch.testing.assert_close(p_defer, p_block, rtol=0, atol=0)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_deferred_allgather_matches_blocking():
+    LocalMultiProcessTest(2, _defer_ag_worker).launch()
