@@ -521,7 +521,7 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
     M, K = x.shape
     N = w.shape[0]
     splitk = 8
-    while K // splitk > 1024:
+    while K // splitk > _sg_kslice_max():
         splitk *= 2
     ws = get_global_memory_buffer().get_tensor(
         (2 * splitk * 16 * N,), torch.float32, "skinny_gemm_ws"
@@ -535,6 +535,18 @@ def maybe_skinny_linear(x: torch.Tensor, w: torch.Tensor,
         # Kept for reference / small-grid experiments only.
         return C.skinny_gemm2(x, w, ws, _skinny_sem(x.device), splitk, residual)
     return C.skinny_gemm(x, w, ws, splitk, residual)
+
+
+def _sg_kslice_max() -> int:
+    """Max K elements per split-K slice (tunes workgroup count); A/B via
+    REALHF_AMD_SG_KSLICE, default 1024."""
+    global _SG_KSLICE
+    if _SG_KSLICE is None:
+        _SG_KSLICE = int(os.environ.get("REALHF_AMD_SG_KSLICE", 1024))
+    return _SG_KSLICE
+
+
+_SG_KSLICE = None
 
 
 def skinny_linear_nc(x: torch.Tensor, w: torch.Tensor) -> Optional[torch.Tensor]:
@@ -563,7 +575,7 @@ def skinny_linear_nc(x: torch.Tensor, w: torch.Tensor) -> Optional[torch.Tensor]
     K = x.shape[1]
     N = w.shape[0]
     splitk = 8
-    while K // splitk > 1024:
+    while K // splitk > _sg_kslice_max():
         splitk *= 2
     ws = get_global_memory_buffer().get_tensor(
         (2 * splitk * 16 * N,), torch.float32, "skinny_gemm_ws"
