@@ -90,3 +90,71 @@ def test_shard_batch_padding_math():
     assert pos0.tolist() == [0, 1, 2]
     assert ids3.tolist() == [9, 0, 0]  # last real token + 2 pad
     assert pos3.tolist() == [5, 0, 1]
+
+
+def _cp_sft_worker():
+    """End-to-end CP TRAINING with a real next-token CE loss: labels are
+    built globally (shift crosses shard boundaries) then sharded with the
+    tokens; summed shard grads must match the single-process step."""
+    import torch.distributed as dist
+    import torch.nn.functional as F
+
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel import cp
+
+    rank, world = dist.get_rank(), dist.get_world_size()
+    cfg = make_test_config(n_layers=4, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=93)
+    rng = np.random.RandomState(29)
+    lens = [10, 14]
+    toks = torch.from_numpy(rng.randint(0, 128, size=sum(lens))).long()
+    cu = torch.tensor([0] + list(np.cumsum(lens)), dtype=torch.int32)
+
+    # global labels: next token within each sequence, -100 at seq ends
+    labels = torch.full((sum(lens),), -100, dtype=torch.long)
+    for i in range(len(lens)):
+        s, e = int(cu[i]), int(cu[i + 1])
+        labels[s:e - 1] = toks[s + 1:e]
+
+    local_ids, local_pos, info = cp.shard_batch(toks, cu, rank, world)
+    pad = info.total - info.orig_total
+    labels_p = torch.cat([labels, torch.full((pad,), -100, dtype=torch.long)])
+    t_loc = info.total // world
+    local_labels = labels_p[rank * t_loc:(rank + 1) * t_loc]
+
+    def ce(logits, lab):
+        return F.cross_entropy(logits.float(), lab, ignore_index=-100,
+                               reduction="sum")
+
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    _fill_model_from_full(m, cfg, sd)
+    m.allocate_grad_buffer()
+    for k, p in m._params.items():
+        p.requires_grad_(True)
+        p.grad = m.grad_view(k)
+    n_pred = int((labels != -100).sum())
+    with cp.context(None, rank, world, info):
+        out = m(packed_input_ids=local_ids, positions=local_pos,
+                cu_seqlens=None, max_seqlen=None)
+    (ce(out, local_labels) / n_pred).backward()
+    g = m.flat_grad.clone()
+    dist.all_reduce(g)
+
+    single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    _fill_model_from_full(single, cfg, sd)
+    single.allocate_grad_buffer()
+    for k, p in single._params.items():
+        p.requires_grad_(True)
+        p.grad = single.grad_view(k)
+    out_ref = single(packed_input_ids=toks, cu_seqlens=cu,
+                     max_seqlen=max(lens))
+    (ce(out_ref, labels) / n_pred).backward()
+    torch.testing.assert_close(g, single.flat_grad, atol=5e-4, rtol=5e-4)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_ulysses_cp2_sft_loss_matches_single():
+    LocalMultiProcessTest(2, _cp_sft_worker).launch()
