@@ -710,6 +710,8 @@ class Trainer:
             eng = self.models[name].module
             m = eng.module
             opt = getattr(eng, "optimizer", None)
+            if opt is not None and hasattr(opt, "finish_allgather"):
+                opt.finish_allgather()  # deferred AG must land before save
             state["models"][str(name)] = {
                 "flat_param": m.flat_param.detach().cpu().clone(),
                 "lora_flat": (m.lora_flat.detach().cpu().clone()
