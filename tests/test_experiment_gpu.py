@@ -33,7 +33,31 @@ def test_tiny_ppo_experiment_gpu(tmp_path):
     cfg.ppo.ppo_n_minibatches = 2
     cfg.exp_ctrl.benchmark_steps = 2
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
-    Trainer(cfg).run()
+    # dims must satisfy the grouped-GEMM alignment (hidden/inter % 64)
+    # so the MoE experts run the native MFMA path, not the rocBLAS loop
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.runtime import trainer as T
+
+    fam = hf_reg.get_family("mixtral")
+    orig = T.build_experiment
+
+    def patched(c, world):
+        built = orig(c, world)
+        for name, rcfg in built.model_cfgs.items():
+            if rcfg.moe is not None:
+                big = fam.make_test_config(
+                    n_layers=2, hidden_dim=64, n_heads=1, n_kv_heads=1,
+                    head_dim=64, intermediate_dim=128, vocab_size=128)
+                big.is_critic = rcfg.is_critic
+                big.dtype = rcfg.dtype
+                built.model_cfgs[name] = big
+        return built
+
+    T.build_experiment = patched
+    try:
+        Trainer(cfg).run()
+    finally:
+        T.build_experiment = orig
 
 
 def test_tiny_sft_gpu(tmp_path):
@@ -158,4 +182,28 @@ def test_tiny_grpo_mixtral_gpu(tmp_path):
     cfg.ppo.ppo_n_minibatches = 2
     cfg.exp_ctrl.benchmark_steps = 2
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
-    Trainer(cfg).run()
+    # dims must satisfy the grouped-GEMM alignment (hidden/inter % 64)
+    # so the MoE experts run the native MFMA path, not the rocBLAS loop
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.runtime import trainer as T
+
+    fam = hf_reg.get_family("mixtral")
+    orig = T.build_experiment
+
+    def patched(c, world):
+        built = orig(c, world)
+        for name, rcfg in built.model_cfgs.items():
+            if rcfg.moe is not None:
+                big = fam.make_test_config(
+                    n_layers=2, hidden_dim=64, n_heads=1, n_kv_heads=1,
+                    head_dim=64, intermediate_dim=128, vocab_size=128)
+                big.is_critic = rcfg.is_critic
+                big.dtype = rcfg.dtype
+                built.model_cfgs[name] = big
+        return built
+
+    T.build_experiment = patched
+    try:
+        Trainer(cfg).run()
+    finally:
+        T.build_experiment = orig
