@@ -226,6 +226,9 @@ def generate(
         and device.type == "cuda"
         and model.dtype == torch.bfloat16
         and cfg.head_dim in (64, 128)
+        # MoE dispatch reads expert counts on the host (counts.cpu());
+        # a D2H sync inside stream capture DEADLOCKS rather than erroring
+        and cfg.moe is None
     )
 
     cur_logits = last_logits

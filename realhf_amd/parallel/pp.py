@@ -350,6 +350,7 @@ class PipelinedEngine(PipelinableEngine):
         use_graph = (
             gconfig.use_hip_graph and dev.type == "cuda"
             and dtype == torch.bfloat16 and cfg.head_dim in (64, 128)
+            and cfg.moe is None  # MoE counts.cpu() deadlocks under capture
         )
 
         last_rank = g.global_rank_of(S - 1, g.dp_rank, g.tp_rank)
