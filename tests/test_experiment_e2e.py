@@ -366,3 +366,41 @@ def test_sft_training_deterministic(tmp_path):
 
     a, b = run("a"), run("b")
     torch.testing.assert_close(a, b, atol=1e-6, rtol=1e-6)
+
+
+def _ppo_dpshard_worker(data, fileroot):
+    """ZeRO-3-style frozen-weight sharding wired through the OffloadHook:
+    rew/ref release to 1/dp shards after their MFC and restore with one
+    all-gather at the next step (to="dp_shard")."""
+    from realhf_amd.api.config import ModelName
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = PPOConfig(experiment_name="t-ppods", trial_name="dist", n_gpus=2)
+    for mc in (cfg.actor, cfg.critic, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+    cfg.allocation_mode = "global"
+    cfg.ref.offload = "dp_shard"
+    cfg.rew.offload = "dp_shard"
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 5
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 2  # 2 steps: shard -> gather -> shard
+    t = Trainer(cfg)
+    t.run()
+    # after the last rew_inf post-hook the frozen models are sharded
+    rew = t.models[ModelName("rew", 0)].module.module
+    assert rew._dp_sharded and rew.flat_param is None
+
+
+@pytest.mark.distributed
+def test_ppo_dp_shard_offload_two_ranks(tmp_path):
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data)
+    LocalMultiProcessTest(2, _ppo_dpshard_worker, data,
+                          str(tmp_path / "root")).launch()
