@@ -533,7 +533,11 @@ class Trainer:
                 models[name] = backend.initialize(model, spec)
             if mc.offload and name not in self.built.trainable:
                 real = models[name].module.model
-                real.async_offload(non_blocking=False)
+                if _offload_style(mc) == "dp_shard":
+                    with constants.model_scope(str(name)):
+                        real.shard_to_dp()
+                else:
+                    real.async_offload(non_blocking=False)
         return models
 
     def _build_dataloader(self, valid: bool = False):
