@@ -33,13 +33,34 @@ def test_tiny_ppo_experiment_gpu(tmp_path):
     cfg.ppo.ppo_n_minibatches = 2
     cfg.exp_ctrl.benchmark_steps = 2
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
-    # NOTE: default tiny dims (hidden 32, head_dim 8) run the torch
-    # fallbacks for attention; the grouped-GEMM/MFMA paths have their own
-    # GPU tests (test_ops_gpu) + the rocprof'd microbench
-    # (profiles/moe_train_kernels_r02.csv).  A variant of this test with
-    # kernel-eligible 1-head dims aborted asynchronously in ways not
-    # reproducible at kernel/module/generate level (see docs/roadmap.md).
-    Trainer(cfg).run()
+    # Kernel-eligible dims (hidden/inter % 64) so the MoE experts run
+    # the MFMA grouped path end-to-end.  EVERY role must share the
+    # actor's vocab: generated token ids feed the reward/ref embeddings,
+    # and an out-of-vocab id is a device-side OOB gather (HW exception —
+    # took a kernel-serialized hunt to find; tools/repro_grpo_e2e.py).
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.runtime import trainer as T
+
+    orig = T.build_experiment
+
+    def patched(c, world):
+        built = orig(c, world)
+        for name, rcfg in built.model_cfgs.items():
+            fam = hf_reg.get_family(rcfg.family or "llama")
+            big = fam.make_test_config(
+                n_layers=2, hidden_dim=64, n_heads=1, n_kv_heads=1,
+                head_dim=64, intermediate_dim=128, vocab_size=128)
+            big.is_critic = rcfg.is_critic
+            big.dtype = rcfg.dtype
+            big.family = rcfg.family
+            built.model_cfgs[name] = big
+        return built
+
+    T.build_experiment = patched
+    try:
+        Trainer(cfg).run()
+    finally:
+        T.build_experiment = orig
 
 
 def test_tiny_sft_gpu(tmp_path):
@@ -164,10 +185,31 @@ def test_tiny_grpo_mixtral_gpu(tmp_path):
     cfg.ppo.ppo_n_minibatches = 2
     cfg.exp_ctrl.benchmark_steps = 2
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
-    # NOTE: default tiny dims (hidden 32, head_dim 8) run the torch
-    # fallbacks for attention; the grouped-GEMM/MFMA paths have their own
-    # GPU tests (test_ops_gpu) + the rocprof'd microbench
-    # (profiles/moe_train_kernels_r02.csv).  A variant of this test with
-    # kernel-eligible 1-head dims aborted asynchronously in ways not
-    # reproducible at kernel/module/generate level (see docs/roadmap.md).
-    Trainer(cfg).run()
+    # Kernel-eligible dims (hidden/inter % 64) so the MoE experts run
+    # the MFMA grouped path end-to-end.  EVERY role must share the
+    # actor's vocab: generated token ids feed the reward/ref embeddings,
+    # and an out-of-vocab id is a device-side OOB gather (HW exception —
+    # took a kernel-serialized hunt to find; tools/repro_grpo_e2e.py).
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.runtime import trainer as T
+
+    orig = T.build_experiment
+
+    def patched(c, world):
+        built = orig(c, world)
+        for name, rcfg in built.model_cfgs.items():
+            fam = hf_reg.get_family(rcfg.family or "llama")
+            big = fam.make_test_config(
+                n_layers=2, hidden_dim=64, n_heads=1, n_kv_heads=1,
+                head_dim=64, intermediate_dim=128, vocab_size=128)
+            big.is_critic = rcfg.is_critic
+            big.dtype = rcfg.dtype
+            big.family = rcfg.family
+            built.model_cfgs[name] = big
+        return built
+
+    T.build_experiment = patched
+    try:
+        Trainer(cfg).run()
+    finally:
+        T.build_experiment = orig
