@@ -80,9 +80,7 @@ typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
 typedef __attribute__((address_space(3))) bf16x4v lds_b64_t;
 
 template <int HD>
-// occupancy 2: a co-resident workgroup covers the stage+sync gaps of
-// the other (A/B'd at 32k: see docs/kernels.md long-context table)
-__global__ __launch_bounds__(64 * AV_WAVES, 2) void attn_varlen_fwd_kernel(
+__global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const int* __restrict__ cu_seqlens,
     const int* __restrict__ blk_offsets, int n_seqs,
