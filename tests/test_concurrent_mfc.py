@@ -141,11 +141,12 @@ def test_disjoint_mesh_mfcs_overlap(tmp_path):
     """Two 0.6 s inference MFCs on disjoint 2-rank meshes: serial
     execution would take >= 1.2 s; the concurrent executor must finish
     the step well under that."""
-    sleep_s = 0.6
+    sleep_s = 1.0
     out = str(tmp_path / "wall.json")
     LocalMultiProcessTest(4, _overlap_worker, sleep_s, out).launch()
     wall = json.load(open(out))["wall"]
-    assert wall < 2 * sleep_s * 0.9, (
+    # concurrent ~= sleep_s + transfer overhead; serial >= 2*sleep_s
+    assert wall < 2 * sleep_s * 0.85, (
         f"step took {wall:.2f}s — disjoint-mesh MFCs did not overlap "
         f"(serial would be ~{2*sleep_s:.1f}s)"
     )
