@@ -28,6 +28,9 @@ def main():
     p.add_argument("--steps", type=int, default=2)
     p.add_argument("--tokens", type=int, default=2048)
     p.add_argument("--model", type=str, default="70b")
+    p.add_argument("--no-offload", dest="offload", action="store_false",
+                   default=True)
+    p.add_argument("--seq", type=int, default=512)
     args = p.parse_args()
 
     from realhf_amd.api.model import ReaLModelConfig
@@ -36,7 +39,11 @@ def main():
     from realhf_amd.models.real_model import ReaLModel
     from realhf_amd.parallel.ddp import OptimizerConfig, ZeRO1Optimizer
 
-    if args.model == "70b":
+    if args.model == "7b":
+        from realhf_amd.models.hf.llama import llama7b_config
+
+        cfg = llama7b_config()
+    elif args.model == "70b":
         cfg = llama70b_config()
     elif args.model == "34b":  # CodeLlama-34b geometry
         cfg = llama70b_config()
@@ -72,7 +79,7 @@ def main():
     with constants.model_scope("m70"):
         opt = ZeRO1Optimizer(
             m, OptimizerConfig(lr=1e-5, warmup_steps_proportion=0.0,
-                               offload=True),
+                               offload=args.offload),
             total_train_steps=100,
         )
     print(f"optimizer (host fp32 states) in {time.time()-t0:.0f}s; "
@@ -80,7 +87,7 @@ def main():
           flush=True)
 
     toks = torch.randint(0, cfg.vocab_size, (args.tokens,), device="cuda")
-    seq = 512
+    seq = args.seq
     cu = torch.arange(0, args.tokens + 1, seq, dtype=torch.int32,
                       device="cuda")
 
