@@ -255,6 +255,217 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Double-buffered long-context variant: KV tiles ping-pong through two LDS
+// buffers; block i+1's global loads issue BEFORE block i's compute and the
+// LDS writes land after it, so the HBM/L2 stream overlaps the MFMA+softmax
+// phases and the two per-block __syncthreads collapse to one.  ~90 KB LDS
+// (2x KV tiles + P), still one workgroup per CU.  Selected for long
+// sequences (REALHF_AMD_ATTN_DB: 0 never, 1 always, default auto at
+// max_seqlen >= 4096); the single-buffer kernel above stays the default
+// for bench-length sequences.
+template <int HD>
+__global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_db_kernel(
+    const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const int* __restrict__ cu_seqlens,
+    const int* __restrict__ blk_offsets, int n_seqs,
+    bf16* __restrict__ out, float* __restrict__ lse,
+    int nq, int nkv, float scale, bool causal, int window) {
+  constexpr int HDCH = HD / 32;
+  const int blk = blockIdx.x;
+  if (blk >= blk_offsets[n_seqs]) return;
+  const int qh = blockIdx.y;
+  const int kvh = qh / (nq / nkv);
+  const int seq = blk_lookup(blk_offsets, n_seqs, blk);
+  const int q0_local = (blk - blk_offsets[seq]) * QBLK;
+  const int s0 = cu_seqlens[seq], s1 = cu_seqlens[seq + 1];
+  const int L = s1 - s0;
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int i16 = lane & 15;
+  const int g = lane >> 4;
+
+  __shared__ __bf16 k_s[2][KVBLK][HD + KPAD];
+  __shared__ __bf16 v_s[2][KVBLK][HD + 16];
+  __shared__ __bf16 p_s[AV_WAVES][QW][KVBLK + VPAD];
+
+  bf16x8 qfrag[HDCH];
+  const int my_qrow_a = q0_local + w * QW + i16;
+  {
+    const bf16* qrow = q + ((long)(s0 + min(my_qrow_a, L - 1)) * nq + qh) * HD;
+    #pragma unroll
+    for (int c = 0; c < HDCH; c++)
+      qfrag[c] = *(const bf16x8*)(qrow + c * 32 + g * 8);
+  }
+
+  float m_r[4], l_r[4];
+  f32x4 o_acc[HD / 16];
+  #pragma unroll
+  for (int r = 0; r < 4; r++) { m_r[r] = -1e30f; l_r[r] = 0.f; }
+  #pragma unroll
+  for (int t = 0; t < HD / 16; t++) o_acc[t] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kv_end = causal ? min(L, q0_local + QBLK) : L;
+  int kv_begin = 0;
+  if (window > 0) {
+    kv_begin = q0_local - window + 1;
+    kv_begin = (kv_begin > 0) ? (kv_begin / KVBLK) * KVBLK : 0;
+  }
+  const int nblocks = (kv_end - kv_begin + KVBLK - 1) / KVBLK;
+  if (nblocks <= 0) return;
+
+  // each thread stages SLOTS 8-elem groups of the 64 x HD tile per tensor
+  constexpr int SLOTS = (KVBLK * (HD / 8)) / (64 * AV_WAVES) > 0
+                            ? (KVBLK * (HD / 8) + 64 * AV_WAVES - 1) /
+                                  (64 * AV_WAVES)
+                            : 1;
+  bf16x8 kreg[SLOTS], vreg[SLOTS];
+
+  auto load_tile = [&](int kv0) {
+    #pragma unroll
+    for (int s = 0; s < SLOTS; s++) {
+      int idx = threadIdx.x + s * 64 * AV_WAVES;
+      if (idx >= KVBLK * (HD / 8)) break;
+      int row = idx / (HD / 8);
+      int col8 = (idx % (HD / 8)) * 8;
+      bf16x8 kv8 = {}, vv8 = {};
+      if (row < kv_end - kv0) {
+        kv8 = *(const bf16x8*)(k + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
+        vv8 = *(const bf16x8*)(v + ((long)(s0 + kv0 + row) * nkv + kvh) * HD + col8);
+      }
+      kreg[s] = kv8;
+      vreg[s] = vv8;
+    }
+  };
+  auto store_tile = [&](int buf) {
+    #pragma unroll
+    for (int s = 0; s < SLOTS; s++) {
+      int idx = threadIdx.x + s * 64 * AV_WAVES;
+      if (idx >= KVBLK * (HD / 8)) break;
+      int row = idx / (HD / 8);
+      int col8 = (idx % (HD / 8)) * 8;
+      *(bf16x8*)(&k_s[buf][row][col8]) = kreg[s];
+      *(bf16x8*)(&v_s[buf][row][col8]) = vreg[s];
+    }
+  };
+
+  load_tile(kv_begin);
+  store_tile(0);
+  __syncthreads();
+
+  for (int b = 0; b < nblocks; b++) {
+    const int kv0 = kv_begin + b * KVBLK;
+    const int cur = b & 1;
+    if (b + 1 < nblocks)
+      load_tile(kv0 + KVBLK);  // global loads in flight during compute
+
+    // ---- S = Q K^T over 4 key subtiles -----------------------------
+    f32x4 s_sub[4];
+    #pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int c = 0; c < HDCH; c++) {
+        bf16x8 bfrag = *(const bf16x8*)(&k_s[cur][ks * 16 + i16][c * 32 + g * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfrag, acc, 0, 0, 0);
+      }
+      s_sub[ks] = acc;
+    }
+
+    // ---- softmax ----------------------------------------------------
+    float tile_max[4];
+    #pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float mx = -1e30f;
+      const int qrow = q0_local + w * QW + g * 4 + r;
+      #pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        int kidx = kv0 + ks * 16 + i16;
+        bool ok = (kidx < kv_end) && (!causal || kidx <= qrow) && (qrow < L)
+                  && (window <= 0 || kidx > qrow - window);
+        float sv = ok ? s_sub[ks][r] * scale : -1e30f;
+        s_sub[ks][r] = sv;
+        mx = fmaxf(mx, sv);
+      }
+      tile_max[r] = group16_max(mx);
+    }
+    #pragma unroll
+    for (int r = 0; r < 4; r++) {
+      float nm = fmaxf(m_r[r], tile_max[r]);
+      if (nm < -1e29f) nm = 0.f;
+      float f = __expf(m_r[r] - nm);
+      if (m_r[r] < -1e29f) f = 0.f;
+      m_r[r] = nm;
+      l_r[r] *= f;
+      #pragma unroll
+      for (int t = 0; t < HD / 16; t++) o_acc[t][r] *= f;
+      float rowsum = 0.f;
+      #pragma unroll
+      for (int ks = 0; ks < 4; ks++) {
+        float p = (s_sub[ks][r] > -1e29f) ? __expf(s_sub[ks][r] - nm) : 0.f;
+        s_sub[ks][r] = p;
+        rowsum += p;
+      }
+      l_r[r] += group16_sum(rowsum);
+    }
+    #pragma unroll
+    for (int ks = 0; ks < 4; ks++) {
+      #pragma unroll
+      for (int r = 0; r < 4; r++)
+        p_s[w][g * 4 + r][ks * 16 + i16] = (__bf16)s_sub[ks][r];
+    }
+    // p_s is per-wave (written and read by the same wave): same-wave LDS
+    // ordering makes it visible without a barrier
+
+    // ---- O += P V ---------------------------------------------------
+    #pragma unroll
+    for (int kk = 0; kk < 2; kk++) {
+      bf16x8 pa = *(const bf16x8*)(&p_s[w][i16][kk * 32 + g * 8]);
+      const int krow0 = kk * 32 + g * 8 + (i16 >> 2);
+      const int vcol4 = (i16 & 3) * 4;
+      #pragma unroll
+      for (int t = 0; t < HD / 16; t++) {
+        bf16x4v r0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_b64_t*)&v_s[cur][krow0][t * 16 + vcol4]);
+        bf16x4v r1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+            (lds_b64_t*)&v_s[cur][krow0 + 4][t * 16 + vcol4]);
+        bf16x8 vb;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) { vb[j] = r0[j]; vb[4 + j] = r1[j]; }
+        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[t], 0, 0, 0);
+      }
+    }
+
+    if (b + 1 < nblocks) {
+      store_tile(cur ^ 1);  // other buffer: no hazard with this block
+      __syncthreads();      // publish before every wave's next S phase
+    }
+  }
+
+  // ---- epilogue ------------------------------------------------------
+  #pragma unroll
+  for (int r = 0; r < 4; r++) {
+    const int qrow = q0_local + w * QW + g * 4 + r;
+    if (qrow >= L) continue;
+    float inv = (l_r[r] > 0.f) ? 1.f / l_r[r] : 0.f;
+    bf16* orow = out + ((long)(s0 + qrow) * nq + qh) * HD;
+    #pragma unroll
+    for (int t = 0; t < HD / 16; t++)
+      orow[t * 16 + i16] = __float2bfloat16(o_acc[t][r] * inv);
+    if (lse && i16 == 0)
+      lse[(long)(s0 + qrow) * nq + qh] = m_r[r] + __logf(fmaxf(l_r[r], 1e-30f));
+  }
+}
+
+static int attn_db_mode() {
+  static int v = [] {
+    const char* e = getenv("REALHF_AMD_ATTN_DB");
+    return e ? atoi(e) : -1;  // -1 = auto
+  }();
+  return v;
+}
+
 std::vector<torch::Tensor> attn_varlen_fwd(
     torch::Tensor q, torch::Tensor k, torch::Tensor v,
     torch::Tensor cu_seqlens, long max_seqlen, bool causal, double scale,
@@ -276,18 +487,34 @@ std::vector<torch::Tensor> attn_varlen_fwd(
   // upper bound on sum(ceil(L/QBLK)) — real tail blocks exit on the
   // device-side count, no host sync anywhere
   dim3 grid((unsigned)(total / QBLK + bs), nq);
+  int dbm = attn_db_mode();
+  bool use_db = (dbm == 1) || (dbm == -1 && max_seqlen >= 4096);
   if (hd == 128) {
-    hipLaunchKernelGGL((attn_varlen_fwd_kernel<128>), grid, dim3(64 * AV_WAVES), 0,
-      cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-      (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
-      blk_off.data_ptr<int>(), bs, (bf16*)out.data_ptr(),
-      lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
+    if (use_db)
+      hipLaunchKernelGGL((attn_varlen_fwd_db_kernel<128>), grid, dim3(64 * AV_WAVES), 0,
+        cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+        (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
+        blk_off.data_ptr<int>(), bs, (bf16*)out.data_ptr(),
+        lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
+    else
+      hipLaunchKernelGGL((attn_varlen_fwd_kernel<128>), grid, dim3(64 * AV_WAVES), 0,
+        cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+        (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
+        blk_off.data_ptr<int>(), bs, (bf16*)out.data_ptr(),
+        lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
   } else if (hd == 64) {
-    hipLaunchKernelGGL((attn_varlen_fwd_kernel<64>), grid, dim3(64 * AV_WAVES), 0,
-      cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
-      (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
-      blk_off.data_ptr<int>(), bs, (bf16*)out.data_ptr(),
-      lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
+    if (use_db)
+      hipLaunchKernelGGL((attn_varlen_fwd_db_kernel<64>), grid, dim3(64 * AV_WAVES), 0,
+        cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+        (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
+        blk_off.data_ptr<int>(), bs, (bf16*)out.data_ptr(),
+        lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
+    else
+      hipLaunchKernelGGL((attn_varlen_fwd_kernel<64>), grid, dim3(64 * AV_WAVES), 0,
+        cur_stream(), (const bf16*)q.data_ptr(), (const bf16*)k.data_ptr(),
+        (const bf16*)v.data_ptr(), cu_dev.data_ptr<int>(),
+        blk_off.data_ptr<int>(), bs, (bf16*)out.data_ptr(),
+        lse.data_ptr<float>(), nq, nkv, (float)scale, causal, (int)window);
   } else {
     TORCH_CHECK(false, "unsupported head_dim ", hd);
   }
