@@ -261,10 +261,11 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_kernel(
 // buffers; block i+1's global loads issue BEFORE block i's compute and the
 // LDS writes land after it, so the HBM/L2 stream overlaps the MFMA+softmax
 // phases and the two per-block __syncthreads collapse to one.  ~90 KB LDS
-// (2x KV tiles + P), still one workgroup per CU.  Selected for long
-// sequences (REALHF_AMD_ATTN_DB: 0 never, 1 always, default auto at
-// max_seqlen >= 4096); the single-buffer kernel above stays the default
-// for bench-length sequences.
+// (2x KV tiles + P), still one workgroup per CU.  MEASURED SLOWER at
+// every shape (317 vs 387 TF/s at 32k): the staging registers + larger
+// LDS hurt more than the saved barrier+overlap help.  Kept behind
+// REALHF_AMD_ATTN_DB=1 as the baseline for future pipelining work;
+// default always uses the single-buffer kernel.
 template <int HD>
 __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_db_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k,
@@ -461,7 +462,7 @@ __global__ __launch_bounds__(64 * AV_WAVES, 1) void attn_varlen_fwd_db_kernel(
 static int attn_db_mode() {
   static int v = [] {
     const char* e = getenv("REALHF_AMD_ATTN_DB");
-    return e ? atoi(e) : -1;  // -1 = auto
+    return e ? atoi(e) : 0;  // default: single-buffer (db measured slower)
   }();
   return v;
 }
