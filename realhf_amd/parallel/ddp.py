@@ -106,6 +106,7 @@ class ZeRO1Optimizer:
         self._sp_repl_ivs = []  # same, but only when SP needs the tp-sum
         self.tp_group = None
         self.tp_size = g.tp_size if g is not None else 1
+        self.pp_size = g.pp_size if g is not None else 1
         self.model_group = g.model_group() if g is not None else None
         if (g is not None and g.tp_size > 1 and hasattr(model, "layout")):
             from realhf_amd.models import param_layout as PL
@@ -395,7 +396,8 @@ class ZeRO1Optimizer:
                             dtype=torch.float32) ** 2
                         sq -= part * (self.tp_size - 1) / self.tp_size
             if self.model_group is not None and (
-                    self.dp_size > 1 or self.tp_size > 1):
+                    self.dp_size > 1 or self.tp_size > 1
+                    or self.pp_size > 1):
                 # model-group reduction (dp x tp x pp): one global norm,
                 # identical clip factor on every rank
                 dist.all_reduce(sq, group=self.model_group)
