@@ -283,6 +283,9 @@ class DFGExecutor:
                 )
                 if src_real._offloaded:
                     src_real.reload_from_offload()
+                if getattr(src_real, "_dp_sharded", False):
+                    with constants.model_scope(str(src_name)):
+                        src_real.gather_from_dp()
             if dst_m is not None:
                 dst_real = (
                     dst_m.module.model if hasattr(dst_m.module, "model")
@@ -290,6 +293,9 @@ class DFGExecutor:
                 )
                 if dst_real._offloaded:
                     dst_real.reload_from_offload()
+                if getattr(dst_real, "_dp_sharded", False):
+                    with constants.model_scope(str(dst_name)):
+                        dst_real.gather_from_dp()
             cfg = (src_real or dst_real).config
             plan = build_realloc_plan(cfg, src_strat, dst_strat)
             execute_realloc(
