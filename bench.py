@@ -184,6 +184,9 @@ def main():
         Abstraction("ppo_actor", {
             "n_minibatches": args.n_minibatches, "gconfig": gconfig,
             "kl_ctl": 0.1, "adv_norm": True,
+            # actor_train is followed by critic_train: the final param
+            # all-gather hides under it (waited at next generate)
+            "defer_final_allgather": world > 1,
         })
     )
     critic_iface = make_interface(
@@ -304,7 +307,8 @@ def main():
     # bucketed ZeRO overlap path: a mis-ordered reduce-scatter would
     # silently diverge the replicas and invalidate the number)
     if world > 1:
-        models["critic"].module.optimizer.finish_allgather()
+        for n in ("actor", "critic"):
+            models[n].module.optimizer.finish_allgather()
         for n in ("actor", "critic"):
             with scope(n):
                 mm = models[n].module.module
