@@ -53,14 +53,28 @@ class GRPOInterface(PPOActorInterface):
         prompt_mask = data.data["prompt_mask"].bool()
         bs = data.bs
 
-        # group-normalized advantages: groups are consecutive group_size seqs
-        g = self.group_size
-        assert bs % g == 0, (bs, g)
-        grouped = score.view(bs // g, g)
-        adv_seq = (grouped - grouped.mean(dim=1, keepdim=True)) / (
-            grouped.std(dim=1, keepdim=True) + 1e-5
-        )
-        adv_seq = adv_seq.flatten()
+        # group-normalized advantages.  Groups are identified by the id
+        # PREFIX ("origid@g{k}" set at generation) — NOT by position: the
+        # balanced DP split can cut a group across ranks or reorder, so
+        # positional view(bs//g, g) would silently mix groups.  Members
+        # co-resident on this shard normalize together; a singleton group
+        # (rest of it on another rank) gets advantage 0.
+        prefixes = [str(i).split("@g")[0] for i in data.ids]
+        uniq = {p: j for j, p in enumerate(dict.fromkeys(prefixes))}
+        gidx = torch.tensor([uniq[p] for p in prefixes], device=score.device)
+        n_groups = len(uniq)
+        cnt = torch.zeros(n_groups, device=score.device).scatter_add_(
+            0, gidx, torch.ones_like(score))
+        gsum = torch.zeros(n_groups, device=score.device).scatter_add_(
+            0, gidx, score)
+        gmean = gsum / cnt.clamp(min=1)
+        gsq = torch.zeros(n_groups, device=score.device).scatter_add_(
+            0, gidx, (score - gmean[gidx]) ** 2)
+        # sample std (ddof=1) to match torch.std's default
+        gstd = (gsq / (cnt - 1).clamp(min=1)).sqrt()
+        adv_seq = (score - gmean[gidx]) / (gstd[gidx] + 1e-5)
+        adv_seq = torch.where(cnt[gidx] > 1, adv_seq,
+                              torch.zeros_like(adv_seq))
 
         shift = build_shift_one_indices(ids.shape[0], cu)
         loss_mask = ~prompt_mask[shift]

@@ -231,3 +231,69 @@ def test_grpo_step():
     st = iface.train_step(actor, rollout)
     assert np.isfinite(st["actor_loss"])
     assert "kl_in_loss" in st
+
+
+def test_grpo_group_norm_id_based():
+    """Group advantages come from id prefixes, not positions: a shuffled
+    or split batch (balanced DP split can cut groups across ranks) must
+    still normalize within the right group; singletons get 0."""
+    import torch
+
+    from realhf_amd.api.data import SequenceSample
+    from realhf_amd.interfaces.grpo import GRPOInterface
+
+    iface = GRPOInterface(group_size=2, n_minibatches=1)
+    # ids deliberately interleaved + one singleton (g1 of "b" elsewhere)
+    ids = ["a@g0", "b@g0", "a@g1", "c@g0", "c@g1"]
+    score = torch.tensor([1.0, 5.0, 3.0, 2.0, 4.0])
+    lens = [4, 4, 4, 4, 4]
+    toks = torch.randint(0, 32, (sum(lens),))
+    pm = torch.zeros(sum(lens), dtype=torch.bool)
+    data = SequenceSample(
+        keys=("packed_input_ids", "packed_logprobs", "packed_ref_logprobs",
+              "rewards", "prompt_mask"),
+        ids=ids,
+        seqlens={
+            "packed_input_ids": [[l] for l in lens],
+            "packed_logprobs": [[l - 1] for l in lens],
+            "packed_ref_logprobs": [[l - 1] for l in lens],
+            "rewards": [[1]] * 5,
+            "prompt_mask": [[l] for l in lens],
+        },
+        data={
+            "packed_input_ids": toks,
+            "packed_logprobs": torch.zeros(sum(lens) - 5),
+            "packed_ref_logprobs": torch.zeros(sum(lens) - 5),
+            "rewards": score,
+            "prompt_mask": pm,
+        },
+    )
+
+    captured = {}
+
+    class _FakeEngine:
+        def train_batch(self, mb, loss_fn, **kw):
+            captured["adv"] = mb.data["advantages"]
+            captured["ids"] = list(mb.ids)
+            return {}
+
+    class _FakeModel:
+        module = _FakeEngine()
+
+        class version:
+            global_step = 0
+
+        @staticmethod
+        def inc_version():
+            pass
+
+    iface.train_step(_FakeModel(), data)
+    adv = captured["adv"]
+    # per-token adv; group "a" = scores (1, 3): normalized to (-1, +1)/std
+    # with ddof=1 std = sqrt(2): first seq tokens ~ -0.707
+    seg = torch.repeat_interleave(torch.arange(5), torch.tensor([3, 3, 3, 3, 3]))
+    per_seq = torch.stack([adv[seg == i][0] for i in range(5)])
+    assert per_seq[1].abs() < 1e-4  # singleton "b@g0" -> 0
+    assert torch.allclose(per_seq[0], -per_seq[2], atol=1e-5)  # group a pair
+    assert torch.allclose(per_seq[3], -per_seq[4], atol=1e-5)  # group c pair
+    assert per_seq[0] < 0 and per_seq[2] > 0
