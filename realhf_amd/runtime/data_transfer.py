@@ -25,7 +25,7 @@ metadata (ids, seqlens, keys — a few hundred bytes) goes through the
 object collectives.  Falls back to pure object transport for non-CUDA
 runs (gloo CPU tests).
 """
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, List, Optional
 
 import torch
 import torch.distributed as dist
