@@ -386,8 +386,26 @@ class ReaLModel(nn.Module):
         n, pad, group = self._dp_shard_meta
         full = torch.empty(n + pad, dtype=self._dp_shard.dtype,
                            device=self._dp_shard.device)
-        dist.all_gather_into_tensor(full, self._dp_shard.contiguous(),
-                                    group=group)
+        try:
+            backend = dist.get_backend(group)
+        except Exception:
+            backend = "nccl"
+        if backend == "gloo" and self._dp_shard.is_cuda:
+            # gloo lacks CUDA all-gather (1-GPU multi-rank tests): one
+            # broadcast per shard; RCCL takes the single-collective path
+            world = dist.get_world_size(group)
+            ranks = (dist.get_process_group_ranks(group)
+                     if group is not None else list(range(world)))
+            l = (n + pad) // world
+            me = dist.get_rank(group)
+            for r in range(world):
+                piece = full[r * l:(r + 1) * l]
+                if r == me:
+                    piece.copy_(self._dp_shard)
+                dist.broadcast(piece, src=ranks[r], group=group)
+        else:
+            dist.all_gather_into_tensor(full, self._dp_shard.contiguous(),
+                                        group=group)
         self.flat_param = full[:n]
         self._dp_shard = None
         self._dp_sharded = False

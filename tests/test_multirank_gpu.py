@@ -36,6 +36,8 @@ def _worker(data, fileroot):
     try:
         cfg = PPOConfig(experiment_name="t-mr", trial_name="gpu", n_gpus=2)
         cfg.allocation_mode = "heuristic"
+        # ZeRO-3-style frozen sharding on CUDA tensors (gloo fallback)
+        cfg.ref.offload = "dp_shard"
         cfg.dataset.type_ = "prompt"
         cfg.dataset.path = data
         cfg.dataset.train_bs_n_seqs = 8
