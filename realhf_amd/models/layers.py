@@ -481,4 +481,10 @@ class OutputHead(nn.Module):
             # fused copy-to-TP + GEMM: bwd all-reduce overlaps dW
             return mappings.column_parallel_linear(h, w, False)
         h = mappings.copy_to_tp_region(h)
+        if not torch.is_grad_enabled():
+            # decode: the vocab head streams 262 MB of weights per token —
+            # the weight-streaming skinny kernel beats hipBLASLt at M<=16
+            o = ops.maybe_skinny_linear(h, w)
+            if o is not None:
+                return o
         return _linear(h, w)  # [total, vocab/tp] — vocab-parallel logits
