@@ -126,3 +126,52 @@ def test_profile_experiment(tmp_path, monkeypatch):
     out = os.path.join(constants.LOG_ROOT("t-prof", "x"),
                        "profile_result.json")
     assert json.load(open(out))["d1t1p1"]
+
+
+def test_sentiment_example_interface(tmp_path, monkeypatch):
+    """Custom external-scorer reward interface (reference:
+    examples/customized_exp/ppo_sentiment.py): a stub classifier stands
+    in for the HF model; the interface must emit per-sequence rewards."""
+    import sys
+    import types
+
+    import torch
+
+    # stub transformers.AutoModelForSequenceClassification
+    class _StubModel(torch.nn.Module):
+        def forward(self, input_ids=None, attention_mask=None):
+            bs = input_ids.shape[0]
+            out = types.SimpleNamespace()
+            out.logits = torch.stack(
+                [torch.zeros(bs), input_ids.float().mean(dim=1)], dim=1)
+            return out
+
+        def parameters(self):
+            yield torch.nn.Parameter(torch.zeros(1))
+
+    import transformers
+
+    monkeypatch.setattr(
+        transformers.AutoModelForSequenceClassification, "from_pretrained",
+        staticmethod(lambda path: _StubModel()))
+    monkeypatch.setenv("SCORER", "stub")
+    sys.path.insert(0, "examples/customized_exp")
+    try:
+        import ppo_sentiment
+
+        iface = ppo_sentiment.SentimentScoringInterface()
+    finally:
+        sys.path.pop(0)
+
+    from realhf_amd.api.data import SequenceSample
+
+    toks = torch.arange(14)
+    data = SequenceSample(
+        keys=("packed_input_ids",), ids=["a", "b"],
+        seqlens={"packed_input_ids": [[6], [8]]},
+        data={"packed_input_ids": toks},
+    )
+    out = iface.inference(None, data)
+    assert out.keys == ("rewards",) and out.data["rewards"].shape == (2,)
+    # stub reward = mean token id of the PADDED row
+    assert out.data["rewards"][1] > out.data["rewards"][0]
