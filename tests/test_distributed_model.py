@@ -381,3 +381,73 @@ def _defer_ag_worker():
 @pytest.mark.distributed
 def test_deferred_allgather_matches_blocking():
     LocalMultiProcessTest(2, _defer_ag_worker).launch()
+
+
+def _tp_gradnorm_worker():
+    """Global grad norm at tp2: replicated params (layernorms) counted
+    ONCE across tp ranks, TP-sharded params summed over ranks — every
+    rank must report the same analytic norm."""
+    import math
+
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models import param_layout as PL
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel.ddp import OptimizerConfig, ZeRO1Optimizer
+
+    init_global_constants(num_dp=1, num_tp=2, num_pp=1, model_name="m")
+    cfg = make_test_config(n_layers=2, hidden_dim=32, n_heads=4, n_kv_heads=4,
+                           vocab_size=64)
+    cfg.dtype = "float32"
+    with constants.model_scope("m"):
+        g = constants.grid()
+        m = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                      tp_rank=g.tp_rank, tp_size=2)
+        m.random_init()
+        opt = ZeRO1Optimizer(
+            m, OptimizerConfig(lr=0.0, gradient_clipping=1.0,
+                               warmup_steps_proportion=0.0))
+        opt.zero_grad()
+        # grads: 2.0 on replicated params (identical across tp, must
+        # count once), 1.0 on tp-sharded params (distinct per rank)
+        expect_sq = 0.0
+        for k in m.layout.keys:
+            kind = PL.key_kind(k)
+            if kind == "head":
+                kind = PL.REPLICATED if cfg.is_critic else PL.VOCAB
+            v = m.grad_view(k)
+            if kind == PL.REPLICATED:
+                v.fill_(2.0)
+                expect_sq += 4.0 * v.numel()  # once, not 2x
+            else:
+                v.fill_(1.0)
+                expect_sq += 1.0 * v.numel() * 2  # both ranks' shards
+        stats = opt.step()
+        # expect_sq computed per rank double-counts... recompute exactly:
+        # replicated numel identical on both ranks; sharded numel per rank
+        # -> sum over ranks.  Build it via an all_reduce of per-rank parts.
+        repl_sq = sum(
+            4.0 * m.grad_view(k).numel() for k in m.layout.keys
+            if (PL.key_kind(k) if PL.key_kind(k) != "head"
+                else (PL.REPLICATED if cfg.is_critic else PL.VOCAB))
+            == PL.REPLICATED
+        )
+        shard_sq_local = sum(
+            1.0 * m.grad_view(k).numel() for k in m.layout.keys
+            if (PL.key_kind(k) if PL.key_kind(k) != "head"
+                else (PL.REPLICATED if cfg.is_critic else PL.VOCAB))
+            != PL.REPLICATED
+        )
+        t = torch.tensor([shard_sq_local])
+        dist.all_reduce(t)
+        want = math.sqrt(repl_sq + float(t))
+        assert abs(stats["grad_norm"] - want) / want < 1e-5, (
+            stats["grad_norm"], want)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_tp2_grad_norm_replicated_once():
+    LocalMultiProcessTest(2, _tp_gradnorm_worker).launch()
