@@ -237,3 +237,34 @@ def test_ppo_heuristic_asymmetric_four_ranks(tmp_path):
     _write_prompt_data(data, n=16)
     LocalMultiProcessTest(4, _ppo_heuristic4_worker, data,
                           str(tmp_path / "root")).launch()
+
+
+def test_ppo_heuristic8_static_plan():
+    """World-8 asymmetric PPO plan (no processes): critic_inf on the
+    first half, rew_inf on the second (where the reward model lives),
+    trains whole-node; the train MFCs receive exactly the cross-half
+    broadcasts of `values` and `rewards`."""
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import build_experiment
+    from realhf_amd.runtime.executor import DFGExecutor
+
+    cfg = PPOConfig(allocation_mode="heuristic")
+    built = build_experiment(cfg, 8)
+    ex = DFGExecutor(built.graph, built.allocations, {}, built.interfaces,
+                     built.model_strategies)
+    ex.world = 8
+    plan = ex._static_plan()
+    assert plan["critic_inf"].mesh == (0, 1, 2, 3)
+    assert plan["rew_inf"].mesh == (4, 5, 6, 7)
+    assert plan["ref_inf"].mesh == tuple(range(8))
+    assert plan["actor_train"].mesh == tuple(range(8))
+    # actor_train: values held by {0..3} -> broadcast to {4..7};
+    # rewards held by {4..7} -> broadcast to {0..3}
+    t = {x.key: x for x in plan["actor_train"].transfers}
+    assert set(t) == {"values", "rewards"}
+    assert set(t["values"].group_ranks) >= {4, 5, 6, 7}
+    assert t["values"].src in (0, 1, 2, 3)
+    assert set(t["rewards"].group_ranks) >= {0, 1, 2, 3}
+    assert t["rewards"].src in (4, 5, 6, 7)
+    # critic_train afterwards needs nothing new (actor_train spread both)
+    assert plan["critic_train"].transfers == []
