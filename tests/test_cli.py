@@ -57,7 +57,8 @@ def test_examples_run(tmp_path):
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     for script in ("examples/load_and_eval_rw.py",
                    "examples/new_algorithms/reinforce.py",
-                   "examples/customized_exp/ppo_ref_ema.py"):
+                   "examples/customized_exp/ppo_ref_ema.py",
+                   "examples/visualize_dfg.py"):
         r = subprocess.run([sys.executable, os.path.join(root, script)],
                            capture_output=True, text=True, timeout=300)
         assert r.returncode == 0, (script, r.stdout[-800:], r.stderr[-800:])
