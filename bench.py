@@ -24,11 +24,16 @@ _TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                       "realhf_amd", "data", "tunableop_gfx950.csv")
 if os.path.exists(_TUNED) and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ:
     # TunableOp appends the device ordinal before .csv on read — stage a
-    # copy per possible ordinal under /tmp
+    # copy per possible ordinal under /tmp.  All torchrun ranks execute
+    # this concurrently: copy to a rank-unique temp then os.replace (atomic
+    # on POSIX) so no rank ever reads a half-written CSV.
     import shutil as _sh
 
+    _pid = os.getpid()
     for _d in range(8):
-        _sh.copy(_TUNED, f"/tmp/realhf_tunableop{_d}.csv")
+        _tmp = f"/tmp/realhf_tunableop{_d}.csv.{_pid}"
+        _sh.copy(_TUNED, _tmp)
+        os.replace(_tmp, f"/tmp/realhf_tunableop{_d}.csv")
     os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
     os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
     os.environ["PYTORCH_TUNABLEOP_FILENAME"] = "/tmp/realhf_tunableop.csv"
