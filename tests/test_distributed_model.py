@@ -451,3 +451,82 @@ def _tp_gradnorm_worker():
 @pytest.mark.distributed
 def test_tp2_grad_norm_replicated_once():
     LocalMultiProcessTest(2, _tp_gradnorm_worker).launch()
+
+
+def _dp8_overlap_worker():
+    """dp8 with a tiny bucket size: the 8-GPU scale bench's geometry
+    (many buckets, last one shorter; align = 256*8) — the bucket math
+    must give the same step as a single process on the combined batch."""
+    import os as _os
+
+    import torch.distributed as dist
+
+    from realhf_amd.api.config import Abstraction, ModelName
+    from realhf_amd.api.data import SequenceSample
+    from realhf_amd.api.model import FinetuneSpec, Model, make_backend, make_interface
+    import realhf_amd.interfaces  # noqa: F401
+    import realhf_amd.runtime.engine  # noqa: F401
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models.real_model import ReaLModel
+
+    _os.environ["REALHF_AMD_ZERO_OVERLAP"] = "1"
+    cfg = make_test_config(n_layers=2, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    cfg.family = "llama"
+    sd = _full_reference_sd(cfg, seed=31)
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    init_global_constants(num_dp=world, num_tp=1, num_pp=1, model_name="m")
+
+    def batch(seed):
+        rng = np.random.RandomState(seed)
+        toks = torch.from_numpy(
+            rng.randint(0, cfg.vocab_size, size=20)).long()
+        pm = torch.zeros(20, dtype=torch.bool)
+        pm[:3] = True
+        return SequenceSample(
+            keys=("packed_input_ids", "prompt_mask"),
+            ids=[f"s{seed}-0", f"s{seed}-1"],
+            seqlens={"packed_input_ids": [[10], [10]], "prompt_mask": [[10], [10]]},
+            data={"packed_input_ids": toks, "prompt_mask": pm},
+        )
+
+    opt_args = {"optimizer": {"lr": 1e-2, "warmup_steps_proportion": 0.0,
+                              "lr_scheduler_type": "constant",
+                              "gradient_clipping": 1.0},
+                "bucket_size": 4096}  # force many buckets (align=2048)
+    with constants.model_scope("m"):
+        m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+        _fill_model_from_full(m, cfg, sd)
+        model = Model(ModelName("m", 0), m, None, torch.device("cpu"),
+                      torch.float32)
+        model = make_backend(Abstraction("zero1", opt_args)).initialize(
+            model, FinetuneSpec(1, 64, 4))
+        assert model.module.optimizer.overlap_comm
+        assert len(model.module.optimizer.buckets) > 4
+        iface = make_interface(Abstraction("sft"))
+        iface.train_step(model, batch(200 + rank))
+        flat = m.flat_param.clone()
+
+    single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    _fill_model_from_full(single, cfg, sd)
+    constants.clear_grids()
+    smodel = Model(ModelName("s", 0), single, None, torch.device("cpu"),
+                   torch.float32)
+    smodel = make_backend(Abstraction("zero1", opt_args)).initialize(
+        smodel, FinetuneSpec(1, 64, 4))
+    combined = SequenceSample.gather([batch(200 + r) for r in range(world)])
+    make_interface(Abstraction("sft")).train_step(smodel, combined)
+    n = flat.numel()
+    torch.testing.assert_close(flat, single.flat_param[:n], atol=1e-5,
+                               rtol=1e-4)
+
+
+@pytest.mark.distributed
+def test_dp8_zero1_overlap_matches_single():
+    """The round-end scale bench runs dp8 — exercise THAT bucket
+    geometry (8-way shard alignment, short last bucket, grad clipping
+    through the model-group norm) on gloo."""
+    LocalMultiProcessTest(8, _dp8_overlap_worker).launch()
