@@ -49,6 +49,9 @@ def _pack_payload(sample: SequenceSample):
     by_dtype: Dict[torch.dtype, List[torch.Tensor]] = {}
     for k in sample.keys:
         t = sample.data[k]
+        if t is None:  # metadata-only key (SequenceSample.gather keeps them)
+            meta["specs"].append((k, None, None))
+            continue
         meta["specs"].append((k, tuple(t.shape), str(t.dtype)))
         by_dtype.setdefault(t.dtype, []).append(t.reshape(-1))
     flats = {str(dt): torch.cat(ts) if len(ts) > 1 else ts[0]
@@ -61,6 +64,9 @@ def _unpack_payload(meta, flats: Dict[str, torch.Tensor]) -> SequenceSample:
     offsets = {dt: 0 for dt in flats}
     data = {}
     for k, shape, dts in meta["specs"]:
+        if shape is None:
+            data[k] = None
+            continue
         n = 1
         for s in shape:
             n *= s

@@ -160,3 +160,32 @@ def test_flops_formula_sanity():
     assert dense_transformer_flops(
         32, 4096, 11008, 32000, 32, 32, 128, 1024, 0.0, backward=True
     ) == 3 * f
+
+
+def test_payload_pack_unpack_roundtrip():
+    """Device-transfer payload codec: mixed dtypes, a 0-d-free mix of
+    shapes, and a None-valued (metadata-only) key must survive the
+    pack -> flat-tensors -> unpack round trip bit-exactly."""
+    from realhf_amd.runtime.data_transfer import _pack_payload, _unpack_payload
+
+    s = SequenceSample(
+        keys=("ids", "logp", "mask", "meta_only"),
+        ids=["a", "b"],
+        seqlens={"ids": [[4], [3]], "logp": [[3], [2]],
+                 "mask": [[4], [3]], "meta_only": [[1], [1]]},
+        data={
+            "ids": torch.arange(7, dtype=torch.long),
+            "logp": torch.randn(5, dtype=torch.float32),
+            "mask": torch.tensor([1, 0, 1, 1, 0, 1, 1], dtype=torch.bool),
+            "meta_only": None,
+        },
+    )
+    meta, flats = _pack_payload(s)
+    # one flat tensor per dtype
+    assert set(flats) == {"torch.int64", "torch.float32", "torch.bool"}
+    out = _unpack_payload(meta, flats)
+    assert out.ids == s.ids and out.keys == s.keys
+    assert out.data["meta_only"] is None
+    for k in ("ids", "logp", "mask"):
+        assert out.data[k].dtype == s.data[k].dtype
+        assert torch.equal(out.data[k], s.data[k])
