@@ -183,6 +183,14 @@ class DFGExecutor:
                         group_ranks=tuple(sorted({src} | need)),
                     ))
                     h |= need
+            if mfc.output_keys:
+                # run_step only merges outputs of non-train MFCs across
+                # the mesh (train/eval return stat dicts) — a train MFC
+                # with output keys would leave stale holder state
+                assert mfc.interface_type in (
+                    ModelInterfaceType.GENERATE, ModelInterfaceType.INFERENCE,
+                ), (f"{mfc.name}: {mfc.interface_type} MFCs cannot produce "
+                    "data outputs (stats only)")
             for k in mfc.output_keys:
                 holders[k] = set(mesh)
             plan[mfc.name] = _MFCPlan(
@@ -317,11 +325,16 @@ class DFGExecutor:
             k: batch.select_keys([k]) for k in batch.keys
         }
         local_stats: Dict[str, dict] = {}
-        device = None
+        # transfer-path choice (device RCCL payloads vs gloo object
+        # collectives) must be IDENTICAL on every rank of a broadcast
+        # group — key off CUDA availability, not off this rank's model
+        # placement (a rank hosting no model must still pick the same path)
+        dev = (torch.device("cuda", torch.cuda.current_device())
+               if torch.cuda.is_available() else None)
         for m in self.models.values():
-            device = m.device
+            if m.device.type != "cuda":
+                dev = None  # CPU-model runs (tests) use object collectives
             break
-        dev = device if (device is not None and device.type == "cuda") else None
 
         for mfc in self._topo_order:
             plan = self._plan[mfc.name]
