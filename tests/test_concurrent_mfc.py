@@ -268,3 +268,22 @@ def test_ppo_heuristic8_static_plan():
     assert t["rewards"].src in (4, 5, 6, 7)
     # critic_train afterwards needs nothing new (actor_train spread both)
     assert plan["critic_train"].transfers == []
+
+
+def test_train_mfc_with_outputs_rejected():
+    """Train MFCs return stat dicts, not data — output keys on one must
+    fail at PLAN time, not silently corrupt holder tracking."""
+    ma = ModelName("a", 0)
+    strat = ParallelStrategy.make(1, 2, 1, ranks=[0, 1])
+    mfcs = [
+        MFCDef(name="t", model_name=ma,
+               interface_type=ModelInterfaceType.TRAIN_STEP,
+               interface_impl=Abstraction("x"), input_keys=("x",),
+               output_keys=("bad",)),
+    ]
+    graph = build_graph(mfcs)
+    with pytest.raises(AssertionError, match="cannot produce"):
+        ex = DFGExecutor(graph, {"t": MFCAllocation(strategy=strat)}, {}, {},
+                         {ma: strat})
+        ex.world = 2
+        ex._static_plan()
