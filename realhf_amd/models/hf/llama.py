@@ -63,6 +63,13 @@ def to_hf(cfg: ReaLModelConfig, sd: Dict[str, torch.Tensor]):
 
 def config_from_hf(hf: dict) -> ReaLModelConfig:
     nh = hf["num_attention_heads"]
+    # HF rope_scaling {"type"|"rope_type": "linear"|"dynamic", "factor": f}
+    # (reference rotary.py:121 supports both) -> rotary_scaling(_type)
+    rs = hf.get("rope_scaling") or {}
+    rs_type = rs.get("type") or rs.get("rope_type")
+    rs_factor = rs.get("factor")
+    if rs_type not in (None, "linear", "dynamic"):
+        raise NotImplementedError(f"rope_scaling type {rs_type!r}")
     return ReaLModelConfig(
         n_layers=hf["num_hidden_layers"],
         hidden_dim=hf["hidden_size"],
@@ -77,6 +84,8 @@ def config_from_hf(hf: dict) -> ReaLModelConfig:
         layer_norm_epsilon=hf.get("rms_norm_eps", 1e-5),
         apply_rotary=True,
         rotary_base=hf.get("rope_theta", 10000.0),
+        rotary_scaling=rs_factor,
+        rotary_scaling_type=rs_type,
         tied_embedding=hf.get("tie_word_embeddings", False),
         use_attention_bias=hf.get("attention_bias", False),
     )
@@ -96,6 +105,11 @@ def config_to_hf(cfg: ReaLModelConfig) -> dict:
         "max_position_embeddings": cfg.max_position_embeddings,
         "rms_norm_eps": cfg.layer_norm_epsilon,
         "rope_theta": cfg.rotary_base,
+        **(
+            {"rope_scaling": {"type": cfg.rotary_scaling_type,
+                              "factor": cfg.rotary_scaling}}
+            if cfg.rotary_scaling_type else {}
+        ),
         "tie_word_embeddings": cfg.tied_embedding,
         "attention_bias": cfg.use_attention_bias,
         "hidden_act": "silu",

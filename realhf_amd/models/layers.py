@@ -243,6 +243,8 @@ class ReaLModelBlock(nn.Module):
             cos, sin = ops.rotary_cache.get(
                 self.hd, int(k_cache.shape[1]), cfg.rotary_base, x.device,
                 scaling=cfg.rotary_scaling,
+                scaling_type=cfg.rotary_scaling_type,
+                orig_max_pos=cfg.max_position_embeddings,
             )
             q = C.rope_qkv_decode(
                 qkv_raw, bias, k_cache, v_cache,
@@ -303,6 +305,8 @@ class ReaLModelBlock(nn.Module):
                 cfg.rotary_base,
                 x.device,
                 scaling=cfg.rotary_scaling,
+                scaling_type=cfg.rotary_scaling_type,
+                orig_max_pos=cfg.max_position_embeddings,
             )
             q = ops.apply_rotary(q, cos, sin, positions, cfg.rotary_interleaved)
             k = ops.apply_rotary(k, cos, sin, positions, cfg.rotary_interleaved)

@@ -82,11 +82,29 @@ class RotaryCache:
         device,
         dtype=torch.float32,
         scaling: Optional[float] = None,
+        scaling_type: Optional[str] = None,
+        orig_max_pos: Optional[int] = None,
     ):
+        """scaling_type: None/"linear" divide positions by `scaling`
+        (position interpolation); "dynamic" rescales `base` NTK-style when
+        max_len exceeds orig_max_pos (reference rotary.py:121
+        `_update_cos_sin_cache` mirrors HF's dynamic-NTK rule).  Dynamic
+        tables are keyed by their exact length — the base depends on it —
+        so no doubling growth is applied there."""
+        dynamic = scaling_type == "dynamic" and scaling is not None
+        if dynamic:
+            orig = int(orig_max_pos or max_len)
+            max_len = max(int(max_len), orig)
+            if max_len > orig:
+                base = base * (
+                    (scaling * max_len / orig) - (scaling - 1)
+                ) ** (head_dim / (head_dim - 2))
+            scaling = None  # dynamic rescales base; positions stay unscaled
         key = (head_dim, base, str(device), dtype, scaling)
         cos, sin, cached_len = self._cache.get(key, (None, None, 0))
         if cached_len < max_len:
-            max_len = max(max_len, 2 * cached_len, 2048)
+            if not dynamic:
+                max_len = max(max_len, 2 * cached_len, 2048)
             inv_freq = 1.0 / (
                 base ** (torch.arange(0, head_dim, 2, dtype=torch.float64) / head_dim)
             )

@@ -144,3 +144,45 @@ def test_save_load_roundtrip(tmp_path):
 
     m = transformers.AutoModelForCausalLM.from_pretrained(str(tmp_path))
     assert m.config.num_hidden_layers == cfg.n_layers
+
+
+@pytest.mark.parametrize("scaling_type", ["linear", "dynamic"])
+def test_rope_scaling_matches_hf(scaling_type):
+    """Linear position-interpolation and dynamic-NTK rope scaling vs
+    transformers (reference modules/rotary.py:121 supports both via
+    `_update_cos_sin_cache`).  Dynamic only binds past
+    max_position_embeddings, so sequences run to 2x that length; HF
+    recomputes its table per forward from that forward's seq_len, so we
+    compare per-sequence with matching max_seqlen."""
+    fam = hf_reg.get_family("llama")
+    cfg = fam.make_test_config(max_position_embeddings=16)
+    cfg.dtype = "float32"
+    cfg.rotary_scaling = 2.0
+    cfg.rotary_scaling_type = scaling_type
+    hf_model = make_hf_model("llama", cfg)
+    assert hf_model.config.rope_scaling["factor"] == 2.0
+
+    model = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    hf_reg.load_from_hf_state_dict(model, "llama", hf_model.state_dict())
+
+    rng = np.random.RandomState(3)
+    for L in (12, 32):  # under and over max_position_embeddings
+        ids = torch.from_numpy(rng.randint(0, cfg.vocab_size, size=L)).long()
+        cu = torch.tensor([0, L], dtype=torch.int32)
+        with torch.no_grad():
+            ours = model(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=L)
+            ref = hf_model(input_ids=ids.unsqueeze(0)).logits[0]
+        torch.testing.assert_close(
+            torch.log_softmax(ours.float(), -1),
+            torch.log_softmax(ref.float(), -1),
+            atol=2e-4,
+            rtol=2e-3,
+        )
+    # the scaling must actually change the logits past the window
+    cfg2 = fam.make_test_config(max_position_embeddings=16)
+    cfg2.dtype = "float32"
+    plain = ReaLModel(cfg2, device="cpu", dtype=torch.float32)
+    hf_reg.load_from_hf_state_dict(plain, "llama", hf_model.state_dict())
+    with torch.no_grad():
+        base = plain(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=32)
+    assert not torch.allclose(ours, base)
