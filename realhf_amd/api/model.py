@@ -36,7 +36,8 @@ class MoEConfig:
     z_loss_coef: float = 0.0
     input_jitter_eps: Optional[float] = None
     capacity_factor: Optional[float] = None
-    token_drop_policy: str = "probs"
+    token_drop_policy: str = "probs"  # "probs" | "position" (model_api.py:120)
+    pad_to_capacity: bool = False  # static [n_experts*cap] dispatch shapes
     use_grouped_gemm: bool = True
     expert_parallel_size: int = 1  # EP over xGMI (absent in reference)
 

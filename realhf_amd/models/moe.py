@@ -17,6 +17,7 @@ matching the reference's grouped_gemm.ops.gmm training path
 CPU/odd-shape fallback (REALHF_AMD_MOE_LOOP=1 forces it, for tests).
 """
 import contextlib
+import math
 from typing import Dict, List
 
 import torch
@@ -80,6 +81,39 @@ def sinkhorn(cost: torch.Tensor, tol: float = 1e-4, n_iter: int = 8):
     return d1.unsqueeze(0) * cost * d0.unsqueeze(1)
 
 
+def apply_expert_capacity(
+    scores: torch.Tensor,   # [tokens, k] routing weights
+    idx: torch.Tensor,      # [tokens, k] global expert ids
+    n_experts: int,
+    capacity_factor: float,
+    policy: str,            # "probs" | "position"
+):
+    """Limit each expert to cap = ceil(tokens*k/n_experts * factor)
+    assignments (reference: utils/moe.py:310 topk_softmax_with_capacity).
+
+    Over-capacity assignments are masked: "probs" keeps the cap
+    highest-weight assignments per expert, "position" the cap earliest
+    tokens.  Returns (masked scores, kept mask [tokens,k], cap).
+    """
+    ntok, k = scores.shape
+    cap = int(math.ceil(ntok * k / n_experts * capacity_factor))
+    sel = torch.zeros(ntok, n_experts, dtype=torch.bool, device=idx.device)
+    sel.scatter_(1, idx, True)
+    if policy == "position":
+        rank = sel.long().cumsum(0) - 1          # arrival order per expert
+    elif policy == "probs":
+        gates = torch.zeros(ntok, n_experts, device=scores.device,
+                            dtype=scores.dtype)
+        gates.scatter_(1, idx, scores)
+        order = torch.argsort(gates, dim=0, descending=True, stable=True)
+        rank = torch.argsort(order, dim=0)       # weight rank per expert
+    else:
+        raise ValueError(f"token_drop_policy {policy!r}")
+    kept_dense = sel & (rank < cap)
+    kept = torch.gather(kept_dense, 1, idx)
+    return scores * kept, kept, cap
+
+
 class TopKRouter(torch.nn.Module):
     def __init__(self, cfg: ReaLModelConfig, weight: torch.Tensor):
         super().__init__()
@@ -114,7 +148,13 @@ class TopKRouter(torch.nn.Module):
             if moe.z_loss_coef > 0 and _collecting():
                 z = torch.logsumexp(logits, dim=-1).square().mean()
                 _AUX_LOSSES.append(moe.z_loss_coef * z)
-        return scores, idx
+        kept = None
+        if moe.capacity_factor is not None:
+            scores, kept, _ = apply_expert_capacity(
+                scores, idx, moe.num_experts, moe.capacity_factor,
+                moe.token_drop_policy,
+            )
+        return scores, idx, kept
 
 
 class _GroupedGemm(torch.autograd.Function):
@@ -274,9 +314,11 @@ class MoELayer(torch.nn.Module):
 
     # ------------------------------------------------------------ forward
     def forward(self, h: torch.Tensor) -> torch.Tensor:
-        scores, idx = self.router(h)  # [tokens, k]
+        scores, idx, kept = self.router(h)  # [tokens, k]
         k = self.cfg.moe.top_k
         h_tp = mappings.copy_to_tp_region(h)
+        if kept is not None:
+            return self._capacity_forward(h, h_tp, scores, idx, kept)
 
         flat_idx = idx.flatten()  # [tokens*k] global expert ids
         sort_order = torch.argsort(flat_idx, stable=True)
@@ -296,6 +338,64 @@ class MoELayer(torch.nn.Module):
         combined.index_add_(0, token_of, out_sorted.float() * w.unsqueeze(-1))
         combined = combined.to(h.dtype)
         return mappings.reduce_from_tp_region(combined)
+
+    def _capacity_forward(self, h, h_tp, scores, idx, kept):
+        """Dispatch with expert-capacity limiting (capacity_factor set).
+
+        Dropped (token, slot) assignments leave the dispatch entirely —
+        their tokens pass through with zero contribution from that slot,
+        like the reference's drop mode (utils/moe.py:380-389).  With
+        pad_to_capacity every expert processes exactly `cap` rows (zero
+        rows for empty slots), so all dispatch shapes are static.
+        """
+        moe = self.cfg.moe
+        ntok, k = scores.shape
+        cap = int(math.ceil(ntok * k / self.n_experts * moe.capacity_factor))
+        flat_keep = kept.flatten()
+        flat_idx = idx.flatten()[flat_keep]
+        flat_tok = torch.arange(
+            ntok, device=idx.device
+        ).repeat_interleave(k)[flat_keep]
+        flat_w = scores.to(torch.float32).flatten()[flat_keep]
+
+        if moe.pad_to_capacity:
+            # static [n_experts*cap] dispatch: slot = e*cap + arrival rank
+            order = torch.argsort(flat_idx, stable=True)
+            e_sorted = flat_idx[order]
+            seg_start = torch.searchsorted(
+                e_sorted, torch.arange(self.n_experts, device=idx.device)
+            )
+            pos = (torch.arange(e_sorted.numel(), device=idx.device)
+                   - seg_start[e_sorted])
+            slots = e_sorted * cap + pos
+            nslot = self.n_experts * cap
+            permuted = torch.zeros(nslot, h_tp.shape[1], dtype=h_tp.dtype,
+                                   device=h_tp.device)
+            permuted[slots] = h_tp[flat_tok[order]]
+            token_of = torch.zeros(nslot, dtype=torch.long, device=h_tp.device)
+            token_of[slots] = flat_tok[order]
+            w = torch.zeros(nslot, dtype=torch.float32, device=h_tp.device)
+            w[slots] = flat_w[order]
+            counts = torch.full((self.n_experts,), cap, dtype=torch.long,
+                                device=h_tp.device)
+        else:
+            order = torch.argsort(flat_idx, stable=True)
+            token_of = flat_tok[order]
+            w = flat_w[order]
+            counts = torch.bincount(flat_idx, minlength=self.n_experts)
+            permuted = h_tp[token_of]
+
+        if self.ep_size > 1:
+            # padded layout is expert-sorted with static per-expert counts
+            # — exactly the exchange's input contract
+            out_sorted, _ = self._ep_exchange_and_compute(permuted, counts)
+        else:
+            counts_cpu = counts.cpu()
+            out_sorted = self._experts_forward(permuted, counts_cpu)
+
+        combined = torch.zeros_like(h_tp, dtype=torch.float32)
+        combined.index_add_(0, token_of, out_sorted.float() * w.unsqueeze(-1))
+        return mappings.reduce_from_tp_region(combined.to(h.dtype))
 
     def _ep_exchange_and_compute(self, permuted, counts):
         """All-to-all token exchange over the EP group, local expert

@@ -1,0 +1,102 @@
+"""Expert-capacity routing tests (reference: utils/moe.py:310
+topk_softmax_with_capacity — capacity_factor / token_drop_policy /
+pad_to_capacity on ReaLMoEConfig, model_api.py:112-123)."""
+import numpy as np
+import pytest
+import torch
+
+from realhf_amd.models.hf import mixtral
+from realhf_amd.models.moe import apply_expert_capacity
+from realhf_amd.models.real_model import ReaLModel
+from tests.test_realloc import _fill_model_from_full, _full_reference_sd
+
+
+def _make_models(capacity_factor=None, policy="probs", pad=False, seed=17):
+    cfg = mixtral.make_test_config(n_layers=2, hidden_dim=64, n_heads=8,
+                                   n_kv_heads=4, vocab_size=128)
+    cfg.dtype = "float32"
+    cfg.moe.capacity_factor = capacity_factor
+    cfg.moe.token_drop_policy = policy
+    cfg.moe.pad_to_capacity = pad
+    sd = _full_reference_sd(cfg, seed=seed)
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    _fill_model_from_full(m, cfg, sd)
+    return cfg, m
+
+
+def _batch(vocab=128, lens=(11, 13), seed=5):
+    rng = np.random.RandomState(seed)
+    packed = torch.from_numpy(rng.randint(0, vocab, size=sum(lens))).long()
+    cu = torch.tensor([0] + list(np.cumsum(lens)), dtype=torch.int32)
+    return packed, cu, max(lens)
+
+
+def test_loose_capacity_is_noop():
+    """cap >= tokens*k/n_experts * factor covers every assignment when the
+    factor equals n_experts — output must equal the dropless model."""
+    packed, cu, mx = _batch()
+    _, dropless = _make_models(None)
+    _, capped = _make_models(float(8))  # cap = tokens*k, nothing can drop
+    dropless.eval(), capped.eval()
+    with torch.no_grad():
+        a = dropless(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+        b = capped(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+    torch.testing.assert_close(a, b, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.parametrize("policy", ["probs", "position"])
+def test_tight_capacity_drops_and_trains(policy):
+    packed, cu, mx = _batch()
+    _, dropless = _make_models(None)
+    _, capped = _make_models(0.5, policy=policy)
+    dropless.eval(), capped.eval()
+    with torch.no_grad():
+        a = dropless(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+        b = capped(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+    assert not torch.allclose(a, b)  # the cap must actually bind
+    # and the capped model still backprops
+    capped.train()
+    capped.allocate_grad_buffer()
+    for k, p in capped._params.items():
+        p.requires_grad_(True)
+        p.grad = capped.grad_view(k)
+    out = capped(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+    out.float().square().mean().backward()
+    assert torch.isfinite(capped.flat_grad).all()
+    assert float(capped.flat_grad.abs().sum()) > 0
+
+
+@pytest.mark.parametrize("policy", ["probs", "position"])
+def test_pad_to_capacity_matches_drop_mode(policy):
+    """Padding only adds zero-weighted rows — logits must match the
+    unpadded drop mode bit-for-bit in fp32 up to summation order."""
+    packed, cu, mx = _batch()
+    _, drop = _make_models(0.5, policy=policy, pad=False)
+    _, pad = _make_models(0.5, policy=policy, pad=True)
+    drop.eval(), pad.eval()
+    with torch.no_grad():
+        a = drop(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+        b = pad(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+    torch.testing.assert_close(a, b, atol=1e-5, rtol=1e-5)
+
+
+def test_apply_expert_capacity_unit():
+    # 4 tokens, 2 experts, k=1: everyone picks expert 0 with descending
+    # weights; cap = ceil(4*1/2 * 0.5) = 1
+    scores = torch.tensor([[0.1], [0.4], [0.3], [0.2]])
+    idx = torch.zeros(4, 1, dtype=torch.long)
+    s, kept, cap = apply_expert_capacity(scores, idx, 2, 0.5, "probs")
+    assert cap == 1
+    assert kept.flatten().tolist() == [False, True, False, False]  # highest
+    assert s.flatten().tolist() == [0.0, pytest.approx(0.4), 0.0, 0.0]
+    s, kept, cap = apply_expert_capacity(scores, idx, 2, 0.5, "position")
+    assert kept.flatten().tolist() == [True, False, False, False]  # earliest
+    # per-expert kept counts never exceed cap under random top-k routing
+    # (top-k picks distinct experts per token, like the real router)
+    g = torch.Generator().manual_seed(0)
+    logits = torch.rand(64, 8, generator=g)
+    scores, idx = torch.topk(torch.softmax(logits, -1), 2, dim=-1)
+    s, kept, cap = apply_expert_capacity(scores, idx, 8, 1.0, "probs")
+    counts = torch.bincount(idx[kept], minlength=8)
+    assert (counts <= cap).all()
+    assert kept.sum() < idx.numel()  # something was dropped at factor 1.0
