@@ -47,6 +47,11 @@ class OptimizerConfig:
     warmup_steps_proportion: float = 0.02
     gradient_clipping: float = 1.0
     offload: bool = False  # optimizer states in host memory (70B tier)
+    # 1 = ZeRO-1 (shard optimizer states); 2 = ZeRO-2 (additionally never
+    # hold a full-model grad buffer: per-microbatch bucket staging +
+    # reduce-scatter accumulation into the shard).  Reference counterpart:
+    # DeepSpeed zero_stage (deepspeed.py:276-359).
+    zero_stage: int = 1
 
 
 class LoRAFacade:
@@ -138,8 +143,33 @@ class ZeRO1Optimizer:
             self._param_padded = new_param
             model._map_params()
             model._build_modules()
-        self.grad_padded = torch.zeros(self.n_pad, dtype=model.flat_param.dtype, device=dev)
-        model.flat_grad = self.grad_padded[:n]
+
+        # ---- ZeRO-2: no full-model grad buffer ever exists -------------
+        # Autograd allocates each param's grad; a post-accumulate hook
+        # stages it into a pooled bucket buffer and frees it.  Complete
+        # buckets reduce-scatter immediately and ACCUMULATE into
+        # grad_shard (one RS per bucket per microbatch — the classic
+        # ZeRO-2 comm/memory trade).  Peak grad memory = a few buckets
+        # instead of a full model copy.
+        self.zero2 = (
+            cfg.zero_stage >= 2
+            and self.dp_size > 1
+            and self.tp_size == 1   # SP tp-reduce needs full-grad intervals
+            and self.pp_size == 1   # pipeline engine drives its own bwd
+            and hasattr(model, "layout")
+            and not self._gloo_cuda()  # needs reduce_scatter_tensor
+        )
+        if cfg.zero_stage >= 2 and not self.zero2:
+            logger.warning(
+                "zero_stage=2 requested but unsupported here "
+                "(needs dp>1, tp==1, pp==1); falling back to ZeRO-1")
+        if self.zero2:
+            self.grad_padded = None
+            model.flat_grad = None
+        else:
+            self.grad_padded = torch.zeros(
+                self.n_pad, dtype=model.flat_param.dtype, device=dev)
+            model.flat_grad = self.grad_padded[:n]
 
         s0 = self.dp_rank * self.shard_size
         s1 = s0 + self.shard_size
@@ -164,10 +194,11 @@ class ZeRO1Optimizer:
             and hasattr(model, "layout")
             and self.tp_size == 1  # tp: SP tp-reduce + norm discount
                                    # need the contiguous shard layout
+            and not self.zero2     # ZeRO-2 has its own per-mb staging
         )
         if self.overlap_comm and self._gloo_cuda():
             self.overlap_comm = False  # async RS unavailable on gloo+CUDA
-        if self.overlap_comm:
+        if self.overlap_comm or self.zero2:
             align = 256 * self.dp_size
             self._bsz = max(align, (bucket_size // align) * align)
             self.buckets = []  # (b0, b1, shard_off)
@@ -196,6 +227,14 @@ class ZeRO1Optimizer:
             self._reduced: List[bool] = []
             self._works: list = []
             self._armed = False
+            if self.zero2:
+                self._z2_pool: List[torch.Tensor] = []     # [bsz] staging
+                self._z2_scratch: List[torch.Tensor] = []  # [bsz/dp] RS out
+                self._z2_active: Dict[int, torch.Tensor] = {}
+                self._z2_left = None   # per-microbatch outstanding counts
+                self._z2_inflight: list = []  # (work, buf, scr, soff, l)
+                self._z2_max_inflight = int(
+                    os.environ.get("REALHF_AMD_Z2_INFLIGHT", 4))
         elif cfg.offload:
             # chunked fp32 conversion: a whole-shard .float() on device
             # would spike 2x shard bytes (275 GB for 70B at dp1)
@@ -257,10 +296,14 @@ class ZeRO1Optimizer:
     # ------------------------------------------------------------------
     def attach_grads(self):
         """Point every param's .grad at its flat-grad view so autograd
-        accumulates in place."""
+        accumulates in place.  ZeRO-2 never attaches views: autograd
+        allocates each grad, the hook stages + frees it."""
         for k, p in self.model._params.items():
             if not p.requires_grad:
                 p.requires_grad_(True)
+            if self.zero2:
+                p.register_post_accumulate_grad_hook(self._make_z2_hook(k))
+                continue
             p.grad = self.model.grad_view(k)
             if self.overlap_comm:
                 bids = self._param_buckets.get(k)
@@ -304,6 +347,94 @@ class ZeRO1Optimizer:
         self._works.append(w)
         self._reduced[b] = True
 
+    # ---------------- ZeRO-2 machinery --------------------------------
+    def _make_z2_hook(self, key):
+        sp = self.model.layout.specs[key]
+        bids = self._param_buckets[key]
+
+        def hook(p):
+            g = p.grad
+            if g is None:
+                return
+            if self._z2_left is None:
+                self._z2_arm()
+            flat = g.reshape(-1)
+            for b in bids:
+                b0, b1, _ = self.buckets[b]
+                buf = self._z2_active.get(b)
+                if buf is None:
+                    buf = self._z2_acquire(b1 - b0)
+                    self._z2_active[b] = buf
+                lo, hi = max(sp.start, b0), min(sp.end, b1)
+                buf[lo - b0:hi - b0].copy_(flat[lo - sp.start:hi - sp.start])
+                self._z2_left[b] -= 1
+                if self._z2_left[b] == 0:
+                    self._z2_rs(b)
+            p.grad = None  # the full grad never outlives its bucket copy
+
+        return hook
+
+    def _z2_arm(self):
+        counts = [0] * len(self.buckets)
+        for k in self.model._params:
+            for b in self._param_buckets.get(k, ()):
+                counts[b] += 1
+        self._z2_left = counts
+
+    def _z2_acquire(self, length):
+        if not self._z2_pool and len(self._z2_inflight) >= self._z2_max_inflight:
+            self._z2_harvest(1)
+        full = (self._z2_pool.pop() if self._z2_pool else torch.empty(
+            self._bsz, dtype=self.model.flat_param.dtype,
+            device=self.model.flat_param.device))
+        buf = full[:length]
+        buf.zero_()  # params missing at flush time contribute zero
+        return buf
+
+    def _z2_rs(self, b):
+        """Reduce-scatter one staged bucket; result accumulates into
+        grad_shard at harvest time (each microbatch's loss is already
+        scaled by 1/n_mbs, so summing per-mb averages is the full-batch
+        average)."""
+        b0, b1, soff = self.buckets[b]
+        buf = self._z2_active.pop(b)
+        l = (b1 - b0) // self.dp_size
+        scr = (self._z2_scratch.pop() if self._z2_scratch else torch.empty(
+            self._bsz // self.dp_size, dtype=buf.dtype, device=buf.device))
+        w = dist.reduce_scatter_tensor(
+            scr[:l], buf, op=dist.ReduceOp.AVG, group=self.dp_group,
+            async_op=True)
+        self._z2_inflight.append((w, buf, scr, soff, l))
+
+    def _z2_harvest(self, k=None):
+        take = len(self._z2_inflight) if k is None else min(
+            k, len(self._z2_inflight))
+        for w, buf, scr, soff, l in self._z2_inflight[:take]:
+            w.wait()
+            self.grad_shard[soff:soff + l] += scr[:l]
+            self._z2_pool.append(self._z2_repool(buf))
+            self._z2_scratch.append(scr)
+        del self._z2_inflight[:take]
+
+    @staticmethod
+    def _z2_repool(buf):
+        # recover the full pooled tensor from a length-limited view
+        # (the last bucket is shorter than _bsz)
+        full = torch.empty(0, dtype=buf.dtype, device=buf.device)
+        full.set_(buf.untyped_storage(), 0,
+                  (buf.untyped_storage().nbytes() // buf.element_size(),))
+        return full
+
+    def end_microbatch(self):
+        """ZeRO-2: called by the engine after every backward — flush
+        partially-filled buckets (missing slices are zero) and re-arm for
+        the next microbatch.  No-op otherwise."""
+        if not self.zero2 or self._z2_left is None:
+            return
+        for b in list(self._z2_active):
+            self._z2_rs(b)
+        self._z2_left = None
+
     def finish_allgather(self):
         for w in self._pending_ag:
             w.wait()
@@ -311,7 +442,13 @@ class ZeRO1Optimizer:
 
     def zero_grad(self):
         self.finish_allgather()
-        self.grad_padded.zero_()
+        if self.zero2:
+            self._z2_harvest()  # drop any stale in-flight RS results
+            self._z2_left = None
+            self._z2_active.clear()
+            self.grad_shard.zero_()
+        else:
+            self.grad_padded.zero_()
         if self.overlap_comm:
             self._armed = False  # a failed backward must not leave stale state
             self._works = []
@@ -344,7 +481,7 @@ class ZeRO1Optimizer:
     def step(self, defer_allgather: bool = False) -> Dict[str, float]:
         self.step_count += 1
         cfg = self.cfg
-        dev = self.grad_padded.device
+        dev = self.model.flat_param.device
 
         # 0. SP: sum replicated params' grads over the tp group
         if self._sp_repl_ivs:
@@ -352,7 +489,12 @@ class ZeRO1Optimizer:
                 dist.all_reduce(self.grad_padded[a:b], group=self.tp_group)
 
         # 1. reduce-scatter grads over DP (average)
-        if self.overlap_comm:
+        if self.zero2:
+            # every microbatch already reduce-scattered; drain the tail
+            self.end_microbatch()
+            self._z2_harvest()
+            gshard = self.grad_shard
+        elif self.overlap_comm:
             # flush buckets the hooks missed (unused params / not armed)
             for b in range(len(self.buckets)):
                 if not self._reduced[b]:
@@ -411,7 +553,7 @@ class ZeRO1Optimizer:
         # 3. AdamW on this rank's shard (bf16 grads read directly)
         lr = self._lr()
         s0, s1 = self.shard_bounds
-        param_shard = (self._shard_bf16 if self.overlap_comm
+        param_shard = (self._shard_bf16 if (self.overlap_comm or self.zero2)
                        else self._param_padded[s0:s1])
         if cfg.offload and dev.type == "cuda":
             self._adamw_offloaded(gshard, param_shard, lr, gscale)
@@ -434,7 +576,7 @@ class ZeRO1Optimizer:
             )
 
         # 4. all-gather updated params
-        if self.overlap_comm:
+        if self.overlap_comm or self.zero2:
             works = []
             for b0, b1, soff in self.buckets:
                 l = (b1 - b0) // self.dp_size

@@ -80,6 +80,9 @@ class PipelinableTrainEngine(PipelinableEngine):
                 # bucketed reduce-scatter overlaps this (last) backward
                 self.optimizer.arm_overlap()
             (loss / len(mbs)).backward()
+            if hasattr(self.optimizer, "end_microbatch"):
+                # ZeRO-2: reduce-scatter this microbatch's residual buckets
+                self.optimizer.end_microbatch()
             for k, v in st.items():
                 stats[k] = stats.get(k, 0.0) + float(v) / len(mbs)
         opt_stats = self.optimizer.step(defer_allgather=defer_allgather)
