@@ -16,6 +16,7 @@ Design (MI355X-first):
   masks EOS early.
 """
 import dataclasses
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -227,8 +228,13 @@ def generate(
         and model.dtype == torch.bfloat16
         and cfg.head_dim in (64, 128)
         # MoE dispatch reads expert counts on the host (counts.cpu());
-        # a D2H sync inside stream capture DEADLOCKS rather than erroring
-        and cfg.moe is None
+        # a D2H sync inside stream capture DEADLOCKS rather than erroring.
+        # pad_to_capacity makes counts compile-time constants (no sync) —
+        # graph capture of MoE decode is then possible, but stays opt-in
+        # (REALHF_AMD_MOE_GRAPH=1) until validated on hardware.
+        and (cfg.moe is None
+             or (cfg.moe.pad_to_capacity
+                 and os.environ.get("REALHF_AMD_MOE_GRAPH") == "1"))
     )
 
     cur_logits = last_logits

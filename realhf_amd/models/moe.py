@@ -389,6 +389,12 @@ class MoELayer(torch.nn.Module):
             # padded layout is expert-sorted with static per-expert counts
             # — exactly the exchange's input contract
             out_sorted, _ = self._ep_exchange_and_compute(permuted, counts)
+        elif moe.pad_to_capacity:
+            # counts are compile-time constants: no device sync (the
+            # counts.cpu() D2H below is what blocks hipGraph capture of
+            # MoE decode)
+            counts_cpu = torch.full((self.n_experts,), cap, dtype=torch.long)
+            out_sorted = self._experts_forward(permuted, counts_cpu)
         else:
             counts_cpu = counts.cpu()
             out_sorted = self._experts_forward(permuted, counts_cpu)
