@@ -588,3 +588,37 @@ def _zero2_worker():
 @pytest.mark.distributed
 def test_zero2_matches_zero1():
     LocalMultiProcessTest(2, _zero2_worker).launch()
+
+
+def _zero2_backend_worker():
+    """zero_stage=2 plumbs through the backend dict path used by the
+    trainer (Abstraction("zero1", {"optimizer": {...}}))."""
+    import torch.distributed as dist
+
+    from realhf_amd.api.config import Abstraction, ModelName
+    from realhf_amd.api.model import FinetuneSpec, Model, make_backend
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models.real_model import ReaLModel
+    import realhf_amd.runtime.engine  # noqa: F401 (registers "zero1")
+
+    init_global_constants(num_dp=2, num_tp=1, num_pp=1, model_name="m")
+    cfg = make_test_config(n_layers=1, hidden_dim=32, n_heads=4, n_kv_heads=4,
+                           vocab_size=64)
+    cfg.dtype = "float32"
+    with constants.model_scope("m"):
+        m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+        m.random_init()
+        model = Model(name=ModelName("m", 0), module=m, tokenizer=None,
+                      device=torch.device("cpu"), dtype=torch.float32)
+        backend = make_backend(Abstraction(
+            "zero1", {"optimizer": {"lr": 1e-3, "zero_stage": 2,
+                                    "warmup_steps_proportion": 0.0}}))
+        model = backend.initialize(model, FinetuneSpec(1, 64, 8))
+        assert model.module.optimizer.zero2
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_zero2_backend_plumbing():
+    LocalMultiProcessTest(2, _zero2_backend_worker).launch()
