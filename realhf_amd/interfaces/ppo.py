@@ -33,6 +33,8 @@ from realhf_amd.ops import functional as ops
 from realhf_amd.parallel.tp import packed_shifted_logprobs
 from realhf_amd.runtime.engine import sample_to_packed
 from realhf_amd.utils.functional import (
+    apply_logits_mask,
+    build_leave_one_indices,
     build_shift_one_indices,
     masked_normalization,
 )
@@ -44,6 +46,24 @@ def _short_cu(cu: torch.Tensor) -> torch.Tensor:
     """cu_seqlens over (len-1)-length per-seq arrays."""
     bs = cu.shape[0] - 1
     return cu - torch.arange(bs + 1, device=cu.device, dtype=cu.dtype)
+
+
+def _warp_logits_like_sampler(logits, cu, mb, temperature: float):
+    """Make a re-forward's logits distribution match the sampler's:
+    temperature scaling + the recorded top-k/p removal mask on the
+    PREDICTING rows (row s..e-2 of each sequence).  In-place on `logits`
+    (safe: GEMM backward does not re-read its output).  Reference:
+    ppo_interface.py:38-54 applies packed_logits_mask the same way."""
+    if temperature != 1.0:
+        logits /= max(temperature, 1e-5)
+    lm = mb.data.get("packed_logits_mask")
+    if lm is None:
+        return logits
+    rows = build_leave_one_indices(int(cu[-1]), cu)
+    sub = logits[rows]                      # advanced-index copy
+    apply_logits_mask(sub, lm.bool())       # in-place on the copy
+    logits[rows] = sub
+    return logits
 
 
 @torch.no_grad()
@@ -132,7 +152,9 @@ class PPOActorInterface(ModelInterface):
         )
         if outs is None:  # pp mid stage: output lives on the last stage
             return None
+        want_mask = not self.gconfig.force_no_logits_mask
         all_ids, all_lp, all_pm, all_noeos, seqlens = [], [], [], [], []
+        all_lm = []
         for gen_out, prompts, cu in outs:
             packed, cu_full, pmask = concat_prompt_to_generation_output(
                 prompts, cu, gen_out
@@ -147,26 +169,42 @@ class PPOActorInterface(ModelInterface):
                 gl = int(gen_out.gen_lengths[i])
                 lps.append(plp[ps:pe].float())
                 lps.append(gen_out.gen_logprobs[i, :gl].float())
+                if want_mask and gen_out.logits_mask is not None:
+                    # rows align with packed_logprobs (l-1 per seq):
+                    # prompt rows allow everything (False = not removed),
+                    # gen rows carry the sampler's top-k/p removals
+                    # (reference stitches this in
+                    # concat_prompt_to_generation_output,
+                    # real_llm_generate.py:451)
+                    lm = gen_out.logits_mask
+                    plen = int(cu[i + 1] - cu[i])
+                    all_lm.append(lm.new_zeros(plen - 1, lm.shape[-1]))
+                    all_lm.append(lm[i, :gl])
             all_ids.append(packed)
             all_lp.append(torch.cat(lps))
             all_pm.append(pmask)
             all_noeos.append(gen_out.no_eos_mask)
             seqlens += [int(cu_full[i + 1] - cu_full[i]) for i in range(bs)]
+        keys = ["packed_input_ids", "packed_logprobs", "prompt_mask",
+                "seq_no_eos_mask"]
+        seql = {
+            "packed_input_ids": [[l] for l in seqlens],
+            "packed_logprobs": [[l - 1] for l in seqlens],
+            "prompt_mask": [[l] for l in seqlens],
+            "seq_no_eos_mask": [[1]] * len(seqlens),
+        }
+        dat = {
+            "packed_input_ids": torch.cat(all_ids),
+            "packed_logprobs": torch.cat(all_lp),
+            "prompt_mask": torch.cat(all_pm),
+            "seq_no_eos_mask": torch.cat(all_noeos),
+        }
+        if all_lm:
+            keys.append("packed_logits_mask")
+            seql["packed_logits_mask"] = [[l - 1] for l in seqlens]
+            dat["packed_logits_mask"] = torch.cat(all_lm)
         res = SequenceSample(
-            keys=("packed_input_ids", "packed_logprobs", "prompt_mask", "seq_no_eos_mask"),
-            ids=list(data.ids),
-            seqlens={
-                "packed_input_ids": [[l] for l in seqlens],
-                "packed_logprobs": [[l - 1] for l in seqlens],
-                "prompt_mask": [[l] for l in seqlens],
-                "seq_no_eos_mask": [[1]] * len(seqlens),
-            },
-            data={
-                "packed_input_ids": torch.cat(all_ids),
-                "packed_logprobs": torch.cat(all_lp),
-                "prompt_mask": torch.cat(all_pm),
-                "seq_no_eos_mask": torch.cat(all_noeos),
-            },
+            keys=tuple(keys), ids=list(data.ids), seqlens=seql, data=dat,
         )
         return res
 
@@ -177,6 +215,7 @@ class PPOActorInterface(ModelInterface):
 
         def post_hook(logits, mb):
             ids, cu, _ = sample_to_packed(mb)
+            _warp_logits_like_sampler(logits, cu, mb, self.gconfig.temperature)
             return packed_shifted_logprobs(logits, cu, ids)
 
         logp = model.module.forward(data, n_mbs=n_mbs, post_hook=post_hook)
@@ -200,9 +239,10 @@ class PPOActorInterface(ModelInterface):
             self.discount, self.gae_lambda, self.adv_norm, None,
         )
         # attach computed tensors for minibatch splitting
-        train_sample = data.select_keys(
-            ["packed_input_ids", "prompt_mask"]
-        )
+        sel = ["packed_input_ids", "prompt_mask"]
+        if "packed_logits_mask" in data.keys:
+            sel.append("packed_logits_mask")
+        train_sample = data.select_keys(sel)
         extra = SequenceSample(
             keys=("old_logp", "advantages", "ppo_loss_mask"),
             ids=list(data.ids),
@@ -256,6 +296,7 @@ class PPOActorInterface(ModelInterface):
 
     def _loss_fn(self, logits, mb: SequenceSample):
         ids, cu, _ = sample_to_packed(mb)
+        _warp_logits_like_sampler(logits, cu, mb, self.gconfig.temperature)
         new_logp = packed_shifted_logprobs(logits, cu, ids)
         loss, stats = ppo_math.actor_loss_fn(
             new_logp,

@@ -36,19 +36,29 @@ class GenerationOutput:
     gen_lengths: torch.Tensor  # [bs] — actual generated length (incl. eos)
     no_eos_mask: torch.Tensor  # [bs] bool — True if never hit EOS
     prompt_logprobs: Optional[torch.Tensor] = None  # [total_prompt - bs] packed
+    # [bs, gen_len, vocab] bool, True = vocab entry REMOVED by top-k/p at
+    # that step (reference genstep logits_mask, real_llm_generate.py:131-136;
+    # emitted only when gconfig.force_no_logits_mask is False)
+    logits_mask: Optional[torch.Tensor] = None
 
 
 def _sample_from_logits(
     logits: torch.Tensor,  # [bs, vocab] fp32
     gconfig: GenerationHyperparameters,
     generator: Optional[torch.Generator],
-) -> Tuple[torch.Tensor, torch.Tensor]:
+    return_mask: bool = False,
+) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
+    """Returns (tokens, logprob, mask).  mask (only when return_mask) is
+    [bs, vocab] bool, True = entry removed by top-k/p; in that mode the
+    logprob is under the FILTERED distribution (reference genstep samples
+    from the warped logits, real_llm_generate.py:80-93) so behavior
+    logprobs line up with masked re-forward logprobs."""
     if gconfig.temperature != 1.0 and not gconfig.greedy:
         logits = logits / max(gconfig.temperature, 1e-5)
     if gconfig.greedy:
         tokens = logits.argmax(dim=-1)
         logp = torch.log_softmax(logits, dim=-1)
-        return tokens, logp.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
+        return tokens, logp.gather(-1, tokens.unsqueeze(-1)).squeeze(-1), None
     V = logits.shape[-1]
     k = gconfig.top_k if 0 < gconfig.top_k < V else V
     if k < V:
@@ -66,8 +76,13 @@ def _sample_from_logits(
         probs = torch.softmax(vals, dim=-1)
     sel = torch.multinomial(probs, 1, generator=generator)
     tokens = idx.gather(-1, sel).squeeze(-1)
+    if return_mask:
+        mask = torch.ones_like(logits, dtype=torch.bool)
+        mask.scatter_(1, idx, vals == float("-inf"))  # kept slots -> False
+        logp_f = torch.log_softmax(vals, dim=-1)  # filtered distribution
+        return tokens, logp_f.gather(-1, sel).squeeze(-1), mask
     logp_all = torch.log_softmax(logits, dim=-1)  # logprob under UNFILTERED dist
-    return tokens, logp_all.gather(-1, tokens.unsqueeze(-1)).squeeze(-1)
+    return tokens, logp_all.gather(-1, tokens.unsqueeze(-1)).squeeze(-1), None
 
 
 class DecodeGraph:
@@ -237,11 +252,21 @@ def generate(
                  and os.environ.get("REALHF_AMD_MOE_GRAPH") == "1"))
     )
 
+    # logits-mask mode (force_no_logits_mask=False): record which vocab
+    # entries top-k/p removed at each step so downstream logprob passes
+    # (ref_inf, actor_train) can mask consistently.  [bs, vocab] bool per
+    # step — memory-heavy by design, hence the reference's opt-out flag.
+    want_mask = not gconfig.force_no_logits_mask and not gconfig.greedy
+    step_masks = [] if want_mask else None
+
     cur_logits = last_logits
     for t in range(max_new):
         if eos_token_id is not None and t < gconfig.min_new_tokens:
             cur_logits[:, eos_token_id] = float("-inf")
-        tokens, logp = _sample_from_logits(cur_logits, gconfig, generator)
+        tokens, logp, smask = _sample_from_logits(
+            cur_logits, gconfig, generator, return_mask=want_mask)
+        if want_mask:
+            step_masks.append(smask)
         tokens = torch.where(done, torch.full_like(tokens, pad_token_id), tokens)
         gen_tokens[:, t] = tokens
         gen_logprobs[:, t] = torch.where(done, torch.zeros_like(logp), logp)
@@ -267,12 +292,22 @@ def generate(
         logits_step = mappings.gather_from_tp_region(logits_step).float()
         cur_logits = logits_step
 
+    gmax = int(gen_lengths.max())
+    logits_mask = None
+    if want_mask and step_masks:
+        logits_mask = torch.stack(step_masks[:gmax], dim=1)  # [bs, <=gmax, V]
+        if logits_mask.shape[1] < gmax:  # early-exit break before step gmax
+            pad_m = logits_mask.new_zeros(
+                logits_mask.shape[0], gmax - logits_mask.shape[1],
+                logits_mask.shape[2])
+            logits_mask = torch.cat([logits_mask, pad_m], dim=1)
     return GenerationOutput(
         gen_tokens=gen_tokens[:, : int(gen_lengths.max())],
         gen_logprobs=gen_logprobs[:, : int(gen_lengths.max())],
         gen_lengths=gen_lengths,
         no_eos_mask=~done,
         prompt_logprobs=prompt_logprobs,
+        logits_mask=logits_mask,
     )
 
 

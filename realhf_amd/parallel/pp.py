@@ -365,7 +365,8 @@ class PipelinedEngine(PipelinableEngine):
             r = s.r
             if eos is not None and r < gconfig.min_new_tokens:
                 cur_logits[:, eos] = float("-inf")
-            tokens, logp = genmod._sample_from_logits(cur_logits, gconfig, gen)
+            tokens, logp, _ = genmod._sample_from_logits(
+                cur_logits, gconfig, gen)
             tokens = torch.where(s.done, torch.full_like(tokens, pad), tokens)
             s.gen_tokens[:, r] = tokens
             s.gen_logprobs[:, r] = torch.where(

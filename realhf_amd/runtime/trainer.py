@@ -345,10 +345,16 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
     if cfg.actor.gen_parallel is not None and cfg.allocation_mode == "manual":
         gen_mc = dataclasses.replace(cfg.actor, parallel=cfg.actor.gen_parallel)
         gen_model, gen_par = add_model("actor", gen_mc, replica=1)
+    # logits-mask mode (gen.force_no_logits_mask=False): the sampler's
+    # top-k/p removal mask travels actor_gen -> ref_inf/actor_train so
+    # re-forward logprobs match the sampled distribution (reference:
+    # ppo_interface.py:245-246 emits packed_logits_mask the same way)
+    mask_keys = ([] if ppo.gen.force_no_logits_mask
+                 else ["packed_logits_mask"])
     add_mfc("actor_gen", gen_model, T.GENERATE, actor_iface,
             ["packed_prompts"],
             ["packed_input_ids", "packed_logprobs", "prompt_mask",
-             "seq_no_eos_mask"], cfg.actor, gen_par)
+             "seq_no_eos_mask"] + mask_keys, cfg.actor, gen_par)
     if gen_model != actor:
         gen_mfc = mfcs[-1]
         gen_mfc.pre_hooks.append(
@@ -359,8 +365,8 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
     # WHOLE node, then critic_inf (first half) and rew_inf (second half)
     # CONCURRENTLY — each sub-mesh rank skips the other MFC and proceeds
     add_mfc("ref_inf", ref, T.INFERENCE, actor_iface,
-            ["packed_input_ids"], ["packed_ref_logprobs"], cfg.ref, refpar,
-            priority=-1)
+            ["packed_input_ids"] + mask_keys, ["packed_ref_logprobs"],
+            cfg.ref, refpar, priority=-1)
     add_mfc("critic_inf", critic, T.INFERENCE, critic_iface,
             ["packed_input_ids"], ["values"], cfg.critic, cpar,
             alloc_strategy=critic_inf_strategy)
@@ -368,8 +374,9 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
             ["packed_input_ids"], ["rewards"], cfg.rew, rewpar)
     train_keys = ["packed_input_ids", "packed_logprobs", "packed_ref_logprobs",
                   "rewards", "values", "prompt_mask", "seq_no_eos_mask"]
-    add_mfc("actor_train", actor, T.TRAIN_STEP, actor_iface, train_keys, [],
-            cfg.actor, apar)
+    # the critic never reads logits — keep the (large) mask off its mesh
+    add_mfc("actor_train", actor, T.TRAIN_STEP, actor_iface,
+            train_keys + mask_keys, [], cfg.actor, apar)
     add_mfc("critic_train", critic, T.TRAIN_STEP, critic_iface, train_keys, [],
             cfg.critic, cpar)
 

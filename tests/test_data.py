@@ -189,3 +189,25 @@ def test_payload_pack_unpack_roundtrip():
     for k in ("ids", "logp", "mask"):
         assert out.data[k].dtype == s.data[k].dtype
         assert torch.equal(out.data[k], s.data[k])
+
+
+def test_payload_roundtrip_2d_bool_logits_mask():
+    """packed_logits_mask is 2-D [rows, vocab] bool — the codec must
+    restore its shape (per-key shape travels in the payload meta)."""
+    from realhf_amd.runtime.data_transfer import _pack_payload, _unpack_payload
+
+    lm = torch.rand(7, 16) > 0.5
+    s = SequenceSample(
+        keys=("packed_logits_mask",),
+        ids=["a", "b"],
+        seqlens={"packed_logits_mask": [[4], [3]]},
+        data={"packed_logits_mask": lm},
+    )
+    meta, flats = _pack_payload(s)
+    out = _unpack_payload(meta, flats)
+    assert out.data["packed_logits_mask"].shape == (7, 16)
+    assert torch.equal(out.data["packed_logits_mask"], lm)
+    # and dim-0 split/gather keeps rows aligned
+    a, b = s.split(2)
+    assert a.data["packed_logits_mask"].shape == (4, 16)
+    assert torch.equal(SequenceSample.gather([a, b]).data["packed_logits_mask"], lm)

@@ -1,0 +1,99 @@
+"""Logits-mask mode (gconfig.force_no_logits_mask=False): the sampler's
+top-k/p removal mask travels actor_gen -> ref_inf/actor_train so every
+logprob is computed under the distribution that was actually sampled
+from (reference: real_llm_generate.py:131-136 genstep logits_mask,
+ppo_interface.py:38-54/245-246)."""
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from realhf_amd.api.model import GenerationHyperparameters
+from realhf_amd.models.generation import _sample_from_logits
+
+
+def test_sampler_mask_and_filtered_logp():
+    g = GenerationHyperparameters(top_k=3, top_p=1.0, temperature=1.0,
+                                  force_no_logits_mask=False)
+    torch.manual_seed(0)
+    logits = torch.randn(5, 16)
+    gen = torch.Generator().manual_seed(1)
+    tokens, logp, mask = _sample_from_logits(logits, g, gen, return_mask=True)
+    assert mask.shape == (5, 16) and mask.dtype == torch.bool
+    assert (mask.sum(-1) == 13).all()  # exactly V - top_k removed
+    # the sampled token is never a removed one
+    assert not mask.gather(1, tokens.unsqueeze(1)).any()
+    # logp is under the FILTERED (renormalized top-k) distribution
+    masked = logits.masked_fill(mask, float("-inf"))
+    want = torch.log_softmax(masked, -1).gather(1, tokens.unsqueeze(1)).squeeze(1)
+    torch.testing.assert_close(logp, want)
+
+
+def test_sampler_mask_with_top_p():
+    g = GenerationHyperparameters(top_k=0, top_p=0.5,
+                                  force_no_logits_mask=False)
+    torch.manual_seed(2)
+    logits = torch.randn(4, 32) * 3  # spiky -> top_p binds
+    gen = torch.Generator().manual_seed(3)
+    tokens, logp, mask = _sample_from_logits(logits, g, gen, return_mask=True)
+    assert mask.any()  # something was removed
+    assert not mask.gather(1, tokens.unsqueeze(1)).any()
+    masked = logits.masked_fill(mask, float("-inf"))
+    want = torch.log_softmax(masked, -1).gather(1, tokens.unsqueeze(1)).squeeze(1)
+    torch.testing.assert_close(logp, want)
+
+
+def test_generation_emits_logits_mask():
+    from realhf_amd.models.generation import generate
+    from realhf_amd.models.hf import llama
+
+    cfg = llama.make_test_config(vocab_size=64)
+    cfg.dtype = "float32"
+    from realhf_amd.models.real_model import ReaLModel
+
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    m.random_init()
+    g = GenerationHyperparameters(max_new_tokens=5, top_k=4,
+                                  use_hip_graph=False,
+                                  force_no_logits_mask=False)
+    rng = np.random.RandomState(0)
+    packed = torch.from_numpy(rng.randint(0, 64, size=14)).long()
+    cu = torch.tensor([0, 6, 14], dtype=torch.int32)
+    out = generate(m, packed, cu, g, eos_token_id=None, pad_token_id=0)
+    lm = out.logits_mask
+    assert lm is not None
+    gmax = int(out.gen_lengths.max())
+    assert lm.shape == (2, gmax, 64)
+    assert (lm.sum(-1) == 60).all()  # V - top_k removed every step
+
+
+def test_ppo_experiment_with_logits_mask(tmp_path):
+    """Full 1-rank PPO step with the mask flowing actor_gen -> ref_inf ->
+    actor_train through the DFG executor."""
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "prompts.jsonl")
+    rng = np.random.RandomState(5)
+    with open(data, "w") as f:
+        for i in range(16):
+            ids = rng.randint(0, 64, size=rng.randint(4, 8)).tolist()
+            f.write(json.dumps({"prompt": "x", "input_ids": ids}) + "\n")
+    cfg = PPOConfig(experiment_name="t-ppo-lmask", trial_name="cpu", n_gpus=1)
+    for mc in (cfg.actor, cfg.critic, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 6
+    cfg.ppo.gen.min_new_tokens = 2
+    cfg.ppo.gen.top_k = 4  # binds: the mask is non-trivial
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.gen.force_no_logits_mask = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
