@@ -97,3 +97,16 @@ def test_ppo_experiment_with_logits_mask(tmp_path):
     cfg.exp_ctrl.benchmark_steps = 1
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
     Trainer(cfg).run()
+
+
+def test_mask_mode_rejects_pp_generation():
+    from realhf_amd.api.config import ParallelismConfig
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import build_experiment
+
+    cfg = PPOConfig(experiment_name="t", trial_name="t", n_gpus=2)
+    cfg.actor.parallel = ParallelismConfig(pipeline_parallel_size=2)
+    cfg.allocation_mode = "manual"
+    cfg.ppo.gen.force_no_logits_mask = False
+    with pytest.raises(ValueError, match="pipeline-parallel generation"):
+        build_experiment(cfg, world=2)
