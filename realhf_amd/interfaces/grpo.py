@@ -85,7 +85,10 @@ class GRPOInterface(PPOActorInterface):
         )
         advantages = adv_seq[seg]
 
-        train_sample = data.select_keys(["packed_input_ids", "prompt_mask"])
+        sel = ["packed_input_ids", "prompt_mask"]
+        if "packed_logits_mask" in data.keys:
+            sel.append("packed_logits_mask")
+        train_sample = data.select_keys(sel)
         L = [[sum(x) - 1] for x in data.seqlens["packed_input_ids"]]
         extra = SequenceSample(
             keys=("old_logp", "ref_logp", "advantages", "ppo_loss_mask"),
@@ -115,7 +118,10 @@ class GRPOInterface(PPOActorInterface):
         return all_stats
 
     def _grpo_loss_fn(self, logits, mb: SequenceSample):
+        from realhf_amd.interfaces.ppo import _warp_logits_like_sampler
+
         ids, cu, _ = sample_to_packed(mb)
+        _warp_logits_like_sampler(logits, cu, mb, self.gconfig.temperature)
         new_logp = packed_shifted_logprobs(logits, cu, ids)
         loss, stats = ppo_math.actor_loss_fn(
             new_logp, mb.data["old_logp"], mb.data["advantages"],

@@ -419,16 +419,23 @@ def _build_grpo(cfg: GRPOConfig, world, add_model, add_mfc, mfcs, trainable):
         "output_scaling": ppo.reward_output_scaling,
         "output_bias": ppo.reward_output_bias,
     })
+    mask_keys = ([] if ppo.gen.force_no_logits_mask
+                 else ["packed_logits_mask"])
+    if mask_keys and apar.pipeline_parallel_size > 1:
+        raise ValueError(
+            "gen.force_no_logits_mask=False is not supported with "
+            "pipeline-parallel generation")
     add_mfc("actor_gen", actor, T.GENERATE, iface, ["packed_prompts"],
             ["packed_input_ids", "packed_logprobs", "prompt_mask",
-             "seq_no_eos_mask"], cfg.actor, apar)
+             "seq_no_eos_mask"] + mask_keys, cfg.actor, apar)
     add_mfc("rew_inf", rew, T.INFERENCE, rw_iface, ["packed_input_ids"],
             ["rewards"], cfg.rew, rewpar)
-    add_mfc("ref_inf", ref, T.INFERENCE, iface, ["packed_input_ids"],
+    add_mfc("ref_inf", ref, T.INFERENCE, iface,
+            ["packed_input_ids"] + mask_keys,
             ["packed_ref_logprobs"], cfg.ref, refpar)
     add_mfc("actor_train", actor, T.TRAIN_STEP, iface,
             ["packed_input_ids", "packed_logprobs", "packed_ref_logprobs",
-             "rewards", "prompt_mask"], [], cfg.actor, apar)
+             "rewards", "prompt_mask"] + mask_keys, [], cfg.actor, apar)
     for m in mfcs:
         if m.name == "rew_inf" and cfg.rew.offload:
             m.post_hooks.append(OffloadHook(to=_offload_style(cfg.rew)))

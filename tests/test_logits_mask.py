@@ -110,3 +110,33 @@ def test_mask_mode_rejects_pp_generation():
     cfg.ppo.gen.force_no_logits_mask = False
     with pytest.raises(ValueError, match="pipeline-parallel generation"):
         build_experiment(cfg, world=2)
+
+
+def test_grpo_experiment_with_logits_mask(tmp_path):
+    """GRPO (no critic) with the mask flowing through its DFG."""
+    from realhf_amd.api.experiment import GRPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "prompts.jsonl")
+    rng = np.random.RandomState(9)
+    with open(data, "w") as f:
+        for i in range(8):
+            ids = rng.randint(0, 64, size=rng.randint(4, 8)).tolist()
+            f.write(json.dumps({"prompt": "x", "input_ids": ids}) + "\n")
+    cfg = GRPOConfig(experiment_name="t-grpo-lmask", trial_name="cpu", n_gpus=1)
+    for mc in (cfg.actor, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+    cfg.rew.is_critic = True
+    cfg.group_size = 2
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 2
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 4
+    cfg.ppo.gen.top_k = 4
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.gen.force_no_logits_mask = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
