@@ -213,3 +213,53 @@ def test_tiny_grpo_mixtral_gpu(tmp_path):
         Trainer(cfg).run()
     finally:
         T.build_experiment = orig
+
+
+def test_ppo_logits_mask_gpu(tmp_path):
+    """PPO with gen.force_no_logits_mask=False on the HIP path: the
+    sampler's top-k mask is recorded outside the decode hipGraph and
+    flows to ref_inf/actor_train."""
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    rng = np.random.RandomState(4)
+    data = str(tmp_path / "prompts.jsonl")
+    with open(data, "w") as f:
+        for _ in range(16):
+            rec = {"input_ids": rng.randint(3, 60, size=rng.randint(6, 10)).tolist()}
+            f.write(json.dumps(rec) + "\n")
+    cfg = PPOConfig(experiment_name="t-ppo-lmask-gpu", trial_name="g", n_gpus=1)
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 8
+    cfg.dataset.max_prompt_len = 10
+    cfg.ppo.gen.max_new_tokens = 16
+    cfg.ppo.gen.min_new_tokens = 4
+    cfg.ppo.gen.top_k = 8  # binds at vocab 128 -> non-trivial mask
+    cfg.ppo.gen.force_no_logits_mask = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 2
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    import realhf_amd.models.hf as hf_reg
+    from realhf_amd.runtime import trainer as T
+
+    orig = T.build_experiment
+
+    def patched(c, world):
+        built = orig(c, world)
+        for name, rcfg in built.model_cfgs.items():
+            fam = hf_reg.get_family(rcfg.family or "llama")
+            big = fam.make_test_config(
+                n_layers=2, hidden_dim=64, n_heads=1, n_kv_heads=1,
+                head_dim=64, intermediate_dim=128, vocab_size=128)
+            big.is_critic = rcfg.is_critic
+            big.dtype = rcfg.dtype
+            big.family = rcfg.family
+            built.model_cfgs[name] = big
+        return built
+
+    T.build_experiment = patched
+    try:
+        Trainer(cfg).run()
+    finally:
+        T.build_experiment = orig

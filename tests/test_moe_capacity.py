@@ -100,3 +100,31 @@ def test_apply_expert_capacity_unit():
     counts = torch.bincount(idx[kept], minlength=8)
     assert (counts <= cap).all()
     assert kept.sum() < idx.numel()  # something was dropped at factor 1.0
+
+
+@pytest.mark.gpu
+def test_capacity_forward_backward_gpu():
+    """Capacity-limited + padded dispatch on the HIP grouped-GEMM path
+    (bf16, kernel-eligible dims)."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs GPU")
+    cfg = mixtral.make_test_config(n_layers=2, hidden_dim=64, n_heads=1,
+                                   n_kv_heads=1, head_dim=64,
+                                   intermediate_dim=128, vocab_size=128)
+    cfg.dtype = "bfloat16"
+    cfg.moe.capacity_factor = 1.0
+    for pad in (False, True):
+        cfg.moe.pad_to_capacity = pad
+        m = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
+        m.random_init()
+        m.allocate_grad_buffer()
+        for k, p in m._params.items():
+            p.requires_grad_(True)
+            p.grad = m.grad_view(k)
+        rng = np.random.RandomState(11)
+        ids = torch.from_numpy(rng.randint(0, 128, size=96)).long().cuda()
+        cu = torch.tensor([0, 48, 96], dtype=torch.int32, device="cuda")
+        out = m(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=48)
+        out.float().square().mean().backward()
+        assert torch.isfinite(m.flat_grad.float()).all()
+        assert float(m.flat_grad.float().abs().sum()) > 0
