@@ -622,3 +622,54 @@ def _zero2_backend_worker():
 @pytest.mark.distributed
 def test_zero2_backend_plumbing():
     LocalMultiProcessTest(2, _zero2_backend_worker).launch()
+
+
+def _zero2_offload_worker():
+    """zero_stage=2 composed with optimizer-state host offload: same
+    parameters as ZeRO-1 + offload."""
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel.ddp import OptimizerConfig, ZeRO1Optimizer
+
+    init_global_constants(num_dp=2, num_tp=1, num_pp=1, model_name="m")
+    cfg = make_test_config(n_layers=2, hidden_dim=32, n_heads=4, n_kv_heads=4,
+                           vocab_size=96)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=131)
+    rank = dist.get_rank()
+
+    def run(stage):
+        with constants.model_scope("m"):
+            m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+            _fill_model_from_full(m, cfg, sd)
+            opt = ZeRO1Optimizer(
+                m, OptimizerConfig(lr=1e-2, warmup_steps_proportion=0.0,
+                                   zero_stage=stage, offload=True),
+                bucket_size=4096,
+            )
+            if stage == 2:
+                assert opt.zero2 and opt.grad_padded is None
+            rng = np.random.RandomState(140 + rank)
+            for _ in range(2):
+                opt.zero_grad()
+                toks = torch.from_numpy(rng.randint(0, 96, size=10)).long()
+                cu = torch.tensor([0, 10], dtype=torch.int32)
+                opt.arm_overlap()
+                out = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=10)
+                out.float().square().mean().backward()
+                opt.end_microbatch()
+                opt.step()
+            return m.flat_param.detach().clone()
+
+    p1 = run(1)
+    p2 = run(2)
+    torch.testing.assert_close(p2, p1, atol=2e-6, rtol=2e-6)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_zero2_offload_matches_zero1_offload():
+    LocalMultiProcessTest(2, _zero2_offload_worker).launch()

@@ -404,3 +404,29 @@ def test_ppo_dp_shard_offload_two_ranks(tmp_path):
     _write_prompt_data(data)
     LocalMultiProcessTest(2, _ppo_dpshard_worker, data,
                           str(tmp_path / "root")).launch()
+
+
+def _sft_zero2_worker(data, fileroot):
+    """SFT at dp2 with optimizer.zero_stage=2 from the experiment config
+    (ZeRO-2: no full-model grad buffer; engine drives end_microbatch)."""
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = SFTConfig(experiment_name="t-sft-z2", trial_name="dist", n_gpus=2)
+    cfg.model.dtype = "float32"
+    cfg.model.optimizer.zero_stage = 2
+    cfg.allocation_mode = "global"
+    cfg.dataset.type_ = "prompt_answer"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.exp_ctrl.benchmark_steps = 2
+    Trainer(cfg).run()
+
+
+@pytest.mark.distributed
+def test_sft_zero2_two_ranks(tmp_path):
+    data = str(tmp_path / "sft.jsonl")
+    _write_sft_data(data)
+    LocalMultiProcessTest(2, _sft_zero2_worker, data,
+                          str(tmp_path / "root")).launch()
