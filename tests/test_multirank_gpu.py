@@ -79,3 +79,63 @@ def test_two_rank_one_gpu_ppo_heuristic(tmp_path):
                 {"input_ids": rng.randint(3, 60, size=8).tolist()}) + "\n")
     LocalMultiProcessTest(2, _worker, data, str(tmp_path / "root"),
                           backend="gloo", timeout_secs=600).launch()
+
+
+def _zero2_cuda_worker():
+    """ZeRO-2 on CUDA tensors over the gloo fallback (2 ranks, 1 GPU):
+    same parameters as ZeRO-1 after 2 steps of 2 microbatches."""
+    import numpy as np
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.models.hf.llama import make_test_config
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel.ddp import OptimizerConfig, ZeRO1Optimizer
+    from tests.test_realloc import _fill_model_from_full, _full_reference_sd
+
+    torch.cuda.set_device(0)
+    init_global_constants(num_dp=2, num_tp=1, num_pp=1, model_name="m")
+    cfg = make_test_config(n_layers=2, hidden_dim=64, n_heads=1, n_kv_heads=1,
+                           head_dim=64, intermediate_dim=128, vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=151)
+    rank = dist.get_rank()
+
+    def run(stage):
+        with constants.model_scope("m"):
+            m = ReaLModel(cfg, device="cuda", dtype=torch.float32)
+            _fill_model_from_full(m, cfg, sd)
+            opt = ZeRO1Optimizer(
+                m, OptimizerConfig(lr=1e-2, warmup_steps_proportion=0.0,
+                                   zero_stage=stage),
+                bucket_size=8192,
+            )
+            if stage == 2:
+                assert opt.zero2 and opt.grad_padded is None
+            rng = np.random.RandomState(160 + rank)
+            for _ in range(2):
+                opt.zero_grad()
+                for i in range(2):
+                    toks = torch.from_numpy(
+                        rng.randint(0, 128, size=12)).long().cuda()
+                    cu = torch.tensor([0, 12], dtype=torch.int32,
+                                      device="cuda")
+                    if i == 1:
+                        opt.arm_overlap()
+                    out = m(packed_input_ids=toks, cu_seqlens=cu,
+                            max_seqlen=12)
+                    (out.float().square().mean() / 2).backward()
+                    opt.end_microbatch()
+                opt.step()
+            return m.flat_param.detach().clone()
+
+    p1 = run(1)
+    p2 = run(2)
+    torch.testing.assert_close(p2, p1, atol=1e-5, rtol=1e-5)
+    dist.barrier()
+
+
+def test_zero2_two_rank_one_gpu():
+    LocalMultiProcessTest(2, _zero2_cuda_worker, backend="gloo",
+                          timeout_secs=600).launch()
