@@ -290,6 +290,13 @@ class ReaLModelBlock(nn.Module):
             # graph-capture-safe length bound: never read positions back
             if decode:
                 rot_len = int(k_cache.shape[1])
+            elif k_cache is not None:
+                # generation prefill: use the SESSION's cache length so
+                # dynamic-NTK picks one base for prefill AND decode (a
+                # per-length base would rotate cached K with a different
+                # basis than the decode queries reading it)
+                rot_len = max(int(k_cache.shape[1]),
+                              cfg.max_position_embeddings)
             else:
                 ms = max_seqlen
                 if ms is None:  # CP: token shard with explicit positions

@@ -186,3 +186,37 @@ def test_rope_scaling_matches_hf(scaling_type):
     with torch.no_grad():
         base = plain(packed_input_ids=ids, cu_seqlens=cu, max_seqlen=32)
     assert not torch.allclose(ours, base)
+
+
+def test_rope_scaling_decode_matches_prefill():
+    """The decode path builds its rotary table from the KV-cache length
+    (layers.py rot_len = cache_len), so dynamic-NTK must produce the
+    same logits step-by-step as a full prefill at the same positions."""
+    from realhf_amd.api.model import GenerationHyperparameters
+    from realhf_amd.models.generation import generate
+
+    fam = hf_reg.get_family("llama")
+    cfg = fam.make_test_config(max_position_embeddings=16)
+    cfg.dtype = "float32"
+    cfg.rotary_scaling = 2.0
+    cfg.rotary_scaling_type = "dynamic"
+    from realhf_amd.models.real_model import ReaLModel
+
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    m.random_init()
+    rng = np.random.RandomState(2)
+    prompt = torch.from_numpy(rng.randint(0, cfg.vocab_size, size=20)).long()
+    cu = torch.tensor([0, 20], dtype=torch.int32)
+    g = GenerationHyperparameters(max_new_tokens=4, greedy=True,
+                                  use_hip_graph=False)
+    out = generate(m, prompt, cu, g, eos_token_id=None, pad_token_id=0)
+    # re-score the generated sequence with a fresh full forward: the
+    # greedy choice at each step must be reproduced
+    full = torch.cat([prompt, out.gen_tokens[0].cpu()])
+    L = full.shape[0]
+    cu2 = torch.tensor([0, L], dtype=torch.int32)
+    with torch.no_grad():
+        logits = m(packed_input_ids=full, cu_seqlens=cu2, max_seqlen=L)
+    for t in range(4):
+        pos = 20 + t - 1  # logits row predicting token 20+t
+        assert int(logits[pos].argmax()) == int(out.gen_tokens[0, t])
