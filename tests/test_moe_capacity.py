@@ -128,3 +128,55 @@ def test_capacity_forward_backward_gpu():
         out.float().square().mean().backward()
         assert torch.isfinite(m.flat_grad.float()).all()
         assert float(m.flat_grad.float().abs().sum()) > 0
+
+
+def _ep_capacity_worker():
+    """Capacity dispatch under EP: the padded/filtered expert-sorted
+    layout feeds the all-to-all exchange (counts per GLOBAL expert)."""
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.topology import ParallelGrid, PipeDataTensorTopology
+    from realhf_amd.models.real_model import ReaLModel
+
+    for pad in (False, True):
+        cfg = mixtral.make_test_config(n_layers=2, hidden_dim=64, n_heads=8,
+                                       n_kv_heads=4, vocab_size=128)
+        cfg.dtype = "float32"
+        cfg.moe.capacity_factor = 0.5
+        cfg.moe.pad_to_capacity = pad
+        sd = _full_reference_sd(cfg, seed=77)
+        topo = PipeDataTensorTopology(num_pp=1, num_dp=2, num_tp=1, ep_size=2)
+        grid = ParallelGrid(topo)
+        constants.set_grid(f"m{pad}", grid)
+        cfg_ep = mixtral.make_test_config(n_layers=2, hidden_dim=64, n_heads=8,
+                                          n_kv_heads=4, vocab_size=128)
+        cfg_ep.dtype = "float32"
+        cfg_ep.moe.capacity_factor = 0.5
+        cfg_ep.moe.pad_to_capacity = pad
+        cfg_ep.moe.expert_parallel_size = 2
+        m = ReaLModel(cfg_ep, device="cpu", dtype=torch.float32,
+                      ep_rank=grid.ep_rank, ep_size=2)
+        _fill_model_from_full(m, cfg_ep, sd)
+        m.eval()
+        single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+        _fill_model_from_full(single, cfg, sd)
+        single.eval()
+        rng = np.random.RandomState(8)
+        lens = [9, 7]
+        packed = torch.from_numpy(rng.randint(0, 128, size=16)).long()
+        cu = torch.tensor([0, 9, 16], dtype=torch.int32)
+        with torch.no_grad():
+            ref = single(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=9)
+            with constants.model_scope(f"m{pad}"):
+                out = m(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=9)
+        torch.testing.assert_close(out, ref, atol=1e-4, rtol=1e-4)
+    import torch.distributed as dist
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_ep2_capacity_matches_replicated():
+    from realhf_amd.base.testing import LocalMultiProcessTest
+
+    LocalMultiProcessTest(2, _ep_capacity_worker).launch()
