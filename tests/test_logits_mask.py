@@ -140,3 +140,48 @@ def test_grpo_experiment_with_logits_mask(tmp_path):
     cfg.exp_ctrl.benchmark_steps = 1
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
     Trainer(cfg).run()
+
+
+def _ppo_mask_dist_worker(data, fileroot):
+    """2-rank PPO with the logits mask crossing mesh boundaries: a tp2
+    gen replica produces it, dp2 ref_inf/actor_train consume it — the
+    2-D bool key rides the executor's device-payload transfer plan."""
+    from realhf_amd.api.config import ParallelismConfig
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = PPOConfig(experiment_name="t-ppo2-lmask", trial_name="dist",
+                    n_gpus=2)
+    for mc in (cfg.actor, cfg.critic, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+        mc.parallel = ParallelismConfig(data_parallel_size=2)
+    cfg.allocation_mode = "manual"
+    cfg.actor.gen_parallel = ParallelismConfig(
+        data_parallel_size=1, tensor_parallel_size=2
+    )
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 5
+    cfg.ppo.gen.top_k = 4
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.gen.force_no_logits_mask = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 1
+    Trainer(cfg).run()
+
+
+@pytest.mark.distributed
+def test_ppo_logits_mask_two_ranks(tmp_path):
+    from realhf_amd.base.testing import LocalMultiProcessTest
+
+    data = str(tmp_path / "prompts.jsonl")
+    rng = np.random.RandomState(13)
+    with open(data, "w") as f:
+        for i in range(16):
+            ids = rng.randint(0, 64, size=rng.randint(4, 8)).tolist()
+            f.write(json.dumps({"prompt": "x", "input_ids": ids}) + "\n")
+    LocalMultiProcessTest(2, _ppo_mask_dist_worker, data,
+                          str(tmp_path / "root")).launch()
