@@ -84,6 +84,9 @@ class PPOHyperparameters:
     gae_lambda: float = 1.0
     adv_norm: bool = True
     value_norm: bool = True
+    value_norm_type: str = "exp"  # exp | ma (reference ppo_exp.py:62-68)
+    value_norm_beta: float = 0.99995
+    value_norm_eps: float = 1e-5
     early_stop_imp_ratio: Optional[float] = None
 
 

@@ -329,11 +329,16 @@ class PPOCriticInterface(ModelInterface):
     discount: float = 1.0
     gae_lambda: float = 1.0
     value_norm: bool = True
+    value_norm_type: str = "exp"  # exp | ma (reference: rms.py:185)
+    value_norm_beta: float = 0.99995
+    value_norm_eps: float = 1e-5
     value_loss_type: str = "huber"
     defer_final_allgather: bool = False
 
     def __post_init__(self):
-        self._rms = ppo_math.ExponentialRunningMeanStd() if self.value_norm else None
+        self._rms = (ppo_math.make_value_norm(
+            self.value_norm_type, self.value_norm_beta, self.value_norm_eps)
+            if self.value_norm else None)
         # The critic must adapt its KL coefficient in lockstep with the actor
         # (reference ppo_interface.py:658) — otherwise value targets are
         # computed from a stale kl_ctl while the actor's rewards drift.

@@ -161,3 +161,68 @@ class ExponentialRunningMeanStd:
 
     def load_state_dict(self, sd):
         self.mean, self.mean_sq, self.debias = sd["mean"], sd["mean_sq"], sd["debias"]
+
+
+class MovingAverageRunningMeanStd:
+    """Cumulative-average value normalization (reference: modules/rms.py:102
+    MovingAverageRunningMeanStd — all history weighted equally, normalize
+    clips to +-5).  Selected by value_norm_type="ma"."""
+
+    def __init__(self, eps: float = 1e-5, high_precision=True):
+        self.eps = eps
+        self.mean = 0.0
+        self.mean_sq = 0.0
+        self.denom = 0.0
+
+    @torch.no_grad()
+    def update(self, x: torch.Tensor, mask=None, group=None):
+        xf = x.float()
+        if mask is not None:
+            m = mask.float()
+            cnt = m.sum()
+            s = (xf * m).sum()
+            sq = (xf * xf * m).sum()
+        else:
+            cnt = torch.tensor(float(xf.numel()), device=x.device)
+            s = xf.sum()
+            sq = (xf * xf).sum()
+        if group is not None and dist.is_initialized():
+            stats = torch.stack([cnt, s, sq])
+            dist.all_reduce(stats, group=group)
+            cnt, s, sq = stats[0], stats[1], stats[2]
+        c = float(cnt)
+        if c < 1:
+            return
+        self.mean = (self.denom * self.mean + float(s)) / (self.denom + c)
+        self.mean_sq = (self.denom * self.mean_sq + float(sq)) / (self.denom + c)
+        self.denom += c
+
+    def mean_std(self):
+        if self.denom == 0:
+            return 0.0, 1.0
+        var = max(self.mean_sq - self.mean * self.mean, 1e-4)
+        return self.mean, var**0.5
+
+    def normalize(self, x: torch.Tensor) -> torch.Tensor:
+        mean, std = self.mean_std()
+        return ((x - mean) / std).clip(-5, 5)
+
+    def denormalize(self, x: torch.Tensor) -> torch.Tensor:
+        mean, std = self.mean_std()
+        return x * std + mean
+
+    def state_dict(self):
+        return dict(mean=self.mean, mean_sq=self.mean_sq, denom=self.denom)
+
+    def load_state_dict(self, sd):
+        self.mean, self.mean_sq, self.denom = sd["mean"], sd["mean_sq"], sd["denom"]
+
+
+def make_value_norm(value_norm_type: str = "exp", beta: float = 0.99995,
+                    eps: float = 1e-5):
+    """reference: modules/rms.py:185 (exp | ma)."""
+    if value_norm_type == "exp":
+        return ExponentialRunningMeanStd(beta=beta, eps=eps)
+    if value_norm_type == "ma":
+        return MovingAverageRunningMeanStd(eps=eps)
+    raise ValueError(f"value_norm_type {value_norm_type!r}")

@@ -331,6 +331,9 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
         "adaptive_kl_ctl": ppo.use_adaptive_kl_ctl,
         "max_reward_clip": ppo.max_reward_clip, "discount": ppo.discount,
         "gae_lambda": ppo.gae_lambda, "value_norm": ppo.value_norm,
+        "value_norm_type": ppo.value_norm_type,
+        "value_norm_beta": ppo.value_norm_beta,
+        "value_norm_eps": ppo.value_norm_eps,
     })
     rw_iface = Abstraction("paired_rw", {
         "output_scaling": ppo.reward_output_scaling,

@@ -297,3 +297,39 @@ def test_grpo_group_norm_id_based():
     assert torch.allclose(per_seq[0], -per_seq[2], atol=1e-5)  # group a pair
     assert torch.allclose(per_seq[3], -per_seq[4], atol=1e-5)  # group c pair
     assert per_seq[0] < 0 and per_seq[2] > 0
+
+
+def test_value_norm_variants():
+    """exp vs ma value normalization (reference: modules/rms.py:16/102,
+    selected by value_norm_type, ppo_exp.py:62-68)."""
+    from realhf_amd.interfaces import ppo_math
+
+    x = torch.tensor([1.0, 2.0, 3.0, 4.0])
+    ma = ppo_math.make_value_norm("ma")
+    ma.update(x)
+    m, s = ma.mean_std()
+    assert abs(m - 2.5) < 1e-6
+    # second update: cumulative average over ALL history, equal weights
+    ma.update(torch.tensor([5.0, 6.0]))
+    m2, _ = ma.mean_std()
+    assert abs(m2 - (1 + 2 + 3 + 4 + 5 + 6) / 6) < 1e-6
+    # round trip
+    y = ma.normalize(x)
+    torch.testing.assert_close(ma.denormalize(y), x, atol=1e-4, rtol=1e-4)
+    # state dict round trip
+    ma2 = ppo_math.make_value_norm("ma")
+    ma2.load_state_dict(ma.state_dict())
+    assert ma2.mean_std() == ma.mean_std()
+
+    exp = ppo_math.make_value_norm("exp", beta=0.9)
+    exp.update(x)
+    m, s = exp.mean_std()
+    assert abs(m - 2.5) < 1e-6  # debiased EMA after one update = batch mean
+
+    # critic interface builds the right normalizer from config
+    from realhf_amd.interfaces.ppo import PPOCriticInterface
+
+    c = PPOCriticInterface(value_norm=True, value_norm_type="ma")
+    assert isinstance(c._rms, ppo_math.MovingAverageRunningMeanStd)
+    with pytest.raises(ValueError):
+        PPOCriticInterface(value_norm=True, value_norm_type="bogus")
