@@ -189,6 +189,9 @@ class GenerationConfig(CommonExperimentConfig):
     gen: GenerationHyperparameters = dataclasses.field(
         default_factory=GenerationHyperparameters
     )
+    # jsonl dump of generations under LOG_ROOT (reference gen_exp.py:49);
+    # None disables
+    output_file: Optional[str] = "output.jsonl"
 
 
 @dataclasses.dataclass

@@ -19,6 +19,11 @@ class GenerationInterface(ModelInterface):
     gconfig: GenerationHyperparameters = dataclasses.field(
         default_factory=GenerationHyperparameters
     )
+    # Dump generations as jsonl under LOG_ROOT (reference:
+    # gen_interface.py:86-152 output_file).  One file per dp rank
+    # (suffix .rankN) instead of the reference's lock-file serialization
+    # — same records, race-free.
+    output_file: str = None
 
     def __post_init__(self):
         if isinstance(self.gconfig, dict):
@@ -31,6 +36,7 @@ class GenerationInterface(ModelInterface):
         if outs is None:  # pp mid stage
             return None
         all_ids, all_pm, seqlens = [], [], []
+        records = []
         for gen_out, prompts, cu in outs:
             packed, cu_full, pmask = concat_prompt_to_generation_output(
                 prompts, cu, gen_out
@@ -38,6 +44,14 @@ class GenerationInterface(ModelInterface):
             all_ids.append(packed)
             all_pm.append(pmask)
             seqlens += [int(cu_full[i + 1] - cu_full[i]) for i in range(cu.shape[0] - 1)]
+            if self.output_file is not None:
+                for i in range(cu.shape[0] - 1):
+                    p = prompts[int(cu[i]):int(cu[i + 1])]
+                    gl = int(gen_out.gen_lengths[i])
+                    a = gen_out.gen_tokens[i, :gl]
+                    records.append((p.cpu(), a.cpu()))
+        if records:
+            self._dump(model, data.ids, records)
         return SequenceSample(
             keys=("packed_input_ids", "prompt_mask"),
             ids=list(data.ids),
@@ -50,6 +64,37 @@ class GenerationInterface(ModelInterface):
                 "prompt_mask": torch.cat(all_pm),
             },
         )
+
+    def _dump(self, model: Model, ids, records):
+        import json
+        import os
+
+        from realhf_amd.base import constants
+
+        if constants.has_current() and (
+            constants.tp_rank() != 0 or not constants.is_last_pipe_stage()
+        ):
+            return  # one writer per dp shard
+        root = constants.LOG_ROOT(constants.experiment_name(),
+                                  constants.trial_name())
+        os.makedirs(root, exist_ok=True)
+        dp = constants.dp_rank() if constants.has_current() else 0
+        path = os.path.join(root, f"{self.output_file}.rank{dp}")
+        tok = model.tokenizer
+        with open(path, "a") as f:
+            for _id, (p, a) in zip(ids, records):
+                if tok is not None:
+                    rec = dict(
+                        id=str(_id),
+                        prompt=tok.decode(p.tolist(), skip_special_tokens=True),
+                        answer=tok.decode(a.tolist(), skip_special_tokens=True),
+                        seq=tok.decode(torch.cat([p, a]).tolist(),
+                                       skip_special_tokens=True),
+                    )
+                else:  # no tokenizer (synthetic runs): dump token ids
+                    rec = dict(id=str(_id), prompt_ids=p.tolist(),
+                               answer_ids=a.tolist())
+                f.write(json.dumps(rec) + "\n")
 
 
 register_interface("generation", GenerationInterface)

@@ -262,7 +262,9 @@ def build_experiment(cfg: CommonExperimentConfig, world: int) -> BuiltExperiment
     elif isinstance(cfg, GenerationConfig):
         name, par = add_model("default", cfg.model)
         add_mfc("gen", name, T.GENERATE,
-                Abstraction("generation", {"gconfig": dataclasses.asdict(cfg.gen)}),
+                Abstraction("generation",
+                            {"gconfig": dataclasses.asdict(cfg.gen),
+                             "output_file": cfg.output_file}),
                 ["packed_prompts"], ["packed_input_ids", "prompt_mask"],
                 cfg.model, par)
     else:

@@ -337,6 +337,15 @@ def test_gen_experiment_single_process(tmp_path):
     cfg.exp_ctrl.benchmark_steps = 1
     os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
     Trainer(cfg).run()
+    # output_file dump (reference gen_exp.py:49): one jsonl per dp rank
+    from realhf_amd.base import constants
+
+    root = constants.LOG_ROOT("t-gen", "cpu")
+    out = os.path.join(root, "output.jsonl.rank0")
+    assert os.path.exists(out), os.listdir(root)
+    recs = [json.loads(l) for l in open(out)]
+    assert len(recs) > 0 and len(recs) % 4 == 0  # multiples of the batch
+    assert all("prompt_ids" in r and "answer_ids" in r for r in recs)
 
 
 def test_sft_training_deterministic(tmp_path):
