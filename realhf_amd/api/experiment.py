@@ -61,9 +61,16 @@ class DatasetConfig:
 
 @dataclasses.dataclass
 class ExperimentSaveEvalControl:
+    """reference: system_api.py:157 — save/eval by epochs, steps or
+    wall-clock seconds, whichever fires first."""
+
     total_train_epochs: int = 1
+    save_freq_epochs: Optional[int] = None
     save_freq_steps: Optional[int] = None
+    save_freq_secs: Optional[float] = None
+    eval_freq_epochs: Optional[int] = None
     eval_freq_steps: Optional[int] = None
+    eval_freq_secs: Optional[float] = None
     benchmark_steps: Optional[int] = None  # early-exit for throughput runs
 
 
@@ -112,7 +119,7 @@ class CommonExperimentConfig:
     slurm: SlurmConfig = dataclasses.field(default_factory=SlurmConfig)
     n_gpus: int = 1
     seed: int = 1
-    allocation_mode: str = "global"  # global | manual | heuristic | d8t1p1-style
+    allocation_mode: str = "global"  # global | manual | heuristic | search | d8t1p1-style
     exp_ctrl: ExperimentSaveEvalControl = dataclasses.field(
         default_factory=ExperimentSaveEvalControl
     )

@@ -617,12 +617,44 @@ class Trainer:
                         {k: round(v, 4) for k, v in out.items()})
         return out
 
+    def _fire(self, fc, **kw) -> bool:
+        """Evaluate a FrequencyControl SPMD-safely.  Step/epoch triggers
+        are deterministic across ranks; a wall-clock trigger is not (save
+        and eval run collectives, so ranks must agree) — rank 0 decides
+        and broadcasts the flag (reference: the master worker decides
+        centrally, master_worker.py:820-838)."""
+        if fc is None:
+            return False
+        fire = fc.check(**kw)
+        if fc.freq_sec is not None and dist.is_initialized():
+            t = torch.tensor([1 if fire else 0])
+            dist.broadcast(t, src=0)
+            fire = bool(t.item())
+            if fire:  # keep non-zero ranks' clocks aligned with rank 0
+                fc._last_time = time.monotonic()
+                fc._last_step = fc._last_epoch = 0
+        return fire
+
     # ----------------------------------------------------------------- run
     def run(self):
         cfg = self.cfg
         dl = self.train_dl
         ctrl = cfg.exp_ctrl
         bench_t0 = None
+        from realhf_amd.base.timeutil import FrequencyControl
+
+        save_ctl = FrequencyControl(
+            freq_epoch=getattr(ctrl, "save_freq_epochs", None),
+            freq_step=ctrl.save_freq_steps,
+            freq_sec=getattr(ctrl, "save_freq_secs", None))
+        eval_ctl = FrequencyControl(
+            freq_epoch=getattr(ctrl, "eval_freq_epochs", None),
+            freq_step=ctrl.eval_freq_steps,
+            freq_sec=getattr(ctrl, "eval_freq_secs", None))
+        save_any = (save_ctl.freq_epoch or save_ctl.freq_step
+                    or save_ctl.freq_sec)
+        eval_any = (eval_ctl.freq_epoch or eval_ctl.freq_step
+                    or eval_ctl.freq_sec)
         recover = self._maybe_load_recover()
         start_epoch, start_step = (recover or (0, 0))
         for epoch in range(start_epoch, ctrl.total_train_epochs):
@@ -658,11 +690,10 @@ class Trainer:
                                                  for k in perf})
                             ),
                         )
-                if ctrl.save_freq_steps and self.global_step % ctrl.save_freq_steps == 0:
+                if save_any and self._fire(save_ctl, steps=1):
                     self.save()
                     self.save_recover_ckpt(epoch, i + 1)
-                if (ctrl.eval_freq_steps
-                        and self.global_step % ctrl.eval_freq_steps == 0):
+                if eval_any and self._fire(eval_ctl, steps=1):
                     self.evaluate()
                 if ctrl.benchmark_steps:
                     if bench_t0 is None:
@@ -680,7 +711,11 @@ class Trainer:
                         self._save_recover_info(epoch, i)
                         return
             self.save_recover_ckpt(epoch + 1, 0)
-        if ctrl.save_freq_steps:
+            if save_any and self._fire(save_ctl, epochs=1):
+                self.save()
+            if eval_any and self._fire(eval_ctl, epochs=1):
+                self.evaluate()
+        if save_any:
             self.save()
 
     def save(self):

@@ -439,3 +439,27 @@ def test_sft_zero2_two_ranks(tmp_path):
     _write_sft_data(data)
     LocalMultiProcessTest(2, _sft_zero2_worker, data,
                           str(tmp_path / "root")).launch()
+
+
+def test_save_eval_freq_epochs_and_secs(tmp_path):
+    """Epoch- and seconds-frequency save/eval triggers (reference:
+    system_api.py:157 ExperimentSaveEvalControl + EpochStepTimeFreqCtl)."""
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.base import constants
+    from realhf_amd.runtime.trainer import Trainer
+
+    data = str(tmp_path / "sft.jsonl")
+    _write_sft_data(data)
+    cfg = SFTConfig(experiment_name="t-sft-freq", trial_name="cpu", n_gpus=1)
+    cfg.model.dtype = "float32"
+    cfg.dataset.type_ = "prompt_answer"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 8
+    cfg.exp_ctrl.total_train_epochs = 2
+    cfg.exp_ctrl.save_freq_epochs = 1
+    cfg.exp_ctrl.eval_freq_secs = 0.0  # fires every step
+    os.environ["REALHF_AMD_FILEROOT"] = str(tmp_path / "root")
+    Trainer(cfg).run()
+    # epoch-frequency saves landed
+    save_root = os.path.join(constants.MODEL_SAVE_ROOT("t-sft-freq", "cpu"))
+    assert os.path.isdir(save_root) and os.listdir(save_root)
