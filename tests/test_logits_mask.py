@@ -185,3 +185,68 @@ def test_ppo_logits_mask_two_ranks(tmp_path):
             f.write(json.dumps({"prompt": "x", "input_ids": ids}) + "\n")
     LocalMultiProcessTest(2, _ppo_mask_dist_worker, data,
                           str(tmp_path / "root")).launch()
+
+
+def test_mask_mode_reforward_matches_sampler_logp():
+    """THE consistency property the mask exists for: re-forwarding the
+    generated sequence with the mask + temperature applied must
+    reproduce the sampler's behavior logprobs on generated tokens
+    (importance ratio == 1 at step 0).  Without the mask, top-k
+    renormalization makes them differ."""
+    from realhf_amd.api.config import ModelName
+    from realhf_amd.api.model import Model
+    from realhf_amd.interfaces.ppo import PPOActorInterface
+    from realhf_amd.models.hf import llama
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.runtime.engine import PipelinableInferenceEngine
+
+    cfg = llama.make_test_config(vocab_size=64)
+    cfg.dtype = "float32"
+    m = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    torch.manual_seed(3)
+    m.random_init()
+    model = Model(name=ModelName("actor", 0), module=PipelinableInferenceEngine(m),
+                  tokenizer=None, device=torch.device("cpu"),
+                  dtype=torch.float32)
+    iface = PPOActorInterface(
+        gconfig={"max_new_tokens": 6, "min_new_tokens": 6, "top_k": 4,
+                 "use_hip_graph": False, "force_no_logits_mask": False},
+    )
+    rng = np.random.RandomState(1)
+    lens = [5, 7]
+    prompts = torch.from_numpy(rng.randint(0, 64, size=sum(lens))).long()
+    sample = __import__("realhf_amd.api.data", fromlist=["SequenceSample"]).SequenceSample(
+        keys=("packed_prompts",),
+        ids=["a", "b"],
+        seqlens={"packed_prompts": [[lens[0]], [lens[1]]]},
+        data={"packed_prompts": prompts},
+    )
+    out = iface.generate(model, sample)
+    assert "packed_logits_mask" in out.keys
+    ref = iface.inference(model, out)  # same weights = "ref" model
+    old = out.data["packed_logprobs"]
+    new = ref.data["packed_ref_logprobs"]
+    pm = out.data["prompt_mask"].bool()
+    # shifted positions: row j of logprobs corresponds to token j+1
+    from realhf_amd.utils.functional import build_shift_one_indices
+
+    total = int(sum(sum(x) for x in out.seqlens["packed_input_ids"]))
+    cu = torch.tensor(
+        [0] + list(np.cumsum([sum(x) for x in out.seqlens["packed_input_ids"]])),
+        dtype=torch.int32)
+    gen_rows = ~pm[build_shift_one_indices(total, cu)]
+    torch.testing.assert_close(new[gen_rows], old[gen_rows],
+                               atol=1e-4, rtol=1e-4)
+    # sanity: the filtered behavior logp differs from the unfiltered one
+    # (top-k renormalization concentrates mass on the kept entries), so
+    # the match above is non-trivial
+    masked_rows = out.data["packed_logits_mask"][gen_rows]
+    assert masked_rows.any()
+    assert (old[gen_rows] > np.log(1.0 / 64) + 0.05).all()  # > uniform-ish
+    # unfiltered re-forward of the SAME sequences (no mask applied)
+    plain = PPOActorInterface(
+        gconfig={"max_new_tokens": 6, "use_hip_graph": False},
+    ).inference(model, out.select_keys(["packed_input_ids"]))
+    assert not torch.allclose(
+        plain.data["packed_ref_logprobs"][gen_rows], old[gen_rows],
+        atol=1e-4)
