@@ -345,6 +345,8 @@ class PipelinedEngine(PipelinableEngine):
                 s.gen_logprobs = torch.zeros(s.bs, max_new,
                                              dtype=torch.float32, device=dev)
                 s.gen_lengths = torch.zeros(s.bs, dtype=torch.long, device=dev)
+                s.step_masks = [] if (not gconfig.force_no_logits_mask
+                                      and not gconfig.greedy) else None
             st.append(s)
 
         use_graph = (
@@ -365,8 +367,11 @@ class PipelinedEngine(PipelinableEngine):
             r = s.r
             if eos is not None and r < gconfig.min_new_tokens:
                 cur_logits[:, eos] = float("-inf")
-            tokens, logp, _ = genmod._sample_from_logits(
-                cur_logits, gconfig, gen)
+            tokens, logp, smask = genmod._sample_from_logits(
+                cur_logits, gconfig, gen,
+                return_mask=s.step_masks is not None)
+            if s.step_masks is not None:
+                s.step_masks.append(smask)
             tokens = torch.where(s.done, torch.full_like(tokens, pad), tokens)
             s.gen_tokens[:, r] = tokens
             s.gen_logprobs[:, r] = torch.where(
@@ -468,6 +473,9 @@ class PipelinedEngine(PipelinableEngine):
                     gen_lengths=s.gen_lengths,
                     no_eos_mask=~s.done,
                     prompt_logprobs=s.prompt_logprobs,
+                    logits_mask=(
+                        torch.stack(s.step_masks[:max_len], dim=1)
+                        if getattr(s, "step_masks", None) else None),
                 ),
                 s.prompts,
                 s.cu,

@@ -356,12 +356,6 @@ def _build_ppo(cfg: PPOConfig, world, add_model, add_mfc, mfcs, trainable):
     # ppo_interface.py:245-246 emits packed_logits_mask the same way)
     mask_keys = ([] if ppo.gen.force_no_logits_mask
                  else ["packed_logits_mask"])
-    if mask_keys and (gen_par if cfg.actor.gen_parallel is None
-                      else cfg.actor.gen_parallel).pipeline_parallel_size > 1:
-        raise ValueError(
-            "gen.force_no_logits_mask=False is not supported with "
-            "pipeline-parallel generation (the token-interleaved PP "
-            "decode does not record per-step masks)")
     add_mfc("actor_gen", gen_model, T.GENERATE, actor_iface,
             ["packed_prompts"],
             ["packed_input_ids", "packed_logprobs", "prompt_mask",
@@ -426,10 +420,6 @@ def _build_grpo(cfg: GRPOConfig, world, add_model, add_mfc, mfcs, trainable):
     })
     mask_keys = ([] if ppo.gen.force_no_logits_mask
                  else ["packed_logits_mask"])
-    if mask_keys and apar.pipeline_parallel_size > 1:
-        raise ValueError(
-            "gen.force_no_logits_mask=False is not supported with "
-            "pipeline-parallel generation")
     add_mfc("actor_gen", actor, T.GENERATE, iface, ["packed_prompts"],
             ["packed_input_ids", "packed_logprobs", "prompt_mask",
              "seq_no_eos_mask"] + mask_keys, cfg.actor, apar)

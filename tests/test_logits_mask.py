@@ -99,17 +99,60 @@ def test_ppo_experiment_with_logits_mask(tmp_path):
     Trainer(cfg).run()
 
 
-def test_mask_mode_rejects_pp_generation():
-    from realhf_amd.api.config import ParallelismConfig
-    from realhf_amd.api.experiment import PPOConfig
-    from realhf_amd.runtime.trainer import build_experiment
+def _pp_mask_worker():
+    """Token-interleaved PP generation records per-step sampler masks
+    (last stage), so the mask mode works under pipeline parallelism."""
+    import torch.distributed as dist
 
-    cfg = PPOConfig(experiment_name="t", trial_name="t", n_gpus=2)
-    cfg.actor.parallel = ParallelismConfig(pipeline_parallel_size=2)
-    cfg.allocation_mode = "manual"
-    cfg.ppo.gen.force_no_logits_mask = False
-    with pytest.raises(ValueError, match="pipeline-parallel generation"):
-        build_experiment(cfg, world=2)
+    from realhf_amd.api.data import SequenceSample
+    from realhf_amd.base import constants
+    from tests.test_pp import _make_pp_engine
+    from tests.test_realloc import _full_reference_sd
+    from realhf_amd.models.hf.llama import make_test_config
+
+    cfg = make_test_config(n_layers=4, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=53)
+    engine, m = _make_pp_engine(cfg, sd, 2)
+    rng = np.random.RandomState(10)
+    lens = [5, 8, 6]
+    toks = torch.from_numpy(rng.randint(0, 128, size=sum(lens))).long()
+    batch = SequenceSample(
+        keys=("packed_prompts",),
+        ids=["g0", "g1", "g2"],
+        seqlens={"packed_prompts": [[l] for l in lens]},
+        data={"packed_prompts": toks},
+    )
+    g = GenerationHyperparameters(max_new_tokens=6, top_k=4,
+                                  use_hip_graph=False,
+                                  force_no_logits_mask=False)
+    with constants.model_scope("m"):
+        outs = engine.generate(batch, gconfig=g, n_mbs=2)
+    if m.is_last_stage:
+        n = 0
+        for gen_out, prompts, cu in outs:
+            lm = gen_out.logits_mask
+            assert lm is not None
+            bs = cu.shape[0] - 1
+            gmax = gen_out.gen_tokens.shape[1]
+            assert lm.shape == (bs, gmax, 128)
+            assert (lm.sum(-1) == 124).all()  # V - top_k removed per step
+            # sampled tokens are never masked (rows where not yet done)
+            for i in range(bs):
+                gl = int(gen_out.gen_lengths[i])
+                got = lm[i, torch.arange(gl), gen_out.gen_tokens[i, :gl]]
+                assert not got.any()
+            n += bs
+        assert n == len(lens)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_pp2_generation_logits_mask():
+    from realhf_amd.base.testing import LocalMultiProcessTest
+
+    LocalMultiProcessTest(2, _pp_mask_worker).launch()
 
 
 def test_grpo_experiment_with_logits_mask(tmp_path):
