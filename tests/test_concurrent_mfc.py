@@ -195,7 +195,7 @@ def _write_prompt_data(path, n=16, vocab=64):
             f.write(json.dumps(rec) + "\n")
 
 
-def _ppo_heuristic4_worker(data, fileroot):
+def _ppo_heuristic4_worker(data, fileroot, logits_mask=False):
     """4-rank PPO with allocation_mode=heuristic: the asymmetric plan
     puts critic_inf on ranks {0,1} and rew_inf (and the reward model
     itself) on {2,3} — concurrent disjoint-mesh inference in a real
@@ -214,6 +214,9 @@ def _ppo_heuristic4_worker(data, fileroot):
     cfg.dataset.max_prompt_len = 8
     cfg.ppo.gen.max_new_tokens = 5
     cfg.ppo.gen.use_hip_graph = False
+    if logits_mask:
+        cfg.ppo.gen.top_k = 4
+        cfg.ppo.gen.force_no_logits_mask = False
     cfg.ppo.ppo_n_minibatches = 2
     cfg.exp_ctrl.benchmark_steps = 1
     t = Trainer(cfg)
@@ -237,6 +240,17 @@ def test_ppo_heuristic_asymmetric_four_ranks(tmp_path):
     _write_prompt_data(data, n=16)
     LocalMultiProcessTest(4, _ppo_heuristic4_worker, data,
                           str(tmp_path / "root")).launch()
+
+
+@pytest.mark.distributed
+def test_ppo_heuristic_four_ranks_logits_mask(tmp_path):
+    """Asymmetric heuristic + logits-mask mode: the 2-D mask key rides
+    the sub-mesh transfer plan (gen whole-node -> ref whole-node /
+    actor_train whole-node, skipping critic/rew halves)."""
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data, n=16)
+    LocalMultiProcessTest(4, _ppo_heuristic4_worker, data,
+                          str(tmp_path / "root2"), True).launch()
 
 
 def test_ppo_heuristic8_static_plan():
