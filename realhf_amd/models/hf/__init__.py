@@ -112,12 +112,23 @@ def load_from_hf(model, family: str, path: str):
             if h not in weight_map:
                 continue
             by_file.setdefault(weight_map[h], []).append(h)
+        from concurrent.futures import ThreadPoolExecutor
+
         from safetensors import safe_open
 
-        for fn, ks in by_file.items():
+        def _load_file(fn_ks):
+            fn, ks = fn_ks
+            out = {}
             with safe_open(os.path.join(path, fn), framework="pt") as f:
                 for h in ks:
-                    hf_tensors[h] = f.get_tensor(h)
+                    out[h] = f.get_tensor(h)
+            return out
+
+        # thread-pool shard loading (reference: hf_registry.py:62 loads
+        # shards concurrently — IO-bound, so threads suffice)
+        with ThreadPoolExecutor(max_workers=min(8, len(by_file))) as ex:
+            for part in ex.map(_load_file, by_file.items()):
+                hf_tensors.update(part)
     else:
         full = torch.load(
             os.path.join(path, "pytorch_model.bin"), map_location="cpu",
