@@ -126,9 +126,10 @@ def load_from_hf(model, family: str, path: str):
 
         # thread-pool shard loading (reference: hf_registry.py:62 loads
         # shards concurrently — IO-bound, so threads suffice)
-        with ThreadPoolExecutor(max_workers=min(8, len(by_file))) as ex:
-            for part in ex.map(_load_file, by_file.items()):
-                hf_tensors.update(part)
+        if by_file:
+            with ThreadPoolExecutor(max_workers=min(8, len(by_file))) as ex:
+                for part in ex.map(_load_file, by_file.items()):
+                    hf_tensors.update(part)
     else:
         full = torch.load(
             os.path.join(path, "pytorch_model.bin"), map_location="cpu",
