@@ -293,3 +293,75 @@ def test_mask_mode_reforward_matches_sampler_logp():
     assert not torch.allclose(
         plain.data["packed_ref_logprobs"][gen_rows], old[gen_rows],
         atol=1e-4)
+
+
+def _tp2_masked_logprob_worker():
+    """Masked + temperature-scaled logprobs under TP vocab sharding:
+    apply_logits_mask slices the full-vocab mask per tp rank and the
+    vocab-parallel logprob matches the single-process masked reference."""
+    import torch.distributed as dist
+
+    from realhf_amd.base import constants
+    from realhf_amd.base.testing import init_global_constants
+    from realhf_amd.interfaces.ppo import _warp_logits_like_sampler
+    from realhf_amd.models.hf.llama import make_test_config
+    from realhf_amd.models.real_model import ReaLModel
+    from realhf_amd.parallel.tp import packed_shifted_logprobs
+    from realhf_amd.api.data import SequenceSample
+    from tests.test_realloc import _fill_model_from_full, _full_reference_sd
+
+    cfg = make_test_config(n_layers=2, hidden_dim=64, n_heads=8, n_kv_heads=4,
+                           vocab_size=128)
+    cfg.dtype = "float32"
+    sd = _full_reference_sd(cfg, seed=57)
+    init_global_constants(num_dp=1, num_tp=2, num_pp=1, model_name="m")
+    g = constants.grid_of("m")
+
+    rng = np.random.RandomState(6)
+    lens = [10, 8]
+    toks = torch.from_numpy(rng.randint(0, 128, size=18)).long()
+    cu = torch.tensor([0, 10, 18], dtype=torch.int32)
+    # random full-vocab mask on the predicting rows, never masking the
+    # next token itself (the sampled one is always kept)
+    lm = torch.rand(16, 128) > 0.4
+    from realhf_amd.utils.functional import build_shift_one_indices
+
+    nxt = toks[build_shift_one_indices(18, cu)]
+    lm[torch.arange(16), nxt] = False
+    mb = SequenceSample(
+        keys=("packed_input_ids", "packed_logits_mask"),
+        ids=["a", "b"],
+        seqlens={"packed_input_ids": [[10], [8]],
+                 "packed_logits_mask": [[9], [7]]},
+        data={"packed_input_ids": toks, "packed_logits_mask": lm},
+    )
+
+    with constants.model_scope("m"):
+        m = ReaLModel(cfg, device="cpu", dtype=torch.float32,
+                      tp_rank=g.tp_rank, tp_size=2)
+        _fill_model_from_full(m, cfg, sd)
+        with torch.no_grad():
+            logits = m(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=10)
+            _warp_logits_like_sampler(logits, cu, mb, temperature=0.7)
+            lp = packed_shifted_logprobs(logits, cu, toks)
+
+    single = ReaLModel(cfg, device="cpu", dtype=torch.float32)
+    _fill_model_from_full(single, cfg, sd)
+    constants.clear_grids()
+    with torch.no_grad():
+        full = single(packed_input_ids=toks, cu_seqlens=cu, max_seqlen=10)
+        full = full / 0.7
+        rows = torch.arange(18)[~torch.isin(
+            torch.arange(18), cu[1:].long() - 1)]  # leave-one rows
+        full[rows] = full[rows].masked_fill(lm, float("-inf"))
+        want = torch.log_softmax(full.float(), -1)
+        want = want[rows].gather(1, nxt.unsqueeze(1)).squeeze(1)
+    torch.testing.assert_close(lp, want, atol=1e-4, rtol=1e-4)
+    dist.barrier()
+
+
+@pytest.mark.distributed
+def test_tp2_masked_logprobs_match_single():
+    from realhf_amd.base.testing import LocalMultiProcessTest
+
+    LocalMultiProcessTest(2, _tp2_masked_logprob_worker).launch()
