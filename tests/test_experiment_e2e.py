@@ -463,3 +463,42 @@ def test_save_eval_freq_epochs_and_secs(tmp_path):
     # epoch-frequency saves landed
     save_root = os.path.join(constants.MODEL_SAVE_ROOT("t-sft-freq", "cpu"))
     assert os.path.isdir(save_root) and os.listdir(save_root)
+
+
+def _ppo_z2_mask_worker(data, fileroot):
+    """PPO at dp2 composing the late round-2 features: ZeRO-2 actors,
+    logits-mask mode, and the tp2 gen replica with param realloc."""
+    from realhf_amd.api.config import ParallelismConfig
+    from realhf_amd.api.experiment import PPOConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    cfg = PPOConfig(experiment_name="t-ppo-z2m", trial_name="dist", n_gpus=2)
+    for mc in (cfg.actor, cfg.critic, cfg.ref, cfg.rew):
+        mc.dtype = "float32"
+        mc.parallel = ParallelismConfig(data_parallel_size=2)
+    cfg.actor.optimizer.zero_stage = 2
+    cfg.critic.optimizer.zero_stage = 2
+    cfg.allocation_mode = "manual"
+    cfg.actor.gen_parallel = ParallelismConfig(
+        data_parallel_size=1, tensor_parallel_size=2
+    )
+    cfg.dataset.type_ = "prompt"
+    cfg.dataset.path = data
+    cfg.dataset.train_bs_n_seqs = 4
+    cfg.dataset.max_prompt_len = 8
+    cfg.ppo.gen.max_new_tokens = 5
+    cfg.ppo.gen.top_k = 4
+    cfg.ppo.gen.use_hip_graph = False
+    cfg.ppo.gen.force_no_logits_mask = False
+    cfg.ppo.ppo_n_minibatches = 2
+    cfg.exp_ctrl.benchmark_steps = 2
+    Trainer(cfg).run()
+
+
+@pytest.mark.distributed
+def test_ppo_zero2_logits_mask_two_ranks(tmp_path):
+    data = str(tmp_path / "prompts.jsonl")
+    _write_prompt_data(data)
+    LocalMultiProcessTest(2, _ppo_z2_mask_worker, data,
+                          str(tmp_path / "root")).launch()
