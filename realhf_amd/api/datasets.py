@@ -157,4 +157,7 @@ class SyntheticPromptDataset(torch.utils.data.Dataset):
 register_dataset("prompt", PromptDataset)
 register_dataset("prompt_answer", PromptAnswerDataset)
 register_dataset("rw_paired", RewardModelingPairedDataset)
+# reference registers this dataset as "rw_pair" (rw_paired_dataset.py) —
+# keep that name working for drop-in configs
+register_dataset("rw_pair", RewardModelingPairedDataset)
 register_dataset("synthetic_prompt", SyntheticPromptDataset)

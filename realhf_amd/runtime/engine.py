@@ -211,3 +211,7 @@ class InferenceBackend(ModelBackend):
 
 register_backend("zero1", ZeRO1Backend)
 register_backend("inference", InferenceBackend)
+# reference name for "wrap without training machinery" (used by custom
+# experiments like ppo_sentiment, system_api "null" backend) — here the
+# inference engine IS that minimal wrap
+register_backend("null", InferenceBackend)
