@@ -53,6 +53,13 @@ def test_cli_override_parsing():
     assert cfg.n_gpus == 4
     assert cfg.ppo.gen.max_new_tokens == 64
     assert cfg.ppo.kl_ctl == 0.05
+    # late round-2 fields stay CLI-drivable
+    _, c2 = parse_cli(["ppo", "actor.optimizer.zero_stage=2",
+                       "ppo.gen.force_no_logits_mask=false",
+                       "ppo.value_norm_type=ma"])
+    assert c2.actor.optimizer.zero_stage == 2
+    assert c2.ppo.gen.force_no_logits_mask is False
+    assert c2.ppo.value_norm_type == "ma"
     assert cfg.actor.optimizer.lr == 1e-4
     assert cfg.exp_ctrl.save_freq_steps is None
     assert cfg.actor.gradient_checkpointing is True
