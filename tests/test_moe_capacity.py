@@ -66,6 +66,22 @@ def test_tight_capacity_drops_and_trains(policy):
     assert float(capped.flat_grad.abs().sum()) > 0
 
 
+def test_pad_to_capacity_backward():
+    """Gradients flow through the padded (static-shape) dispatch — the
+    zero pad rows contribute nothing."""
+    packed, cu, mx = _batch()
+    _, pad = _make_models(0.5, pad=True)
+    pad.train()
+    pad.allocate_grad_buffer()
+    for k, p in pad._params.items():
+        p.requires_grad_(True)
+        p.grad = pad.grad_view(k)
+    out = pad(packed_input_ids=packed, cu_seqlens=cu, max_seqlen=mx)
+    out.float().square().mean().backward()
+    assert torch.isfinite(pad.flat_grad).all()
+    assert float(pad.flat_grad.abs().sum()) > 0
+
+
 @pytest.mark.parametrize("policy", ["probs", "position"])
 def test_pad_to_capacity_matches_drop_mode(policy):
     """Padding only adds zero-weighted rows — logits must match the
