@@ -98,13 +98,13 @@ def _zero2_cuda_worker():
     init_global_constants(num_dp=2, num_tp=1, num_pp=1, model_name="m")
     cfg = make_test_config(n_layers=2, hidden_dim=64, n_heads=1, n_kv_heads=1,
                            head_dim=64, intermediate_dim=128, vocab_size=128)
-    cfg.dtype = "float32"
+    cfg.dtype = "bfloat16"  # the HIP kernels are bf16-only
     sd = _full_reference_sd(cfg, seed=151)
     rank = dist.get_rank()
 
     def run(stage):
         with constants.model_scope("m"):
-            m = ReaLModel(cfg, device="cuda", dtype=torch.float32)
+            m = ReaLModel(cfg, device="cuda", dtype=torch.bfloat16)
             _fill_model_from_full(m, cfg, sd)
             opt = ZeRO1Optimizer(
                 m, OptimizerConfig(lr=1e-2, warmup_steps_proportion=0.0,
@@ -132,7 +132,10 @@ def _zero2_cuda_worker():
 
     p1 = run(1)
     p2 = run(2)
-    torch.testing.assert_close(p2, p1, atol=1e-5, rtol=1e-5)
+    # bf16 grads round differently per-microbatch (z2 reduce-scatters
+    # each mb; z1 accumulates then reduces once) — exact equivalence is
+    # pinned by the fp32 CPU test; here the tolerance is a few bf16 ulps
+    torch.testing.assert_close(p2.float(), p1.float(), atol=3e-2, rtol=3e-2)
     dist.barrier()
 
 
