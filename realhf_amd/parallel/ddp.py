@@ -155,13 +155,12 @@ class ZeRO1Optimizer:
             cfg.zero_stage >= 2
             and self.dp_size > 1
             and self.tp_size == 1   # SP tp-reduce needs full-grad intervals
-            and self.pp_size == 1   # pipeline engine drives its own bwd
             and hasattr(model, "layout")
         )
         if cfg.zero_stage >= 2 and not self.zero2:
             logger.warning(
                 "zero_stage=2 requested but unsupported here "
-                "(needs dp>1, tp==1, pp==1); falling back to ZeRO-1")
+                "(needs dp>1, tp==1); falling back to ZeRO-1")
         if self.zero2:
             self.grad_padded = None
             model.flat_grad = None
@@ -223,7 +222,9 @@ class ZeRO1Optimizer:
                 self._param_buckets[k] = list(
                     range(sp.start // self._bsz, (sp.end - 1) // self._bsz + 1))
             self._bucket_left: List[int] = []
-            self._reduced: List[bool] = []
+            # safe default: an engine that never arms (or a backward that
+            # skipped arming) flushes every bucket synchronously at step()
+            self._reduced: List[bool] = [False] * len(self.buckets)
             self._works: list = []
             self._armed = False
             if self.zero2:
@@ -514,6 +515,8 @@ class ZeRO1Optimizer:
                 w.wait()
             self._works = []
             self._armed = False
+            # ready for the next step even if the engine never re-arms
+            self._reduced = [False] * len(self.buckets)
             gshard = self.grad_shard
         elif self.dp_size > 1:
             if self._gloo_cuda():

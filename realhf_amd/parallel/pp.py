@@ -221,6 +221,10 @@ class PipelinedEngine(PipelinableEngine):
             fwd_state[i] = (hidden, out, mb, aux)
 
         def do_bwd(i):
+            if i == n_mbs - 1 and hasattr(self.optimizer, "arm_overlap"):
+                # ZeRO-1 overlap: grads are complete after the LAST
+                # microbatch's backward — bucket reduce-scatters overlap it
+                self.optimizer.arm_overlap()
             hidden, out, mb, aux = fwd_state.pop(i)
             if self.model.is_last_stage:
                 loss, st = loss_fn(out, mb)
@@ -240,6 +244,10 @@ class PipelinedEngine(PipelinableEngine):
                     torch.autograd.backward(out, grad_tensors=gout)
             if hidden is not None:
                 _send(hidden.grad, g.pp_prev_global_rank())
+            if hasattr(self.optimizer, "end_microbatch"):
+                # ZeRO-2: this stage's residual grad buckets reduce-scatter
+                # now, overlapped with the next microbatch's fwd/bwd
+                self.optimizer.end_microbatch()
 
         fi = bi = 0
         for _ in range(warmup):
