@@ -509,3 +509,35 @@ def test_ppo_zero2_logits_mask_two_ranks(tmp_path):
     _write_prompt_data(data)
     LocalMultiProcessTest(2, _ppo_z2_mask_worker, data,
                           str(tmp_path / "root")).launch()
+
+
+def _sft_dp2pp2_worker(data, fileroot):
+    """4-rank SFT at dp2 x pp2 through the full trainer — the config
+    class that exposed the unarmed-overlap crash (ZeRO-1 bucketed RS +
+    PipelinedEngine); runs both zero stages."""
+    from realhf_amd.api.config import ParallelismConfig
+    from realhf_amd.api.experiment import SFTConfig
+    from realhf_amd.runtime.trainer import Trainer
+
+    os.environ["REALHF_AMD_FILEROOT"] = fileroot
+    for stage in (1, 2):
+        cfg = SFTConfig(experiment_name=f"t-sft-dp2pp2-z{stage}",
+                        trial_name="dist", n_gpus=4)
+        cfg.model.dtype = "float32"
+        cfg.model.parallel = ParallelismConfig(
+            data_parallel_size=2, pipeline_parallel_size=2)
+        cfg.model.optimizer.zero_stage = stage
+        cfg.allocation_mode = "manual"
+        cfg.dataset.type_ = "prompt_answer"
+        cfg.dataset.path = data
+        cfg.dataset.train_bs_n_seqs = 4
+        cfg.exp_ctrl.benchmark_steps = 2
+        Trainer(cfg).run()
+
+
+@pytest.mark.distributed
+def test_sft_dp2_pp2_two_stages(tmp_path):
+    data = str(tmp_path / "sft.jsonl")
+    _write_sft_data(data)
+    LocalMultiProcessTest(4, _sft_dp2pp2_worker, data,
+                          str(tmp_path / "root")).launch()
