@@ -9,9 +9,12 @@
 # (mfc, rank, step); REALHF_AMD_DUMP_MEMORY=1 dumps allocator snapshots.
 MODEL_FAMILY=${MODEL_FAMILY:-llama}
 
+MODEL_PATH_ARG=""
+[ -n "${SFT_MODEL_PATH:-}" ] && MODEL_PATH_ARG="model.path=$SFT_MODEL_PATH"
+
 python -m realhf_amd.apps.quickstart profile \
     experiment_name=profile-example trial_name=t0 n_gpus=8 \
-    model.family=$MODEL_FAMILY model.path=${SFT_MODEL_PATH:-} \
+    model.family=$MODEL_FAMILY $MODEL_PATH_ARG \
     interfaces=inference,train_step,generate \
     strategies="d8;d4t2;d2t4;d1t8;d4p2" \
     n_seqs=128 seq_len=512 gen_tokens=128
