@@ -333,3 +333,33 @@ def test_value_norm_variants():
     assert isinstance(c._rms, ppo_math.MovingAverageRunningMeanStd)
     with pytest.raises(ValueError):
         PPOCriticInterface(value_norm=True, value_norm_type="bogus")
+
+
+def test_reference_registry_name_aliases(tmp_path):
+    """Reference configs use dataset "rw_pair" and backend "null" — both
+    names must resolve here (drop-in config compatibility)."""
+    import json
+
+    import numpy as np
+
+    import realhf_amd.api.datasets  # noqa: F401 (registers dataset names)
+    from realhf_amd.api.data import make_dataset
+
+    rng = np.random.RandomState(0)
+    path = tmp_path / "rw.jsonl"
+    with path.open("w") as f:
+        for i in range(4):
+            f.write(json.dumps({
+                "prompt": "p",
+                "pos_ids": rng.randint(0, 50, size=9).tolist(),
+                "neg_ids": rng.randint(0, 50, size=9).tolist(),
+            }) + "\n")
+    ds = make_dataset(Abstraction("rw_pair", {"path": str(path),
+                                              "max_seqlen": 32}),
+                      seed=0, dp_rank=0, world_size=1)
+    assert len(ds) == 4
+    # "null" backend: minimal inference wrap
+    model = make_model()
+    wrapped = make_backend(Abstraction("null")).initialize(
+        model, FinetuneSpec(1, 64, 8))
+    assert wrapped.backend_name == "inference"
